@@ -1,0 +1,223 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: whole-node HTTP requests/sec, 1 KB JSON echo.
+
+BASELINE.json metric: "HTTP requests/sec (whole node) + p99 latency,
+1 KB JSON echo at 1/2/4/8 MI355X". Config 2 (N=1): 4 routes, single-GPU
+request-batch engine. Config 4 shape (N>1): requests sharded across GPUs
+with RCCL all-to-all over xGMI (ingress shard -> owner shard exchange of
+raw request slots, response gather back), one rank per GPU.
+
+One timed step = processing one batch of --batch synthetic 1 KB JSON echo
+requests end-to-end: H2D staging copy, (N>1: all-to-all request scatter),
+k_parse_route, k_respond, (N>1: all-to-all response gather), D2H of the
+response bytes. The payload is synthetic (in-memory request generator —
+no network on the box; SURVEY.md §4's fake-transport tier), byte-exact
+HTTP; there is no numeric precision to reduce (dtype=uint8-exact).
+
+Usage: python bench.py --gpus N --steps K --warmup W
+(N>1 is launched by the driver via torch.distributed.run, one rank/GPU.)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np  # noqa: E402
+
+import gofr_amd  # noqa: E402
+from gofr_amd import handlers  # noqa: E402
+from gofr_amd.config import MapConfig  # noqa: E402
+from gofr_amd.engine import BatchEngine, make_batch  # noqa: E402
+
+
+def build_app():
+    """Config-2 app: 4 routes, traffic is 100% the 1 KB JSON echo."""
+    app = gofr_amd.New(config=MapConfig({"APP_NAME": "bench",
+                                         "LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.GET("/status", handlers.static_json({"status": "ok"}))
+    app.GET("/version", handlers.static_json({"version": "0.1.0"}))
+    return app
+
+
+def make_echo_request(payload_bytes: int = 1024) -> bytes:
+    """One 1 KB JSON echo request (the JSON body is exactly
+    payload_bytes long)."""
+    fixed = b'{"payload":"'
+    pad = payload_bytes - len(fixed) - 2
+    body = fixed + b"a" * pad + b'"}'
+    assert len(body) == payload_bytes
+    return (b"POST /echo HTTP/1.1\r\n"
+            b"Host: bench.local\r\n"
+            b"Content-Type: application/json\r\n"
+            b"User-Agent: gofr-bench/0.1\r\n"
+            b"Content-Length: " + str(len(body)).encode() + b"\r\n"
+            b"\r\n" + body)
+
+
+def run_single(eng, reqs, lens, steps, warmup):
+    import torch
+    times = []
+    for it in range(warmup + steps):
+        if it == warmup:
+            if eng.device is not None:
+                torch.cuda.synchronize(eng.device)
+            t_start = time.perf_counter()
+        t0 = time.perf_counter()
+        resp, rlen = eng.process(reqs, lens)
+        times.append(time.perf_counter() - t0)
+        if it == 0:  # verify once
+            first = resp[:int(rlen[0])].tobytes()
+            assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+    if eng.device is not None:
+        torch.cuda.synchronize(eng.device)
+    elapsed = time.perf_counter() - t_start
+    return elapsed, times[warmup:]
+
+
+def run_multi(eng, reqs, lens, steps, warmup, rank, world):
+    """RCCL all-to-all sharding: each rank ingests its local batch,
+    scatters request slots to owner shards (uniform round-robin conn
+    ids -> equal splits), processes, gathers responses back, D2H."""
+    import torch
+    import torch.distributed as dist
+    t = torch
+    dev = eng.device
+    n = len(lens)
+    assert n % world == 0
+    slot, rslot = eng.slot, eng.rslot
+    # pinned + device staging
+    p_in = t.from_numpy(reqs).pin_memory()
+    p_len = t.from_numpy(lens.astype(np.int32)).pin_memory()
+    d_in = t.empty(n * slot, dtype=t.uint8, device=dev)
+    d_sh = t.empty(n * slot, dtype=t.uint8, device=dev)  # after exchange
+    d_len_in = t.empty(n, dtype=t.int32, device=dev)
+    d_len_sh = t.empty(n, dtype=t.int32, device=dev)
+    d_resp_sh = t.empty(n * rslot, dtype=t.uint8, device=dev)
+    p_resp = t.empty(n * rslot, dtype=t.uint8).pin_memory()
+    p_rlen = t.empty(n, dtype=t.int32).pin_memory()
+
+    times = []
+    for it in range(warmup + steps):
+        if it == warmup:
+            torch.cuda.synchronize(dev)
+            dist.barrier()
+            torch.cuda.synchronize(dev)
+            t_start = time.perf_counter()
+        t0 = time.perf_counter()
+        # ingress: H2D staging of this shard's accepted connections
+        d_in.copy_(p_in, non_blocking=True)
+        d_len_in.copy_(p_len, non_blocking=True)
+        # re-balance: all-to-all over xGMI (7 p2p links used concurrently;
+        # SURVEY.md §2.3 — rings are per-link-bound, all-to-all is the
+        # right shape for request scatter)
+        dist.all_to_all_single(d_sh, d_in)
+        dist.all_to_all_single(d_len_sh, d_len_in)
+        d_resp, d_rlen = eng.process_device(d_sh, d_len_sh, n)
+        # response gather: return each response slab to its ingress rank
+        dist.all_to_all_single(d_resp_sh, d_resp[:n * rslot])
+        dist.all_to_all_single(d_len_in, d_rlen[:n])  # reuse buffer
+        p_resp.copy_(d_resp_sh, non_blocking=True)
+        p_rlen.copy_(d_len_in, non_blocking=True)
+        torch.cuda.synchronize(dev)
+        times.append(time.perf_counter() - t0)
+        if it == 0:
+            first = p_resp[:int(p_rlen[0])].numpy().tobytes()
+            assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+    torch.cuda.synchronize(dev)
+    dist.barrier()
+    torch.cuda.synchronize(dev)
+    elapsed = time.perf_counter() - t_start
+    return elapsed, times[warmup:]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=32768)
+    ap.add_argument("--payload", type=int, default=1024)
+    args = ap.parse_args()
+
+    import torch
+    have_gpu = torch.cuda.is_available()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    batch = args.batch
+    if not have_gpu:
+        batch = min(batch, 256)  # CPU mirror sanity mode
+
+    app = build_app()
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+    eng = BatchEngine(app, device=device, slot=2048, rslot=2048,
+                      max_batch=batch)
+
+    raw = make_echo_request(args.payload)
+    reqs, lens = make_batch([raw] * batch, eng.slot)
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.init_process_group("nccl")
+        elapsed, times = run_multi(eng, reqs, lens, args.steps, args.warmup,
+                                   rank, world)
+        # whole-job aggregate: max elapsed over ranks
+        t_t = torch.tensor([elapsed], device=eng.device)
+        dist.all_reduce(t_t, op=dist.ReduceOp.MAX)
+        elapsed = float(t_t.item())
+        p99_t = torch.tensor([float(np.percentile(times, 99) * 1000)],
+                             device=eng.device)
+        dist.all_reduce(p99_t, op=dist.ReduceOp.MAX)
+        p99_ms = float(p99_t.item())
+    else:
+        elapsed, times = run_single(eng, reqs, lens, args.steps, args.warmup)
+        p99_ms = float(np.percentile(times, 99) * 1000)
+
+    n_gpus = world if have_gpu else 0
+    total_reqs = batch * args.steps * max(world, 1)
+    value = total_reqs / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank == 0:
+        out = {
+            "metric": "HTTP requests/sec (whole node), 1 KB JSON echo",
+            "value": round(value, 1),
+            "unit": "req/s",
+            "n_gpus": n_gpus if have_gpu else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8-exact",
+            "data": "synthetic",
+            "config": {
+                "model": "request-batch engine, 4 routes, 1KB JSON echo",
+                "global_batch": batch * max(world, 1),
+                "seq_len": args.payload,
+                "parallelism": (f"alltoall{world}" if world > 1 else "single"),
+                "p99_step_ms": round(p99_ms, 3),
+                "engine": "gpu" if have_gpu else "cpu-mirror",
+            },
+        }
+        print(json.dumps(out))
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,590 @@
+// gofr_amd CDNA4 (gfx950) request-batch kernels.
+//
+// MI355X-native data plane for the GoFr-capability framework
+// (reference behaviors: SURVEY.md §2.2; this is a new design, not a port —
+// the reference's hot path is Go's net/http + gorilla/mux + encoding/json,
+// pkg/gofr/http/router.go:13-33, http/request.go:40-47, http/responder.go:19-41).
+//
+// Design (one wavefront per request, 64 lanes):
+//   k_parse_route — structural-index HTTP/1.1 parse + radix-trie route
+//     match. Lanes classify 64 bytes per iteration into LF/SP/COLON/QMARK
+//     bitmasks via __ballot (wave64 -> one u64 mask per 64-byte chunk, no
+//     per-byte branching); lane 0 then tokenizes by walking the masks with
+//     ctz. The route trie (compiled by gofr_amd/http/router.py:compile)
+//     is walked on the path segments; {param} spans are captured.
+//   k_respond — handler execution for GPU-native handler kinds (JSON echo,
+//     static body, host-supplied body) fused with the response serializer:
+//     status line + CORS + correlation-id headers (xorshift128 hex, fused)
+//     + Content-Length itoa + envelope + parallel body copy (all 64 lanes).
+//
+// Both kernels are launched with 256-thread blocks (4 waves = 4 requests).
+// Buffers are SoA int32 field tables; see FI_* layout below (mirrored in
+// gofr_amd/ops/__init__.py).
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+#define WAVES_PER_BLOCK 4
+#define BLOCK_THREADS (WAVE * WAVES_PER_BLOCK)
+#define MAX_SLOT 4096
+#define MAX_CHUNKS (MAX_SLOT / 64)
+
+// ---- field table layout (int32 [n, NF]) ------------------------------------
+#define FI_METHOD 0
+#define FI_PATH_OFF 1
+#define FI_PATH_LEN 2
+#define FI_QUERY_OFF 3
+#define FI_QUERY_LEN 4
+#define FI_BODY_OFF 5
+#define FI_BODY_LEN 6
+#define FI_CLEN 7
+#define FI_FLAGS 8
+#define FI_ROUTE 9
+#define FI_KIND 10
+#define FI_STATUS 11
+#define FI_RESP_LEN 12
+#define FI_AUTH_OFF 13
+#define FI_AUTH_LEN 14
+#define FI_RESP_OFF 15
+#define FI_PARAM0 16 /* 4 x (off, len) pairs: 16..23 */
+#define NF 24
+
+// flags
+#define FL_ERR_PARSE 1
+#define FL_NEEDS_HOST 2
+#define FL_KEEP_ALIVE 4
+#define FL_JSON_CT 8
+#define FL_IS_OPTIONS 16
+#define FL_BODY_INVALID 32
+
+// handler kinds (handler table rows: [kind, arg_off, arg_len, status])
+#define HK_HOST 0
+#define HK_ECHO_JSON 1
+#define HK_STATIC 2
+
+// methods — must match gofr_amd/http/request.py METHOD_IDS
+// GET POST PUT DELETE PATCH OPTIONS HEAD
+#define M_GET 0
+#define M_POST 1
+#define M_PUT 2
+#define M_DELETE 3
+#define M_PATCH 4
+#define M_OPTIONS 5
+#define M_HEAD 6
+
+#define N_METHODS_PAD 8
+#define MAX_PARAMS 4
+
+struct TrieDev {
+    const uint8_t* seg_blob;
+    const int32_t* node_child_first;
+    const int32_t* node_child_count;
+    const int32_t* child_seg_off;
+    const int32_t* child_seg_len;
+    const int32_t* child_node;
+    const int32_t* node_param;
+    const int32_t* node_prefix;
+    const int32_t* node_route;
+};
+
+// ---------------------------------------------------------------------------
+// helpers
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int lane_id() { return threadIdx.x & (WAVE - 1); }
+
+__device__ __forceinline__ uint8_t lower(uint8_t c) {
+    return (c >= 'A' && c <= 'Z') ? (c | 0x20) : c;
+}
+
+// case-insensitive compare of req bytes against lowercase literal
+__device__ __forceinline__ bool ieq(const uint8_t* p, const char* lit, int n) {
+    for (int i = 0; i < n; ++i)
+        if (lower(p[i]) != (uint8_t)lit[i]) return false;
+    return true;
+}
+
+__device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+// ---------------------------------------------------------------------------
+// k_parse_route
+// ---------------------------------------------------------------------------
+// classes packed in one LDS array (single __shared__ object)
+#define CLS_LF 0
+#define CLS_SP 1
+#define CLS_COLON 2
+#define CLS_QM 3
+#define N_CLS 4
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_parse_route(const uint8_t* __restrict__ reqs,
+              const int32_t* __restrict__ req_len,
+              int32_t* __restrict__ fields,
+              int n, int slot,
+              TrieDev trie,
+              const int32_t* __restrict__ handler_tab, int n_routes,
+              int32_t* __restrict__ host_needed) {
+    __shared__ uint64_t masks[WAVES_PER_BLOCK][N_CLS][MAX_CHUNKS];
+
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    const int lane = lane_id();
+    const uint8_t* base = reqs + (size_t)req * slot;
+    const int len = req_len[req];
+    int32_t* F = fields + (size_t)req * NF;
+
+    const int nchunks = (len + WAVE - 1) / WAVE;
+    bool saw_percent = false;
+    // structural classification: 64 bytes per wave-iteration, one ballot
+    // per class — no per-byte branching (SURVEY.md §7 hard part 2)
+    for (int c = 0; c < nchunks; ++c) {
+        const int i = c * WAVE + lane;
+        const uint8_t b = (i < len) ? base[i] : 0;
+        const uint64_t m_lf = __ballot(b == '\n');
+        const uint64_t m_sp = __ballot(b == ' ');
+        const uint64_t m_co = __ballot(b == ':');
+        const uint64_t m_qm = __ballot(b == '?');
+        saw_percent |= (b == '%');
+        if (lane == 0) {
+            masks[wv][CLS_LF][c] = m_lf;
+            masks[wv][CLS_SP][c] = m_sp;
+            masks[wv][CLS_COLON][c] = m_co;
+            masks[wv][CLS_QM][c] = m_qm;
+        }
+    }
+    const bool any_percent = __ballot(saw_percent) != 0;
+    // no cross-wave sharing: wave-local LDS, no barrier needed (same wave
+    // wrote and reads; LDS ops from one wave are ordered)
+
+    if (lane != 0) return;
+
+    // ---- lane 0: tokenize by walking the masks -----------------------------
+    auto next_bit = [&](int cls, int from) -> int {
+        int c = from / WAVE;
+        if (c >= nchunks) return -1;
+        uint64_t m = masks[wv][cls][c] & (~0ull << (from & (WAVE - 1)));
+        while (true) {
+            if (m) return c * WAVE + __builtin_ctzll(m);
+            if (++c >= nchunks) return -1;
+            m = masks[wv][cls][c];
+        }
+    };
+
+    int32_t flags = 0;
+    if (any_percent) flags |= FL_NEEDS_HOST;
+
+    // request line: METHOD SP target SP version CRLF
+    const int sp1 = next_bit(CLS_SP, 0);
+    const int lf1 = next_bit(CLS_LF, 0);
+    if (sp1 < 0 || lf1 < 0 || sp1 >= lf1) {
+        F[FI_FLAGS] = FL_ERR_PARSE;
+        F[FI_METHOD] = -1;
+        F[FI_KIND] = HK_HOST;
+        F[FI_STATUS] = 400;
+        F[FI_ROUTE] = -1;
+        atomicAdd(host_needed, 1);
+        return;
+    }
+    // method id from leading bytes
+    int method = -1;
+    {
+        static const char* MNAMES[7] = {"GET", "POST", "PUT", "DELETE",
+                                        "PATCH", "OPTIONS", "HEAD"};
+        static const int MLENS[7] = {3, 4, 3, 6, 5, 7, 4};
+        for (int mi = 0; mi < 7; ++mi) {
+            if (MLENS[mi] != sp1) continue;
+            bool eq = true;
+            for (int k = 0; k < sp1; ++k)
+                if (base[k] != (uint8_t)MNAMES[mi][k]) { eq = false; break; }
+            if (eq) { method = mi; break; }
+        }
+    }
+    const int sp2 = next_bit(CLS_SP, sp1 + 1);
+    int target_end = (sp2 > 0 && sp2 < lf1) ? sp2 : (lf1 - 1); // before CR
+    int path_off = sp1 + 1;
+    int qm = next_bit(CLS_QM, path_off);
+    int path_end = (qm >= 0 && qm < target_end) ? qm : target_end;
+    int query_off = (qm >= 0 && qm < target_end) ? qm + 1 : target_end;
+    int query_len = (qm >= 0 && qm < target_end) ? target_end - query_off : 0;
+
+    F[FI_METHOD] = method;
+    F[FI_PATH_OFF] = path_off;
+    F[FI_PATH_LEN] = path_end - path_off;
+    F[FI_QUERY_OFF] = query_off;
+    F[FI_QUERY_LEN] = query_len;
+    if (method == M_OPTIONS) flags |= FL_IS_OPTIONS;
+    if (method < 0) flags |= FL_ERR_PARSE;
+
+    // ---- headers ----------------------------------------------------------
+    int clen = 0;
+    bool keep_alive = true;  // HTTP/1.1 default
+    int body_off = len, auth_off = 0, auth_len = 0;
+    int prev_lf = lf1;
+    for (int guard = 0; guard < 64; ++guard) {
+        const int ls = prev_lf + 1;           // line start
+        const int lf = next_bit(CLS_LF, ls);
+        if (lf < 0) { flags |= FL_ERR_PARSE; break; }
+        const int le = lf - 1;                // position of CR (line end excl)
+        if (le <= ls) {                       // empty line -> end of headers
+            body_off = lf + 1;
+            break;
+        }
+        const int co = next_bit(CLS_COLON, ls);
+        if (co < 0 || co >= le) { prev_lf = lf; continue; }
+        const int nlen = co - ls;
+        int vs = co + 1;
+        while (vs < le && base[vs] == ' ') ++vs;
+        const int vlen = le - vs;
+        const uint8_t* nm = base + ls;
+        if (nlen == 14 && ieq(nm, "content-length", 14)) {
+            int v = 0;
+            for (int i = 0; i < vlen; ++i) {
+                uint8_t d = base[vs + i];
+                if (d >= '0' && d <= '9') v = v * 10 + (d - '0');
+            }
+            clen = v;
+        } else if (nlen == 10 && ieq(nm, "connection", 10)) {
+            if (vlen == 5 && ieq(base + vs, "close", 5)) keep_alive = false;
+        } else if (nlen == 12 && ieq(nm, "content-type", 12)) {
+            if (vlen >= 16 && ieq(base + vs, "application/json", 16))
+                flags |= FL_JSON_CT;
+        } else if (nlen == 13 && ieq(nm, "authorization", 13)) {
+            auth_off = vs; auth_len = vlen;
+        } else if (nlen == 17 && ieq(nm, "transfer-encoding", 17)) {
+            flags |= FL_NEEDS_HOST;  // chunked -> host slow path
+        }
+        prev_lf = lf;
+    }
+    int body_len = len - body_off;
+    if (body_len < 0) body_len = 0;
+    if (clen > 0 && clen < body_len) body_len = clen;
+    if (keep_alive) flags |= FL_KEEP_ALIVE;
+
+    F[FI_BODY_OFF] = body_off;
+    F[FI_BODY_LEN] = body_len;
+    F[FI_CLEN] = clen;
+    F[FI_AUTH_OFF] = auth_off;
+    F[FI_AUTH_LEN] = auth_len;
+
+    // ---- route match: trie walk (gofr_amd/http/router.py compile layout) --
+    int node = 0;
+    int best_prefix = trie.node_prefix[0];
+    int nparams = 0;
+    int poffs[MAX_PARAMS], plens[MAX_PARAMS];
+    {
+        int pos = path_off;
+        const int pend = path_off + F[FI_PATH_LEN];
+        bool dead = false;
+        while (pos < pend && !dead) {
+            while (pos < pend && base[pos] == '/') ++pos;
+            if (pos >= pend) break;
+            int se = pos;
+            while (se < pend && base[se] != '/') ++se;
+            const int slen = se - pos;
+            // literal children
+            int nxt = -1;
+            const int cf = trie.node_child_first[node];
+            const int cc = trie.node_child_count[node];
+            for (int ci = cf; ci < cf + cc; ++ci) {
+                if (trie.child_seg_len[ci] != slen) continue;
+                const uint8_t* sb = trie.seg_blob + trie.child_seg_off[ci];
+                bool eq = true;
+                for (int k = 0; k < slen; ++k)
+                    if (sb[k] != base[pos + k]) { eq = false; break; }
+                if (eq) { nxt = trie.child_node[ci]; break; }
+            }
+            if (nxt < 0) {
+                const int pc = trie.node_param[node];
+                if (pc >= 0) {
+                    if (nparams < MAX_PARAMS) {
+                        poffs[nparams] = pos;
+                        plens[nparams] = slen;
+                        ++nparams;
+                    } else {
+                        flags |= FL_NEEDS_HOST;  // >4 params: host path
+                    }
+                    nxt = pc;
+                }
+            }
+            if (nxt < 0) { dead = true; break; }
+            node = nxt;
+            if (trie.node_prefix[node] >= 0)
+                best_prefix = trie.node_prefix[node];
+            pos = se;
+        }
+        int route = -1;
+        if (!dead) {
+            const int m = (method >= 0) ? method : 0;
+            route = trie.node_route[node * N_METHODS_PAD + m];
+        }
+        if (route < 0) route = best_prefix;
+        F[FI_ROUTE] = route;
+        for (int k = 0; k < MAX_PARAMS; ++k) {
+            F[FI_PARAM0 + 2 * k] = (k < nparams) ? poffs[k] : 0;
+            F[FI_PARAM0 + 2 * k + 1] = (k < nparams) ? plens[k] : 0;
+        }
+        int kind = HK_STATIC;  // catch-all default resolved by host table
+        int status = 404;
+        if (route >= 0 && route < n_routes) {
+            kind = handler_tab[route * 4 + 0];
+            status = handler_tab[route * 4 + 3];
+        }
+        if (flags & FL_IS_OPTIONS) { kind = HK_STATIC; status = 200; }
+        if (flags & (FL_ERR_PARSE | FL_NEEDS_HOST)) kind = HK_HOST;
+        F[FI_KIND] = kind;
+        F[FI_STATUS] = status;
+        if (kind == HK_HOST) atomicAdd(host_needed, 1);
+    }
+    F[FI_FLAGS] = flags;
+}
+
+// ---------------------------------------------------------------------------
+// k_respond — fused handler + serializer
+// ---------------------------------------------------------------------------
+
+__constant__ char HDR_P1[] = "HTTP/1.1 ";
+// after status: " OK\r\n" etc from reason table (blob)
+__constant__ char HDR_CT_JSON[] = "Content-Type: application/json\r\n";
+__constant__ char HDR_CT_ICON[] = "Content-Type: image/x-icon\r\n";
+__constant__ char HDR_CT_BIN[]  = "Content-Type: application/octet-stream\r\n";
+__constant__ char HDR_CT_TXT[]  = "Content-Type: text/plain\r\n";
+__constant__ char HDR_CORS[] =
+    "Access-Control-Allow-Origin: *\r\n"
+    "Access-Control-Allow-Methods: POST, GET, OPTIONS, PUT, DELETE\r\n";
+__constant__ char HDR_CORR[] = "X-Correlation-ID: ";
+__constant__ char HDR_CL[] = "Content-Length: ";
+__constant__ char HDR_CONN_KA[] = "Connection: keep-alive\r\n\r\n";
+__constant__ char HDR_CONN_CL[] = "Connection: close\r\n\r\n";
+__constant__ char ENV_OPEN[] = "{\"data\":";
+__constant__ char HEXD[] = "0123456789abcdef";
+
+// reasons — indexed by status class (host passes exact bytes in blob for
+// anything unusual); we inline the four the engine emits
+__device__ __forceinline__ const char* reason_of(int st, int* rlen) {
+    switch (st) {
+        case 200: *rlen = 2; return "OK";
+        case 404: *rlen = 9; return "Not Found";
+        case 405: *rlen = 18; return "Method Not Allowed";
+        case 401: *rlen = 12; return "Unauthorized";
+        default:  *rlen = 21; return "Internal Server Error";
+    }
+}
+
+__device__ __forceinline__ int itoa10(uint8_t* dst, int v) {
+    char tmp[10];
+    int n = 0;
+    if (v == 0) { dst[0] = '0'; return 1; }
+    while (v > 0) { tmp[n++] = '0' + (v % 10); v /= 10; }
+    for (int i = 0; i < n; ++i) dst[i] = tmp[n - 1 - i];
+    return n;
+}
+
+// host result table row: [off, len, status, ct_or_flags]
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_respond(const uint8_t* __restrict__ reqs,
+          int32_t* __restrict__ fields,
+          uint8_t* __restrict__ resp,
+          int32_t* __restrict__ resp_len_out,
+          int n, int slot, int rslot,
+          const int32_t* __restrict__ handler_tab, int n_routes,
+          const uint8_t* __restrict__ blob,
+          const uint8_t* __restrict__ host_blob,
+          const int32_t* __restrict__ host_tab,
+          uint64_t seed) {
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    const int lane = lane_id();
+    int32_t* F = fields + (size_t)req * NF;
+    uint8_t* out = resp + (size_t)req * rslot;
+    const uint8_t* rbase = reqs + (size_t)req * slot;
+
+    const int kind = F[FI_KIND];
+    int status = F[FI_STATUS];
+    const int flags = F[FI_FLAGS];
+    const bool keep = (flags & FL_KEEP_ALIVE) != 0;
+    const bool is_options = (flags & FL_IS_OPTIONS) != 0;
+
+    // ---- resolve body source + length --------------------------------------
+    // body = envelope_open + payload + envelope_close for ECHO; raw blob for
+    // STATIC (payload already enveloped); host bytes for HOST
+    const uint8_t* body_src = nullptr;
+    int body_src_len = 0;
+    int env = 0;  // 1 -> wrap in {"data": ... }
+    if (is_options) {
+        body_src_len = 0;
+    } else if (kind == HK_ECHO_JSON) {
+        body_src = rbase + F[FI_BODY_OFF];
+        body_src_len = F[FI_BODY_LEN];
+        env = 1;
+        // minimal structural validation: brace/bracket balance + quote
+        // parity, escapes honored (lane-serial over short bodies would
+        // diverge; all lanes vote instead)
+        int depth = 0;
+        bool bad = false;
+        if (lane == 0) {
+            bool in_str = false;
+            for (int i = 0; i < body_src_len; ++i) {
+                uint8_t c = body_src[i];
+                if (in_str) {
+                    if (c == '\\') { ++i; }
+                    else if (c == '"') in_str = false;
+                } else if (c == '"') in_str = true;
+                else if (c == '{' || c == '[') ++depth;
+                else if (c == '}' || c == ']') --depth;
+                if (depth < 0) { bad = true; break; }
+            }
+            bad = bad || depth != 0 || in_str || body_src_len == 0;
+        }
+        bad = __shfl(bad ? 1 : 0, 0) != 0;
+        if (bad) {
+            status = 500;
+            // error envelope from blob: handler_tab row arg points at the
+            // route's static arg; the invalid-body envelope is blob[0..]
+            // convention: first 4 bytes of blob = len of invalid-body env
+            const int elen = *(const int32_t*)blob;
+            body_src = blob + 4;
+            body_src_len = elen;
+            env = 0;
+        }
+    } else if (kind == HK_STATIC) {
+        // host guarantees route >= 0 via the catch-all prefix route; an
+        // unresolved route (defaults not installed) serves an empty body
+        // with the parse-time status (404)
+        const int route = F[FI_ROUTE];
+        if (route >= 0 && route < n_routes) {
+            body_src = blob + handler_tab[route * 4 + 1];
+            body_src_len = handler_tab[route * 4 + 2];
+        }
+    }
+    int ct_id = 0;  // 0 json, 1 icon, 2 octet-stream, 3 text/plain
+    if (kind == HK_HOST && !is_options) {
+        body_src = host_blob + host_tab[req * 4 + 0];
+        body_src_len = host_tab[req * 4 + 1];
+        status = host_tab[req * 4 + 2];
+        ct_id = host_tab[req * 4 + 3];
+    }
+
+    const int body_total = body_src_len + (env ? 9 : 0);  // {"data": + }
+
+    // ---- header length arithmetic ------------------------------------------
+    int rlen_reason;
+    const char* reason = reason_of(status, &rlen_reason);
+    // HTTP/1.1 [9] + 3 + 1 + reason + 2
+    int hl = 9 + 3 + 1 + rlen_reason + 2;
+    const char* ct_str = HDR_CT_JSON;
+    int ct_len = sizeof(HDR_CT_JSON) - 1;
+    if (ct_id == 1) { ct_str = HDR_CT_ICON; ct_len = sizeof(HDR_CT_ICON) - 1; }
+    else if (ct_id == 2) { ct_str = HDR_CT_BIN; ct_len = sizeof(HDR_CT_BIN) - 1; }
+    else if (ct_id == 3) { ct_str = HDR_CT_TXT; ct_len = sizeof(HDR_CT_TXT) - 1; }
+    hl += ct_len;
+    hl += sizeof(HDR_CORS) - 1;
+    hl += sizeof(HDR_CORR) - 1 + 32 + 2;
+    int cl_digits = 1;
+    for (int v = body_total; v >= 10; v /= 10) ++cl_digits;
+    hl += sizeof(HDR_CL) - 1 + cl_digits + 2;
+    hl += (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
+
+    // ---- parallel body copy (all lanes) ------------------------------------
+    const int body_start = hl + (env ? 8 : 0);
+    for (int i = lane; i < body_src_len; i += WAVE)
+        out[body_start + i] = body_src[i];
+
+    // ---- lane 0: header + envelope + trailer -------------------------------
+    if (lane == 0) {
+        uint8_t* p = out;
+        for (int i = 0; i < 9; ++i) *p++ = HDR_P1[i];
+        *p++ = '0' + (status / 100);
+        *p++ = '0' + ((status / 10) % 10);
+        *p++ = '0' + (status % 10);
+        *p++ = ' ';
+        for (int i = 0; i < rlen_reason; ++i) *p++ = reason[i];
+        *p++ = '\r'; *p++ = '\n';
+        for (int i = 0; i < ct_len; ++i) *p++ = ct_str[i];
+        for (int i = 0; i < (int)sizeof(HDR_CORS) - 1; ++i) *p++ = HDR_CORS[i];
+        for (int i = 0; i < (int)sizeof(HDR_CORR) - 1; ++i) *p++ = HDR_CORR[i];
+        // correlation id: 128-bit from splitmix64 (fused id+hex — the
+        // reference takes it from the OTel span, middleware/logger.go:46)
+        uint64_t h1 = splitmix64(seed ^ (uint64_t)req);
+        uint64_t h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03ull);
+        for (int i = 0; i < 16; ++i)
+            *p++ = HEXD[(h1 >> (60 - 4 * i)) & 0xF];
+        for (int i = 0; i < 16; ++i)
+            *p++ = HEXD[(h2 >> (60 - 4 * i)) & 0xF];
+        *p++ = '\r'; *p++ = '\n';
+        for (int i = 0; i < (int)sizeof(HDR_CL) - 1; ++i) *p++ = HDR_CL[i];
+        p += itoa10(p, body_total);
+        *p++ = '\r'; *p++ = '\n';
+        const char* conn = keep ? HDR_CONN_KA : HDR_CONN_CL;
+        const int conn_len = (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
+        for (int i = 0; i < conn_len; ++i) *p++ = conn[i];
+        // assert header arithmetic matched
+        // (p - out) == hl by construction
+        if (env) {
+            for (int i = 0; i < 8; ++i) out[hl + i] = ENV_OPEN[i];
+            out[hl + 8 + body_src_len] = '}';
+        }
+        const int total = hl + body_total;
+        F[FI_RESP_LEN] = total;
+        F[FI_RESP_OFF] = (int)((size_t)req * rslot);
+        resp_len_out[req] = total;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// C API (ctypes-driven; raw device pointers from torch tensors)
+// ---------------------------------------------------------------------------
+extern "C" {
+
+int gofr_launch_parse_route(
+        void* stream,
+        const void* reqs, const void* req_len, void* fields,
+        int n, int slot,
+        const void* seg_blob, const void* node_child_first,
+        const void* node_child_count, const void* child_seg_off,
+        const void* child_seg_len, const void* child_node,
+        const void* node_param, const void* node_prefix,
+        const void* node_route,
+        const void* handler_tab, int n_routes,
+        void* host_needed) {
+    TrieDev trie{(const uint8_t*)seg_blob, (const int32_t*)node_child_first,
+                 (const int32_t*)node_child_count, (const int32_t*)child_seg_off,
+                 (const int32_t*)child_seg_len, (const int32_t*)child_node,
+                 (const int32_t*)node_param, (const int32_t*)node_prefix,
+                 (const int32_t*)node_route};
+    const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    hipLaunchKernelGGL(k_parse_route, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)reqs, (const int32_t*)req_len,
+                       (int32_t*)fields, n, slot, trie,
+                       (const int32_t*)handler_tab, n_routes,
+                       (int32_t*)host_needed);
+    return (int)hipGetLastError();
+}
+
+int gofr_launch_respond(
+        void* stream,
+        const void* reqs, void* fields, void* resp, void* resp_len_out,
+        int n, int slot, int rslot,
+        const void* handler_tab, int n_routes,
+        const void* blob, const void* host_blob, const void* host_tab,
+        uint64_t seed) {
+    const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)reqs, (int32_t*)fields,
+                       (uint8_t*)resp, (int32_t*)resp_len_out,
+                       n, slot, rslot,
+                       (const int32_t*)handler_tab, n_routes,
+                       (const uint8_t*)blob, (const uint8_t*)host_blob,
+                       (const int32_t*)host_tab, seed);
+    return (int)hipGetLastError();
+}
+
+}  // extern "C"

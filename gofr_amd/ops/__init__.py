@@ -1,0 +1,414 @@
+"""Device op bindings + CPU golden mirrors.
+
+`HipOps` drives the gfx950 kernels in gofr_amd/_gofr_hip.so (built from
+native/hip/gofr_kernels.hip) through ctypes on raw torch device pointers.
+`cpu_parse_route` / `cpu_respond` are byte-exact Python mirrors of the
+kernels: the GPU tests (tests/test_gpu_engine.py) assert kernel output ==
+mirror output on the same buffers, and the engine uses the mirrors as its
+CPU fallback so every code path runs (slowly) without a GPU.
+
+On a GPU box the extension is REQUIRED: HipOps raises if the .so is
+missing rather than silently falling back (driver policy: native code
+must be what actually runs).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+
+import numpy as np
+
+# ---- field table layout (mirror of gofr_kernels.hip) ------------------------
+FI_METHOD = 0
+FI_PATH_OFF = 1
+FI_PATH_LEN = 2
+FI_QUERY_OFF = 3
+FI_QUERY_LEN = 4
+FI_BODY_OFF = 5
+FI_BODY_LEN = 6
+FI_CLEN = 7
+FI_FLAGS = 8
+FI_ROUTE = 9
+FI_KIND = 10
+FI_STATUS = 11
+FI_RESP_LEN = 12
+FI_AUTH_OFF = 13
+FI_AUTH_LEN = 14
+FI_RESP_OFF = 15
+FI_PARAM0 = 16
+NF = 24
+
+FL_ERR_PARSE = 1
+FL_NEEDS_HOST = 2
+FL_KEEP_ALIVE = 4
+FL_JSON_CT = 8
+FL_IS_OPTIONS = 16
+
+HK_HOST = 0
+HK_ECHO_JSON = 1
+HK_STATIC = 2
+
+MAX_PARAMS = 4
+N_METHODS_PAD = 8
+
+_METHOD_NAMES = ["GET", "POST", "PUT", "DELETE", "PATCH", "OPTIONS", "HEAD"]
+
+_SO_PATH = os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "_gofr_hip.so")
+
+
+class HipOps:
+    """ctypes driver for the device kernels."""
+
+    def __init__(self, so_path: str = _SO_PATH):
+        if not os.path.exists(so_path):
+            raise FileNotFoundError(
+                f"HIP extension not built: {so_path}. Run "
+                f"`python setup.py build_hip` or __graft_entry__.build().")
+        self.lib = ctypes.CDLL(so_path)
+        self.lib.gofr_launch_parse_route.restype = ctypes.c_int
+        self.lib.gofr_launch_parse_route.argtypes = [ctypes.c_void_p] * 3 + [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int] + \
+            [ctypes.c_void_p] * 9 + [ctypes.c_void_p, ctypes.c_int,
+                                     ctypes.c_void_p]
+        self.lib.gofr_launch_respond.restype = ctypes.c_int
+        self.lib.gofr_launch_respond.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_void_p, ctypes.c_int,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_uint64]
+
+    def parse_route(self, stream, reqs_t, req_len_t, fields_t, n, slot,
+                    trie_t: dict, handler_tab_t, n_routes, host_needed_t):
+        rc = self.lib.gofr_launch_parse_route(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(reqs_t.data_ptr()),
+            ctypes.c_void_p(req_len_t.data_ptr()),
+            ctypes.c_void_p(fields_t.data_ptr()),
+            n, slot,
+            ctypes.c_void_p(trie_t["seg_blob"].data_ptr()),
+            ctypes.c_void_p(trie_t["node_child_first"].data_ptr()),
+            ctypes.c_void_p(trie_t["node_child_count"].data_ptr()),
+            ctypes.c_void_p(trie_t["child_seg_off"].data_ptr()),
+            ctypes.c_void_p(trie_t["child_seg_len"].data_ptr()),
+            ctypes.c_void_p(trie_t["child_node"].data_ptr()),
+            ctypes.c_void_p(trie_t["node_param"].data_ptr()),
+            ctypes.c_void_p(trie_t["node_prefix"].data_ptr()),
+            ctypes.c_void_p(trie_t["node_route"].data_ptr()),
+            ctypes.c_void_p(handler_tab_t.data_ptr()), n_routes,
+            ctypes.c_void_p(host_needed_t.data_ptr()))
+        if rc != 0:
+            raise RuntimeError(f"k_parse_route launch failed: hipError {rc}")
+
+    def respond(self, stream, reqs_t, fields_t, resp_t, resp_len_t, n, slot,
+                rslot, handler_tab_t, n_routes, blob_t, host_blob_t,
+                host_tab_t, seed):
+        rc = self.lib.gofr_launch_respond(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(reqs_t.data_ptr()),
+            ctypes.c_void_p(fields_t.data_ptr()),
+            ctypes.c_void_p(resp_t.data_ptr()),
+            ctypes.c_void_p(resp_len_t.data_ptr()),
+            n, slot, rslot,
+            ctypes.c_void_p(handler_tab_t.data_ptr()), n_routes,
+            ctypes.c_void_p(blob_t.data_ptr()),
+            ctypes.c_void_p(host_blob_t.data_ptr()),
+            ctypes.c_void_p(host_tab_t.data_ptr()),
+            ctypes.c_uint64(seed))
+        if rc != 0:
+            raise RuntimeError(f"k_respond launch failed: hipError {rc}")
+
+
+# ---------------------------------------------------------------------------
+# CPU golden mirrors (byte-exact models of the kernels)
+# ---------------------------------------------------------------------------
+
+M64 = (1 << 64) - 1
+
+
+def splitmix64(x: int) -> int:
+    x = (x + 0x9E3779B97F4A7C15) & M64
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & M64
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & M64
+    return x ^ (x >> 31)
+
+
+def _lower(c: int) -> int:
+    return c | 0x20 if ord("A") <= c <= ord("Z") else c
+
+
+def _ieq(buf, off, lit: bytes) -> bool:
+    return all(_lower(buf[off + i]) == lit[i] for i in range(len(lit)))
+
+
+def cpu_parse_route(reqs: np.ndarray, req_len: np.ndarray, slot: int,
+                    trie: dict, handler_tab: np.ndarray) -> np.ndarray:
+    """Mirror of k_parse_route over a [n*slot] uint8 buffer."""
+    n = len(req_len)
+    fields = np.zeros((n, NF), np.int32)
+    n_routes = len(handler_tab) // 4
+    for r in range(n):
+        base = r * slot
+        ln = int(req_len[r])
+        buf = reqs[base:base + ln].tobytes()
+        F = fields[r]
+        flags = 0
+        if b"%" in buf:
+            flags |= FL_NEEDS_HOST
+        sp1 = buf.find(b" ")
+        lf1 = buf.find(b"\n")
+        if sp1 < 0 or lf1 < 0 or sp1 >= lf1:
+            F[FI_FLAGS] = FL_ERR_PARSE
+            F[FI_METHOD] = -1
+            F[FI_KIND] = HK_HOST
+            F[FI_STATUS] = 400
+            F[FI_ROUTE] = -1
+            continue
+        method = -1
+        head = buf[:sp1]
+        for mi, mn in enumerate(_METHOD_NAMES):
+            if head == mn.encode():
+                method = mi
+                break
+        sp2 = buf.find(b" ", sp1 + 1)
+        target_end = sp2 if (0 < sp2 < lf1) else lf1 - 1
+        path_off = sp1 + 1
+        qm = buf.find(b"?", path_off)
+        path_end = qm if (0 <= qm < target_end) else target_end
+        query_off = qm + 1 if (0 <= qm < target_end) else target_end
+        query_len = target_end - query_off if (0 <= qm < target_end) else 0
+        F[FI_METHOD] = method
+        F[FI_PATH_OFF] = path_off
+        F[FI_PATH_LEN] = path_end - path_off
+        F[FI_QUERY_OFF] = query_off
+        F[FI_QUERY_LEN] = query_len
+        if method == 5:  # OPTIONS
+            flags |= FL_IS_OPTIONS
+        if method < 0:
+            flags |= FL_ERR_PARSE
+
+        clen = 0
+        keep_alive = True
+        body_off = ln
+        auth_off = auth_len = 0
+        prev_lf = lf1
+        for _ in range(64):
+            ls = prev_lf + 1
+            lf = buf.find(b"\n", ls)
+            if lf < 0:
+                flags |= FL_ERR_PARSE
+                break
+            le = lf - 1
+            if le <= ls:
+                body_off = lf + 1
+                break
+            co = buf.find(b":", ls)
+            if co < 0 or co >= le:
+                prev_lf = lf
+                continue
+            nlen = co - ls
+            vs = co + 1
+            while vs < le and buf[vs] == 0x20:
+                vs += 1
+            vlen = le - vs
+            if nlen == 14 and _ieq(buf, ls, b"content-length"):
+                v = 0
+                for i in range(vlen):
+                    d = buf[vs + i]
+                    if 0x30 <= d <= 0x39:
+                        v = v * 10 + d - 0x30
+                clen = v
+            elif nlen == 10 and _ieq(buf, ls, b"connection"):
+                if vlen == 5 and _ieq(buf, vs, b"close"):
+                    keep_alive = False
+            elif nlen == 12 and _ieq(buf, ls, b"content-type"):
+                if vlen >= 16 and _ieq(buf, vs, b"application/json"):
+                    flags |= FL_JSON_CT
+            elif nlen == 13 and _ieq(buf, ls, b"authorization"):
+                auth_off, auth_len = vs, vlen
+            elif nlen == 17 and _ieq(buf, ls, b"transfer-encoding"):
+                flags |= FL_NEEDS_HOST
+            prev_lf = lf
+        body_len = max(0, ln - body_off)
+        if 0 < clen < body_len:
+            body_len = clen
+        if keep_alive:
+            flags |= FL_KEEP_ALIVE
+        F[FI_BODY_OFF] = body_off
+        F[FI_BODY_LEN] = body_len
+        F[FI_CLEN] = clen
+        F[FI_AUTH_OFF] = auth_off
+        F[FI_AUTH_LEN] = auth_len
+
+        # trie walk
+        node = 0
+        best_prefix = int(trie["node_prefix"][0])
+        nparams = 0
+        pspans = []
+        pos, pend = path_off, path_off + int(F[FI_PATH_LEN])
+        dead = False
+        while pos < pend and not dead:
+            while pos < pend and buf[pos] == ord("/"):
+                pos += 1
+            if pos >= pend:
+                break
+            se = pos
+            while se < pend and buf[se] != ord("/"):
+                se += 1
+            slen = se - pos
+            nxt = -1
+            cf = int(trie["node_child_first"][node])
+            cc = int(trie["node_child_count"][node])
+            for ci in range(cf, cf + cc):
+                if int(trie["child_seg_len"][ci]) != slen:
+                    continue
+                so = int(trie["child_seg_off"][ci])
+                if bytes(trie["seg_blob"][so:so + slen]) == buf[pos:se]:
+                    nxt = int(trie["child_node"][ci])
+                    break
+            if nxt < 0:
+                pc = int(trie["node_param"][node])
+                if pc >= 0:
+                    if nparams < MAX_PARAMS:
+                        pspans.append((pos, slen))
+                        nparams += 1
+                    else:
+                        flags |= FL_NEEDS_HOST
+                    nxt = pc
+            if nxt < 0:
+                dead = True
+                break
+            node = nxt
+            if int(trie["node_prefix"][node]) >= 0:
+                best_prefix = int(trie["node_prefix"][node])
+            pos = se
+        route = -1
+        if not dead:
+            m = method if method >= 0 else 0
+            route = int(trie["node_route"][node * N_METHODS_PAD + m])
+        if route < 0:
+            route = best_prefix
+        F[FI_ROUTE] = route
+        for k in range(MAX_PARAMS):
+            if k < len(pspans):
+                F[FI_PARAM0 + 2 * k] = pspans[k][0]
+                F[FI_PARAM0 + 2 * k + 1] = pspans[k][1]
+        kind, status = HK_STATIC, 404
+        if 0 <= route < n_routes:
+            kind = int(handler_tab[route * 4 + 0])
+            status = int(handler_tab[route * 4 + 3])
+        if flags & FL_IS_OPTIONS:
+            kind, status = HK_STATIC, 200
+        if flags & (FL_ERR_PARSE | FL_NEEDS_HOST):
+            kind = HK_HOST
+        F[FI_KIND] = kind
+        F[FI_STATUS] = status
+        F[FI_FLAGS] = flags
+    return fields
+
+
+_REASONS_DEV = {200: b"OK", 404: b"Not Found", 405: b"Method Not Allowed",
+                401: b"Unauthorized"}
+_CT_STRS = [b"Content-Type: application/json\r\n",
+            b"Content-Type: image/x-icon\r\n",
+            b"Content-Type: application/octet-stream\r\n",
+            b"Content-Type: text/plain\r\n"]
+_CORS = (b"Access-Control-Allow-Origin: *\r\n"
+         b"Access-Control-Allow-Methods: POST, GET, OPTIONS, PUT, DELETE\r\n")
+
+
+def _dev_reason(st: int) -> bytes:
+    return _REASONS_DEV.get(st, b"Internal Server Error")
+
+
+def _json_body_valid(body: bytes) -> bool:
+    """Mirror of the kernel's structural validation (NOT full JSON)."""
+    depth = 0
+    in_str = False
+    i = 0
+    while i < len(body):
+        c = body[i:i + 1]
+        if in_str:
+            if c == b"\\":
+                i += 1
+            elif c == b'"':
+                in_str = False
+        elif c == b'"':
+            in_str = True
+        elif c in (b"{", b"["):
+            depth += 1
+        elif c in (b"}", b"]"):
+            depth -= 1
+            if depth < 0:
+                return False
+        i += 1
+    return depth == 0 and not in_str and len(body) > 0
+
+
+def cpu_respond(reqs: np.ndarray, fields: np.ndarray, slot: int, rslot: int,
+                handler_tab: np.ndarray, blob: bytes,
+                host_blob: bytes, host_tab: np.ndarray,
+                seed: int):
+    """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
+    n = len(fields)
+    host_tab = np.asarray(host_tab, np.int32).reshape(-1)
+    n_routes = len(handler_tab) // 4
+    resp = np.zeros(n * rslot, np.uint8)
+    resp_len = np.zeros(n, np.int32)
+    for r in range(n):
+        F = fields[r]
+        base = r * slot
+        kind = int(F[FI_KIND])
+        status = int(F[FI_STATUS])
+        flags = int(F[FI_FLAGS])
+        keep = bool(flags & FL_KEEP_ALIVE)
+        is_options = bool(flags & FL_IS_OPTIONS)
+        body_src = b""
+        env = False
+        ct_id = 0
+        if is_options:
+            pass
+        elif kind == HK_ECHO_JSON:
+            bo, bl = int(F[FI_BODY_OFF]), int(F[FI_BODY_LEN])
+            body_src = reqs[base + bo:base + bo + bl].tobytes()
+            env = True
+            if not _json_body_valid(body_src):
+                status = 500
+                elen = int.from_bytes(blob[:4], "little")
+                body_src = blob[4:4 + elen]
+                env = False
+        elif kind == HK_STATIC:
+            route = int(F[FI_ROUTE])
+            if 0 <= route < n_routes:
+                off = int(handler_tab[route * 4 + 1])
+                ln = int(handler_tab[route * 4 + 2])
+                body_src = blob[off:off + ln]
+        if kind == HK_HOST and not is_options:
+            off, ln, status, ct_id = (int(host_tab[r * 4 + i])
+                                      for i in range(4))
+            body_src = host_blob[off:off + ln]
+
+        body_total = len(body_src) + (9 if env else 0)
+        reason = _dev_reason(status)
+        h1 = splitmix64(seed ^ r)
+        h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03)
+        corr = f"{h1:016x}{h2:016x}".encode()
+        head = (b"HTTP/1.1 " + f"{status:03d}".encode() + b" " + reason +
+                b"\r\n" + _CT_STRS[ct_id] + _CORS +
+                b"X-Correlation-ID: " + corr + b"\r\n" +
+                b"Content-Length: " + str(body_total).encode() + b"\r\n" +
+                (b"Connection: keep-alive\r\n\r\n" if keep
+                 else b"Connection: close\r\n\r\n"))
+        if env:
+            payload = head + b'{"data":' + body_src + b"}"
+        else:
+            payload = head + body_src
+        resp[r * rslot:r * rslot + len(payload)] = np.frombuffer(
+            payload, np.uint8)
+        resp_len[r] = len(payload)
+        fields[r][FI_RESP_LEN] = len(payload)
+        fields[r][FI_RESP_OFF] = r * rslot
+    return resp, resp_len

@@ -1,0 +1,170 @@
+"""Batch-engine tests on the CPU mirrors (byte-exact models of the
+gfx950 kernels). The GPU tier (test_gpu_engine.py) asserts the kernels
+produce the same bytes as these mirrors."""
+
+import json
+
+import numpy as np
+import pytest
+
+import gofr_amd
+from gofr_amd import handlers
+from gofr_amd.engine import BatchEngine, make_batch
+
+
+def build_app(map_config):
+    app = gofr_amd.New(config=map_config)
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.GET("/user/{id}", lambda ctx: {"id": ctx.PathParam("id")})
+    return app
+
+
+def http_req(method="GET", path="/", body=b"", headers=None):
+    h = dict(headers or {})
+    h.setdefault("Host", "localhost")
+    if body:
+        h.setdefault("Content-Type", "application/json")
+        h["Content-Length"] = str(len(body))
+    head = f"{method} {path} HTTP/1.1\r\n" + "".join(
+        f"{k}: {v}\r\n" for k, v in h.items()) + "\r\n"
+    return head.encode() + body
+
+
+def parse_http_response(raw: bytes):
+    head, _, body = raw.partition(b"\r\n\r\n")
+    lines = head.split(b"\r\n")
+    status = int(lines[0].split(b" ")[1])
+    hdrs = {}
+    for line in lines[1:]:
+        k, _, v = line.partition(b": ")
+        hdrs[k.decode().lower()] = v.decode()
+    return status, hdrs, body
+
+
+@pytest.fixture()
+def engine(map_config):
+    app = build_app(map_config)
+    return BatchEngine(app, device="cpu")
+
+
+def run_one(engine, raw):
+    reqs, lens = make_batch([raw], engine.slot)
+    resp, rlen = engine.process(reqs, lens)
+    return parse_http_response(resp[:int(rlen[0])].tobytes())
+
+
+def test_echo_roundtrip(engine):
+    payload = json.dumps({"msg": "x" * 100, "n": 42}).encode()
+    st, hdrs, body = run_one(engine, http_req("POST", "/echo", payload))
+    assert st == 200
+    assert json.loads(body) == {"data": {"msg": "x" * 100, "n": 42}}
+    assert hdrs["content-type"] == "application/json"
+    assert hdrs["access-control-allow-origin"] == "*"
+    assert len(hdrs["x-correlation-id"]) == 32
+    assert int(hdrs["content-length"]) == len(body)
+
+
+def test_echo_invalid_json_500(engine):
+    st, _, body = run_one(engine, http_req("POST", "/echo", b"{broken"))
+    assert st == 500
+    assert json.loads(body) == {"error": {"message": "invalid JSON body"}}
+
+
+def test_static_route(engine):
+    st, _, body = run_one(engine, http_req("GET", "/greet"))
+    assert st == 200
+    assert json.loads(body) == {"data": "Hello World!"}
+
+
+def test_host_route_with_params(engine):
+    st, _, body = run_one(engine, http_req("GET", "/user/melody"))
+    assert st == 200
+    assert json.loads(body) == {"data": {"id": "melody"}}
+
+
+def test_catch_all_404(engine):
+    st, _, body = run_one(engine, http_req("GET", "/nope/nothing"))
+    assert st == 404
+    assert json.loads(body) == {"error": {"message": "http: no such file"}}
+
+
+def test_options_short_circuit(engine):
+    st, hdrs, body = run_one(engine, http_req("OPTIONS", "/echo"))
+    assert st == 200 and body == b""
+    assert hdrs["access-control-allow-methods"] == \
+        "POST, GET, OPTIONS, PUT, DELETE"
+
+
+def test_health_route_host_path(engine):
+    st, _, body = run_one(engine, http_req("GET", "/.well-known/health"))
+    assert st == 200
+    assert "data" in json.loads(body)
+
+
+def test_favicon_host_path(engine):
+    st, hdrs, body = run_one(engine, http_req("GET", "/favicon.ico"))
+    assert st == 200
+    assert hdrs["content-type"] == "image/x-icon"
+    assert body[:4] == b"\x00\x00\x01\x00"
+
+
+def test_connection_close_honored(engine):
+    st, hdrs, _ = run_one(engine, http_req(
+        "GET", "/greet", headers={"Connection": "close"}))
+    assert hdrs["connection"] == "close"
+
+
+def test_mixed_batch(engine):
+    payload = json.dumps({"k": 1}).encode()
+    raws = [http_req("POST", "/echo", payload),
+            http_req("GET", "/greet"),
+            http_req("GET", "/user/7"),
+            http_req("GET", "/missing"),
+            http_req("OPTIONS", "/greet")]
+    reqs, lens = make_batch(raws, engine.slot)
+    resp, rlen = engine.process(reqs, lens)
+    outs = [parse_http_response(
+        resp[i * engine.rslot:i * engine.rslot + int(rlen[i])].tobytes())
+        for i in range(len(raws))]
+    assert [o[0] for o in outs] == [200, 200, 200, 404, 200]
+    assert json.loads(outs[0][2]) == {"data": {"k": 1}}
+    assert json.loads(outs[2][2]) == {"data": {"id": "7"}}
+    # correlation ids differ per request
+    assert outs[0][1]["x-correlation-id"] != outs[1][1]["x-correlation-id"]
+
+
+def test_parse_fields_match_reference_parser(engine):
+    """Structural parse mirror vs the Python reference parser."""
+    from gofr_amd import ops
+    raws = [
+        http_req("POST", "/echo?a=1&b=2", b'{"x":[1,2,3]}'),
+        http_req("GET", "/user/42",
+                 headers={"Authorization": "Bearer tok123",
+                          "Connection": "close"}),
+        http_req("DELETE", "/user/42/x"),
+    ]
+    reqs, lens = make_batch(raws, engine.slot)
+    fields = ops.cpu_parse_route(reqs, lens, engine.slot,
+                                 engine.program.trie,
+                                 engine.program.handler_tab)
+    from gofr_amd.http.request import parse_request_bytes
+    for i, raw in enumerate(raws):
+        ref = parse_request_bytes(raw)
+        F = fields[i]
+        base = i * engine.slot
+        path = reqs[base + F[ops.FI_PATH_OFF]:
+                    base + F[ops.FI_PATH_OFF] + F[ops.FI_PATH_LEN]].tobytes()
+        assert path.decode() == ref.path
+        body = reqs[base + F[ops.FI_BODY_OFF]:
+                    base + F[ops.FI_BODY_OFF] + F[ops.FI_BODY_LEN]].tobytes()
+        assert body == ref.body
+        from gofr_amd.http.request import METHOD_IDS
+        assert F[ops.FI_METHOD] == METHOD_IDS[ref.method]
+    # auth header captured
+    F = fields[1]
+    base = engine.slot
+    auth = reqs[base + F[ops.FI_AUTH_OFF]:
+                base + F[ops.FI_AUTH_OFF] + F[ops.FI_AUTH_LEN]].tobytes()
+    assert auth == b"Bearer tok123"
+    assert not (F[ops.FI_FLAGS] & ops.FL_KEEP_ALIVE)
