@@ -155,22 +155,33 @@ class BatchEngine:
 
     def _process_gpu(self, reqs, req_len, n):
         t = self.torch
-        slot, rslot = self.slot, self.rslot
-        nb_req = n * slot
-        self.p_reqs[:nb_req] = t.from_numpy(
-            reqs[:nb_req].view(np.uint8))
+        nb_req = n * self.slot
+        self.p_reqs[:nb_req] = t.from_numpy(reqs[:nb_req].view(np.uint8))
         self.p_req_len[:n] = t.from_numpy(req_len.astype(np.int32,
                                                          copy=False))
-        self.d_reqs[:nb_req].copy_(self.p_reqs[:nb_req], non_blocking=True)
+        p_resp, p_rlen = self.process_pinned(n, host_reqs=reqs,
+                                             host_req_len=req_len)
+        return p_resp.numpy().copy(), p_rlen.numpy().copy()
+
+    def process_pinned(self, n, host_reqs=None, host_req_len=None):
+        """Process requests already staged in the pinned ingress ring
+        (self.p_reqs / self.p_req_len[:n]) — the real serving dataflow:
+        the socket layer recv()s directly into the pinned ring, so this
+        H2D copy + kernels + D2H into the pinned egress ring is the whole
+        per-batch pipeline. Responses land in self.p_resp/self.p_resp_len;
+        returned as pinned tensor views (write them to sockets directly)."""
+        t = self.torch
+        slot, rslot = self.slot, self.rslot
+        self.d_reqs[:n * slot].copy_(self.p_reqs[:n * slot],
+                                     non_blocking=True)
         self.d_req_len[:n].copy_(self.p_req_len[:n], non_blocking=True)
         d_resp, d_resp_len = self.process_device(
-            self.d_reqs, self.d_req_len, n, host_reqs=reqs,
-            host_req_len=req_len)
+            self.d_reqs, self.d_req_len, n, host_reqs=host_reqs,
+            host_req_len=host_req_len)
         self.p_resp[:n * rslot].copy_(d_resp[:n * rslot], non_blocking=True)
         self.p_resp_len[:n].copy_(d_resp_len[:n], non_blocking=True)
         t.cuda.synchronize(self.device)
-        return (self.p_resp[:n * rslot].numpy().copy(),
-                self.p_resp_len[:n].numpy().copy())
+        return self.p_resp[:n * rslot], self.p_resp_len[:n]
 
     def process_device(self, d_reqs, d_req_len, n, host_reqs=None,
                        host_req_len=None):

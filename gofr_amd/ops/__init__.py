@@ -66,6 +66,14 @@ class HipOps:
             raise FileNotFoundError(
                 f"HIP extension not built: {so_path}. Run "
                 f"`python setup.py build_hip` or __graft_entry__.build().")
+        # Load torch FIRST so its libamdhip64 (same soname) is the HIP
+        # runtime our .so binds to. Loading /opt/rocm's copy first and
+        # torch's second puts two HIP runtimes in the process and kernel
+        # launches fail with hipErrorNoDevice(100).
+        try:
+            import torch  # noqa: F401
+        except ImportError:
+            pass
         self.lib = ctypes.CDLL(so_path)
         self.lib.gofr_launch_parse_route.restype = ctypes.c_int
         self.lib.gofr_launch_parse_route.argtypes = [ctypes.c_void_p] * 3 + [
