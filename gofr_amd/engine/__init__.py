@@ -69,16 +69,49 @@ class RouteProgram:
         self.blob = bytes(blob)
 
 
+class _Lane:
+    """One pipeline stage: its own HIP stream + device/pinned buffer set.
+
+    The serving loop round-robins lanes so the H2D of batch i+1 and the
+    D2H of batch i-1 overlap the kernels of batch i (separate streams,
+    pinned staging both ways).
+    """
+
+    def __init__(self, t, dev, nb, slot, rslot, host_blob_cap=4 << 20):
+        self.stream = t.cuda.Stream(device=dev)
+        self.event = t.cuda.Event()
+        self.d_reqs = t.empty(nb * slot, dtype=t.uint8, device=dev)
+        self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
+        self.d_fields = t.zeros(nb * ops.NF, dtype=t.int32, device=dev)
+        self.d_resp = t.empty(nb * rslot, dtype=t.uint8, device=dev)
+        self.d_resp_len = t.empty(nb, dtype=t.int32, device=dev)
+        self.d_host_needed = t.zeros(1, dtype=t.int32, device=dev)
+        self.d_host_tab = t.zeros(nb * 4, dtype=t.int32, device=dev)
+        self.d_host_blob = t.zeros(host_blob_cap, dtype=t.uint8, device=dev)
+        self.p_reqs = t.empty(nb * slot, dtype=t.uint8).pin_memory()
+        self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
+        self.p_resp = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
+        self.p_resp_len = t.empty(nb, dtype=t.int32).pin_memory()
+        self.p_fields = t.empty(nb * ops.NF, dtype=t.int32).pin_memory()
+        self.p_host_needed = t.zeros(1, dtype=t.int32).pin_memory()
+        self.n = 0
+        self.seed = 0
+
+
 class BatchEngine:
     """Processes request batches through the GPU kernels (or CPU mirrors)."""
 
     def __init__(self, app, device=None, slot: int = 2048,
-                 rslot: int = 4096, max_batch: int = 65536,
-                 require_gpu: bool = False):
+                 rslot: int = 0, max_batch: int = 65536,
+                 require_gpu: bool = False, pipeline: int = 1):
         self.app = app
         self.slot = slot
-        self.rslot = rslot
+        # response slot must hold worst-case: headers(~260) + envelope(9) +
+        # a body as large as the request slot
+        self.rslot = rslot if rslot else slot + 512
+        assert self.rslot >= slot + 512, "rslot too small for worst case"
         self.max_batch = max_batch
+        self.pipeline = max(1, pipeline)
         self.program = RouteProgram(app)
         self._seed = 0x6F667247414D4421  # advanced per batch
         self.device = None
@@ -104,14 +137,6 @@ class BatchEngine:
     def _alloc_device(self):
         t, dev = self.torch, self.device
         nb, slot, rslot = self.max_batch, self.slot, self.rslot
-        self.d_reqs = t.empty(nb * slot, dtype=t.uint8, device=dev)
-        self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
-        self.d_fields = t.zeros(nb * ops.NF, dtype=t.int32, device=dev)
-        self.d_resp = t.empty(nb * rslot, dtype=t.uint8, device=dev)
-        self.d_resp_len = t.empty(nb, dtype=t.int32, device=dev)
-        self.d_host_needed = t.zeros(1, dtype=t.int32, device=dev)
-        self.d_host_tab = t.zeros(nb * 4, dtype=t.int32, device=dev)
-        self.d_host_blob = t.zeros(8 << 20, dtype=t.uint8, device=dev)
         tr = {}
         for k, v in self.program.trie.items():
             if k == "n_nodes":
@@ -121,12 +146,18 @@ class BatchEngine:
         self.d_handler_tab = t.as_tensor(self.program.handler_tab).to(dev)
         self.d_blob = t.as_tensor(
             np.frombuffer(self.program.blob, np.uint8).copy()).to(dev)
-        # pinned staging
-        self.p_reqs = t.empty(nb * slot, dtype=t.uint8).pin_memory()
-        self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
-        self.p_resp = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
-        self.p_resp_len = t.empty(nb, dtype=t.int32).pin_memory()
-        self.p_fields = t.empty(nb * ops.NF, dtype=t.int32).pin_memory()
+        self.lanes = [_Lane(t, dev, nb, slot, rslot)
+                      for _ in range(self.pipeline)]
+        # lane-0 aliases: the synchronous API and the multi-GPU path
+        ln = self.lanes[0]
+        self.d_reqs, self.d_req_len = ln.d_reqs, ln.d_req_len
+        self.d_fields, self.d_resp = ln.d_fields, ln.d_resp
+        self.d_resp_len = ln.d_resp_len
+        self.d_host_needed = ln.d_host_needed
+        self.d_host_tab, self.d_host_blob = ln.d_host_tab, ln.d_host_blob
+        self.p_reqs, self.p_req_len = ln.p_reqs, ln.p_req_len
+        self.p_resp, self.p_resp_len = ln.p_resp, ln.p_resp_len
+        self.p_fields = ln.p_fields
 
     def _next_seed(self) -> int:
         self._seed = ops.splitmix64(self._seed)
@@ -159,52 +190,108 @@ class BatchEngine:
         self.p_reqs[:nb_req] = t.from_numpy(reqs[:nb_req].view(np.uint8))
         self.p_req_len[:n] = t.from_numpy(req_len.astype(np.int32,
                                                          copy=False))
-        p_resp, p_rlen = self.process_pinned(n, host_reqs=reqs,
-                                             host_req_len=req_len)
+        p_resp, p_rlen = self.process_pinned(n)
         return p_resp.numpy().copy(), p_rlen.numpy().copy()
 
-    def process_pinned(self, n, host_reqs=None, host_req_len=None):
-        """Process requests already staged in the pinned ingress ring
-        (self.p_reqs / self.p_req_len[:n]) — the real serving dataflow:
-        the socket layer recv()s directly into the pinned ring, so this
-        H2D copy + kernels + D2H into the pinned egress ring is the whole
-        per-batch pipeline. Responses land in self.p_resp/self.p_resp_len;
-        returned as pinned tensor views (write them to sockets directly)."""
+    # -- pipelined API --------------------------------------------------------
+    def submit(self, n: int, lane_idx: int = 0) -> None:
+        """Enqueue one batch (already staged in lane.p_reqs/p_req_len[:n])
+        on the lane's stream: H2D -> parse -> respond (optimistic) -> D2H.
+        Never blocks. complete() finishes it."""
         t = self.torch
+        ln = self.lanes[lane_idx]
         slot, rslot = self.slot, self.rslot
-        self.d_reqs[:n * slot].copy_(self.p_reqs[:n * slot],
-                                     non_blocking=True)
-        self.d_req_len[:n].copy_(self.p_req_len[:n], non_blocking=True)
-        d_resp, d_resp_len = self.process_device(
-            self.d_reqs, self.d_req_len, n, host_reqs=host_reqs,
-            host_req_len=host_req_len)
-        self.p_resp[:n * rslot].copy_(d_resp[:n * rslot], non_blocking=True)
-        self.p_resp_len[:n].copy_(d_resp_len[:n], non_blocking=True)
-        t.cuda.synchronize(self.device)
-        return self.p_resp[:n * rslot], self.p_resp_len[:n]
+        ln.n = n
+        ln.seed = self._next_seed()
+        with t.cuda.stream(ln.stream):
+            ln.d_reqs[:n * slot].copy_(ln.p_reqs[:n * slot],
+                                       non_blocking=True)
+            ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
+            ln.d_host_needed.zero_()
+            cs = ln.stream.cuda_stream
+            self.hip.parse_route(cs, ln.d_reqs, ln.d_req_len, ln.d_fields,
+                                 n, slot, self.d_trie, self.d_handler_tab,
+                                 self.program.n_routes, ln.d_host_needed)
+            ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
+            # optimistic respond: host rows render a 500 fallback that the
+            # fixup pass overwrites before the responses are released
+            self.hip.respond(cs, ln.d_reqs, ln.d_fields, ln.d_resp,
+                             ln.d_resp_len, n, slot, rslot,
+                             self.d_handler_tab, self.program.n_routes,
+                             self.d_blob, ln.d_host_blob, ln.d_host_tab,
+                             ln.seed)
+            ln.p_resp[:n * rslot].copy_(ln.d_resp[:n * rslot],
+                                        non_blocking=True)
+            ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
+            ln.event.record(ln.stream)
+
+    def complete(self, lane_idx: int = 0):
+        """Wait for the lane's in-flight batch; run the host fixup pass if
+        any row needed the trampoline. Returns pinned (resp, resp_len)."""
+        t = self.torch
+        ln = self.lanes[lane_idx]
+        n, slot, rslot = ln.n, self.slot, self.rslot
+        ln.event.synchronize()
+        if int(ln.p_host_needed[0]):
+            # fixup: run Python handlers for HK_HOST rows, re-serialize
+            with t.cuda.stream(ln.stream):
+                ln.p_fields[:n * ops.NF].copy_(ln.d_fields[:n * ops.NF],
+                                               non_blocking=True)
+            ln.stream.synchronize()
+            fields = ln.p_fields[:n * ops.NF].numpy().reshape(n, ops.NF)
+            host_reqs = ln.p_reqs[:n * slot].numpy()
+            host_req_len = ln.p_req_len[:n].numpy()
+            host_blob, host_tab = self._run_host_rows(
+                fields, host_reqs, host_req_len)
+            hb = np.frombuffer(host_blob, np.uint8)
+            with t.cuda.stream(ln.stream):
+                if len(hb):
+                    ln.d_host_blob[:len(hb)].copy_(
+                        t.from_numpy(hb.copy()), non_blocking=True)
+                ln.d_host_tab[:n * 4].copy_(
+                    t.from_numpy(host_tab.reshape(-1).copy()),
+                    non_blocking=True)
+                self.hip.respond(ln.stream.cuda_stream, ln.d_reqs,
+                                 ln.d_fields, ln.d_resp, ln.d_resp_len,
+                                 n, slot, rslot, self.d_handler_tab,
+                                 self.program.n_routes, self.d_blob,
+                                 ln.d_host_blob, ln.d_host_tab, ln.seed)
+                ln.p_resp[:n * rslot].copy_(ln.d_resp[:n * rslot],
+                                            non_blocking=True)
+                ln.p_resp_len[:n].copy_(ln.d_resp_len[:n],
+                                        non_blocking=True)
+            ln.stream.synchronize()
+        return ln.p_resp[:n * rslot], ln.p_resp_len[:n]
+
+    def process_pinned(self, n, lane_idx: int = 0):
+        """Synchronous one-batch pipeline on a lane (requests staged in
+        lane.p_reqs/p_req_len). The socket layer recv()s directly into the
+        pinned ring, so H2D + kernels + D2H is the whole per-batch path."""
+        self.submit(n, lane_idx)
+        return self.complete(lane_idx)
 
     def process_device(self, d_reqs, d_req_len, n, host_reqs=None,
                        host_req_len=None):
         """Run the kernel pipeline on request bytes already resident on the
         device (the multi-GPU all-to-all path hands exchanged slabs in
-        directly). Returns (d_resp, d_resp_len) device tensors; no D2H of
-        the responses. Host trampoline rows need host_reqs (or are re-read
-        from the device when not provided)."""
+        directly). Runs on the CALLER's current stream; returns
+        (d_resp, d_resp_len) device tensors, no D2H."""
         t = self.torch
+        ln = self.lanes[0]
         slot, rslot = self.slot, self.rslot
         seed = self._next_seed()
         stream = t.cuda.current_stream(self.device).cuda_stream
-        self.d_host_needed.zero_()
+        ln.d_host_needed.zero_()
         self.hip.parse_route(stream, d_reqs, d_req_len,
-                             self.d_fields, n, slot, self.d_trie,
+                             ln.d_fields, n, slot, self.d_trie,
                              self.d_handler_tab, self.program.n_routes,
-                             self.d_host_needed)
+                             ln.d_host_needed)
         # host trampoline only when some row needs it (4-byte D2H + sync)
-        host_needed = int(self.d_host_needed.item())
+        host_needed = int(ln.d_host_needed.item())
         if host_needed:
-            self.p_fields[:n * ops.NF].copy_(self.d_fields[:n * ops.NF])
+            ln.p_fields[:n * ops.NF].copy_(ln.d_fields[:n * ops.NF])
             t.cuda.synchronize(self.device)
-            fields = self.p_fields[:n * ops.NF].numpy().reshape(n, ops.NF)
+            fields = ln.p_fields[:n * ops.NF].numpy().reshape(n, ops.NF)
             if host_reqs is None:
                 host_reqs = d_reqs[:n * slot].cpu().numpy()
                 host_req_len = d_req_len[:n].cpu().numpy()
@@ -212,17 +299,18 @@ class BatchEngine:
                 fields, host_reqs, host_req_len)
             hb = np.frombuffer(host_blob, np.uint8)
             if len(hb):
-                self.d_host_blob[:len(hb)].copy_(
+                ln.d_host_blob[:len(hb)].copy_(
                     t.from_numpy(hb.copy()), non_blocking=True)
-            self.d_host_tab[:n * 4].copy_(
+            ln.d_host_tab[:n * 4].copy_(
                 t.from_numpy(host_tab.reshape(-1).copy()),
                 non_blocking=True)
-        self.hip.respond(stream, d_reqs, self.d_fields, self.d_resp,
-                         self.d_resp_len, n, slot, rslot,
+        self.hip.respond(stream, d_reqs, ln.d_fields, ln.d_resp,
+                         ln.d_resp_len, n, slot, rslot,
                          self.d_handler_tab, self.program.n_routes,
-                         self.d_blob, self.d_host_blob, self.d_host_tab,
+                         self.d_blob, ln.d_host_blob, ln.d_host_tab,
                          seed)
-        return self.d_resp, self.d_resp_len
+        return ln.d_resp, ln.d_resp_len
+
 
     # -- host trampoline ------------------------------------------------------
     def _run_host_rows(self, fields, reqs, req_len):

@@ -385,7 +385,14 @@ __device__ __forceinline__ int itoa10(uint8_t* dst, int v) {
     return n;
 }
 
-// host result table row: [off, len, status, ct_or_flags]
+// host result table row: [off, len, status, ct_id]
+//
+// The whole response (headers + body) is assembled in LDS per wave, then
+// copied to global with one coalesced 16 B/lane sweep. Header segments are
+// written cooperatively (each segment is <=64 B: one strided store per
+// lane); the JSON-echo body is validated with a ballot structural pass
+// (quote/backslash/brace bitmasks; lane 0 only touches the ~tens of
+// structural positions, not every byte) before being staged.
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
 k_respond(const uint8_t* __restrict__ reqs,
           int32_t* __restrict__ fields,
@@ -397,6 +404,8 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ host_blob,
           const int32_t* __restrict__ host_tab,
           uint64_t seed) {
+    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
+
     const int wv = threadIdx.x / WAVE;
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
     if (req >= n) return;
@@ -404,6 +413,7 @@ k_respond(const uint8_t* __restrict__ reqs,
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
     const uint8_t* rbase = reqs + (size_t)req * slot;
+    uint8_t* obuf = obuf_all + wv * MAX_SLOT;
 
     const int kind = F[FI_KIND];
     int status = F[FI_STATUS];
@@ -412,8 +422,6 @@ k_respond(const uint8_t* __restrict__ reqs,
     const bool is_options = (flags & FL_IS_OPTIONS) != 0;
 
     // ---- resolve body source + length --------------------------------------
-    // body = envelope_open + payload + envelope_close for ECHO; raw blob for
-    // STATIC (payload already enveloped); host bytes for HOST
     const uint8_t* body_src = nullptr;
     int body_src_len = 0;
     int env = 0;  // 1 -> wrap in {"data": ... }
@@ -423,40 +431,67 @@ k_respond(const uint8_t* __restrict__ reqs,
         body_src = rbase + F[FI_BODY_OFF];
         body_src_len = F[FI_BODY_LEN];
         env = 1;
-        // minimal structural validation: brace/bracket balance + quote
-        // parity, escapes honored (lane-serial over short bodies would
-        // diverge; all lanes vote instead)
+        // structural validation: per-64B-chunk ballots; lane 0 walks only
+        // the structural bit positions with escape-parity handling —
+        // byte-equivalent to the serial model in gofr_amd/ops
+        bool bad = (body_src_len == 0);
         int depth = 0;
-        bool bad = false;
-        if (lane == 0) {
-            bool in_str = false;
-            for (int i = 0; i < body_src_len; ++i) {
-                uint8_t c = body_src[i];
-                if (in_str) {
-                    if (c == '\\') { ++i; }
-                    else if (c == '"') in_str = false;
-                } else if (c == '"') in_str = true;
-                else if (c == '{' || c == '[') ++depth;
-                else if (c == '}' || c == ']') --depth;
-                if (depth < 0) { bad = true; break; }
+        bool in_str = false;
+        int bs_suffix = 0;  // backslash run carried across chunk boundary
+        const int nch = (body_src_len + WAVE - 1) / WAVE;
+        for (int c = 0; c < nch && !bad; ++c) {
+            const int i = c * WAVE + lane;
+            const uint8_t b = (i < body_src_len) ? body_src[i] : 0;
+            const uint64_t m_q = __ballot(b == '"');
+            const uint64_t m_bs = __ballot(b == '\\');
+            const uint64_t m_op = __ballot(b == '{' || b == '[');
+            const uint64_t m_cl = __ballot(b == '}' || b == ']');
+            if (lane == 0) {
+                uint64_t strct = m_q | m_op | m_cl;
+                while (strct) {
+                    const int p = __builtin_ctzll(strct);
+                    strct &= strct - 1;
+                    const uint64_t bit = 1ull << p;
+                    if (m_q & bit) {
+                        // escaped iff odd backslash run ends at p-1
+                        int run;
+                        if (p == 0) {
+                            run = bs_suffix;
+                        } else {
+                            const uint64_t below = m_bs & (bit - 1);
+                            // count consecutive set bits ending at p-1
+                            run = 0;
+                            uint64_t probe = 1ull << (p - 1);
+                            while ((below & probe) && run < p) {
+                                ++run;
+                                probe >>= 1;
+                            }
+                            if (run == p) run += bs_suffix;
+                        }
+                        const bool escaped = in_str && (run & 1);
+                        if (!escaped) in_str = !in_str;
+                    } else if (!in_str) {
+                        if (m_op & bit) ++depth;
+                        else if (--depth < 0) { bad = true; break; }
+                    }
+                }
+                // trailing backslash run of this chunk
+                if (m_bs == ~0ull) bs_suffix += 64;
+                else bs_suffix = __builtin_clzll(~m_bs);
             }
-            bad = bad || depth != 0 || in_str || body_src_len == 0;
+            bad = __shfl(bad ? 1 : 0, 0) != 0;
         }
+        if (lane == 0 && (depth != 0 || in_str)) bad = true;
         bad = __shfl(bad ? 1 : 0, 0) != 0;
         if (bad) {
             status = 500;
-            // error envelope from blob: handler_tab row arg points at the
-            // route's static arg; the invalid-body envelope is blob[0..]
-            // convention: first 4 bytes of blob = len of invalid-body env
+            // invalid-body envelope: blob[0:4]=len, blob[4:4+len]=bytes
             const int elen = *(const int32_t*)blob;
             body_src = blob + 4;
             body_src_len = elen;
             env = 0;
         }
     } else if (kind == HK_STATIC) {
-        // host guarantees route >= 0 via the catch-all prefix route; an
-        // unresolved route (defaults not installed) serves an empty body
-        // with the parse-time status (404)
         const int route = F[FI_ROUTE];
         if (route >= 0 && route < n_routes) {
             body_src = blob + handler_tab[route * 4 + 1];
@@ -469,71 +504,101 @@ k_respond(const uint8_t* __restrict__ reqs,
         body_src_len = host_tab[req * 4 + 1];
         status = host_tab[req * 4 + 2];
         ct_id = host_tab[req * 4 + 3];
+        if (status == 0) {  // trampoline never ran (optimistic mode)
+            status = 500;
+            const int elen = *(const int32_t*)blob;
+            body_src = blob + 4;
+            body_src_len = elen;
+        }
     }
 
     const int body_total = body_src_len + (env ? 9 : 0);  // {"data": + }
 
-    // ---- header length arithmetic ------------------------------------------
+    // ---- header segments (lengths first, then cooperative writes) ---------
     int rlen_reason;
     const char* reason = reason_of(status, &rlen_reason);
-    // HTTP/1.1 [9] + 3 + 1 + reason + 2
-    int hl = 9 + 3 + 1 + rlen_reason + 2;
     const char* ct_str = HDR_CT_JSON;
     int ct_len = sizeof(HDR_CT_JSON) - 1;
     if (ct_id == 1) { ct_str = HDR_CT_ICON; ct_len = sizeof(HDR_CT_ICON) - 1; }
     else if (ct_id == 2) { ct_str = HDR_CT_BIN; ct_len = sizeof(HDR_CT_BIN) - 1; }
     else if (ct_id == 3) { ct_str = HDR_CT_TXT; ct_len = sizeof(HDR_CT_TXT) - 1; }
-    hl += ct_len;
-    hl += sizeof(HDR_CORS) - 1;
-    hl += sizeof(HDR_CORR) - 1 + 32 + 2;
     int cl_digits = 1;
     for (int v = body_total; v >= 10; v /= 10) ++cl_digits;
-    hl += sizeof(HDR_CL) - 1 + cl_digits + 2;
-    hl += (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
+    const char* conn = keep ? HDR_CONN_KA : HDR_CONN_CL;
+    const int conn_len =
+        (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
 
-    // ---- parallel body copy (all lanes) ------------------------------------
+    // segment offsets (all lanes compute identically)
+    const int o_status = 9;                       // after "HTTP/1.1 "
+    const int o_reason = o_status + 3 + 1;        // "xxx "
+    const int o_ct = o_reason + rlen_reason + 2;  // reason \r\n
+    const int o_cors = o_ct + ct_len;
+    const int o_corr = o_cors + (int)sizeof(HDR_CORS) - 1;
+    const int o_corrhex = o_corr + (int)sizeof(HDR_CORR) - 1;
+    const int o_cl = o_corrhex + 32 + 2;
+    const int o_cld = o_cl + (int)sizeof(HDR_CL) - 1;
+    const int o_conn = o_cld + cl_digits + 2;
+    const int hl = o_conn + conn_len;
+
+    // cooperative writes into LDS: each segment <=64B -> one store per lane
+    if (lane < 9) obuf[lane] = HDR_P1[lane];
+    if (lane < (unsigned)rlen_reason) obuf[o_reason + lane] = reason[lane];
+    if (lane < (unsigned)ct_len) obuf[o_ct + lane] = ct_str[lane];
+    // CORS is 95 bytes: two strided stores
+    {
+        const int cors_len = (int)sizeof(HDR_CORS) - 1;
+        for (int i = lane; i < cors_len; i += WAVE)
+            obuf[o_cors + i] = HDR_CORS[i];
+    }
+    if (lane < (int)sizeof(HDR_CORR) - 1) obuf[o_corr + lane] = HDR_CORR[lane];
+    if (lane < (int)sizeof(HDR_CL) - 1) obuf[o_cl + lane] = HDR_CL[lane];
+    if (lane < (unsigned)conn_len) obuf[o_conn + lane] = conn[lane];
+    // correlation id: every lane computes the 128-bit id and emits its char
+    {
+        const uint64_t h1 = splitmix64(seed ^ (uint64_t)req);
+        const uint64_t h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03ull);
+        if (lane < 32) {
+            const uint64_t h = (lane < 16) ? h1 : h2;
+            const int sh = 60 - 4 * (lane & 15);
+            obuf[o_corrhex + lane] = HEXD[(h >> sh) & 0xF];
+        }
+    }
+    if (lane == 0) {
+        obuf[o_status] = '0' + (status / 100);
+        obuf[o_status + 1] = '0' + ((status / 10) % 10);
+        obuf[o_status + 2] = '0' + (status % 10);
+        obuf[o_status + 3] = ' ';
+        obuf[o_reason + rlen_reason] = '\r';
+        obuf[o_reason + rlen_reason + 1] = '\n';
+        obuf[o_corrhex + 32] = '\r';
+        obuf[o_corrhex + 33] = '\n';
+        itoa10(obuf + o_cld, body_total);
+        obuf[o_cld + cl_digits] = '\r';
+        obuf[o_cld + cl_digits + 1] = '\n';
+        if (env) {
+            for (int i = 0; i < 8; ++i) obuf[hl + i] = ENV_OPEN[i];
+            obuf[hl + 8 + body_src_len] = '}';
+        }
+    }
+
+    // ---- body into LDS (coalesced byte lanes) ------------------------------
     const int body_start = hl + (env ? 8 : 0);
     for (int i = lane; i < body_src_len; i += WAVE)
-        out[body_start + i] = body_src[i];
+        obuf[body_start + i] = body_src[i];
 
-    // ---- lane 0: header + envelope + trailer -------------------------------
+    const int total = hl + body_total;
+    // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
+    if (total <= rslot) {
+        const int nv = (total + 15) >> 4;
+        const uint4* src = (const uint4*)obuf;
+        uint4* dst = (uint4*)out;
+        for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+    }
     if (lane == 0) {
-        uint8_t* p = out;
-        for (int i = 0; i < 9; ++i) *p++ = HDR_P1[i];
-        *p++ = '0' + (status / 100);
-        *p++ = '0' + ((status / 10) % 10);
-        *p++ = '0' + (status % 10);
-        *p++ = ' ';
-        for (int i = 0; i < rlen_reason; ++i) *p++ = reason[i];
-        *p++ = '\r'; *p++ = '\n';
-        for (int i = 0; i < ct_len; ++i) *p++ = ct_str[i];
-        for (int i = 0; i < (int)sizeof(HDR_CORS) - 1; ++i) *p++ = HDR_CORS[i];
-        for (int i = 0; i < (int)sizeof(HDR_CORR) - 1; ++i) *p++ = HDR_CORR[i];
-        // correlation id: 128-bit from splitmix64 (fused id+hex — the
-        // reference takes it from the OTel span, middleware/logger.go:46)
-        uint64_t h1 = splitmix64(seed ^ (uint64_t)req);
-        uint64_t h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03ull);
-        for (int i = 0; i < 16; ++i)
-            *p++ = HEXD[(h1 >> (60 - 4 * i)) & 0xF];
-        for (int i = 0; i < 16; ++i)
-            *p++ = HEXD[(h2 >> (60 - 4 * i)) & 0xF];
-        *p++ = '\r'; *p++ = '\n';
-        for (int i = 0; i < (int)sizeof(HDR_CL) - 1; ++i) *p++ = HDR_CL[i];
-        p += itoa10(p, body_total);
-        *p++ = '\r'; *p++ = '\n';
-        const char* conn = keep ? HDR_CONN_KA : HDR_CONN_CL;
-        const int conn_len = (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
-        for (int i = 0; i < conn_len; ++i) *p++ = conn[i];
-        // assert header arithmetic matched
-        // (p - out) == hl by construction
-        if (env) {
-            for (int i = 0; i < 8; ++i) out[hl + i] = ENV_OPEN[i];
-            out[hl + 8 + body_src_len] = '}';
-        }
-        const int total = hl + body_total;
-        F[FI_RESP_LEN] = total;
+        const int tl = (total <= rslot) ? total : 0;
+        F[FI_RESP_LEN] = tl;
         F[FI_RESP_OFF] = (int)((size_t)req * rslot);
-        resp_len_out[req] = total;
+        resp_len_out[req] = tl;
     }
 }
 
