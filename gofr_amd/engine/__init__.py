@@ -1,16 +1,21 @@
 """GPU request-batch engine.
 
-The MI355X data plane (SURVEY.md §7 phase 2): request bytes are staged
-into pinned ring buffers, one H2D copy moves the batch onto the device,
-k_parse_route + k_respond (native/hip/gofr_kernels.hip) parse, route and
-serialize the whole batch, and one D2H copy brings the response bytes
-back. Handlers with a GPU spec (gofr_amd/handlers.py) never touch the
-host; the rest run through the host trampoline between the two kernels
-(only for the requests that need it — a device-side counter tells the
-host whether any do, so the pure-GPU path never blocks on Python).
+The MI355X data plane (SURVEY.md §7 phase 2): raw request bytes are
+PACKED back-to-back into a pinned ingress ring (offset + length per
+request — no slot padding on the bus), one H2D copy moves the batch onto
+the device, k_parse_route + k_respond (native/hip/gofr_kernels.hip)
+parse, route and serialize the whole batch, k_compact gathers the
+response slots into a contiguous 16B-aligned stream, and one D2H copy
+brings exactly the response bytes back. Handlers with a GPU spec
+(gofr_amd/handlers.py) never touch the host; the rest run through the
+host trampoline fixup pass (only when the device-side counter says some
+request needs it — pure-GPU batches never block on Python).
 
-Without a GPU the engine runs the byte-exact CPU mirrors
-(gofr_amd/ops), so every test of this module runs on the CPU box.
+Multiple lanes (stream + buffer set each) let the serving loop overlap
+H2D of batch i+1 and D2H of batch i-1 with the kernels of batch i.
+
+Without a GPU the engine runs the byte-exact CPU mirrors (gofr_amd/ops),
+so every test of this module runs on the CPU box.
 """
 
 from __future__ import annotations
@@ -69,32 +74,58 @@ class RouteProgram:
         self.blob = bytes(blob)
 
 
+def pack_batch(payloads: list[bytes]):
+    """Pack raw request byte strings back-to-back.
+    Returns (buf uint8, off int64, len int32)."""
+    n = len(payloads)
+    lens = np.asarray([len(p) for p in payloads], np.int32)
+    offs = np.zeros(n, np.int64)
+    if n > 1:
+        np.cumsum(lens[:-1], out=offs[1:])
+    buf = np.frombuffer(b"".join(payloads), np.uint8).copy()
+    return buf, offs, lens
+
+
+def make_batch(payloads: list[bytes], slot: int):
+    """Slot layout packer (multi-GPU fixed-size exchange path)."""
+    n = len(payloads)
+    reqs = np.zeros(n * slot, np.uint8)
+    lens = np.zeros(n, np.int32)
+    for i, p in enumerate(payloads):
+        assert len(p) <= slot, "request exceeds slot size"
+        reqs[i * slot:i * slot + len(p)] = np.frombuffer(p, np.uint8)
+        lens[i] = len(p)
+    return reqs, lens
+
+
 class _Lane:
-    """One pipeline stage: its own HIP stream + device/pinned buffer set.
+    """One pipeline stage: its own HIP stream + device/pinned buffer set."""
 
-    The serving loop round-robins lanes so the H2D of batch i+1 and the
-    D2H of batch i-1 overlap the kernels of batch i (separate streams,
-    pinned staging both ways).
-    """
-
-    def __init__(self, t, dev, nb, slot, rslot, host_blob_cap=4 << 20):
+    def __init__(self, t, dev, nb, max_bytes, rslot, host_blob_cap=4 << 20):
         self.stream = t.cuda.Stream(device=dev)
         self.event = t.cuda.Event()
-        self.d_reqs = t.empty(nb * slot, dtype=t.uint8, device=dev)
+        self.d_reqs = t.empty(max_bytes, dtype=t.uint8, device=dev)
+        self.d_req_off = t.empty(nb, dtype=t.int64, device=dev)
         self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
         self.d_fields = t.zeros(nb * ops.NF, dtype=t.int32, device=dev)
         self.d_resp = t.empty(nb * rslot, dtype=t.uint8, device=dev)
         self.d_resp_len = t.empty(nb, dtype=t.int32, device=dev)
+        self.d_resp_off = t.empty(nb, dtype=t.int32, device=dev)
+        self.d_out = t.empty(nb * rslot, dtype=t.uint8, device=dev)
         self.d_host_needed = t.zeros(1, dtype=t.int32, device=dev)
         self.d_host_tab = t.zeros(nb * 4, dtype=t.int32, device=dev)
         self.d_host_blob = t.zeros(host_blob_cap, dtype=t.uint8, device=dev)
-        self.p_reqs = t.empty(nb * slot, dtype=t.uint8).pin_memory()
+        self.p_reqs = t.empty(max_bytes, dtype=t.uint8).pin_memory()
+        self.p_req_off = t.empty(nb, dtype=t.int64).pin_memory()
         self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
-        self.p_resp = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
+        self.p_out = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
         self.p_resp_len = t.empty(nb, dtype=t.int32).pin_memory()
+        self.p_resp_off = t.empty(nb, dtype=t.int32).pin_memory()
+        self.p_total = t.empty(1, dtype=t.int32).pin_memory()
         self.p_fields = t.empty(nb * ops.NF, dtype=t.int32).pin_memory()
         self.p_host_needed = t.zeros(1, dtype=t.int32).pin_memory()
         self.n = 0
+        self.nbytes = 0
         self.seed = 0
 
 
@@ -105,12 +136,13 @@ class BatchEngine:
                  rslot: int = 0, max_batch: int = 65536,
                  require_gpu: bool = False, pipeline: int = 1):
         self.app = app
+        # `slot` = max single-request size; packed layout means it no
+        # longer costs bus bytes, only worst-case device buffer sizing
         self.slot = slot
-        # response slot must hold worst-case: headers(~260) + envelope(9) +
-        # a body as large as the request slot
         self.rslot = rslot if rslot else slot + 512
         assert self.rslot >= slot + 512, "rslot too small for worst case"
         self.max_batch = max_batch
+        self.max_bytes = max_batch * slot
         self.pipeline = max(1, pipeline)
         self.program = RouteProgram(app)
         self._seed = 0x6F667247414D4421  # advanced per batch
@@ -135,8 +167,8 @@ class BatchEngine:
 
     # -- device state --------------------------------------------------------
     def _alloc_device(self):
-        t, dev = self.torch, self.device
-        nb, slot, rslot = self.max_batch, self.slot, self.rslot
+        t = self.torch
+        dev = self.device
         tr = {}
         for k, v in self.program.trie.items():
             if k == "n_nodes":
@@ -146,91 +178,116 @@ class BatchEngine:
         self.d_handler_tab = t.as_tensor(self.program.handler_tab).to(dev)
         self.d_blob = t.as_tensor(
             np.frombuffer(self.program.blob, np.uint8).copy()).to(dev)
-        self.lanes = [_Lane(t, dev, nb, slot, rslot)
-                      for _ in range(self.pipeline)]
-        # lane-0 aliases: the synchronous API and the multi-GPU path
-        ln = self.lanes[0]
-        self.d_reqs, self.d_req_len = ln.d_reqs, ln.d_req_len
-        self.d_fields, self.d_resp = ln.d_fields, ln.d_resp
-        self.d_resp_len = ln.d_resp_len
-        self.d_host_needed = ln.d_host_needed
-        self.d_host_tab, self.d_host_blob = ln.d_host_tab, ln.d_host_blob
-        self.p_reqs, self.p_req_len = ln.p_reqs, ln.p_req_len
-        self.p_resp, self.p_resp_len = ln.p_resp, ln.p_resp_len
-        self.p_fields = ln.p_fields
+        self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
+                            self.rslot) for _ in range(self.pipeline)]
 
     def _next_seed(self) -> int:
         self._seed = ops.splitmix64(self._seed)
         return self._seed
 
-    # -- main entry ----------------------------------------------------------
-    def process(self, reqs: np.ndarray, req_len: np.ndarray):
-        """reqs: uint8 [n*slot], req_len: int32 [n].
-        Returns (resp uint8 [n*rslot], resp_len int32 [n])."""
-        n = len(req_len)
+    # -- main entries ---------------------------------------------------------
+    def process(self, payloads: list[bytes]) -> list[bytes]:
+        """Convenience: pack, run, slice. Returns one response byte string
+        per request."""
+        buf, offs, lens = pack_batch(payloads)
+        out, roffs, rlens = self.process_packed(buf, offs, lens)
+        return [bytes(out[int(roffs[i]):int(roffs[i]) + int(rlens[i])])
+                for i in range(len(payloads))]
+
+    def process_packed(self, buf: np.ndarray, offs: np.ndarray,
+                       lens: np.ndarray):
+        """buf: packed request bytes; offs/lens per request.
+        Returns (out np.uint8, resp_off int32, resp_len int32) — the
+        responses are 16B-aligned contiguous in `out` at resp_off."""
+        n = len(lens)
         assert n <= self.max_batch
         if self.device is None:
-            return self._process_cpu(reqs, req_len)
-        return self._process_gpu(reqs, req_len, n)
+            return self._process_cpu(buf, offs, lens)
+        t = self.torch
+        ln = self.lanes[0]
+        nbytes = int(offs[-1] + lens[-1]) if n else 0
+        ln.p_reqs[:nbytes] = t.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n] = t.from_numpy(offs.astype(np.int64, copy=False))
+        ln.p_req_len[:n] = t.from_numpy(lens.astype(np.int32, copy=False))
+        self.submit(n, nbytes, 0)
+        out_t, roff_t, rlen_t = self.complete(0)
+        return (out_t.numpy().copy(), roff_t.numpy().copy(),
+                rlen_t.numpy().copy())
 
     # CPU fallback: byte-exact mirrors (never used on a GPU box)
-    def _process_cpu(self, reqs, req_len):
+    def _process_cpu(self, buf, offs, lens):
         seed = self._next_seed()
-        fields = ops.cpu_parse_route(reqs, req_len, self.slot,
-                                     self.program.trie,
+        fields = ops.cpu_parse_route(buf, offs, lens, self.program.trie,
                                      self.program.handler_tab)
-        host_blob, host_tab = self._run_host_rows(fields, reqs, req_len)
-        return ops.cpu_respond(
-            reqs, fields, self.slot, self.rslot, self.program.handler_tab,
+        host_blob, host_tab = self._run_host_rows(fields, buf, offs, lens)
+        resp_slots, resp_len = ops.cpu_respond(
+            buf, offs, fields, self.rslot, self.program.handler_tab,
             self.program.blob, host_blob, host_tab, seed)
-
-    def _process_gpu(self, reqs, req_len, n):
-        t = self.torch
-        nb_req = n * self.slot
-        self.p_reqs[:nb_req] = t.from_numpy(reqs[:nb_req].view(np.uint8))
-        self.p_req_len[:n] = t.from_numpy(req_len.astype(np.int32,
-                                                         copy=False))
-        p_resp, p_rlen = self.process_pinned(n)
-        return p_resp.numpy().copy(), p_rlen.numpy().copy()
+        # compaction mirror (same round16 layout as k_compact)
+        n = len(lens)
+        pads = (resp_len + 15) & ~15
+        roffs = np.zeros(n, np.int32)
+        if n > 1:
+            np.cumsum(pads[:-1], out=roffs[1:])
+        out = np.zeros(int(pads.sum()), np.uint8)
+        for i in range(n):
+            o = int(roffs[i])
+            out[o:o + int(resp_len[i])] = resp_slots[
+                i * self.rslot:i * self.rslot + int(resp_len[i])]
+        return out, roffs, resp_len
 
     # -- pipelined API --------------------------------------------------------
-    def submit(self, n: int, lane_idx: int = 0) -> None:
-        """Enqueue one batch (already staged in lane.p_reqs/p_req_len[:n])
-        on the lane's stream: H2D -> parse -> respond (optimistic) -> D2H.
-        Never blocks. complete() finishes it."""
+    def submit(self, n: int, nbytes: int, lane_idx: int = 0) -> None:
+        """Enqueue one batch (already staged in lane.p_reqs[:nbytes] /
+        p_req_off / p_req_len) on the lane's stream: H2D -> parse ->
+        respond (optimistic) -> pad16 cumsum -> compact -> D2H of
+        lens/offs/total. Never blocks; complete() finishes it."""
         t = self.torch
         ln = self.lanes[lane_idx]
-        slot, rslot = self.slot, self.rslot
-        ln.n = n
+        ln.n, ln.nbytes = n, nbytes
         ln.seed = self._next_seed()
         with t.cuda.stream(ln.stream):
-            ln.d_reqs[:n * slot].copy_(ln.p_reqs[:n * slot],
-                                       non_blocking=True)
+            ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
+            ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
             ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
             ln.d_host_needed.zero_()
             cs = ln.stream.cuda_stream
-            self.hip.parse_route(cs, ln.d_reqs, ln.d_req_len, ln.d_fields,
-                                 n, slot, self.d_trie, self.d_handler_tab,
-                                 self.program.n_routes, ln.d_host_needed)
+            self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
+                                 ln.d_fields, n, self.d_trie,
+                                 self.d_handler_tab, self.program.n_routes,
+                                 ln.d_host_needed)
             ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
             # optimistic respond: host rows render a 500 fallback that the
             # fixup pass overwrites before the responses are released
-            self.hip.respond(cs, ln.d_reqs, ln.d_fields, ln.d_resp,
-                             ln.d_resp_len, n, slot, rslot,
-                             self.d_handler_tab, self.program.n_routes,
-                             self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                             ln.seed)
-            ln.p_resp[:n * rslot].copy_(ln.d_resp[:n * rslot],
-                                        non_blocking=True)
-            ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
+            self._respond_compact(ln, n)
             ln.event.record(ln.stream)
+
+    def _respond_compact(self, ln, n):
+        """respond + pad16 cumsum + compact + D2H of lens/offs/total on
+        the lane's stream (caller holds the stream context)."""
+        t = self.torch
+        cs = ln.stream.cuda_stream
+        self.hip.respond(cs, ln.d_reqs, ln.d_req_off, ln.d_fields,
+                         ln.d_resp, ln.d_resp_len, n, self.rslot,
+                         self.d_handler_tab, self.program.n_routes,
+                         self.d_blob, ln.d_host_blob, ln.d_host_tab,
+                         ln.seed)
+        pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
+        csum = t.cumsum(pads, 0, dtype=t.int32)
+        ln.d_resp_off[:n].copy_(csum - pads)
+        self.hip.compact(cs, ln.d_resp, ln.d_resp_len, ln.d_resp_off,
+                         ln.d_out, n, self.rslot)
+        ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
+        ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
+        ln.p_total.copy_(csum[-1:], non_blocking=True)
 
     def complete(self, lane_idx: int = 0):
         """Wait for the lane's in-flight batch; run the host fixup pass if
-        any row needed the trampoline. Returns pinned (resp, resp_len)."""
+        any row needed the trampoline; D2H the compact response stream.
+        Returns pinned (out, resp_off, resp_len) tensor views."""
         t = self.torch
         ln = self.lanes[lane_idx]
-        n, slot, rslot = ln.n, self.slot, self.rslot
+        n = ln.n
         ln.event.synchronize()
         if int(ln.p_host_needed[0]):
             # fixup: run Python handlers for HK_HOST rows, re-serialize
@@ -239,10 +296,9 @@ class BatchEngine:
                                                non_blocking=True)
             ln.stream.synchronize()
             fields = ln.p_fields[:n * ops.NF].numpy().reshape(n, ops.NF)
-            host_reqs = ln.p_reqs[:n * slot].numpy()
-            host_req_len = ln.p_req_len[:n].numpy()
             host_blob, host_tab = self._run_host_rows(
-                fields, host_reqs, host_req_len)
+                fields, ln.p_reqs.numpy(), ln.p_req_off[:n].numpy(),
+                ln.p_req_len[:n].numpy())
             hb = np.frombuffer(host_blob, np.uint8)
             with t.cuda.stream(ln.stream):
                 if len(hb):
@@ -251,52 +307,38 @@ class BatchEngine:
                 ln.d_host_tab[:n * 4].copy_(
                     t.from_numpy(host_tab.reshape(-1).copy()),
                     non_blocking=True)
-                self.hip.respond(ln.stream.cuda_stream, ln.d_reqs,
-                                 ln.d_fields, ln.d_resp, ln.d_resp_len,
-                                 n, slot, rslot, self.d_handler_tab,
-                                 self.program.n_routes, self.d_blob,
-                                 ln.d_host_blob, ln.d_host_tab, ln.seed)
-                ln.p_resp[:n * rslot].copy_(ln.d_resp[:n * rslot],
-                                            non_blocking=True)
-                ln.p_resp_len[:n].copy_(ln.d_resp_len[:n],
-                                        non_blocking=True)
+                self._respond_compact(ln, n)
             ln.stream.synchronize()
-        return ln.p_resp[:n * rslot], ln.p_resp_len[:n]
+        total = int(ln.p_total[0])
+        with t.cuda.stream(ln.stream):
+            ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
+        ln.stream.synchronize()
+        return ln.p_out[:total], ln.p_resp_off[:n], ln.p_resp_len[:n]
 
-    def process_pinned(self, n, lane_idx: int = 0):
-        """Synchronous one-batch pipeline on a lane (requests staged in
-        lane.p_reqs/p_req_len). The socket layer recv()s directly into the
-        pinned ring, so H2D + kernels + D2H is the whole per-batch path."""
-        self.submit(n, lane_idx)
-        return self.complete(lane_idx)
-
-    def process_device(self, d_reqs, d_req_len, n, host_reqs=None,
-                       host_req_len=None):
-        """Run the kernel pipeline on request bytes already resident on the
-        device (the multi-GPU all-to-all path hands exchanged slabs in
-        directly). Runs on the CALLER's current stream; returns
-        (d_resp, d_resp_len) device tensors, no D2H."""
+    def process_device(self, d_reqs, d_req_off, d_req_len, n):
+        """Kernel pipeline on device-resident packed requests (multi-GPU
+        all-to-all path). Responses stay SLOT-shaped on device (fixed-size
+        all-to-all return). Runs on the caller's current stream; returns
+        (d_resp slots, d_resp_len)."""
         t = self.torch
         ln = self.lanes[0]
-        slot, rslot = self.slot, self.rslot
         seed = self._next_seed()
         stream = t.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
-        self.hip.parse_route(stream, d_reqs, d_req_len,
-                             ln.d_fields, n, slot, self.d_trie,
+        self.hip.parse_route(stream, d_reqs, d_req_off, d_req_len,
+                             ln.d_fields, n, self.d_trie,
                              self.d_handler_tab, self.program.n_routes,
                              ln.d_host_needed)
-        # host trampoline only when some row needs it (4-byte D2H + sync)
         host_needed = int(ln.d_host_needed.item())
         if host_needed:
             ln.p_fields[:n * ops.NF].copy_(ln.d_fields[:n * ops.NF])
             t.cuda.synchronize(self.device)
             fields = ln.p_fields[:n * ops.NF].numpy().reshape(n, ops.NF)
-            if host_reqs is None:
-                host_reqs = d_reqs[:n * slot].cpu().numpy()
-                host_req_len = d_req_len[:n].cpu().numpy()
+            host_reqs = d_reqs.cpu().numpy()
+            host_off = d_req_off[:n].cpu().numpy()
+            host_len = d_req_len[:n].cpu().numpy()
             host_blob, host_tab = self._run_host_rows(
-                fields, host_reqs, host_req_len)
+                fields, host_reqs, host_off, host_len)
             hb = np.frombuffer(host_blob, np.uint8)
             if len(hb):
                 ln.d_host_blob[:len(hb)].copy_(
@@ -304,25 +346,23 @@ class BatchEngine:
             ln.d_host_tab[:n * 4].copy_(
                 t.from_numpy(host_tab.reshape(-1).copy()),
                 non_blocking=True)
-        self.hip.respond(stream, d_reqs, ln.d_fields, ln.d_resp,
-                         ln.d_resp_len, n, slot, rslot,
+        self.hip.respond(stream, d_reqs, d_req_off, ln.d_fields,
+                         ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
-                         self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         seed)
+                         self.d_blob, ln.d_host_blob, ln.d_host_tab, seed)
         return ln.d_resp, ln.d_resp_len
 
-
     # -- host trampoline ------------------------------------------------------
-    def _run_host_rows(self, fields, reqs, req_len):
+    def _run_host_rows(self, fields, reqs, offs, lens):
         """Run Python handlers for HK_HOST rows; returns (blob, tab)."""
-        n = len(req_len)
+        n = len(lens)
         host_tab = np.zeros((n, 4), np.int32)
         blob = bytearray()
         for r in range(n):
             if fields[r][ops.FI_KIND] != ops.HK_HOST:
                 continue
-            raw = reqs[r * self.slot:r * self.slot + int(req_len[r])] \
-                .tobytes()
+            o = int(offs[r])
+            raw = np.asarray(reqs[o:o + int(lens[r])]).tobytes()
             try:
                 request = parse_request_bytes(raw)
                 resp = dispatch(self.app, request)
@@ -337,18 +377,6 @@ class BatchEngine:
             blob += body
             host_tab[r] = (off, len(body), status, _CT_IDS.get(ct, 0))
         return bytes(blob), host_tab
-
-
-def make_batch(payloads: list[bytes], slot: int):
-    """Pack raw request byte strings into the engine's slot layout."""
-    n = len(payloads)
-    reqs = np.zeros(n * slot, np.uint8)
-    lens = np.zeros(n, np.int32)
-    for i, p in enumerate(payloads):
-        assert len(p) <= slot, "request exceeds slot size"
-        reqs[i * slot:i * slot + len(p)] = np.frombuffer(p, np.uint8)
-        lens[i] = len(p)
-    return reqs, lens
 
 
 class GPUServer:
@@ -380,8 +408,9 @@ class GPUServer:
         sock.bind(("0.0.0.0", self.port))
         sock.listen(1024)
         self._listener = sock
-        threading.Thread(target=self._accept_loop, daemon=True).start()
-        threading.Thread(target=self._batch_loop, daemon=True).start()
+        import threading as th
+        th.Thread(target=self._accept_loop, daemon=True).start()
+        th.Thread(target=self._batch_loop, daemon=True).start()
 
     def _accept_loop(self):
         import threading
@@ -451,12 +480,9 @@ class GPUServer:
                     items.append(self._q.get(timeout=remain))
                 except queue.Empty:
                     break
-            payloads = [it[0] for it in items]
-            reqs, lens = make_batch(payloads, self.engine.slot)
-            resp, resp_len = self.engine.process(reqs, lens)
-            for i, (_, done, slotref) in enumerate(items):
-                o = i * self.engine.rslot
-                slotref["resp"] = resp[o:o + int(resp_len[i])].tobytes()
+            outs = self.engine.process([it[0] for it in items])
+            for out, (_, done, slotref) in zip(outs, items):
+                slotref["resp"] = out
                 done.set()
 
     def stop(self):

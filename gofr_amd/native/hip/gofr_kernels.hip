@@ -123,9 +123,10 @@ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
 
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
 k_parse_route(const uint8_t* __restrict__ reqs,
+              const int64_t* __restrict__ req_off,
               const int32_t* __restrict__ req_len,
               int32_t* __restrict__ fields,
-              int n, int slot,
+              int n,
               TrieDev trie,
               const int32_t* __restrict__ handler_tab, int n_routes,
               int32_t* __restrict__ host_needed) {
@@ -135,8 +136,9 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
     if (req >= n) return;
     const int lane = lane_id();
-    const uint8_t* base = reqs + (size_t)req * slot;
-    const int len = req_len[req];
+    const uint8_t* base = reqs + req_off[req];
+    int len = req_len[req];
+    if (len > MAX_SLOT) len = MAX_SLOT;
     int32_t* F = fields + (size_t)req * NF;
 
     const int nchunks = (len + WAVE - 1) / WAVE;
@@ -395,10 +397,11 @@ __device__ __forceinline__ int itoa10(uint8_t* dst, int v) {
 // structural positions, not every byte) before being staged.
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
 k_respond(const uint8_t* __restrict__ reqs,
+          const int64_t* __restrict__ req_off,
           int32_t* __restrict__ fields,
           uint8_t* __restrict__ resp,
           int32_t* __restrict__ resp_len_out,
-          int n, int slot, int rslot,
+          int n, int rslot,
           const int32_t* __restrict__ handler_tab, int n_routes,
           const uint8_t* __restrict__ blob,
           const uint8_t* __restrict__ host_blob,
@@ -412,7 +415,7 @@ k_respond(const uint8_t* __restrict__ reqs,
     const int lane = lane_id();
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
-    const uint8_t* rbase = reqs + (size_t)req * slot;
+    const uint8_t* rbase = reqs + req_off[req];
     uint8_t* obuf = obuf_all + wv * MAX_SLOT;
 
     const int kind = F[FI_KIND];
@@ -603,14 +606,36 @@ k_respond(const uint8_t* __restrict__ reqs,
 }
 
 // ---------------------------------------------------------------------------
+// k_compact — gather response slots into a contiguous 16B-aligned stream
+// (halves the D2H bytes vs slot-strided responses; offsets are the
+// exclusive cumsum of round16(resp_len), computed host-side via torch)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_compact(const uint8_t* __restrict__ resp_slots,
+          const int32_t* __restrict__ resp_len,
+          const int32_t* __restrict__ resp_off,
+          uint8_t* __restrict__ out,
+          int n, int rslot) {
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    const int lane = lane_id();
+    const int len = resp_len[req];
+    const int nv = (len + 15) >> 4;
+    const uint4* src = (const uint4*)(resp_slots + (size_t)req * rslot);
+    uint4* dst = (uint4*)(out + (size_t)resp_off[req]);
+    for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+}
+
+// ---------------------------------------------------------------------------
 // C API (ctypes-driven; raw device pointers from torch tensors)
 // ---------------------------------------------------------------------------
 extern "C" {
 
 int gofr_launch_parse_route(
         void* stream,
-        const void* reqs, const void* req_len, void* fields,
-        int n, int slot,
+        const void* reqs, const void* req_off, const void* req_len,
+        void* fields, int n,
         const void* seg_blob, const void* node_child_first,
         const void* node_child_count, const void* child_seg_off,
         const void* child_seg_len, const void* child_node,
@@ -626,8 +651,9 @@ int gofr_launch_parse_route(
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_parse_route, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
-                       (const uint8_t*)reqs, (const int32_t*)req_len,
-                       (int32_t*)fields, n, slot, trie,
+                       (const uint8_t*)reqs, (const int64_t*)req_off,
+                       (const int32_t*)req_len,
+                       (int32_t*)fields, n, trie,
                        (const int32_t*)handler_tab, n_routes,
                        (int32_t*)host_needed);
     return (int)hipGetLastError();
@@ -635,20 +661,33 @@ int gofr_launch_parse_route(
 
 int gofr_launch_respond(
         void* stream,
-        const void* reqs, void* fields, void* resp, void* resp_len_out,
-        int n, int slot, int rslot,
+        const void* reqs, const void* req_off, void* fields, void* resp,
+        void* resp_len_out, int n, int rslot,
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
         uint64_t seed) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
-                       (const uint8_t*)reqs, (int32_t*)fields,
+                       (const uint8_t*)reqs, (const int64_t*)req_off,
+                       (int32_t*)fields,
                        (uint8_t*)resp, (int32_t*)resp_len_out,
-                       n, slot, rslot,
+                       n, rslot,
                        (const int32_t*)handler_tab, n_routes,
                        (const uint8_t*)blob, (const uint8_t*)host_blob,
                        (const int32_t*)host_tab, seed);
+    return (int)hipGetLastError();
+}
+
+int gofr_launch_compact(
+        void* stream,
+        const void* resp_slots, const void* resp_len, const void* resp_off,
+        void* out, int n, int rslot) {
+    const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)resp_slots, (const int32_t*)resp_len,
+                       (const int32_t*)resp_off, (uint8_t*)out, n, rslot);
     return (int)hipGetLastError();
 }
 

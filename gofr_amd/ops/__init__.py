@@ -76,27 +76,32 @@ class HipOps:
             pass
         self.lib = ctypes.CDLL(so_path)
         self.lib.gofr_launch_parse_route.restype = ctypes.c_int
-        self.lib.gofr_launch_parse_route.argtypes = [ctypes.c_void_p] * 3 + [
-            ctypes.c_void_p, ctypes.c_int, ctypes.c_int] + \
+        self.lib.gofr_launch_parse_route.argtypes = \
+            [ctypes.c_void_p] * 5 + [ctypes.c_int] + \
             [ctypes.c_void_p] * 9 + [ctypes.c_void_p, ctypes.c_int,
                                      ctypes.c_void_p]
         self.lib.gofr_launch_respond.restype = ctypes.c_int
         self.lib.gofr_launch_respond.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-            ctypes.c_void_p, ctypes.c_void_p,
-            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_int, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
             ctypes.c_uint64]
+        self.lib.gofr_launch_compact.restype = ctypes.c_int
+        self.lib.gofr_launch_compact.argtypes = \
+            [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
 
-    def parse_route(self, stream, reqs_t, req_len_t, fields_t, n, slot,
-                    trie_t: dict, handler_tab_t, n_routes, host_needed_t):
+    def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
+                    n, trie_t: dict, handler_tab_t, n_routes,
+                    host_needed_t):
         rc = self.lib.gofr_launch_parse_route(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
+            ctypes.c_void_p(req_off_t.data_ptr()),
             ctypes.c_void_p(req_len_t.data_ptr()),
             ctypes.c_void_p(fields_t.data_ptr()),
-            n, slot,
+            n,
             ctypes.c_void_p(trie_t["seg_blob"].data_ptr()),
             ctypes.c_void_p(trie_t["node_child_first"].data_ptr()),
             ctypes.c_void_p(trie_t["node_child_count"].data_ptr()),
@@ -111,16 +116,17 @@ class HipOps:
         if rc != 0:
             raise RuntimeError(f"k_parse_route launch failed: hipError {rc}")
 
-    def respond(self, stream, reqs_t, fields_t, resp_t, resp_len_t, n, slot,
-                rslot, handler_tab_t, n_routes, blob_t, host_blob_t,
-                host_tab_t, seed):
+    def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
+                resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
+                host_blob_t, host_tab_t, seed):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
+            ctypes.c_void_p(req_off_t.data_ptr()),
             ctypes.c_void_p(fields_t.data_ptr()),
             ctypes.c_void_p(resp_t.data_ptr()),
             ctypes.c_void_p(resp_len_t.data_ptr()),
-            n, slot, rslot,
+            n, rslot,
             ctypes.c_void_p(handler_tab_t.data_ptr()), n_routes,
             ctypes.c_void_p(blob_t.data_ptr()),
             ctypes.c_void_p(host_blob_t.data_ptr()),
@@ -128,6 +134,18 @@ class HipOps:
             ctypes.c_uint64(seed))
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
+
+    def compact(self, stream, resp_slots_t, resp_len_t, resp_off_t, out_t,
+                n, rslot):
+        rc = self.lib.gofr_launch_compact(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(resp_slots_t.data_ptr()),
+            ctypes.c_void_p(resp_len_t.data_ptr()),
+            ctypes.c_void_p(resp_off_t.data_ptr()),
+            ctypes.c_void_p(out_t.data_ptr()),
+            n, rslot)
+        if rc != 0:
+            raise RuntimeError(f"k_compact launch failed: hipError {rc}")
 
 
 # ---------------------------------------------------------------------------
@@ -152,14 +170,15 @@ def _ieq(buf, off, lit: bytes) -> bool:
     return all(_lower(buf[off + i]) == lit[i] for i in range(len(lit)))
 
 
-def cpu_parse_route(reqs: np.ndarray, req_len: np.ndarray, slot: int,
+def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
+                    req_len: np.ndarray,
                     trie: dict, handler_tab: np.ndarray) -> np.ndarray:
-    """Mirror of k_parse_route over a [n*slot] uint8 buffer."""
+    """Mirror of k_parse_route over a packed uint8 buffer + offsets."""
     n = len(req_len)
     fields = np.zeros((n, NF), np.int32)
     n_routes = len(handler_tab) // 4
     for r in range(n):
-        base = r * slot
+        base = int(req_off[r])
         ln = int(req_len[r])
         buf = reqs[base:base + ln].tobytes()
         F = fields[r]
@@ -356,7 +375,8 @@ def _json_body_valid(body: bytes) -> bool:
     return depth == 0 and not in_str and len(body) > 0
 
 
-def cpu_respond(reqs: np.ndarray, fields: np.ndarray, slot: int, rslot: int,
+def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
+                fields: np.ndarray, rslot: int,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
                 seed: int):
@@ -368,7 +388,7 @@ def cpu_respond(reqs: np.ndarray, fields: np.ndarray, slot: int, rslot: int,
     resp_len = np.zeros(n, np.int32)
     for r in range(n):
         F = fields[r]
-        base = r * slot
+        base = int(req_off[r])
         kind = int(F[FI_KIND])
         status = int(F[FI_STATUS])
         flags = int(F[FI_FLAGS])

@@ -9,7 +9,7 @@ import pytest
 
 import gofr_amd
 from gofr_amd import handlers
-from gofr_amd.engine import BatchEngine, make_batch
+from gofr_amd.engine import BatchEngine, pack_batch
 
 
 def build_app(map_config):
@@ -49,9 +49,7 @@ def engine(map_config):
 
 
 def run_one(engine, raw):
-    reqs, lens = make_batch([raw], engine.slot)
-    resp, rlen = engine.process(reqs, lens)
-    return parse_http_response(resp[:int(rlen[0])].tobytes())
+    return parse_http_response(engine.process([raw])[0])
 
 
 def test_echo_roundtrip(engine):
@@ -122,11 +120,7 @@ def test_mixed_batch(engine):
             http_req("GET", "/user/7"),
             http_req("GET", "/missing"),
             http_req("OPTIONS", "/greet")]
-    reqs, lens = make_batch(raws, engine.slot)
-    resp, rlen = engine.process(reqs, lens)
-    outs = [parse_http_response(
-        resp[i * engine.rslot:i * engine.rslot + int(rlen[i])].tobytes())
-        for i in range(len(raws))]
+    outs = [parse_http_response(r) for r in engine.process(raws)]
     assert [o[0] for o in outs] == [200, 200, 200, 404, 200]
     assert json.loads(outs[0][2]) == {"data": {"k": 1}}
     assert json.loads(outs[2][2]) == {"data": {"id": "7"}}
@@ -144,15 +138,15 @@ def test_parse_fields_match_reference_parser(engine):
                           "Connection": "close"}),
         http_req("DELETE", "/user/42/x"),
     ]
-    reqs, lens = make_batch(raws, engine.slot)
-    fields = ops.cpu_parse_route(reqs, lens, engine.slot,
+    reqs, offs, lens = pack_batch(raws)
+    fields = ops.cpu_parse_route(reqs, offs, lens,
                                  engine.program.trie,
                                  engine.program.handler_tab)
     from gofr_amd.http.request import parse_request_bytes
     for i, raw in enumerate(raws):
         ref = parse_request_bytes(raw)
         F = fields[i]
-        base = i * engine.slot
+        base = int(offs[i])
         path = reqs[base + F[ops.FI_PATH_OFF]:
                     base + F[ops.FI_PATH_OFF] + F[ops.FI_PATH_LEN]].tobytes()
         assert path.decode() == ref.path
@@ -163,7 +157,7 @@ def test_parse_fields_match_reference_parser(engine):
         assert F[ops.FI_METHOD] == METHOD_IDS[ref.method]
     # auth header captured
     F = fields[1]
-    base = engine.slot
+    base = int(offs[1])
     auth = reqs[base + F[ops.FI_AUTH_OFF]:
                 base + F[ops.FI_AUTH_OFF] + F[ops.FI_AUTH_LEN]].tobytes()
     assert auth == b"Bearer tok123"

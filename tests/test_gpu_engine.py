@@ -13,7 +13,7 @@ import pytest
 import gofr_amd
 from gofr_amd import handlers
 from gofr_amd.config import MapConfig
-from gofr_amd.engine import BatchEngine, make_batch
+from gofr_amd.engine import BatchEngine, pack_batch
 
 pytestmark = pytest.mark.gpu
 
@@ -71,17 +71,12 @@ def test_kernels_match_cpu_mirror_bytes():
     cpu._seed = gpu._seed
 
     raws = mixed_payloads(512)
-    reqs, lens = make_batch(raws, gpu.slot)
-    g_resp, g_len = gpu.process(reqs.copy(), lens.copy())
-    c_resp, c_len = cpu.process(reqs.copy(), lens.copy())
-
-    assert np.array_equal(g_len, c_len), "response lengths differ"
-    for i in range(len(raws)):
-        go = g_resp[i * gpu.rslot:i * gpu.rslot + int(g_len[i])]
-        co = c_resp[i * cpu.rslot:i * cpu.rslot + int(c_len[i])]
-        assert np.array_equal(go, co), \
-            f"req {i}: GPU bytes != mirror bytes\nGPU: {go.tobytes()!r}\n" \
-            f"CPU: {co.tobytes()!r}\nREQ: {raws[i]!r}"
+    g_out = gpu.process(list(raws))
+    c_out = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g_out, c_out)):
+        assert go == co, \
+            f"req {i}: GPU bytes != mirror bytes\nGPU: {go!r}\n" \
+            f"CPU: {co!r}\nREQ: {raws[i]!r}"
 
 
 def test_gpu_pure_batch_no_host_sync():
@@ -90,7 +85,6 @@ def test_gpu_pure_batch_no_host_sync():
     eng = BatchEngine(app, device="cuda", max_batch=8192)
     body = json.dumps({"k": list(range(50))}).encode()
     raws = [http_req("POST", "/echo", body)] * 4096
-    reqs, lens = make_batch(raws, eng.slot)
     called = []
     orig = eng._run_host_rows
 
@@ -98,10 +92,9 @@ def test_gpu_pure_batch_no_host_sync():
         called.append(1)
         return orig(*a, **kw)
     eng._run_host_rows = spy
-    resp, rlen = eng.process(reqs, lens)
+    outs = eng.process(raws)
     assert not called, "host trampoline ran on a pure-GPU batch"
-    st = resp[:int(rlen[0])].tobytes().split(b" ", 2)[1]
-    assert st == b"200"
+    assert outs[0].split(b" ", 2)[1] == b"200"
 
 
 def test_gpu_large_echo_batch_correct():
@@ -109,12 +102,10 @@ def test_gpu_large_echo_batch_correct():
     eng = BatchEngine(app, device="cuda", max_batch=32768)
     body = (b'{"payload":"' + b"x" * 950 + b'"}')
     raws = [http_req("POST", "/echo", body)] * 16384
-    reqs, lens = make_batch(raws, eng.slot)
-    resp, rlen = eng.process(reqs, lens)
+    outs = eng.process(raws)
     want = b'{"data":{"payload":"' + b"x" * 950 + b'"}}'
     for i in (0, 1, 8191, 16383):
-        raw = resp[i * eng.rslot:i * eng.rslot + int(rlen[i])].tobytes()
-        head, _, got = raw.partition(b"\r\n\r\n")
+        head, _, got = outs[i].partition(b"\r\n\r\n")
         assert got == want, f"row {i} body wrong"
         assert head.startswith(b"HTTP/1.1 200 OK\r\n")
 
