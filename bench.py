@@ -92,6 +92,9 @@ def run_single(eng, payloads, steps, warmup):
         ln.p_req_off[:n] = torch.from_numpy(offs)
         ln.p_req_len[:n] = torch.from_numpy(lens)
     P = len(eng.lanes)
+    # capture the per-lane batch pipeline as a hipGraph (replayed below)
+    for li in range(P):
+        eng.capture_graph(n, nbytes, li)
     # warmup: serial batches
     for _ in range(max(1, warmup)):
         eng.submit(n, nbytes, 0)

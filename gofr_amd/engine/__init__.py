@@ -124,9 +124,12 @@ class _Lane:
         self.p_total = t.empty(1, dtype=t.int32).pin_memory()
         self.p_fields = t.empty(nb * ops.NF, dtype=t.int32).pin_memory()
         self.p_host_needed = t.zeros(1, dtype=t.int32).pin_memory()
+        self.d_seed = t.zeros(1, dtype=t.int64, device=dev)
+        self.p_seed = t.zeros(1, dtype=t.int64).pin_memory()
         self.n = 0
         self.nbytes = 0
-        self.seed = 0
+        self.graph = None
+        self.graph_key = None
 
 
 class BatchEngine:
@@ -245,22 +248,54 @@ class BatchEngine:
         t = self.torch
         ln = self.lanes[lane_idx]
         ln.n, ln.nbytes = n, nbytes
-        ln.seed = self._next_seed()
-        with t.cuda.stream(ln.stream):
-            ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
-            ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
-            ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
-            ln.d_host_needed.zero_()
-            cs = ln.stream.cuda_stream
-            self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
-                                 ln.d_fields, n, self.d_trie,
-                                 self.d_handler_tab, self.program.n_routes,
-                                 ln.d_host_needed)
-            ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
-            # optimistic respond: host rows render a 500 fallback that the
-            # fixup pass overwrites before the responses are released
-            self._respond_compact(ln, n)
+        seed = self._next_seed()
+        ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
+        if ln.graph is not None and ln.graph_key == (n, nbytes):
+            # hipGraph replay: one launch for the whole batch pipeline
+            # (the captured H2D copies re-read the pinned rings, so fresh
+            # request bytes and the fresh seed flow through the replay)
+            ln.graph.replay()
             ln.event.record(ln.stream)
+            return
+        with t.cuda.stream(ln.stream):
+            self._submit_body(ln, n, nbytes)
+            ln.event.record(ln.stream)
+
+    def _submit_body(self, ln, n, nbytes):
+        cs = ln.stream.cuda_stream
+        ln.d_seed.copy_(ln.p_seed, non_blocking=True)
+        ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
+        ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
+        ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
+        ln.d_host_needed.zero_()
+        self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
+                             ln.d_fields, n, self.d_trie,
+                             self.d_handler_tab, self.program.n_routes,
+                             ln.d_host_needed)
+        ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
+        # optimistic respond: host rows render a 500 fallback that the
+        # fixup pass overwrites before the responses are released
+        self._respond_compact(ln, n)
+
+    def capture_graph(self, n: int, nbytes: int, lane_idx: int = 0) -> bool:
+        """Capture the lane's batch pipeline into a hipGraph for (n,
+        nbytes)-shaped batches; submit() then replays it (one host call
+        instead of ~10 launches/copies per batch). Returns False when the
+        torch build lacks graph support."""
+        t = self.torch
+        ln = self.lanes[lane_idx]
+        if not hasattr(t.cuda, "CUDAGraph"):
+            return False
+        # warmup pass (allocations settle) then capture
+        with t.cuda.stream(ln.stream):
+            self._submit_body(ln, n, nbytes)
+        ln.stream.synchronize()
+        g = t.cuda.CUDAGraph()
+        with t.cuda.graph(g, stream=ln.stream):
+            self._submit_body(ln, n, nbytes)
+        ln.graph = g
+        ln.graph_key = (n, nbytes)
+        return True
 
     def _respond_compact(self, ln, n):
         """respond + pad16 cumsum + compact + D2H of lens/offs/total on
@@ -271,7 +306,7 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.seed)
+                         ln.d_seed)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
@@ -323,6 +358,8 @@ class BatchEngine:
         t = self.torch
         ln = self.lanes[0]
         seed = self._next_seed()
+        ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
+        ln.d_seed.copy_(ln.p_seed, non_blocking=True)
         stream = t.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
         self.hip.parse_route(stream, d_reqs, d_req_off, d_req_len,
@@ -349,7 +386,8 @@ class BatchEngine:
         self.hip.respond(stream, d_reqs, d_req_off, ln.d_fields,
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
-                         self.d_blob, ln.d_host_blob, ln.d_host_tab, seed)
+                         self.d_blob, ln.d_host_blob, ln.d_host_tab,
+                         ln.d_seed)
         return ln.d_resp, ln.d_resp_len
 
     # -- host trampoline ------------------------------------------------------

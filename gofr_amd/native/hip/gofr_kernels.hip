@@ -406,7 +406,8 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ blob,
           const uint8_t* __restrict__ host_blob,
           const int32_t* __restrict__ host_tab,
-          uint64_t seed) {
+          const uint64_t* __restrict__ seed_ptr) {
+    const uint64_t seed = *seed_ptr;
     __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
 
     const int wv = threadIdx.x / WAVE;
@@ -665,7 +666,7 @@ int gofr_launch_respond(
         void* resp_len_out, int n, int rslot,
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
-        uint64_t seed) {
+        const void* seed_ptr) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
@@ -675,7 +676,7 @@ int gofr_launch_respond(
                        n, rslot,
                        (const int32_t*)handler_tab, n_routes,
                        (const uint8_t*)blob, (const uint8_t*)host_blob,
-                       (const int32_t*)host_tab, seed);
+                       (const int32_t*)host_tab, (const uint64_t*)seed_ptr);
     return (int)hipGetLastError();
 }
 

@@ -87,7 +87,7 @@ class HipOps:
             ctypes.c_int, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-            ctypes.c_uint64]
+            ctypes.c_void_p]
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
@@ -118,7 +118,7 @@ class HipOps:
 
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
-                host_blob_t, host_tab_t, seed):
+                host_blob_t, host_tab_t, seed_t):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -131,7 +131,7 @@ class HipOps:
             ctypes.c_void_p(blob_t.data_ptr()),
             ctypes.c_void_p(host_blob_t.data_ptr()),
             ctypes.c_void_p(host_tab_t.data_ptr()),
-            ctypes.c_uint64(seed))
+            ctypes.c_void_p(seed_t.data_ptr()))
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
 
