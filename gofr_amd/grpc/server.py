@@ -1,0 +1,274 @@
+"""gRPC unary server over the from-scratch HTTP/2 transport.
+
+Reference: pkg/gofr/grpc.go:16-47 (grpcServer with panic-recovery +
+logging interceptor chain, serving registered protoc services) and
+pkg/gofr/grpc/log.go:15-50 (RPCLog with trace id, start time, µs
+response time, method).
+
+Services are registered via App.RegisterService(ServiceDesc, impl);
+impl methods have the signature method(ctx, request_dict) -> dict.
+gRPC handlers get a plain context (not *gofr.Context), matching the
+reference (SURVEY.md §3.5 note).
+"""
+
+from __future__ import annotations
+
+import socket
+import struct
+import threading
+import time
+
+from .codec import MessageDesc, decode_message, encode_message
+from . import http2 as h2
+
+
+class RPCLog:
+    """Reference: grpc/log.go:15-25."""
+
+    __slots__ = ("id", "start_time", "response_time_us", "method")
+
+    def __init__(self, id_, start_time, response_time_us, method):
+        self.id = id_
+        self.start_time = start_time
+        self.response_time_us = response_time_us
+        self.method = method
+
+    def to_dict(self):
+        return {"id": self.id, "startTime": self.start_time,
+                "responseTime": self.response_time_us,
+                "method": self.method}
+
+    def pretty(self) -> str:
+        return f"RPC    {self.response_time_us:8.0f}µs  {self.method}"
+
+
+class ServiceDesc:
+    """methods: {name: (request MessageDesc, response MessageDesc)}."""
+
+    def __init__(self, name: str, methods: dict):
+        self.name = name
+        self.methods = methods
+
+
+class GRPCServer:
+    def __init__(self, app, port: int):
+        self.app = app
+        self.port = port
+        self._services: dict[str, tuple[ServiceDesc, object]] = {}
+        for service, impl in app._grpc_services:
+            self._services[service.name] = (service, impl)
+        self._stop = threading.Event()
+        self._sock = None
+
+    def start(self) -> None:
+        sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        sock.bind(("0.0.0.0", self.port))
+        sock.listen(256)
+        self._sock = sock
+        threading.Thread(target=self._accept_loop, daemon=True).start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._sock is not None:
+            try:
+                self._sock.close()
+            except OSError:
+                pass
+
+    def _accept_loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                conn, _ = self._sock.accept()
+            except OSError:
+                return
+            threading.Thread(target=self._serve_conn, args=(conn,),
+                             daemon=True).start()
+
+    # -- one HTTP/2 connection ------------------------------------------------
+    def _serve_conn(self, conn: socket.socket) -> None:
+        conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        buf = bytearray()
+
+        def read_exact(n: int) -> bytes:
+            while len(buf) < n:
+                chunk = conn.recv(65536)
+                if not chunk:
+                    raise ConnectionError("closed")
+                buf.extend(chunk)
+            out = bytes(buf[:n])
+            del buf[:n]
+            return out
+
+        try:
+            preface = read_exact(len(h2.PREFACE))
+            if preface != h2.PREFACE:
+                return
+            conn.sendall(h2.pack_frame(h2.FT_SETTINGS, 0, 0, b""))
+            decoder = h2.HpackDecoder()
+            streams: dict[int, dict] = {}
+            while not self._stop.is_set():
+                ftype, flags, sid, payload = h2.read_frame(read_exact)
+                if ftype == h2.FT_SETTINGS:
+                    if not flags & h2.FLAG_ACK:
+                        conn.sendall(h2.pack_frame(h2.FT_SETTINGS,
+                                                   h2.FLAG_ACK, 0, b""))
+                elif ftype == h2.FT_PING:
+                    if not flags & h2.FLAG_ACK:
+                        conn.sendall(h2.pack_frame(h2.FT_PING, h2.FLAG_ACK,
+                                                   0, payload))
+                elif ftype == h2.FT_HEADERS:
+                    pos = 0
+                    if flags & h2.FLAG_PADDED:
+                        pos += 1
+                    if flags & h2.FLAG_PRIORITY:
+                        pos += 5
+                    headers = decoder.decode(payload[pos:])
+                    streams[sid] = {"headers": dict(headers),
+                                    "data": bytearray()}
+                    if flags & h2.FLAG_END_STREAM:
+                        self._dispatch(conn, sid, streams.pop(sid))
+                elif ftype == h2.FT_DATA:
+                    st = streams.get(sid)
+                    if st is None:
+                        continue
+                    st["data"].extend(payload)
+                    if flags & h2.FLAG_END_STREAM:
+                        self._dispatch(conn, sid, streams.pop(sid))
+                elif ftype == h2.FT_GOAWAY:
+                    return
+                elif ftype in (h2.FT_WINDOW_UPDATE, h2.FT_RST_STREAM,
+                               h2.FT_CONTINUATION):
+                    pass
+        except (ConnectionError, OSError, ValueError):
+            return
+        finally:
+            try:
+                conn.close()
+            except OSError:
+                pass
+
+    # -- unary dispatch with interceptors -------------------------------------
+    def _dispatch(self, conn, sid: int, stream: dict) -> None:
+        path = stream["headers"].get(":path", "")
+        start = time.time()
+        t0 = time.perf_counter_ns()
+        tracer = self.app.tracer
+        span = tracer.start_span(f"grpc{path}") if tracer else None
+        grpc_status, msg_bytes, err_msg = 0, b"", ""
+        try:
+            service_name, _, method_name = path.lstrip("/").partition("/")
+            entry = self._services.get(service_name)
+            if entry is None or method_name not in entry[0].methods:
+                grpc_status, err_msg = 12, f"unknown method {path}"  # UNIMPL.
+            else:
+                desc, impl = entry
+                req_desc, resp_desc = desc.methods[method_name]
+                data = bytes(stream["data"])
+                if len(data) < 5:
+                    raise ValueError("short gRPC frame")
+                compressed = data[0]
+                mlen = struct.unpack(">I", data[1:5])[0]
+                if compressed:
+                    grpc_status, err_msg = 12, "compression not supported"
+                else:
+                    req = decode_message(data[5:5 + mlen], req_desc)
+                    fn = getattr(impl, method_name)
+                    resp = fn(None, req)
+                    msg_bytes = encode_message(resp or {}, resp_desc)
+        except Exception as e:  # noqa: BLE001 — recovery interceptor
+            # reference: grpc_recovery -> codes.Internal (grpc.go:25)
+            grpc_status, err_msg = 13, str(e) or "internal error"
+        finally:
+            if span is not None:
+                span.End()
+            dur_us = (time.perf_counter_ns() - t0) / 1000.0
+            from ..http.middleware import rfc3339nano
+            self.app.container.logger.info_record(RPCLog(
+                span.trace_id if span else "", rfc3339nano(start),
+                dur_us, path))
+
+        # response: HEADERS + DATA + trailers
+        hdr = h2.HpackEncoder.encode([
+            (":status", "200"), ("content-type", "application/grpc")])
+        conn.sendall(h2.pack_frame(h2.FT_HEADERS, h2.FLAG_END_HEADERS,
+                                   sid, hdr))
+        if grpc_status == 0:
+            frame = bytes([0]) + struct.pack(">I", len(msg_bytes)) + msg_bytes
+            conn.sendall(h2.pack_frame(h2.FT_DATA, 0, sid, frame))
+        trailers = [("grpc-status", str(grpc_status))]
+        if err_msg:
+            trailers.append(("grpc-message", err_msg))
+        conn.sendall(h2.pack_frame(
+            h2.FT_HEADERS, h2.FLAG_END_HEADERS | h2.FLAG_END_STREAM,
+            sid, h2.HpackEncoder.encode(trailers)))
+
+
+class GRPCClient:
+    """Minimal unary client (tests + inter-service calls)."""
+
+    def __init__(self, host: str, port: int, timeout: float = 5.0):
+        self.sock = socket.create_connection((host, port), timeout=timeout)
+        self.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        self.sock.sendall(h2.PREFACE)
+        self.sock.sendall(h2.pack_frame(h2.FT_SETTINGS, 0, 0, b""))
+        self._buf = bytearray()
+        self._next_sid = 1
+        self._decoder = h2.HpackDecoder()
+
+    def _read_exact(self, n: int) -> bytes:
+        while len(self._buf) < n:
+            chunk = self.sock.recv(65536)
+            if not chunk:
+                raise ConnectionError("closed")
+            self._buf.extend(chunk)
+        out = bytes(self._buf[:n])
+        del self._buf[:n]
+        return out
+
+    def call(self, service: str, method: str, request: dict,
+             req_desc: MessageDesc, resp_desc: MessageDesc):
+        """Returns (response dict or None, grpc_status, message)."""
+        sid = self._next_sid
+        self._next_sid += 2
+        headers = h2.HpackEncoder.encode([
+            (":method", "POST"), (":scheme", "http"),
+            (":path", f"/{service}/{method}"), (":authority", "localhost"),
+            ("content-type", "application/grpc"), ("te", "trailers")])
+        self.sock.sendall(h2.pack_frame(h2.FT_HEADERS, h2.FLAG_END_HEADERS,
+                                        sid, headers))
+        payload = encode_message(request, req_desc)
+        frame = bytes([0]) + struct.pack(">I", len(payload)) + payload
+        self.sock.sendall(h2.pack_frame(h2.FT_DATA, h2.FLAG_END_STREAM,
+                                        sid, frame))
+        resp_msg = None
+        status = -1
+        message = ""
+        while True:
+            ftype, flags, fsid, payload = h2.read_frame(self._read_exact)
+            if ftype == h2.FT_SETTINGS:
+                if not flags & h2.FLAG_ACK:
+                    self.sock.sendall(h2.pack_frame(h2.FT_SETTINGS,
+                                                    h2.FLAG_ACK, 0, b""))
+                continue
+            if fsid != sid:
+                continue
+            if ftype == h2.FT_HEADERS:
+                hdrs = dict(self._decoder.decode(payload))
+                if "grpc-status" in hdrs:
+                    status = int(hdrs["grpc-status"])
+                    message = hdrs.get("grpc-message", "")
+                if flags & h2.FLAG_END_STREAM:
+                    return resp_msg, status, message
+            elif ftype == h2.FT_DATA:
+                if len(payload) >= 5:
+                    mlen = struct.unpack(">I", payload[1:5])[0]
+                    resp_msg = decode_message(payload[5:5 + mlen], resp_desc)
+                if flags & h2.FLAG_END_STREAM:
+                    return resp_msg, status, message
+
+    def close(self):
+        try:
+            self.sock.close()
+        except OSError:
+            pass

@@ -1,0 +1,171 @@
+"""gRPC layer tests: codec round trips, HPACK, real-socket unary calls.
+
+Mirrors the reference's grpc tier (examples/grpc-server/main_test.go:
+start server, drive a real unary call) plus codec unit tests the
+reference gets from the protobuf runtime.
+"""
+
+import threading
+import time
+
+import pytest
+
+import gofr_amd
+from gofr_amd.config import MapConfig
+from gofr_amd.grpc import MessageDesc, decode_message, encode_message
+from gofr_amd.grpc.codec import HELLO_REQUEST, HELLO_RESPONSE
+from gofr_amd.grpc.http2 import HpackDecoder, HpackEncoder, huffman_decode
+from gofr_amd.grpc.server import GRPCClient, ServiceDesc
+
+
+# -- codec -------------------------------------------------------------------
+
+def test_codec_roundtrip_scalars():
+    desc = MessageDesc("T", {
+        1: ("s", "string"), 2: ("i", "int64"), 3: ("b", "bool"),
+        4: ("d", "double"), 5: ("raw", "bytes"), 6: ("u", "uint32"),
+    })
+    msg = {"s": "héllo", "i": -42, "b": True, "d": 3.5,
+           "raw": b"\x00\x01", "u": 7}
+    assert decode_message(encode_message(msg, desc), desc) == msg
+
+
+def test_codec_nested_and_repeated():
+    inner = MessageDesc("Inner", {1: ("x", "int32")})
+    desc = MessageDesc("T", {
+        1: ("items", "repeated_string"),
+        2: ("child", "message", inner),
+        3: ("nums", "repeated_int32"),
+    })
+    msg = {"items": ["a", "b"], "child": {"x": -5}, "nums": [1, 2, 3]}
+    assert decode_message(encode_message(msg, desc), desc) == msg
+
+
+def test_codec_unknown_field_skipped():
+    d1 = MessageDesc("A", {1: ("a", "string"), 2: ("b", "int32")})
+    d2 = MessageDesc("B", {1: ("a", "string")})
+    data = encode_message({"a": "x", "b": 9}, d1)
+    assert decode_message(data, d2) == {"a": "x"}
+
+
+def test_codec_proto3_defaults():
+    data = encode_message({"name": ""}, HELLO_REQUEST)
+    assert data == b""  # default string elided
+    assert decode_message(b"", HELLO_REQUEST) == {"name": ""}
+
+
+# -- HPACK -------------------------------------------------------------------
+
+def test_hpack_roundtrip_via_own_encoder():
+    headers = [(":method", "POST"), (":path", "/hello.Hello/SayHello"),
+               ("content-type", "application/grpc"),
+               ("x-long", "v" * 300)]
+    data = HpackEncoder.encode(headers)
+    assert HpackDecoder().decode(data) == headers
+
+
+def test_hpack_static_indexed():
+    # 0x82 = indexed entry 2 (:method GET), 0x86 = (:scheme http)
+    assert HpackDecoder().decode(b"\x82\x86") == [
+        (":method", "GET"), (":scheme", "http")]
+
+
+def test_hpack_huffman_rfc_vector():
+    # RFC 7541 C.4.1: "www.example.com"
+    data = bytes.fromhex("f1e3c2e5f23a6ba0ab90f4ff")
+    assert huffman_decode(data) == b"www.example.com"
+
+
+def test_hpack_literal_incremental_and_reuse():
+    dec = HpackDecoder()
+    # literal with incremental indexing, new name "x-a": "1"
+    block = b"\x40" + b"\x03x-a" + b"\x011"
+    assert dec.decode(block) == [("x-a", "1")]
+    # now indexed from the dynamic table (index 62)
+    assert dec.decode(b"\xbe") == [("x-a", "1")]
+
+
+# -- real-socket unary server -------------------------------------------------
+
+class HelloImpl:
+    def SayHello(self, ctx, req):
+        name = req.get("name") or "World"
+        return {"message": f"Hello {name}!"}
+
+    def Boom(self, ctx, req):
+        raise RuntimeError("kaboom")
+
+
+@pytest.fixture()
+def grpc_app():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    svc = ServiceDesc("hello.Hello", {
+        "SayHello": (HELLO_REQUEST, HELLO_RESPONSE),
+        "Boom": (HELLO_REQUEST, HELLO_RESPONSE),
+    })
+    app.RegisterService(svc, HelloImpl())
+    from gofr_amd.grpc.server import GRPCServer
+    server = GRPCServer(app, 0)
+    import socket as s
+    sock = s.socket(s.AF_INET, s.SOCK_STREAM)
+    sock.setsockopt(s.SOL_SOCKET, s.SO_REUSEADDR, 1)
+    sock.bind(("127.0.0.1", 0))
+    sock.listen(64)
+    server._sock = sock
+    server.port = sock.getsockname()[1]
+    threading.Thread(target=server._accept_loop, daemon=True).start()
+    yield server
+    server.stop()
+
+
+def test_grpc_unary_echo(grpc_app):
+    c = GRPCClient("127.0.0.1", grpc_app.port)
+    resp, status, msg = c.call("hello.Hello", "SayHello", {"name": "gofr"},
+                               HELLO_REQUEST, HELLO_RESPONSE)
+    assert status == 0 and resp == {"message": "Hello gofr!"}
+    # empty name -> default (reference: examples/grpc-server/grpc/server.go)
+    resp, status, _ = c.call("hello.Hello", "SayHello", {},
+                             HELLO_REQUEST, HELLO_RESPONSE)
+    assert resp == {"message": "Hello World!"}
+    c.close()
+
+
+def test_grpc_panic_recovery_internal(grpc_app):
+    c = GRPCClient("127.0.0.1", grpc_app.port)
+    resp, status, msg = c.call("hello.Hello", "Boom", {"name": "x"},
+                               HELLO_REQUEST, HELLO_RESPONSE)
+    assert status == 13 and "kaboom" in msg  # codes.Internal
+    c.close()
+
+
+def test_grpc_unknown_method(grpc_app):
+    c = GRPCClient("127.0.0.1", grpc_app.port)
+    resp, status, _ = c.call("hello.Hello", "Nope", {},
+                             HELLO_REQUEST, HELLO_RESPONSE)
+    assert status == 12  # UNIMPLEMENTED
+    c.close()
+
+
+def test_grpc_rpc_logged():
+    from gofr_amd.testutil import MockLogger
+    from gofr_amd import logging as gl
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.container.logger = MockLogger(level=gl.DEBUG)
+    svc = ServiceDesc("hello.Hello",
+                      {"SayHello": (HELLO_REQUEST, HELLO_RESPONSE)})
+    app.RegisterService(svc, HelloImpl())
+    from gofr_amd.grpc.server import GRPCServer
+    import socket as s
+    server = GRPCServer(app, 0)
+    sock = s.socket(s.AF_INET, s.SOCK_STREAM)
+    sock.bind(("127.0.0.1", 0))
+    sock.listen(8)
+    server._sock = sock
+    threading.Thread(target=server._accept_loop, daemon=True).start()
+    c = GRPCClient("127.0.0.1", sock.getsockname()[1])
+    c.call("hello.Hello", "SayHello", {"name": "log"},
+           HELLO_REQUEST, HELLO_RESPONSE)
+    c.close()
+    time.sleep(0.05)
+    assert "/hello.Hello/SayHello" in app.container.logger.stdout
+    server.stop()
