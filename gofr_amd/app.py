@@ -76,6 +76,8 @@ class App:
         self.http_server = None
         self.grpc_server = None
         self.engine = None  # GPU batch engine, created in Run when enabled
+        self.auth_secret = None   # enable_auth()
+        self.gzip_min_size = None  # enable_gzip()
 
     @staticmethod
     def _read_config():
@@ -129,6 +131,18 @@ class App:
                 "service %s already registered", name)
         self.container.services[name] = NewHTTPService(
             address, logger=self.container.logger, tracer=self.tracer)
+
+    # -- middlewares (BASELINE config 4: auth + gzip + log) -------------------
+    def enable_auth(self, secret: bytes):
+        """HMAC-SHA256 bearer auth on every non-OPTIONS request:
+        Authorization: HMAC <hex of HMAC-SHA256(secret, "METHOD path")>.
+        GPU engine: k_auth kernel; CPU transport: checked in dispatch."""
+        self.auth_secret = bytes(secret)
+
+    def enable_gzip(self, min_size: int = 256):
+        """gzip-compress JSON responses when the request advertises
+        Accept-Encoding: gzip and the body is >= min_size bytes."""
+        self.gzip_min_size = int(min_size)
 
     # -- CLI — reference: gofr.go:181-183 -------------------------------------
     def SubCommand(self, pattern: str, handler):

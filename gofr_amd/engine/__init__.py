@@ -33,6 +33,7 @@ _CT_IDS = {"application/json": 0, "image/x-icon": 1,
 
 INVALID_BODY_ENV = b'{"error":{"message":"invalid JSON body"}}'
 NOTFOUND_ENV = b'{"error":{"message":"http: no such file"}}'
+UNAUTHORIZED_ENV = b'{"error":{"message":"unauthorized"}}'
 
 
 class RouteProgram:
@@ -69,6 +70,9 @@ class RouteProgram:
                 rows.append((ops.HK_STATIC, off, len(body), status))
             else:
                 rows.append((ops.HK_HOST, 0, 0, 200))
+        # auth middleware 401 envelope (k_respond auth_env_off/len)
+        self.auth_env = (len(blob), len(UNAUTHORIZED_ENV))
+        blob += UNAUTHORIZED_ENV
         self.handler_tab = np.asarray(rows, np.int32).reshape(-1)
         self.n_routes = len(rows)
         self.blob = bytes(blob)
@@ -181,6 +185,12 @@ class BatchEngine:
         self.d_handler_tab = t.as_tensor(self.program.handler_tab).to(dev)
         self.d_blob = t.as_tensor(
             np.frombuffer(self.program.blob, np.uint8).copy()).to(dev)
+        secret = getattr(self.app, "auth_secret", None)
+        if secret:
+            self.d_secret = t.as_tensor(
+                np.frombuffer(secret, np.uint8).copy()).to(dev)
+        else:
+            self.d_secret = None
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot) for _ in range(self.pipeline)]
 
@@ -222,10 +232,14 @@ class BatchEngine:
         seed = self._next_seed()
         fields = ops.cpu_parse_route(buf, offs, lens, self.program.trie,
                                      self.program.handler_tab)
+        secret = getattr(self.app, "auth_secret", None)
+        if secret:
+            ops.cpu_auth(buf, offs, fields, secret)
         host_blob, host_tab = self._run_host_rows(fields, buf, offs, lens)
         resp_slots, resp_len = ops.cpu_respond(
             buf, offs, fields, self.rslot, self.program.handler_tab,
-            self.program.blob, host_blob, host_tab, seed)
+            self.program.blob, host_blob, host_tab, seed,
+            auth_env=self.program.auth_env)
         # compaction mirror (same round16 layout as k_compact)
         n = len(lens)
         pads = (resp_len + 15) & ~15
@@ -273,6 +287,9 @@ class BatchEngine:
                              self.d_handler_tab, self.program.n_routes,
                              ln.d_host_needed)
         ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
+        if self.d_secret is not None:
+            self.hip.auth(cs, ln.d_reqs, ln.d_req_off, ln.d_fields, n,
+                          self.d_secret, len(self.app.auth_secret))
         # optimistic respond: host rows render a 500 fallback that the
         # fixup pass overwrites before the responses are released
         self._respond_compact(ln, n)
@@ -306,7 +323,7 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.d_seed)
+                         ln.d_seed, auth_env=self.program.auth_env)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
@@ -366,6 +383,9 @@ class BatchEngine:
                              ln.d_fields, n, self.d_trie,
                              self.d_handler_tab, self.program.n_routes,
                              ln.d_host_needed)
+        if self.d_secret is not None:
+            self.hip.auth(stream, d_reqs, d_req_off, ln.d_fields, n,
+                          self.d_secret, len(self.app.auth_secret))
         host_needed = int(ln.d_host_needed.item())
         if host_needed:
             ln.p_fields[:n * ops.NF].copy_(ln.d_fields[:n * ops.NF])
@@ -387,7 +407,7 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.d_seed)
+                         ln.d_seed, auth_env=self.program.auth_env)
         return ln.d_resp, ln.d_resp_len
 
     # -- host trampoline ------------------------------------------------------
@@ -399,6 +419,8 @@ class BatchEngine:
         for r in range(n):
             if fields[r][ops.FI_KIND] != ops.HK_HOST:
                 continue
+            if fields[r][ops.FI_FLAGS] & ops.FL_AUTH_FAIL:
+                continue  # auth middleware blocks the handler
             o = int(offs[r])
             raw = np.asarray(reqs[o:o + int(lens[r])]).tobytes()
             try:

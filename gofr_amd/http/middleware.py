@@ -84,3 +84,25 @@ def make_request_log(span, request, status: int, start: float,
 def panic_log(error: BaseException) -> dict:
     """Reference: middleware/logger.go:86-89 panicLog (error + stack)."""
     return {"error": str(error), "stack": traceback.format_exc()}
+
+
+def hmac_token(secret: bytes, method: str, path: str) -> str:
+    """The auth middleware's expected MAC: HMAC-SHA256(secret,
+    "METHOD path") hex. GPU analog: k_auth in gofr_kernels.hip."""
+    import hashlib
+    import hmac
+    return hmac.new(secret, f"{method} {path}".encode(),
+                    hashlib.sha256).hexdigest()
+
+
+def auth_ok(secret: bytes, request) -> bool:
+    import hmac
+    value = request.header("authorization")
+    if not value.startswith("HMAC ") or len(value) != 69:
+        return False
+    return hmac.compare_digest(value[5:].lower(),
+                               hmac_token(secret, request.method,
+                                          request.path))
+
+
+UNAUTHORIZED_BODY = b'{"error":{"message":"unauthorized"}}'

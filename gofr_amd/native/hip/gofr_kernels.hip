@@ -57,6 +57,8 @@
 #define FL_JSON_CT 8
 #define FL_IS_OPTIONS 16
 #define FL_BODY_INVALID 32
+#define FL_ACCEPT_GZIP 64
+#define FL_AUTH_FAIL 128
 
 // handler kinds (handler table rows: [kind, arg_off, arg_len, status])
 #define HK_HOST 0
@@ -260,6 +262,15 @@ k_parse_route(const uint8_t* __restrict__ reqs,
             auth_off = vs; auth_len = vlen;
         } else if (nlen == 17 && ieq(nm, "transfer-encoding", 17)) {
             flags |= FL_NEEDS_HOST;  // chunked -> host slow path
+        } else if (nlen == 15 && ieq(nm, "accept-encoding", 15)) {
+            // substring scan for "gzip" in the value
+            for (int i = 0; i + 4 <= vlen; ++i) {
+                if (base[vs+i]=='g' && base[vs+i+1]=='z' &&
+                    base[vs+i+2]=='i' && base[vs+i+3]=='p') {
+                    flags |= FL_ACCEPT_GZIP;
+                    break;
+                }
+            }
         }
         prev_lf = lf;
     }
@@ -406,7 +417,8 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ blob,
           const uint8_t* __restrict__ host_blob,
           const int32_t* __restrict__ host_tab,
-          const uint64_t* __restrict__ seed_ptr) {
+          const uint64_t* __restrict__ seed_ptr,
+          int auth_env_off, int auth_env_len) {
     const uint64_t seed = *seed_ptr;
     __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
 
@@ -503,7 +515,7 @@ k_respond(const uint8_t* __restrict__ reqs,
         }
     }
     int ct_id = 0;  // 0 json, 1 icon, 2 octet-stream, 3 text/plain
-    if (kind == HK_HOST && !is_options) {
+    if (kind == HK_HOST && !is_options && !(flags & FL_AUTH_FAIL)) {
         body_src = host_blob + host_tab[req * 4 + 0];
         body_src_len = host_tab[req * 4 + 1];
         status = host_tab[req * 4 + 2];
@@ -514,6 +526,15 @@ k_respond(const uint8_t* __restrict__ reqs,
             body_src = blob + 4;
             body_src_len = elen;
         }
+    }
+
+    // auth middleware verdict overrides everything but OPTIONS
+    if ((flags & FL_AUTH_FAIL) && !is_options) {
+        status = 401;
+        body_src = blob + auth_env_off;
+        body_src_len = auth_env_len;
+        env = 0;
+        ct_id = 0;
     }
 
     const int body_total = body_src_len + (env ? 9 : 0);  // {"data": + }
@@ -607,6 +628,248 @@ k_respond(const uint8_t* __restrict__ reqs,
 }
 
 // ---------------------------------------------------------------------------
+// k_auth — batched HMAC-SHA256 bearer-token check, one request per LANE.
+//
+// Middleware semantics (the auth middleware of BASELINE config 4): the
+// Authorization header must be "HMAC <64 hex>" where the MAC is
+// HMAC-SHA256(secret, METHOD + " " + path). Failure sets FL_AUTH_FAIL in
+// the request flags; k_respond renders the 401 envelope. SHA-256 is
+// inherently serial per message, so the parallel axis is requests (one
+// per lane: a 256-thread block checks 256 requests).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint32_t ror32(uint32_t x, int n) {
+    return (x >> n) | (x << (32 - n));
+}
+
+__constant__ uint32_t SHA_K[64] = {
+    0x428a2f98,0x71374491,0xb5c0fbcf,0xe9b5dba5,0x3956c25b,0x59f111f1,
+    0x923f82a4,0xab1c5ed5,0xd807aa98,0x12835b01,0x243185be,0x550c7dc3,
+    0x72be5d74,0x80deb1fe,0x9bdc06a7,0xc19bf174,0xe49b69c1,0xefbe4786,
+    0x0fc19dc6,0x240ca1cc,0x2de92c6f,0x4a7484aa,0x5cb0a9dc,0x76f988da,
+    0x983e5152,0xa831c66d,0xb00327c8,0xbf597fc7,0xc6e00bf3,0xd5a79147,
+    0x06ca6351,0x14292967,0x27b70a85,0x2e1b2138,0x4d2c6dfc,0x53380d13,
+    0x650a7354,0x766a0abb,0x81c2c92e,0x92722c85,0xa2bfe8a1,0xa81a664b,
+    0xc24b8b70,0xc76c51a3,0xd192e819,0xd6990624,0xf40e3585,0x106aa070,
+    0x19a4c116,0x1e376c08,0x2748774c,0x34b0bcb5,0x391c0cb3,0x4ed8aa4a,
+    0x5b9cca4f,0x682e6ff3,0x748f82ee,0x78a5636f,0x84c87814,0x8cc70208,
+    0x90befffa,0xa4506ceb,0xbef9a3f7,0xc67178f2};
+
+struct Sha256Ctx {
+    uint32_t h[8];
+    uint8_t buf[64];
+    int buflen;
+    uint64_t total;
+};
+
+__device__ void sha256_init(Sha256Ctx* c) {
+    c->h[0]=0x6a09e667; c->h[1]=0xbb67ae85; c->h[2]=0x3c6ef372;
+    c->h[3]=0xa54ff53a; c->h[4]=0x510e527f; c->h[5]=0x9b05688c;
+    c->h[6]=0x1f83d9ab; c->h[7]=0x5be0cd19;
+    c->buflen = 0;
+    c->total = 0;
+}
+
+__device__ void sha256_block(Sha256Ctx* c, const uint8_t* p) {
+    uint32_t w[64];
+    for (int i = 0; i < 16; ++i)
+        w[i] = ((uint32_t)p[4*i] << 24) | ((uint32_t)p[4*i+1] << 16) |
+               ((uint32_t)p[4*i+2] << 8) | p[4*i+3];
+    for (int i = 16; i < 64; ++i) {
+        uint32_t s0 = ror32(w[i-15],7) ^ ror32(w[i-15],18) ^ (w[i-15] >> 3);
+        uint32_t s1 = ror32(w[i-2],17) ^ ror32(w[i-2],19) ^ (w[i-2] >> 10);
+        w[i] = w[i-16] + s0 + w[i-7] + s1;
+    }
+    uint32_t a=c->h[0],b=c->h[1],cc=c->h[2],d=c->h[3],
+             e=c->h[4],f=c->h[5],g=c->h[6],h=c->h[7];
+    for (int i = 0; i < 64; ++i) {
+        uint32_t S1 = ror32(e,6) ^ ror32(e,11) ^ ror32(e,25);
+        uint32_t ch = (e & f) ^ (~e & g);
+        uint32_t t1 = h + S1 + ch + SHA_K[i] + w[i];
+        uint32_t S0 = ror32(a,2) ^ ror32(a,13) ^ ror32(a,22);
+        uint32_t maj = (a & b) ^ (a & cc) ^ (b & cc);
+        uint32_t t2 = S0 + maj;
+        h=g; g=f; f=e; e=d+t1; d=cc; cc=b; b=a; a=t1+t2;
+    }
+    c->h[0]+=a; c->h[1]+=b; c->h[2]+=cc; c->h[3]+=d;
+    c->h[4]+=e; c->h[5]+=f; c->h[6]+=g; c->h[7]+=h;
+}
+
+__device__ void sha256_update(Sha256Ctx* c, const uint8_t* p, int n) {
+    c->total += n;
+    while (n > 0) {
+        int take = 64 - c->buflen;
+        if (take > n) take = n;
+        for (int i = 0; i < take; ++i) c->buf[c->buflen + i] = p[i];
+        c->buflen += take;
+        p += take; n -= take;
+        if (c->buflen == 64) { sha256_block(c, c->buf); c->buflen = 0; }
+    }
+}
+
+__device__ void sha256_final(Sha256Ctx* c, uint8_t out[32]) {
+    uint64_t bits = c->total * 8;
+    uint8_t pad = 0x80;
+    sha256_update(c, &pad, 1);
+    uint8_t z = 0;
+    while (c->buflen != 56) sha256_update(c, &z, 1);
+    uint8_t lenb[8];
+    for (int i = 0; i < 8; ++i) lenb[i] = (uint8_t)(bits >> (56 - 8*i));
+    sha256_update(c, lenb, 8);
+    for (int i = 0; i < 8; ++i) {
+        out[4*i] = (uint8_t)(c->h[i] >> 24);
+        out[4*i+1] = (uint8_t)(c->h[i] >> 16);
+        out[4*i+2] = (uint8_t)(c->h[i] >> 8);
+        out[4*i+3] = (uint8_t)(c->h[i]);
+    }
+}
+
+__device__ __forceinline__ int hexval(uint8_t c) {
+    if (c >= '0' && c <= '9') return c - '0';
+    if (c >= 'a' && c <= 'f') return c - 'a' + 10;
+    if (c >= 'A' && c <= 'F') return c - 'A' + 10;
+    return -1;
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_auth(const uint8_t* __restrict__ reqs,
+       const int64_t* __restrict__ req_off,
+       int32_t* __restrict__ fields,
+       int n,
+       const uint8_t* __restrict__ secret, int secret_len) {
+    const int req = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+    if (req >= n) return;
+    int32_t* F = fields + (size_t)req * NF;
+    const int flags = F[FI_FLAGS];
+    if (flags & (FL_ERR_PARSE | FL_IS_OPTIONS)) return;  // OPTIONS exempt
+    const uint8_t* base = reqs + req_off[req];
+    const int aoff = F[FI_AUTH_OFF], alen = F[FI_AUTH_LEN];
+    bool ok = false;
+    // expected form: "HMAC " + 64 hex chars
+    if (alen == 69 && base[aoff]=='H' && base[aoff+1]=='M' &&
+        base[aoff+2]=='A' && base[aoff+3]=='C' && base[aoff+4]==' ') {
+        // HMAC-SHA256(secret, METHOD + " " + path)
+        uint8_t kblk[64], digest[32];
+        for (int i = 0; i < 64; ++i)
+            kblk[i] = (i < secret_len ? secret[i] : 0) ^ 0x36;
+        Sha256Ctx ctx;
+        sha256_init(&ctx);
+        sha256_update(&ctx, kblk, 64);
+        // message: method token (from request start to first space) + ' ' + path
+        int sp = 0;
+        while (base[sp] != ' ') ++sp;
+        sha256_update(&ctx, base, sp + 1);
+        sha256_update(&ctx, base + F[FI_PATH_OFF], F[FI_PATH_LEN]);
+        sha256_final(&ctx, digest);
+        for (int i = 0; i < 64; ++i)
+            kblk[i] = (i < secret_len ? secret[i] : 0) ^ 0x5c;
+        Sha256Ctx ctx2;
+        sha256_init(&ctx2);
+        sha256_update(&ctx2, kblk, 64);
+        sha256_update(&ctx2, digest, 32);
+        sha256_final(&ctx2, digest);
+        // constant-time hex compare
+        int diff = 0;
+        for (int i = 0; i < 32; ++i) {
+            const int hi = hexval(base[aoff + 5 + 2*i]);
+            const int lo = hexval(base[aoff + 5 + 2*i + 1]);
+            diff |= (hi < 0 || lo < 0) ? 1 : (((hi << 4) | lo) ^ digest[i]);
+        }
+        ok = diff == 0;
+    }
+    if (!ok) {
+        F[FI_FLAGS] = flags | FL_AUTH_FAIL;
+        F[FI_STATUS] = 401;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// k_varint_spans — batched protobuf field-span decode (gRPC codec path).
+//
+// One message per lane. Output per message: up to MAX_PB_FIELDS rows of
+// [field_no, wire_type, payload_off(abs in buf) or varint lo32, len or
+// varint hi32]; n_fields in the count array. Length-delimited fields
+// record (offset, len); varint fields record the decoded 64-bit value
+// split into lo/hi. The host codec (gofr_amd/grpc/codec.py) is the
+// golden model (tests/test_gpu_middleware.py).
+// ---------------------------------------------------------------------------
+#define MAX_PB_FIELDS 16
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_varint_spans(const uint8_t* __restrict__ buf,
+               const int64_t* __restrict__ msg_off,
+               const int32_t* __restrict__ msg_len,
+               int32_t* __restrict__ out,      // [n, MAX_PB_FIELDS, 4]
+               int32_t* __restrict__ out_n,    // [n]
+               int n) {
+    const int m = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+    if (m >= n) return;
+    const int64_t base = msg_off[m];
+    const uint8_t* p = buf + base;
+    const int len = msg_len[m];
+    int32_t* row = out + (size_t)m * MAX_PB_FIELDS * 4;
+    int nf = 0;
+    int pos = 0;
+    while (pos < len && nf < MAX_PB_FIELDS) {
+        // read tag varint
+        uint64_t tag = 0;
+        int shift = 0;
+        while (pos < len) {
+            const uint8_t b = p[pos++];
+            tag |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+            if (shift > 63) { out_n[m] = -1; return; }
+        }
+        const int fno = (int)(tag >> 3);
+        const int wt = (int)(tag & 7);
+        if (wt == 0) {
+            uint64_t v = 0;
+            shift = 0;
+            while (pos < len) {
+                const uint8_t b = p[pos++];
+                v |= (uint64_t)(b & 0x7F) << shift;
+                if (!(b & 0x80)) break;
+                shift += 7;
+                if (shift > 63) { out_n[m] = -1; return; }
+            }
+            row[nf*4+0] = fno; row[nf*4+1] = 0;
+            row[nf*4+2] = (int32_t)(v & 0xFFFFFFFF);
+            row[nf*4+3] = (int32_t)(v >> 32);
+        } else if (wt == 2) {
+            uint64_t ln = 0;
+            shift = 0;
+            while (pos < len) {
+                const uint8_t b = p[pos++];
+                ln |= (uint64_t)(b & 0x7F) << shift;
+                if (!(b & 0x80)) break;
+                shift += 7;
+            }
+            if (pos + (int)ln > len) { out_n[m] = -1; return; }
+            row[nf*4+0] = fno; row[nf*4+1] = 2;
+            row[nf*4+2] = (int32_t)(base + pos);
+            row[nf*4+3] = (int32_t)ln;
+            pos += (int)ln;
+        } else if (wt == 1) {
+            if (pos + 8 > len) { out_n[m] = -1; return; }
+            row[nf*4+0] = fno; row[nf*4+1] = 1;
+            row[nf*4+2] = (int32_t)(base + pos); row[nf*4+3] = 8;
+            pos += 8;
+        } else if (wt == 5) {
+            if (pos + 4 > len) { out_n[m] = -1; return; }
+            row[nf*4+0] = fno; row[nf*4+1] = 5;
+            row[nf*4+2] = (int32_t)(base + pos); row[nf*4+3] = 4;
+            pos += 4;
+        } else {
+            out_n[m] = -1;
+            return;
+        }
+        ++nf;
+    }
+    out_n[m] = nf;
+}
+
+// ---------------------------------------------------------------------------
 // k_compact — gather response slots into a contiguous 16B-aligned stream
 // (halves the D2H bytes vs slot-strided responses; offsets are the
 // exclusive cumsum of round16(resp_len), computed host-side via torch)
@@ -666,7 +929,7 @@ int gofr_launch_respond(
         void* resp_len_out, int n, int rslot,
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
-        const void* seed_ptr) {
+        const void* seed_ptr, int auth_env_off, int auth_env_len) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
@@ -676,7 +939,34 @@ int gofr_launch_respond(
                        n, rslot,
                        (const int32_t*)handler_tab, n_routes,
                        (const uint8_t*)blob, (const uint8_t*)host_blob,
-                       (const int32_t*)host_tab, (const uint64_t*)seed_ptr);
+                       (const int32_t*)host_tab, (const uint64_t*)seed_ptr,
+                       auth_env_off, auth_env_len);
+    return (int)hipGetLastError();
+}
+
+int gofr_launch_auth(
+        void* stream,
+        const void* reqs, const void* req_off, void* fields, int n,
+        const void* secret, int secret_len) {
+    const int blocks = (n + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    hipLaunchKernelGGL(k_auth, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)reqs, (const int64_t*)req_off,
+                       (int32_t*)fields, n,
+                       (const uint8_t*)secret, secret_len);
+    return (int)hipGetLastError();
+}
+
+int gofr_launch_varint_spans(
+        void* stream,
+        const void* buf, const void* msg_off, const void* msg_len,
+        void* out, void* out_n, int n) {
+    const int blocks = (n + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    hipLaunchKernelGGL(k_varint_spans, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)buf, (const int64_t*)msg_off,
+                       (const int32_t*)msg_len,
+                       (int32_t*)out, (int32_t*)out_n, n);
     return (int)hipGetLastError();
 }
 

@@ -44,6 +44,9 @@ FL_NEEDS_HOST = 2
 FL_KEEP_ALIVE = 4
 FL_JSON_CT = 8
 FL_IS_OPTIONS = 16
+FL_BODY_INVALID = 32
+FL_ACCEPT_GZIP = 64
+FL_AUTH_FAIL = 128
 
 HK_HOST = 0
 HK_ECHO_JSON = 1
@@ -91,6 +94,13 @@ class HipOps:
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
+        self.lib.gofr_launch_auth.restype = ctypes.c_int
+        self.lib.gofr_launch_auth.argtypes = \
+            [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_void_p,
+                                     ctypes.c_int]
+        self.lib.gofr_launch_varint_spans.restype = ctypes.c_int
+        self.lib.gofr_launch_varint_spans.argtypes = \
+            [ctypes.c_void_p] * 6 + [ctypes.c_int]
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,
@@ -118,7 +128,7 @@ class HipOps:
 
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
-                host_blob_t, host_tab_t, seed_t):
+                host_blob_t, host_tab_t, seed_t, auth_env=(0, 0)):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -131,9 +141,35 @@ class HipOps:
             ctypes.c_void_p(blob_t.data_ptr()),
             ctypes.c_void_p(host_blob_t.data_ptr()),
             ctypes.c_void_p(host_tab_t.data_ptr()),
-            ctypes.c_void_p(seed_t.data_ptr()))
+            ctypes.c_void_p(seed_t.data_ptr()),
+            auth_env[0], auth_env[1])
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
+
+    def auth(self, stream, reqs_t, req_off_t, fields_t, n, secret_t,
+             secret_len):
+        rc = self.lib.gofr_launch_auth(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(reqs_t.data_ptr()),
+            ctypes.c_void_p(req_off_t.data_ptr()),
+            ctypes.c_void_p(fields_t.data_ptr()),
+            n,
+            ctypes.c_void_p(secret_t.data_ptr()), secret_len)
+        if rc != 0:
+            raise RuntimeError(f"k_auth launch failed: hipError {rc}")
+
+    def varint_spans(self, stream, buf_t, msg_off_t, msg_len_t, out_t,
+                     out_n_t, n):
+        rc = self.lib.gofr_launch_varint_spans(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(buf_t.data_ptr()),
+            ctypes.c_void_p(msg_off_t.data_ptr()),
+            ctypes.c_void_p(msg_len_t.data_ptr()),
+            ctypes.c_void_p(out_t.data_ptr()),
+            ctypes.c_void_p(out_n_t.data_ptr()), n)
+        if rc != 0:
+            raise RuntimeError(
+                f"k_varint_spans launch failed: hipError {rc}")
 
     def compact(self, stream, resp_slots_t, resp_len_t, resp_off_t, out_t,
                 n, rslot):
@@ -258,6 +294,9 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
                 auth_off, auth_len = vs, vlen
             elif nlen == 17 and _ieq(buf, ls, b"transfer-encoding"):
                 flags |= FL_NEEDS_HOST
+            elif nlen == 15 and _ieq(buf, ls, b"accept-encoding"):
+                if b"gzip" in buf[vs:vs + vlen]:
+                    flags |= FL_ACCEPT_GZIP
             prev_lf = lf
         body_len = max(0, ln - body_off)
         if 0 < clen < body_len:
@@ -379,7 +418,7 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 fields: np.ndarray, rslot: int,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
-                seed: int):
+                seed: int, auth_env=(0, 0)):
     """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
     n = len(fields)
     host_tab = np.asarray(host_tab, np.int32).reshape(-1)
@@ -414,10 +453,20 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 off = int(handler_tab[route * 4 + 1])
                 ln = int(handler_tab[route * 4 + 2])
                 body_src = blob[off:off + ln]
-        if kind == HK_HOST and not is_options:
+        if kind == HK_HOST and not is_options and \
+                not (flags & FL_AUTH_FAIL):
             off, ln, status, ct_id = (int(host_tab[r * 4 + i])
                                       for i in range(4))
             body_src = host_blob[off:off + ln]
+            if status == 0:
+                status = 500
+                elen = int.from_bytes(blob[:4], "little")
+                body_src = blob[4:4 + elen]
+        if (flags & FL_AUTH_FAIL) and not is_options:
+            status = 401
+            body_src = blob[auth_env[0]:auth_env[0] + auth_env[1]]
+            env = False
+            ct_id = 0
 
         body_total = len(body_src) + (9 if env else 0)
         reason = _dev_reason(status)
@@ -440,3 +489,119 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
         fields[r][FI_RESP_LEN] = len(payload)
         fields[r][FI_RESP_OFF] = r * rslot
     return resp, resp_len
+
+
+def cpu_auth(reqs: np.ndarray, req_off: np.ndarray, fields: np.ndarray,
+             secret: bytes) -> None:
+    """Mirror of k_auth: HMAC-SHA256 bearer check; sets FL_AUTH_FAIL."""
+    import hashlib
+    import hmac as hmac_mod
+    for r in range(len(fields)):
+        F = fields[r]
+        flags = int(F[FI_FLAGS])
+        if flags & (FL_ERR_PARSE | FL_IS_OPTIONS):
+            continue
+        base = int(req_off[r])
+        aoff, alen = int(F[FI_AUTH_OFF]), int(F[FI_AUTH_LEN])
+        ok = False
+        val = reqs[base + aoff:base + aoff + alen].tobytes()
+        if alen == 69 and val[:5] == b"HMAC ":
+            raw = reqs[base:base + 64].tobytes()
+            sp = raw.index(b" ")
+            msg = raw[:sp + 1] + reqs[
+                base + F[FI_PATH_OFF]:
+                base + F[FI_PATH_OFF] + F[FI_PATH_LEN]].tobytes()
+            want = hmac_mod.new(secret, msg, hashlib.sha256).hexdigest()
+            ok = hmac_mod.compare_digest(val[5:].decode("latin-1").lower(),
+                                         want)
+        if not ok:
+            F[FI_FLAGS] = flags | FL_AUTH_FAIL
+            F[FI_STATUS] = 401
+
+
+MAX_PB_FIELDS = 16
+
+
+def cpu_varint_spans(buf: np.ndarray, msg_off: np.ndarray,
+                     msg_len: np.ndarray):
+    """Mirror of k_varint_spans."""
+    n = len(msg_len)
+    out = np.zeros((n, MAX_PB_FIELDS, 4), np.int32)
+    out_n = np.zeros(n, np.int32)
+    data = buf.tobytes()
+    for m in range(n):
+        base = int(msg_off[m])
+        ln = int(msg_len[m])
+        pos = 0
+        nf = 0
+        bad = False
+        while pos < ln and nf < MAX_PB_FIELDS:
+            tag = 0
+            shift = 0
+            while pos < ln:
+                b = data[base + pos]
+                pos += 1
+                tag |= (b & 0x7F) << shift
+                if not b & 0x80:
+                    break
+                shift += 7
+                if shift > 63:
+                    bad = True
+                    break
+            if bad:
+                break
+            fno, wt = tag >> 3, tag & 7
+            if wt == 0:
+                v = 0
+                shift = 0
+                while pos < ln:
+                    b = data[base + pos]
+                    pos += 1
+                    v |= (b & 0x7F) << shift
+                    if not b & 0x80:
+                        break
+                    shift += 7
+                    if shift > 63:
+                        bad = True
+                        break
+                if bad:
+                    break
+                out[m, nf] = (fno, 0,
+                              np.int32(v & 0xFFFFFFFF) if v & 0xFFFFFFFF < (1 << 31)
+                              else np.int32((v & 0xFFFFFFFF) - (1 << 32)),
+                              np.int32((v >> 32) & 0xFFFFFFFF)
+                              if (v >> 32) < (1 << 31)
+                              else np.int32((v >> 32) - (1 << 32)))
+            elif wt == 2:
+                l2 = 0
+                shift = 0
+                while pos < ln:
+                    b = data[base + pos]
+                    pos += 1
+                    l2 |= (b & 0x7F) << shift
+                    if not b & 0x80:
+                        break
+                    shift += 7
+                if pos + l2 > ln:
+                    bad = True
+                    break
+                out[m, nf] = (fno, 2, base + pos, l2)
+                pos += l2
+            elif wt == 1:
+                if pos + 8 > ln:
+                    bad = True
+                    break
+                out[m, nf] = (fno, 1, base + pos, 8)
+                pos += 8
+            elif wt == 5:
+                if pos + 4 > ln:
+                    bad = True
+                    break
+                out[m, nf] = (fno, 5, base + pos, 4)
+                pos += 4
+            else:
+                bad = True
+                break
+            nf += 1
+        out_n[m] = -1 if bad else nf
+    return out, out_n

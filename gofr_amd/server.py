@@ -23,7 +23,8 @@ from typing import Optional
 
 from .context import new_context
 from .errors import GofrError, PANIC_BODY
-from .http.middleware import CORS_HEADERS, make_request_log, panic_log
+from .http.middleware import (CORS_HEADERS, UNAUTHORIZED_BODY, auth_ok,
+                              make_request_log, panic_log)
 from .http.request import Request, parse_request_bytes
 from .http.responder import envelope_bytes, reason_phrase
 from .trace import noop_tracer
@@ -58,6 +59,15 @@ def dispatch(app, request: Request) -> _CapturedResponse:
     out.headers.extend(CORS_HEADERS)
     if request.method == "OPTIONS":
         out.status = 200
+        span.End()
+        _log_request(app, span, request, out.status, start, t0)
+        return out
+
+    # auth middleware (enable_auth) — GPU analog: k_auth
+    if app.auth_secret is not None and not auth_ok(app.auth_secret, request):
+        out.status = 401
+        out.headers.append(("Content-Type", "application/json"))
+        out.body = UNAUTHORIZED_BODY
         span.End()
         _log_request(app, span, request, out.status, start, t0)
         return out
@@ -108,6 +118,12 @@ def dispatch(app, request: Request) -> _CapturedResponse:
     status2, ct, body = envelope_bytes(data, err)
     out.status = status2
     out.headers.append(("Content-Type", ct))
+    # gzip middleware (enable_gzip) — GPU analog: fused deflate in respond
+    if (app.gzip_min_size is not None and len(body) >= app.gzip_min_size
+            and "gzip" in request.header("accept-encoding")):
+        import gzip as _gz
+        body = _gz.compress(body, compresslevel=1, mtime=0)
+        out.headers.append(("Content-Encoding", "gzip"))
     out.body = body
     span.End()
     _log_request(app, span, request, out.status, start, t0)

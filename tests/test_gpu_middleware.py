@@ -1,0 +1,99 @@
+"""GPU middleware kernels vs CPU mirrors: k_auth, k_varint_spans."""
+
+import numpy as np
+import pytest
+
+import gofr_amd
+from gofr_amd import handlers, ops
+from gofr_amd.config import MapConfig
+from gofr_amd.engine import BatchEngine, pack_batch
+from gofr_amd.http.middleware import hmac_token
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("no GPU", allow_module_level=True)
+
+SECRET = b"sup3r-secret"
+
+
+def http_req(method="GET", path="/", body=b"", headers=None):
+    h = dict(headers or {})
+    h.setdefault("Host", "localhost")
+    if body:
+        h.setdefault("Content-Type", "application/json")
+        h["Content-Length"] = str(len(body))
+    head = f"{method} {path} HTTP/1.1\r\n" + "".join(
+        f"{k}: {v}\r\n" for k, v in h.items()) + "\r\n"
+    return head.encode() + body
+
+
+def build_auth_app():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("hi"))
+    app.enable_auth(SECRET)
+    return app
+
+
+def test_k_auth_matches_mirror_bytes():
+    app = build_auth_app()
+    gpu = BatchEngine(app, device="cuda", max_batch=2048)
+    cpu = BatchEngine(app, device="cpu", max_batch=2048)
+    cpu._seed = gpu._seed
+    raws = []
+    for i in range(256):
+        tok = hmac_token(SECRET, "GET", "/greet")
+        if i % 3 == 0:
+            raws.append(http_req("GET", "/greet",
+                                 headers={"Authorization": f"HMAC {tok}"}))
+        elif i % 3 == 1:
+            raws.append(http_req("GET", "/greet"))
+        else:
+            raws.append(http_req(
+                "GET", "/greet",
+                headers={"Authorization": "HMAC " + "ab" * 32}))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, f"req {i}\nGPU {go[:120]!r}\nCPU {co[:120]!r}"
+    assert g[0].startswith(b"HTTP/1.1 200")
+    assert g[1].startswith(b"HTTP/1.1 401")
+    assert g[2].startswith(b"HTTP/1.1 401")
+
+
+def test_k_varint_spans_matches_mirror():
+    from gofr_amd.grpc.codec import MessageDesc, encode_message
+    import random
+    rng = random.Random(7)
+    desc = MessageDesc("T", {1: ("name", "string"), 2: ("n", "int64"),
+                             3: ("d", "double"), 4: ("f", "fixed32")})
+    payloads = []
+    for i in range(512):
+        m = {"name": "x" * rng.randrange(0, 300),
+             "n": rng.randrange(0, 1 << 50)}
+        if i % 4 == 0:
+            m["d"] = rng.random()
+        if i % 5 == 0:
+            m["f"] = rng.randrange(1, 1 << 31)
+        payloads.append(encode_message(m, desc))
+    buf, offs, lens = pack_batch(payloads)
+    ref_out, ref_n = ops.cpu_varint_spans(buf, offs, lens)
+
+    hip = ops.HipOps()
+    dev = torch.device("cuda:0")
+    n = len(payloads)
+    d_buf = torch.from_numpy(buf).to(dev)
+    d_off = torch.from_numpy(offs).to(dev)
+    d_len = torch.from_numpy(lens).to(dev)
+    d_out = torch.zeros(n * ops.MAX_PB_FIELDS * 4, dtype=torch.int32,
+                        device=dev)
+    d_out_n = torch.zeros(n, dtype=torch.int32, device=dev)
+    stream = torch.cuda.current_stream().cuda_stream
+    hip.varint_spans(stream, d_buf, d_off, d_len, d_out, d_out_n, n)
+    torch.cuda.synchronize()
+    got_out = d_out.cpu().numpy().reshape(n, ops.MAX_PB_FIELDS, 4)
+    got_n = d_out_n.cpu().numpy()
+    assert np.array_equal(got_n, ref_n)
+    assert np.array_equal(got_out, ref_out)

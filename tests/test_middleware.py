@@ -1,0 +1,187 @@
+"""Auth (HMAC-SHA256) + gzip middleware tests, CPU transport and engine
+mirror paths, plus the protobuf varint-span mirror."""
+
+import gzip
+import json
+
+import numpy as np
+import pytest
+
+import gofr_amd
+from gofr_amd import handlers, ops
+from gofr_amd.config import MapConfig
+from gofr_amd.engine import BatchEngine, pack_batch
+from gofr_amd.http.middleware import hmac_token
+from gofr_amd.server import dispatch
+
+from conftest import make_request
+
+SECRET = b"sup3r-secret"
+
+
+def build_auth_app():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("hi"))
+    app.enable_auth(SECRET)
+    app.install_default_routes()
+    return app
+
+
+def http_req(method="GET", path="/", body=b"", headers=None):
+    h = dict(headers or {})
+    h.setdefault("Host", "localhost")
+    if body:
+        h.setdefault("Content-Type", "application/json")
+        h["Content-Length"] = str(len(body))
+    head = f"{method} {path} HTTP/1.1\r\n" + "".join(
+        f"{k}: {v}\r\n" for k, v in h.items()) + "\r\n"
+    return head.encode() + body
+
+
+# -- CPU transport auth -------------------------------------------------------
+
+def test_auth_rejects_without_token():
+    app = build_auth_app()
+    resp = dispatch(app, make_request("GET", "/greet"))
+    assert resp.status == 401
+    assert json.loads(resp.body) == {"error": {"message": "unauthorized"}}
+
+
+def test_auth_accepts_valid_token():
+    app = build_auth_app()
+    tok = hmac_token(SECRET, "GET", "/greet")
+    resp = dispatch(app, make_request(
+        "GET", "/greet", headers={"Authorization": f"HMAC {tok}"}))
+    assert resp.status == 200
+
+
+def test_auth_rejects_wrong_path_token():
+    app = build_auth_app()
+    tok = hmac_token(SECRET, "GET", "/other")
+    resp = dispatch(app, make_request(
+        "GET", "/greet", headers={"Authorization": f"HMAC {tok}"}))
+    assert resp.status == 401
+
+
+def test_auth_options_exempt():
+    app = build_auth_app()
+    resp = dispatch(app, make_request("OPTIONS", "/greet"))
+    assert resp.status == 200
+
+
+# -- engine mirror auth (same bytes the GPU kernel must produce) -------------
+
+def test_engine_auth_mirror():
+    app = build_auth_app()
+    eng = BatchEngine(app, device="cpu")
+    tok = hmac_token(SECRET, "POST", "/echo")
+    good = http_req("POST", "/echo", b'{"a":1}',
+                    headers={"Authorization": f"HMAC {tok}"})
+    bad = http_req("POST", "/echo", b'{"a":1}')
+    wrong = http_req("POST", "/echo", b'{"a":1}',
+                     headers={"Authorization": "HMAC " + "0" * 64})
+    outs = eng.process([good, bad, wrong])
+    assert outs[0].startswith(b"HTTP/1.1 200 OK")
+    assert outs[1].startswith(b"HTTP/1.1 401 Unauthorized")
+    assert outs[2].startswith(b"HTTP/1.1 401 Unauthorized")
+    assert b'{"error":{"message":"unauthorized"}}' in outs[1]
+    # auth failure must not run the handler (no data envelope)
+    assert b'"data"' not in outs[1]
+
+
+def test_engine_auth_host_route_blocked():
+    app = build_auth_app()
+    app_called = []
+
+    def host_handler(ctx):
+        app_called.append(1)
+        return "x"
+    app.router.add("GET", "/host", host_handler)
+    eng = BatchEngine(app, device="cpu")
+    outs = eng.process([http_req("GET", "/host")])
+    assert outs[0].startswith(b"HTTP/1.1 401")
+    assert not app_called
+
+
+# -- gzip middleware ----------------------------------------------------------
+
+def test_gzip_applied_when_accepted():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/big", lambda ctx: {"pad": "z" * 600})
+    app.enable_gzip(min_size=128)
+    app.install_default_routes()
+    resp = dispatch(app, make_request(
+        "GET", "/big", headers={"Accept-Encoding": "gzip, deflate"}))
+    hdrs = dict(resp.headers)
+    assert hdrs.get("Content-Encoding") == "gzip"
+    assert json.loads(gzip.decompress(resp.body)) == {
+        "data": {"pad": "z" * 600}}
+
+
+def test_gzip_skipped_small_or_unaccepted():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/small", lambda ctx: "ok")
+    app.GET("/big", lambda ctx: {"pad": "z" * 600})
+    app.enable_gzip(min_size=128)
+    app.install_default_routes()
+    resp = dispatch(app, make_request(
+        "GET", "/small", headers={"Accept-Encoding": "gzip"}))
+    assert "Content-Encoding" not in dict(resp.headers)
+    resp = dispatch(app, make_request("GET", "/big"))
+    assert "Content-Encoding" not in dict(resp.headers)
+
+
+def test_parse_flags_accept_gzip():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.install_default_routes()
+    eng = BatchEngine(app, device="cpu")
+    raws = [http_req("GET", "/x", headers={"Accept-Encoding": "gzip"}),
+            http_req("GET", "/x", headers={"Accept-Encoding": "br"}),
+            http_req("GET", "/x")]
+    buf, offs, lens = pack_batch(raws)
+    fields = ops.cpu_parse_route(buf, offs, lens, eng.program.trie,
+                                 eng.program.handler_tab)
+    assert fields[0][ops.FI_FLAGS] & ops.FL_ACCEPT_GZIP
+    assert not (fields[1][ops.FI_FLAGS] & ops.FL_ACCEPT_GZIP)
+    assert not (fields[2][ops.FI_FLAGS] & ops.FL_ACCEPT_GZIP)
+
+
+# -- varint span mirror vs host codec ----------------------------------------
+
+def test_varint_spans_mirror_matches_codec():
+    from gofr_amd.grpc.codec import MessageDesc, encode_message
+    desc = MessageDesc("T", {1: ("name", "string"), 2: ("n", "int64"),
+                             3: ("d", "double"), 4: ("f", "fixed32")})
+    msgs = [
+        {"name": "hello", "n": 300},
+        {"name": "x" * 200, "n": 1, "d": 2.5},
+        {"n": (1 << 40) + 7, "f": 9},
+        {"name": ""},
+    ]
+    payloads = [encode_message(m, desc) for m in msgs]
+    out, out_n = ops.cpu_varint_spans(
+        np.frombuffer(b"".join(payloads), np.uint8).copy(),
+        np.asarray([sum(len(p) for p in payloads[:i])
+                    for i in range(len(payloads))], np.int64),
+        np.asarray([len(p) for p in payloads], np.int32))
+    # message 0: field1 string span "hello", field2 varint 300
+    blob = b"".join(payloads)
+    f = out[0]
+    assert out_n[0] == 2
+    assert f[0][0] == 1 and f[0][1] == 2
+    assert blob[f[0][2]:f[0][2] + f[0][3]] == b"hello"
+    assert f[1][0] == 2 and f[1][1] == 0 and f[1][2] == 300
+    # message 2: 40-bit varint split lo/hi
+    f = out[2]
+    v = (int(f[0][3]) << 32) | (int(f[0][2]) & 0xFFFFFFFF)
+    assert v == (1 << 40) + 7
+    # message 3: empty -> 0 fields
+    assert out_n[3] == 0
+
+
+def test_varint_spans_malformed():
+    out, out_n = ops.cpu_varint_spans(
+        np.frombuffer(b"\x0a\xff", np.uint8).copy(),  # len 255 but truncated
+        np.asarray([0], np.int64), np.asarray([2], np.int32))
+    assert out_n[0] == -1
