@@ -239,7 +239,8 @@ class BatchEngine:
         resp_slots, resp_len = ops.cpu_respond(
             buf, offs, fields, self.rslot, self.program.handler_tab,
             self.program.blob, host_blob, host_tab, seed,
-            auth_env=self.program.auth_env)
+            auth_env=self.program.auth_env,
+            gzip_min=self.app.gzip_min_size or 0)
         # compaction mirror (same round16 layout as k_compact)
         n = len(lens)
         pads = (resp_len + 15) & ~15
@@ -323,7 +324,8 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.d_seed, auth_env=self.program.auth_env)
+                         ln.d_seed, auth_env=self.program.auth_env,
+                         gzip_min=self.app.gzip_min_size or 0)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
@@ -407,7 +409,8 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.d_seed, auth_env=self.program.auth_env)
+                         ln.d_seed, auth_env=self.program.auth_env,
+                         gzip_min=self.app.gzip_min_size or 0)
         return ln.d_resp, ln.d_resp_len
 
     # -- host trampoline ------------------------------------------------------

@@ -398,38 +398,249 @@ __device__ __forceinline__ int itoa10(uint8_t* dst, int v) {
     return n;
 }
 
+// ---------------------------------------------------------------------------
+// Fused gzip (DEFLATE, static Huffman) — BASELINE config 4's gzip
+// middleware, fused into the response serializer (HBM guideline: compress
+// in the producing kernel instead of re-reading the body).
+//
+// One response per wave. The LZ77 greedy parse runs as a whole-wave loop:
+// lane 0 owns the position/hash bookkeeping and bit emission; match
+// LENGTH determination is wave-parallel (lane l compares input[cand+l]
+// vs input[pos+l], ballot -> ctz). Hash table and bit output live in LDS.
+// CRC32 uses a block-shared LDS table built cooperatively at launch.
+//
+// Byte-exact mirror: gofr_amd/ops/__init__.py gzip_static_mirror().
+// ---------------------------------------------------------------------------
+
+#define GZ_HASH_BITS 9
+#define GZ_HASH_SIZE (1 << GZ_HASH_BITS)
+#define GZ_MIN_MATCH 3
+#define GZ_MAX_DIST 2048  /* window = slot size; responses are < 4 KB */
+
+struct BitWriter {
+    uint8_t* out;
+    int bytepos;
+    uint64_t bitbuf;
+    int nbits;
+};
+
+__device__ __forceinline__ void bw_init(BitWriter* w, uint8_t* out) {
+    w->out = out;
+    w->bytepos = 0;
+    w->bitbuf = 0;
+    w->nbits = 0;
+}
+
+// append `len` bits of `bits` LSB-first (DEFLATE bit order)
+__device__ __forceinline__ void bw_put(BitWriter* w, uint32_t bits, int len) {
+    w->bitbuf |= (uint64_t)bits << w->nbits;
+    w->nbits += len;
+    while (w->nbits >= 8) {
+        w->out[w->bytepos++] = (uint8_t)(w->bitbuf & 0xFF);
+        w->bitbuf >>= 8;
+        w->nbits -= 8;
+    }
+}
+
+__device__ __forceinline__ void bw_flush(BitWriter* w) {
+    if (w->nbits > 0) {
+        w->out[w->bytepos++] = (uint8_t)(w->bitbuf & 0xFF);
+        w->bitbuf = 0;
+        w->nbits = 0;
+    }
+}
+
+// reverse the low `len` bits (huffman codes are emitted MSB-first)
+__device__ __forceinline__ uint32_t bitrev(uint32_t v, int len) {
+    uint32_t r = 0;
+    for (int i = 0; i < len; ++i) {
+        r = (r << 1) | (v & 1);
+        v >>= 1;
+    }
+    return r;
+}
+
+// static-huffman literal/length-symbol code (already reversed for bw_put)
+__device__ __forceinline__ void fixed_lit_code(int sym, uint32_t* code,
+                                               int* len) {
+    if (sym < 144)      { *len = 8; *code = bitrev(0x30 + sym, 8); }
+    else if (sym < 256) { *len = 9; *code = bitrev(0x190 + sym - 144, 9); }
+    else if (sym < 280) { *len = 7; *code = bitrev(sym - 256, 7); }
+    else                { *len = 8; *code = bitrev(0xC0 + sym - 280, 8); }
+}
+
+// length -> (code 257..285, extra bits, extra value)
+__device__ __forceinline__ void length_code(int mlen, int* sym, int* ebits,
+                                            int* eval) {
+    static const int base[29] = {3,4,5,6,7,8,9,10,11,13,15,17,19,23,27,31,
+                                 35,43,51,59,67,83,99,115,131,163,195,227,258};
+    static const int extra[29] = {0,0,0,0,0,0,0,0,1,1,1,1,2,2,2,2,
+                                  3,3,3,3,4,4,4,4,5,5,5,5,0};
+    for (int i = 28; i >= 0; --i) {
+        if (mlen >= base[i]) {
+            *sym = 257 + i;
+            *ebits = extra[i];
+            *eval = mlen - base[i];
+            return;
+        }
+    }
+    *sym = 257; *ebits = 0; *eval = 0;
+}
+
+// distance -> (code 0..29, extra bits, extra value)
+__device__ __forceinline__ void dist_code(int dist, int* sym, int* ebits,
+                                          int* eval) {
+    static const int base[30] = {1,2,3,4,5,7,9,13,17,25,33,49,65,97,129,193,
+                                 257,385,513,769,1025,1537,2049,3073,4097,
+                                 6145,8193,12289,16385,24577};
+    static const int extra[30] = {0,0,0,0,1,1,2,2,3,3,4,4,5,5,6,6,
+                                  7,7,8,8,9,9,10,10,11,11,12,12,13,13};
+    for (int i = 29; i >= 0; --i) {
+        if (dist >= base[i]) {
+            *sym = i;
+            *ebits = extra[i];
+            *eval = dist - base[i];
+            return;
+        }
+    }
+    *sym = 0; *ebits = 0; *eval = 0;
+}
+
+// whole-wave gzip of src[0..len) in LDS -> dst (LDS); returns total gzip
+// bytes (negative on overflow of cap). hash: per-wave LDS u16 table.
+// crc_tab: block-shared 256-entry LDS CRC32 table.
+__device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
+                                 int cap, uint16_t* hash,
+                                 const uint32_t* crc_tab, int lane) {
+    // cooperative hash clear
+    for (int i = lane; i < GZ_HASH_SIZE; i += WAVE) hash[i] = 0;
+    // gzip header (lane 0)
+    BitWriter w;
+    int total = -1;
+    if (lane == 0) {
+        uint8_t* p = dst;
+        p[0]=0x1f; p[1]=0x8b; p[2]=8; p[3]=0;
+        p[4]=p[5]=p[6]=p[7]=0; p[8]=0; p[9]=255;
+        bw_init(&w, dst + 10);
+        bw_put(&w, 1, 1);   // BFINAL
+        bw_put(&w, 1, 2);   // BTYPE=01 static
+    }
+    int pos = 0;
+    uint32_t crc = 0xFFFFFFFFu;
+    // whole-wave greedy LZ77 loop: lane 0 decides; the wave measures
+    while (true) {
+        const int p0 = __shfl(pos, 0);
+        if (p0 >= len) break;
+        // candidate from hash (lane 0 computes, broadcasts)
+        int cand = -1;
+        if (lane == 0 && p0 + GZ_MIN_MATCH <= len) {
+            const uint32_t h = ((src[p0] | (src[p0+1] << 8) |
+                                (src[p0+2] << 16)) * 0x9E3779B1u)
+                               >> (32 - GZ_HASH_BITS);
+            const int stored = (int)hash[h] - 1;
+            if (stored >= 0 && p0 - stored <= GZ_MAX_DIST && stored < p0)
+                cand = stored;
+            hash[h] = (uint16_t)(p0 + 1);
+        }
+        cand = __shfl(cand, 0);
+        int mlen = 0;
+        if (cand >= 0) {
+            // wave-parallel match length, up to 258 in 64-byte strides
+            int agree = 0;
+            for (int base = 0; base < 258 && agree == base; base += WAVE) {
+                const int i = base + lane;
+                const bool eq = (p0 + i < len) && (i < 258) &&
+                                (src[cand + i] == src[p0 + i]);
+                const uint64_t m = __ballot(eq);
+                const int run = (m == ~0ull) ? WAVE
+                                             : __builtin_ctzll(~m);
+                agree += run;
+                if (run < WAVE) break;
+            }
+            mlen = agree > 258 ? 258 : agree;
+        }
+        if (lane == 0) {
+            if (mlen >= GZ_MIN_MATCH) {
+                int sym, ebits, eval;
+                length_code(mlen, &sym, &ebits, &eval);
+                uint32_t code; int clen;
+                fixed_lit_code(sym, &code, &clen);
+                bw_put(&w, code, clen);
+                if (ebits) bw_put(&w, (uint32_t)eval, ebits);
+                dist_code(p0 - cand, &sym, &ebits, &eval);
+                bw_put(&w, bitrev((uint32_t)sym, 5), 5);
+                if (ebits) bw_put(&w, (uint32_t)eval, ebits);
+                // crc + hash inserts over the matched span
+                for (int i = 0; i < mlen; ++i) {
+                    const uint8_t b = src[p0 + i];
+                    crc = (crc >> 8) ^ crc_tab[(crc ^ b) & 0xFF];
+                    if (i > 0 && p0 + i + GZ_MIN_MATCH <= len) {
+                        const int q = p0 + i;
+                        const uint32_t h2 = ((src[q] | (src[q+1] << 8) |
+                                              (src[q+2] << 16))
+                                             * 0x9E3779B1u)
+                                            >> (32 - GZ_HASH_BITS);
+                        hash[h2] = (uint16_t)(q + 1);
+                    }
+                }
+                pos = p0 + mlen;
+            } else {
+                const uint8_t b = src[p0];
+                uint32_t code; int clen;
+                fixed_lit_code(b, &code, &clen);
+                bw_put(&w, code, clen);
+                crc = (crc >> 8) ^ crc_tab[(crc ^ b) & 0xFF];
+                pos = p0 + 1;
+            }
+            if (w.bytepos > cap - 24) pos = len + 1;  // overflow guard
+        }
+        pos = __shfl(pos, 0);
+        if (pos > len) return -1;
+    }
+    if (lane == 0) {
+        uint32_t code; int clen;
+        fixed_lit_code(256, &code, &clen);  // end of block
+        bw_put(&w, code, clen);
+        bw_flush(&w);
+        uint8_t* p = dst + 10 + w.bytepos;
+        const uint32_t c = crc ^ 0xFFFFFFFFu;
+        p[0]=(uint8_t)c; p[1]=(uint8_t)(c>>8);
+        p[2]=(uint8_t)(c>>16); p[3]=(uint8_t)(c>>24);
+        p[4]=(uint8_t)len; p[5]=(uint8_t)(len>>8);
+        p[6]=(uint8_t)(len>>16); p[7]=(uint8_t)(len>>24);
+        total = 10 + w.bytepos + 8;
+    }
+    return __shfl(total, 0);
+}
+
+__constant__ char HDR_CE[] = "Content-Encoding: gzip\r\n";
+
 // host result table row: [off, len, status, ct_id]
 //
 // The whole response (headers + body) is assembled in LDS per wave, then
 // copied to global with one coalesced 16 B/lane sweep. Header segments are
-// written cooperatively (each segment is <=64 B: one strided store per
-// lane); the JSON-echo body is validated with a ballot structural pass
-// (quote/backslash/brace bitmasks; lane 0 only touches the ~tens of
-// structural positions, not every byte) before being staged.
-extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
-k_respond(const uint8_t* __restrict__ reqs,
-          const int64_t* __restrict__ req_off,
-          int32_t* __restrict__ fields,
-          uint8_t* __restrict__ resp,
-          int32_t* __restrict__ resp_len_out,
-          int n, int rslot,
-          const int32_t* __restrict__ handler_tab, int n_routes,
-          const uint8_t* __restrict__ blob,
-          const uint8_t* __restrict__ host_blob,
-          const int32_t* __restrict__ host_tab,
-          const uint64_t* __restrict__ seed_ptr,
-          int auth_env_off, int auth_env_len) {
-    const uint64_t seed = *seed_ptr;
-    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
-
-    const int wv = threadIdx.x / WAVE;
-    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
-    if (req >= n) return;
-    const int lane = lane_id();
+// written cooperatively; the JSON-echo body is validated with a ballot
+// structural pass. The GZ=true instantiation (k_respond_gz) additionally
+// deflate-compresses the body in-LDS when the request advertised
+// Accept-Encoding: gzip and the plain body >= gzip_min.
+template <bool GZ>
+__device__ __forceinline__ void respond_impl(
+        const uint8_t* __restrict__ reqs,
+        const int64_t* __restrict__ req_off,
+        int32_t* __restrict__ fields,
+        uint8_t* __restrict__ resp,
+        int32_t* __restrict__ resp_len_out,
+        int n, int rslot,
+        const int32_t* __restrict__ handler_tab, int n_routes,
+        const uint8_t* __restrict__ blob,
+        const uint8_t* __restrict__ host_blob,
+        const int32_t* __restrict__ host_tab,
+        uint64_t seed, int auth_env_off, int auth_env_len,
+        int gzip_min, uint8_t* obuf, uint8_t* plainbuf, uint16_t* hash,
+        const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
     const uint8_t* rbase = reqs + req_off[req];
-    uint8_t* obuf = obuf_all + wv * MAX_SLOT;
 
     const int kind = F[FI_KIND];
     int status = F[FI_STATUS];
@@ -475,7 +686,6 @@ k_respond(const uint8_t* __restrict__ reqs,
                             run = bs_suffix;
                         } else {
                             const uint64_t below = m_bs & (bit - 1);
-                            // count consecutive set bits ending at p-1
                             run = 0;
                             uint64_t probe = 1ull << (p - 1);
                             while ((below & probe) && run < p) {
@@ -491,7 +701,6 @@ k_respond(const uint8_t* __restrict__ reqs,
                         else if (--depth < 0) { bad = true; break; }
                     }
                 }
-                // trailing backslash run of this chunk
                 if (m_bs == ~0ull) bs_suffix += 64;
                 else bs_suffix = __builtin_clzll(~m_bs);
             }
@@ -501,7 +710,6 @@ k_respond(const uint8_t* __restrict__ reqs,
         bad = __shfl(bad ? 1 : 0, 0) != 0;
         if (bad) {
             status = 500;
-            // invalid-body envelope: blob[0:4]=len, blob[4:4+len]=bytes
             const int elen = *(const int32_t*)blob;
             body_src = blob + 4;
             body_src_len = elen;
@@ -537,7 +745,32 @@ k_respond(const uint8_t* __restrict__ reqs,
         ct_id = 0;
     }
 
-    const int body_total = body_src_len + (env ? 9 : 0);  // {"data": + }
+    int body_total = body_src_len + (env ? 9 : 0);  // {"data": + }
+
+    // ---- fused gzip (GZ instantiation only) --------------------------------
+    int content_enc = 0;
+    if (GZ) {
+        if ((flags & FL_ACCEPT_GZIP) && !is_options &&
+            body_total >= gzip_min && body_total <= MAX_SLOT) {
+            // assemble the plain body (envelope included) into plainbuf
+            const int boff = env ? 8 : 0;
+            if (env && lane < 8) plainbuf[lane] = ENV_OPEN[lane];
+            for (int i = lane; i < body_src_len; i += WAVE)
+                plainbuf[boff + i] = body_src[i];
+            if (env && lane == 0) plainbuf[boff + body_src_len] = '}';
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            const int gz_total = deflate_gzip_wave(
+                plainbuf, body_total, obuf + 512, MAX_SLOT - 512,
+                hash, crc_tab, lane);
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            if (gz_total > 0) {
+                content_enc = 1;
+                body_total = gz_total;
+                env = 0;
+            }
+            // gz_total < 0 (incompressible/overflow): serve uncompressed
+        }
+    }
 
     // ---- header segments (lengths first, then cooperative writes) ---------
     int rlen_reason;
@@ -552,12 +785,14 @@ k_respond(const uint8_t* __restrict__ reqs,
     const char* conn = keep ? HDR_CONN_KA : HDR_CONN_CL;
     const int conn_len =
         (keep ? sizeof(HDR_CONN_KA) : sizeof(HDR_CONN_CL)) - 1;
+    const int ce_len = content_enc ? (int)sizeof(HDR_CE) - 1 : 0;
 
     // segment offsets (all lanes compute identically)
     const int o_status = 9;                       // after "HTTP/1.1 "
     const int o_reason = o_status + 3 + 1;        // "xxx "
     const int o_ct = o_reason + rlen_reason + 2;  // reason \r\n
-    const int o_cors = o_ct + ct_len;
+    const int o_ce = o_ct + ct_len;
+    const int o_cors = o_ce + ce_len;
     const int o_corr = o_cors + (int)sizeof(HDR_CORS) - 1;
     const int o_corrhex = o_corr + (int)sizeof(HDR_CORR) - 1;
     const int o_cl = o_corrhex + 32 + 2;
@@ -569,7 +804,7 @@ k_respond(const uint8_t* __restrict__ reqs,
     if (lane < 9) obuf[lane] = HDR_P1[lane];
     if (lane < (unsigned)rlen_reason) obuf[o_reason + lane] = reason[lane];
     if (lane < (unsigned)ct_len) obuf[o_ct + lane] = ct_str[lane];
-    // CORS is 95 bytes: two strided stores
+    if (content_enc && lane < ce_len) obuf[o_ce + lane] = HDR_CE[lane];
     {
         const int cors_len = (int)sizeof(HDR_CORS) - 1;
         for (int i = lane; i < cors_len; i += WAVE)
@@ -606,13 +841,25 @@ k_respond(const uint8_t* __restrict__ reqs,
         }
     }
 
-    // ---- body into LDS (coalesced byte lanes) ------------------------------
-    const int body_start = hl + (env ? 8 : 0);
-    for (int i = lane; i < body_src_len; i += WAVE)
-        obuf[body_start + i] = body_src[i];
+    // ---- body into LDS -----------------------------------------------------
+    if (GZ && content_enc) {
+        // shift the deflate output from obuf+512 down to obuf+hl.
+        // dst < src and the wave is lockstep (loads of an iteration all
+        // execute before its stores), so a forward sweep is safe.
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        for (int i = lane; i < body_total; i += WAVE) {
+            const uint8_t v = obuf[512 + i];
+            obuf[hl + i] = v;
+        }
+    } else {
+        const int body_start = hl + (env ? 8 : 0);
+        for (int i = lane; i < body_src_len; i += WAVE)
+            obuf[body_start + i] = body_src[i];
+    }
 
     const int total = hl + body_total;
     // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     if (total <= rslot) {
         const int nv = (total + 15) >> 4;
         const uint4* src = (const uint4*)obuf;
@@ -625,6 +872,70 @@ k_respond(const uint8_t* __restrict__ reqs,
         F[FI_RESP_OFF] = (int)((size_t)req * rslot);
         resp_len_out[req] = tl;
     }
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_respond(const uint8_t* __restrict__ reqs,
+          const int64_t* __restrict__ req_off,
+          int32_t* __restrict__ fields,
+          uint8_t* __restrict__ resp,
+          int32_t* __restrict__ resp_len_out,
+          int n, int rslot,
+          const int32_t* __restrict__ handler_tab, int n_routes,
+          const uint8_t* __restrict__ blob,
+          const uint8_t* __restrict__ host_blob,
+          const int32_t* __restrict__ host_tab,
+          const uint64_t* __restrict__ seed_ptr,
+          int auth_env_off, int auth_env_len) {
+    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    respond_impl<false>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
+                        handler_tab, n_routes, blob, host_blob, host_tab,
+                        *seed_ptr, auth_env_off, auth_env_len, 0,
+                        obuf_all + wv * MAX_SLOT, nullptr, nullptr, nullptr,
+                        req, lane_id());
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_respond_gz(const uint8_t* __restrict__ reqs,
+             const int64_t* __restrict__ req_off,
+             int32_t* __restrict__ fields,
+             uint8_t* __restrict__ resp,
+             int32_t* __restrict__ resp_len_out,
+             int n, int rslot,
+             const int32_t* __restrict__ handler_tab, int n_routes,
+             const uint8_t* __restrict__ blob,
+             const uint8_t* __restrict__ host_blob,
+             const int32_t* __restrict__ host_tab,
+             const uint64_t* __restrict__ seed_ptr,
+             int auth_env_off, int auth_env_len, int gzip_min) {
+    // single __shared__ block (cdna guide §5 trap 4a)
+    __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
+                           WAVES_PER_BLOCK * GZ_HASH_SIZE * 2 + 256 * 4];
+    uint8_t* obuf_all = lds;
+    uint8_t* plain_all = lds + WAVES_PER_BLOCK * MAX_SLOT;
+    uint16_t* hash_all = (uint16_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2);
+    uint32_t* crc_tab = (uint32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2 +
+                                    WAVES_PER_BLOCK * GZ_HASH_SIZE * 2);
+    // build the CRC32 table cooperatively BEFORE any thread can exit
+    for (int i = threadIdx.x; i < 256; i += BLOCK_THREADS) {
+        uint32_t c = (uint32_t)i;
+        for (int k = 0; k < 8; ++k)
+            c = (c & 1) ? (c >> 1) ^ 0xEDB88320u : (c >> 1);
+        crc_tab[i] = c;
+    }
+    __syncthreads();
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    respond_impl<true>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
+                       handler_tab, n_routes, blob, host_blob, host_tab,
+                       *seed_ptr, auth_env_off, auth_env_len, gzip_min,
+                       obuf_all + wv * MAX_SLOT, plain_all + wv * MAX_SLOT,
+                       hash_all + wv * GZ_HASH_SIZE, crc_tab,
+                       req, lane_id());
 }
 
 // ---------------------------------------------------------------------------
@@ -929,18 +1240,34 @@ int gofr_launch_respond(
         void* resp_len_out, int n, int rslot,
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
-        const void* seed_ptr, int auth_env_off, int auth_env_len) {
+        const void* seed_ptr, int auth_env_off, int auth_env_len,
+        int gzip_min) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
-    hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
-                       (hipStream_t)stream,
-                       (const uint8_t*)reqs, (const int64_t*)req_off,
-                       (int32_t*)fields,
-                       (uint8_t*)resp, (int32_t*)resp_len_out,
-                       n, rslot,
-                       (const int32_t*)handler_tab, n_routes,
-                       (const uint8_t*)blob, (const uint8_t*)host_blob,
-                       (const int32_t*)host_tab, (const uint64_t*)seed_ptr,
-                       auth_env_off, auth_env_len);
+    if (gzip_min > 0) {
+        hipLaunchKernelGGL(k_respond_gz, dim3(blocks), dim3(BLOCK_THREADS),
+                           0, (hipStream_t)stream,
+                           (const uint8_t*)reqs, (const int64_t*)req_off,
+                           (int32_t*)fields,
+                           (uint8_t*)resp, (int32_t*)resp_len_out,
+                           n, rslot,
+                           (const int32_t*)handler_tab, n_routes,
+                           (const uint8_t*)blob, (const uint8_t*)host_blob,
+                           (const int32_t*)host_tab,
+                           (const uint64_t*)seed_ptr,
+                           auth_env_off, auth_env_len, gzip_min);
+    } else {
+        hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                           (hipStream_t)stream,
+                           (const uint8_t*)reqs, (const int64_t*)req_off,
+                           (int32_t*)fields,
+                           (uint8_t*)resp, (int32_t*)resp_len_out,
+                           n, rslot,
+                           (const int32_t*)handler_tab, n_routes,
+                           (const uint8_t*)blob, (const uint8_t*)host_blob,
+                           (const int32_t*)host_tab,
+                           (const uint64_t*)seed_ptr,
+                           auth_env_off, auth_env_len);
+    }
     return (int)hipGetLastError();
 }
 

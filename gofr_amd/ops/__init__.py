@@ -90,7 +90,7 @@ class HipOps:
             ctypes.c_int, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-            ctypes.c_void_p]
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int]
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
@@ -128,7 +128,8 @@ class HipOps:
 
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
-                host_blob_t, host_tab_t, seed_t, auth_env=(0, 0)):
+                host_blob_t, host_tab_t, seed_t, auth_env=(0, 0),
+                gzip_min=0):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -142,7 +143,7 @@ class HipOps:
             ctypes.c_void_p(host_blob_t.data_ptr()),
             ctypes.c_void_p(host_tab_t.data_ptr()),
             ctypes.c_void_p(seed_t.data_ptr()),
-            auth_env[0], auth_env[1])
+            auth_env[0], auth_env[1], gzip_min)
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
 
@@ -418,7 +419,7 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 fields: np.ndarray, rslot: int,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
-                seed: int, auth_env=(0, 0)):
+                seed: int, auth_env=(0, 0), gzip_min: int = 0):
     """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
     n = len(fields)
     host_tab = np.asarray(host_tab, np.int32).reshape(-1)
@@ -469,12 +470,24 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
             ct_id = 0
 
         body_total = len(body_src) + (9 if env else 0)
+        content_enc = False
+        if (gzip_min > 0 and (flags & FL_ACCEPT_GZIP) and not is_options
+                and gzip_min <= body_total <= MAX_SLOT):
+            plain = (b'{"data":' + body_src + b"}") if env else body_src
+            gz = gzip_static_mirror(plain)
+            if gz is not None:
+                content_enc = True
+                body_src = gz
+                body_total = len(gz)
+                env = False
         reason = _dev_reason(status)
         h1 = splitmix64(seed ^ r)
         h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03)
         corr = f"{h1:016x}{h2:016x}".encode()
         head = (b"HTTP/1.1 " + f"{status:03d}".encode() + b" " + reason +
-                b"\r\n" + _CT_STRS[ct_id] + _CORS +
+                b"\r\n" + _CT_STRS[ct_id] +
+                (b"Content-Encoding: gzip\r\n" if content_enc else b"") +
+                _CORS +
                 b"X-Correlation-ID: " + corr + b"\r\n" +
                 b"Content-Length: " + str(body_total).encode() + b"\r\n" +
                 (b"Connection: keep-alive\r\n\r\n" if keep
@@ -489,6 +502,119 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
         fields[r][FI_RESP_LEN] = len(payload)
         fields[r][FI_RESP_OFF] = r * rslot
     return resp, resp_len
+
+
+GZ_HASH_BITS = 9
+GZ_MIN_MATCH = 3
+GZ_MAX_DIST = 2048
+MAX_SLOT = 4096
+
+_LEN_BASE = [3,4,5,6,7,8,9,10,11,13,15,17,19,23,27,31,
+             35,43,51,59,67,83,99,115,131,163,195,227,258]
+_LEN_EXTRA = [0,0,0,0,0,0,0,0,1,1,1,1,2,2,2,2,3,3,3,3,4,4,4,4,5,5,5,5,0]
+_DIST_BASE = [1,2,3,4,5,7,9,13,17,25,33,49,65,97,129,193,257,385,513,769,
+              1025,1537,2049,3073,4097,6145,8193,12289,16385,24577]
+_DIST_EXTRA = [0,0,0,0,1,1,2,2,3,3,4,4,5,5,6,6,7,7,8,8,9,9,10,10,11,11,
+               12,12,13,13]
+
+import zlib as _zlib
+
+
+def _bitrev(v, ln):
+    r = 0
+    for _ in range(ln):
+        r = (r << 1) | (v & 1)
+        v >>= 1
+    return r
+
+
+def _fixed_lit(sym):
+    if sym < 144:
+        return _bitrev(0x30 + sym, 8), 8
+    if sym < 256:
+        return _bitrev(0x190 + sym - 144, 9), 9
+    if sym < 280:
+        return _bitrev(sym - 256, 7), 7
+    return _bitrev(0xC0 + sym - 280, 8), 8
+
+
+def gzip_static_mirror(data: bytes, cap: int = MAX_SLOT - 512):
+    """Byte-exact model of deflate_gzip_wave (greedy LZ77 + static
+    huffman + gzip framing). Returns None when the output would exceed
+    `cap` (the kernel falls back to uncompressed)."""
+    n = len(data)
+    out = bytearray(b"\x1f\x8b\x08\x00\x00\x00\x00\x00\x00\xff")
+    bitbuf = 0
+    nbits = 0
+
+    def put(bits, ln):
+        nonlocal bitbuf, nbits
+        bitbuf |= bits << nbits
+        nbits += ln
+        while nbits >= 8:
+            out.append(bitbuf & 0xFF)
+            bitbuf >>= 8
+            nbits -= 8
+
+    put(1, 1)
+    put(1, 2)
+    hash_tab = [0] * (1 << GZ_HASH_BITS)
+    pos = 0
+    overflow = False
+    while pos < n:
+        cand = -1
+        if pos + GZ_MIN_MATCH <= n:
+            h = (((data[pos] | (data[pos + 1] << 8) |
+                   (data[pos + 2] << 16)) * 0x9E3779B1) & 0xFFFFFFFF) \
+                >> (32 - GZ_HASH_BITS)
+            stored = hash_tab[h] - 1
+            if stored >= 0 and pos - stored <= GZ_MAX_DIST and stored < pos:
+                cand = stored
+            hash_tab[h] = pos + 1
+        mlen = 0
+        if cand >= 0:
+            while mlen < 258 and pos + mlen < n and \
+                    data[cand + mlen] == data[pos + mlen]:
+                mlen += 1
+        if mlen >= GZ_MIN_MATCH:
+            # largest i with mlen >= base[i] (mirror of the C loop)
+            idx = max(i for i in range(29) if mlen >= _LEN_BASE[i])
+            code, clen = _fixed_lit(257 + idx)
+            put(code, clen)
+            if _LEN_EXTRA[idx]:
+                put(mlen - _LEN_BASE[idx], _LEN_EXTRA[idx])
+            dist = pos - cand
+            didx = max(i for i in range(30) if dist >= _DIST_BASE[i])
+            put(_bitrev(didx, 5), 5)
+            if _DIST_EXTRA[didx]:
+                put(dist - _DIST_BASE[didx], _DIST_EXTRA[didx])
+            for i in range(1, mlen):
+                q = pos + i
+                if q + GZ_MIN_MATCH <= n:
+                    h2 = (((data[q] | (data[q + 1] << 8) |
+                            (data[q + 2] << 16)) * 0x9E3779B1)
+                          & 0xFFFFFFFF) >> (32 - GZ_HASH_BITS)
+                    hash_tab[h2] = q + 1
+            pos += mlen
+        else:
+            code, clen = _fixed_lit(data[pos])
+            put(code, clen)
+            pos += 1
+        if len(out) - 10 > cap - 24:
+            overflow = True
+            break
+    if overflow:
+        return None
+    code, clen = _fixed_lit(256)
+    put(code, clen)
+    if nbits > 0:
+        out.append(bitbuf & 0xFF)
+    crc = _zlib.crc32(data) & 0xFFFFFFFF
+    out += crc.to_bytes(4, "little")
+    out += (n & 0xFFFFFFFF).to_bytes(4, "little")
+    if len(out) > cap:
+        return None
+    return bytes(out)
 
 
 def cpu_auth(reqs: np.ndarray, req_off: np.ndarray, fields: np.ndarray,

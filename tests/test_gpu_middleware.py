@@ -97,3 +97,38 @@ def test_k_varint_spans_matches_mirror():
     got_n = d_out_n.cpu().numpy()
     assert np.array_equal(got_n, ref_n)
     assert np.array_equal(got_out, ref_out)
+
+
+def test_k_respond_gz_matches_mirror_bytes():
+    """Fused gzip kernel vs the byte-exact Python mirror + decompression."""
+    import gzip as _gz
+    import json
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/big", handlers.static_json({"pad": "y" * 700}))
+    app.enable_gzip(min_size=128)
+    gpu = BatchEngine(app, device="cuda", max_batch=2048)
+    cpu = BatchEngine(app, device="cpu", max_batch=2048)
+    cpu._seed = gpu._seed
+    import random
+    rng = random.Random(99)
+    raws = []
+    for i in range(256):
+        body = json.dumps(
+            {"i": i, "pad": "ab" * rng.randrange(10, 400),
+             "r": str(rng.random())}).encode()
+        hdrs = {"Accept-Encoding": "gzip"} if i % 3 else {}
+        raws.append(http_req("POST", "/echo", body, headers=hdrs))
+        if i % 7 == 0:
+            raws.append(http_req("GET", "/big",
+                                 headers={"Accept-Encoding": "gzip"}))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    n_gz = 0
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, f"req {i}\nGPU {go[:200]!r}\nCPU {co[:200]!r}"
+        head, _, body = go.partition(b"\r\n\r\n")
+        if b"Content-Encoding: gzip" in head:
+            n_gz += 1
+            _gz.decompress(body)  # must be a valid gzip stream
+    assert n_gz > 100

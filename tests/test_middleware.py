@@ -185,3 +185,50 @@ def test_varint_spans_malformed():
         np.frombuffer(b"\x0a\xff", np.uint8).copy(),  # len 255 but truncated
         np.asarray([0], np.int64), np.asarray([2], np.int32))
     assert out_n[0] == -1
+
+
+# -- fused gzip engine path ---------------------------------------------------
+
+def test_gzip_mirror_valid_gzip_stream():
+    from gofr_amd.ops import gzip_static_mirror
+    for data in [b"hello world hello world hello world",
+                 b'{"data":{"pad":"' + b"z" * 600 + b'"}}',
+                 bytes(range(256)) * 4,
+                 b"a",
+                 b'{"x":' + b"12345 " * 100 + b'}']:
+        gz = gzip_static_mirror(data)
+        assert gz is not None
+        assert gzip.decompress(gz) == data
+    # repetitive data must actually compress
+    data = b'{"data":{"pad":"' + b"z" * 1000 + b'"}}'
+    gz = gzip_static_mirror(data)
+    assert len(gz) < len(data) // 4
+
+
+def test_engine_gzip_fused():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/big", handlers.static_json({"pad": "y" * 500}))
+    app.enable_gzip(min_size=128)
+    eng = BatchEngine(app, device="cpu")
+    body = b'{"pad":"' + b"x" * 800 + b'"}'
+    gz_req = http_req("POST", "/echo", body,
+                      headers={"Accept-Encoding": "gzip"})
+    plain_req = http_req("POST", "/echo", body)
+    outs = eng.process([gz_req, plain_req,
+                        http_req("GET", "/big",
+                                 headers={"Accept-Encoding": "gzip"})])
+    head, _, rbody = outs[0].partition(b"\r\n\r\n")
+    assert b"Content-Encoding: gzip" in head
+    assert json.loads(gzip.decompress(rbody)) == {"data": json.loads(body)}
+    assert int(dict(
+        l.split(b": ") for l in head.split(b"\r\n")[1:]
+    )[b"Content-Length"]) == len(rbody)
+    # no accept-encoding -> plain
+    head2, _, rbody2 = outs[1].partition(b"\r\n\r\n")
+    assert b"Content-Encoding" not in head2
+    assert json.loads(rbody2) == {"data": json.loads(body)}
+    # static route compressed too
+    head3, _, rbody3 = outs[2].partition(b"\r\n\r\n")
+    assert b"Content-Encoding: gzip" in head3
+    assert json.loads(gzip.decompress(rbody3)) == {"data": {"pad": "y" * 500}}
