@@ -443,116 +443,81 @@ class BatchEngine:
 
 
 class GPUServer:
-    """Socket front-end serving through the batch engine.
+    """Native serving front-end: C++ epoll ingress + GPU batch engine.
 
-    Accepts connections with the CPU listener machinery, forms batches
-    with an adaptive deadline, processes them through BatchEngine, and
-    writes responses back. This is the serving path of App.Run(engine=gpu);
-    bench.py drives BatchEngine directly (synthetic in-memory load).
+    The epoll loop (gofr_amd/native/core/epoll_server.cpp) reads complete
+    HTTP requests off the sockets and harvest() copies a batch of them
+    straight into the engine's pinned ingress ring; the engine runs the
+    kernel pipeline; send() writes the compact egress ring back to the
+    sockets. No Python in the byte path — Python only orchestrates
+    batches (and runs host-trampoline handlers when a route needs it).
+
+    Without a GPU the same loop runs on the CPU-mirror engine (numpy
+    staging buffers), so the server is testable on the CPU box.
     """
 
     def __init__(self, app, port: int, batch_window_us: int = 200,
-                 max_batch: int = 4096):
+                 max_batch: int = 8192):
         self.app = app
         self.port = port
         self.engine = BatchEngine(app, max_batch=max_batch)
         self.batch_window_us = batch_window_us
+        self._core = None
+        self._thread = None
         self._stop = None
-        self._listener = None
 
     def start(self):
-        import queue
-        import socket
         import threading
+        from .. import _core  # built by setup.py / __graft_entry__.build()
+        self._core = _core.EpollServer(self.port)
+        self._core.start()
+        self.port = self._core.port()
         self._stop = threading.Event()
-        self._q = queue.Queue()
-        sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
-        sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
-        sock.bind(("0.0.0.0", self.port))
-        sock.listen(1024)
-        self._listener = sock
-        import threading as th
-        th.Thread(target=self._accept_loop, daemon=True).start()
-        th.Thread(target=self._batch_loop, daemon=True).start()
+        self._thread = threading.Thread(target=self._serve_loop,
+                                        daemon=True)
+        self._thread.start()
 
-    def _accept_loop(self):
-        import threading
+    def _serve_loop(self):
+        eng = self.engine
+        nb = eng.max_batch
+        conn_ids = np.zeros(nb, np.uint64)
+        conn_ptr = conn_ids.ctypes.data
+        gpu = eng.device is not None
+        if gpu:
+            ln = eng.lanes[0]
+            buf_ptr = ln.p_reqs.data_ptr()
+            cap = eng.max_bytes
+            off_ptr = ln.p_req_off.data_ptr()
+            len_ptr = ln.p_req_len.data_ptr()
+        else:
+            buf_np = np.zeros(eng.max_bytes, np.uint8)
+            off_np = np.zeros(nb, np.int64)
+            len_np = np.zeros(nb, np.int32)
+            buf_ptr = buf_np.ctypes.data
+            cap = eng.max_bytes
+            off_ptr = off_np.ctypes.data
+            len_ptr = len_np.ctypes.data
         while not self._stop.is_set():
-            try:
-                conn, _ = self._listener.accept()
-            except OSError:
-                return
-            threading.Thread(target=self._conn_loop, args=(conn,),
-                             daemon=True).start()
-
-    def _conn_loop(self, conn):
-        import socket as _s
-        import threading
-        conn.setsockopt(_s.IPPROTO_TCP, _s.TCP_NODELAY, 1)
-        buf = b""
-        try:
-            while not self._stop.is_set():
-                while b"\r\n\r\n" not in buf:
-                    chunk = conn.recv(65536)
-                    if not chunk:
-                        return
-                    buf += chunk
-                he = buf.index(b"\r\n\r\n") + 4
-                clen = 0
-                for line in buf[:he].split(b"\r\n")[1:]:
-                    if line[:15].lower() == b"content-length:":
-                        clen = int(line.split(b":", 1)[1].strip() or b"0")
-                        break
-                while len(buf) < he + clen:
-                    chunk = conn.recv(65536)
-                    if not chunk:
-                        return
-                    buf += chunk
-                raw, buf = buf[:he + clen], buf[he + clen:]
-                done = threading.Event()
-                slotref = {}
-                self._q.put((raw, done, slotref))
-                done.wait(timeout=30)
-                resp = slotref.get("resp", b"")
-                if resp:
-                    conn.sendall(resp)
-                else:
-                    return
-        except (OSError, ValueError):
-            return
-        finally:
-            try:
-                conn.close()
-            except OSError:
-                pass
-
-    def _batch_loop(self):
-        import queue
-        while not self._stop.is_set():
-            try:
-                first = self._q.get(timeout=0.2)
-            except queue.Empty:
+            n, nbytes = self._core.harvest(
+                buf_ptr, cap, off_ptr, len_ptr, conn_ptr, nb,
+                self.batch_window_us)
+            if n == 0:
                 continue
-            items = [first]
-            deadline = time.perf_counter() + self.batch_window_us / 1e6
-            while len(items) < self.engine.max_batch:
-                remain = deadline - time.perf_counter()
-                if remain <= 0:
-                    break
-                try:
-                    items.append(self._q.get(timeout=remain))
-                except queue.Empty:
-                    break
-            outs = self.engine.process([it[0] for it in items])
-            for out, (_, done, slotref) in zip(outs, items):
-                slotref["resp"] = out
-                done.set()
+            if gpu:
+                eng.submit(n, nbytes, 0)
+                out_t, roff_t, rlen_t = eng.complete(0)
+                self._core.send(conn_ptr, n, out_t.data_ptr(),
+                                roff_t.data_ptr(), rlen_t.data_ptr())
+            else:
+                out, roffs, rlens = eng.process_packed(
+                    buf_np, off_np[:n], len_np[:n])
+                self._core.send(conn_ptr, n, out.ctypes.data,
+                                roffs.ctypes.data, rlens.ctypes.data)
 
     def stop(self):
         if self._stop is not None:
             self._stop.set()
-        if self._listener is not None:
-            try:
-                self._listener.close()
-            except OSError:
-                pass
+        if self._core is not None:
+            self._core.stop()
+        if self._thread is not None:
+            self._thread.join(timeout=2)

@@ -1,0 +1,356 @@
+// gofr_amd native ingress: epoll HTTP/1.1 listener staging request bytes
+// for the GPU batch engine.
+//
+// Role of the reference's net/http accept loop (pkg/gofr/httpServer.go:
+// 24-36, goroutine per connection) re-designed for the MI355X data plane:
+// instead of a handler per connection, the event loop reads COMPLETE
+// HTTP requests (headers + content-length body) into per-connection
+// buffers and batch-harvests them straight into the engine's pinned
+// ingress ring (no Python in the byte path). Responses are written back
+// from the engine's pinned egress buffer, honoring keep-alive.
+//
+// Python surface (pybind11 module gofr_amd._core):
+//   s = EpollServer(port, max_conn_buf)
+//   s.start(nthreads)
+//   n, nbytes = s.harvest(buf_ptr, buf_cap, off_ptr, len_ptr, conn_ptr,
+//                         max_n, window_us)      # blocks <= window_us
+//   s.send(conn_ptr, n, out_ptr, roff_ptr, rlen_ptr)
+//   s.stop()
+
+#include <pybind11/pybind11.h>
+
+#include <arpa/inet.h>
+#include <cerrno>
+#include <cstring>
+#include <deque>
+#include <fcntl.h>
+#include <mutex>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <string>
+#include <sys/epoll.h>
+#include <sys/socket.h>
+#include <thread>
+#include <unistd.h>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+struct Conn {
+    int fd = -1;
+    std::string rbuf;      // bytes read, not yet parsed into a request
+    std::string wbuf;      // bytes pending write
+    bool close_after_write = false;
+    bool dead = false;
+    uint64_t id = 0;
+};
+
+struct PendingReq {
+    uint64_t conn_id;
+    std::string bytes;
+};
+
+class EpollServer {
+public:
+    EpollServer(int port, size_t max_req = 1 << 20)
+        : port_(port), max_req_(max_req) {}
+
+    ~EpollServer() { stop(); }
+
+    void start() {
+        listen_fd_ = ::socket(AF_INET, SOCK_STREAM | SOCK_NONBLOCK, 0);
+        if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+        int one = 1;
+        setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+        sockaddr_in addr{};
+        addr.sin_family = AF_INET;
+        addr.sin_addr.s_addr = INADDR_ANY;
+        addr.sin_port = htons((uint16_t)port_);
+        if (bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0)
+            throw std::runtime_error("bind() failed: " +
+                                     std::string(strerror(errno)));
+        if (port_ == 0) {
+            socklen_t alen = sizeof(addr);
+            getsockname(listen_fd_, (sockaddr*)&addr, &alen);
+            port_ = ntohs(addr.sin_port);
+        }
+        if (listen(listen_fd_, 4096) != 0)
+            throw std::runtime_error("listen() failed");
+        ep_ = epoll_create1(0);
+        epoll_event ev{};
+        ev.events = EPOLLIN;
+        ev.data.u64 = 0;  // listener marker
+        epoll_ctl(ep_, EPOLL_CTL_ADD, listen_fd_, &ev);
+        running_ = true;
+        loop_ = std::thread([this] { event_loop(); });
+    }
+
+    void stop() {
+        if (!running_) return;
+        running_ = false;
+        if (loop_.joinable()) loop_.join();
+        for (auto& kv : conns_) ::close(kv.second.fd);
+        conns_.clear();
+        if (listen_fd_ >= 0) ::close(listen_fd_);
+        if (ep_ >= 0) ::close(ep_);
+        listen_fd_ = ep_ = -1;
+    }
+
+    int port() const { return port_; }
+
+    // Harvest up to max_n complete requests into the caller's buffers
+    // (the engine's pinned ingress ring). Blocks up to window_us for the
+    // FIRST request, then drains whatever is ready. GIL released.
+    std::pair<int, long> harvest(uintptr_t buf_ptr, long buf_cap,
+                                 uintptr_t off_ptr, uintptr_t len_ptr,
+                                 uintptr_t conn_ptr, int max_n,
+                                 int window_us) {
+        py::gil_scoped_release rel;
+        uint8_t* buf = (uint8_t*)buf_ptr;
+        int64_t* offs = (int64_t*)off_ptr;
+        int32_t* lens = (int32_t*)len_ptr;
+        uint64_t* cids = (uint64_t*)conn_ptr;
+        int n = 0;
+        long pos = 0;
+        const auto deadline = std::chrono::steady_clock::now() +
+                              std::chrono::microseconds(window_us);
+        while (n < max_n) {
+            std::unique_lock<std::mutex> lk(mu_);
+            if (ready_.empty()) {
+                lk.unlock();
+                if (n > 0 ||
+                    std::chrono::steady_clock::now() >= deadline)
+                    break;
+                std::this_thread::sleep_for(std::chrono::microseconds(50));
+                continue;
+            }
+            PendingReq req = std::move(ready_.front());
+            ready_.pop_front();
+            lk.unlock();
+            const long sz = (long)req.bytes.size();
+            if (pos + sz > buf_cap) {
+                // ring full: requeue and stop
+                std::lock_guard<std::mutex> lk2(mu_);
+                ready_.push_front(std::move(req));
+                break;
+            }
+            memcpy(buf + pos, req.bytes.data(), sz);
+            offs[n] = pos;
+            lens[n] = (int32_t)sz;
+            cids[n] = req.conn_id;
+            pos += sz;
+            ++n;
+        }
+        return {n, pos};
+    }
+
+    // Write responses back (engine egress pinned buffer). Partial writes
+    // are queued on the connection and drained by the event loop.
+    void send(uintptr_t conn_ptr, int n, uintptr_t out_ptr,
+              uintptr_t roff_ptr, uintptr_t rlen_ptr) {
+        py::gil_scoped_release rel;
+        const uint64_t* cids = (const uint64_t*)conn_ptr;
+        const uint8_t* out = (const uint8_t*)out_ptr;
+        const int32_t* roffs = (const int32_t*)roff_ptr;
+        const int32_t* rlens = (const int32_t*)rlen_ptr;
+        std::lock_guard<std::mutex> lk(wmu_);
+        for (int i = 0; i < n; ++i) {
+            auto it = conn_index_.find(cids[i]);
+            if (it == conn_index_.end()) continue;
+            Conn* c = it->second;
+            c->wbuf.append((const char*)out + roffs[i], (size_t)rlens[i]);
+            pending_writes_.push_back(cids[i]);
+        }
+        // wake the loop via the self-pipe-less approach: loop polls with
+        // a short timeout, so queued writes drain within ~1 ms.
+    }
+
+    long ready_count() {
+        std::lock_guard<std::mutex> lk(mu_);
+        return (long)ready_.size();
+    }
+
+private:
+    void event_loop() {
+        std::vector<epoll_event> events(1024);
+        std::string tmp(1 << 16, '\0');
+        while (running_) {
+            drain_pending_writes();
+            const int k = epoll_wait(ep_, events.data(),
+                                     (int)events.size(), 1);
+            for (int i = 0; i < k; ++i) {
+                const uint64_t id = events[i].data.u64;
+                if (id == 0) {
+                    accept_new();
+                    continue;
+                }
+                auto it = conns_.find(id);
+                if (it == conns_.end()) continue;
+                Conn& c = it->second;
+                if (events[i].events & (EPOLLHUP | EPOLLERR)) {
+                    close_conn(c);
+                    continue;
+                }
+                if (events[i].events & EPOLLIN) {
+                    if (!read_ready(c, tmp)) {
+                        close_conn(c);
+                        continue;
+                    }
+                }
+                if (events[i].events & EPOLLOUT) flush_conn(c);
+            }
+            gc_dead();
+        }
+    }
+
+    void accept_new() {
+        while (true) {
+            const int fd = accept4(listen_fd_, nullptr, nullptr,
+                                   SOCK_NONBLOCK);
+            if (fd < 0) return;
+            int one = 1;
+            setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+            const uint64_t id = next_id_++;
+            Conn& c = conns_[id];
+            c.fd = fd;
+            c.id = id;
+            {
+                std::lock_guard<std::mutex> lk(wmu_);
+                conn_index_[id] = &c;
+            }
+            epoll_event ev{};
+            ev.events = EPOLLIN;
+            ev.data.u64 = id;
+            epoll_ctl(ep_, EPOLL_CTL_ADD, fd, &ev);
+        }
+    }
+
+    bool read_ready(Conn& c, std::string& tmp) {
+        while (true) {
+            const ssize_t r = recv(c.fd, tmp.data(), tmp.size(), 0);
+            if (r == 0) return false;
+            if (r < 0) {
+                if (errno == EAGAIN || errno == EWOULDBLOCK) break;
+                return false;
+            }
+            c.rbuf.append(tmp.data(), (size_t)r);
+            if (c.rbuf.size() > max_req_) return false;
+        }
+        // slice out complete requests
+        while (true) {
+            const size_t he = c.rbuf.find("\r\n\r\n");
+            if (he == std::string::npos) break;
+            size_t clen = 0;
+            // find content-length (case-insensitive) in the header block
+            for (size_t p = 0; p < he;) {
+                size_t eol = c.rbuf.find("\r\n", p);
+                if (eol == std::string::npos || eol > he) eol = he;
+                if (eol - p > 15) {
+                    static const char k[] = "content-length:";
+                    bool match = true;
+                    for (int j = 0; j < 15; ++j) {
+                        char ch = c.rbuf[p + j];
+                        if (ch >= 'A' && ch <= 'Z') ch |= 0x20;
+                        if (ch != k[j]) { match = false; break; }
+                    }
+                    if (match)
+                        clen = strtoul(c.rbuf.c_str() + p + 15, nullptr, 10);
+                }
+                p = eol + 2;
+            }
+            const size_t total = he + 4 + clen;
+            if (c.rbuf.size() < total) break;
+            PendingReq req;
+            req.conn_id = c.id;
+            req.bytes = c.rbuf.substr(0, total);
+            c.rbuf.erase(0, total);
+            std::lock_guard<std::mutex> lk(mu_);
+            ready_.push_back(std::move(req));
+        }
+        return true;
+    }
+
+    void drain_pending_writes() {
+        std::vector<uint64_t> ids;
+        {
+            std::lock_guard<std::mutex> lk(wmu_);
+            ids.swap(pending_writes_);
+        }
+        for (uint64_t id : ids) {
+            auto it = conns_.find(id);
+            if (it != conns_.end()) flush_conn(it->second);
+        }
+    }
+
+    void flush_conn(Conn& c) {
+        while (!c.wbuf.empty()) {
+            const ssize_t w = ::send(c.fd, c.wbuf.data(), c.wbuf.size(),
+                                     MSG_NOSIGNAL);
+            if (w < 0) {
+                if (errno == EAGAIN || errno == EWOULDBLOCK) {
+                    epoll_event ev{};
+                    ev.events = EPOLLIN | EPOLLOUT;
+                    ev.data.u64 = c.id;
+                    epoll_ctl(ep_, EPOLL_CTL_MOD, c.fd, &ev);
+                    return;
+                }
+                close_conn(c);
+                return;
+            }
+            c.wbuf.erase(0, (size_t)w);
+        }
+        epoll_event ev{};
+        ev.events = EPOLLIN;
+        ev.data.u64 = c.id;
+        epoll_ctl(ep_, EPOLL_CTL_MOD, c.fd, &ev);
+        if (c.close_after_write) close_conn(c);
+    }
+
+    void close_conn(Conn& c) {
+        if (c.dead) return;
+        c.dead = true;
+        epoll_ctl(ep_, EPOLL_CTL_DEL, c.fd, nullptr);
+        ::close(c.fd);
+        std::lock_guard<std::mutex> lk(wmu_);
+        conn_index_.erase(c.id);
+    }
+
+    void gc_dead() {
+        for (auto it = conns_.begin(); it != conns_.end();) {
+            if (it->second.dead) it = conns_.erase(it);
+            else ++it;
+        }
+    }
+
+    int port_;
+    size_t max_req_;
+    int listen_fd_ = -1;
+    int ep_ = -1;
+    std::atomic<bool> running_{false};
+    std::thread loop_;
+    uint64_t next_id_ = 1;
+    std::unordered_map<uint64_t, Conn> conns_;
+    std::unordered_map<uint64_t, Conn*> conn_index_;
+    std::mutex mu_;    // guards ready_
+    std::mutex wmu_;   // guards conn_index_ + pending_writes_
+    std::deque<PendingReq> ready_;
+    std::vector<uint64_t> pending_writes_;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+    m.doc() = "gofr_amd native epoll ingress";
+    py::class_<EpollServer>(m, "EpollServer")
+        .def(py::init<int, size_t>(), py::arg("port"),
+             py::arg("max_req") = (size_t)1 << 20)
+        .def("start", &EpollServer::start)
+        .def("stop", &EpollServer::stop)
+        .def("port", &EpollServer::port)
+        .def("ready_count", &EpollServer::ready_count)
+        .def("harvest", &EpollServer::harvest)
+        .def("send", &EpollServer::send);
+}
