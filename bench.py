@@ -122,27 +122,24 @@ def run_single(eng, payloads, steps, warmup):
 
 
 def run_multi(eng, payloads, steps, warmup, rank, world):
-    """RCCL all-to-all sharding: each rank ingests its local batch,
-    scatters request slots to owner shards (uniform round-robin conn
-    ids -> equal splits), processes, gathers responses back, D2H."""
+    """RCCL all-to-all sharding (AllToAllSharder — the same class the
+    gloo multi-process tests cover): scatter request slots to owner
+    shards, process on the local engine, gather responses back."""
     import torch
+    from gofr_amd.engine.shard import AllToAllSharder
     import torch.distributed as dist
     t = torch
     dev = eng.device
-    # fixed-size slot exchange (equal splits across ranks)
     reqs, lens = make_batch(payloads, eng.slot)
     n = len(lens)
     assert n % world == 0
-    slot, rslot = eng.slot, eng.rslot
+    sharder = AllToAllSharder(eng, world)
+    sharder.alloc(n)
     p_in = t.from_numpy(reqs).pin_memory()
     p_len = t.from_numpy(lens.astype(np.int32)).pin_memory()
-    d_in = t.empty(n * slot, dtype=t.uint8, device=dev)
-    d_sh = t.empty(n * slot, dtype=t.uint8, device=dev)  # after exchange
+    d_in = t.empty(n * eng.slot, dtype=t.uint8, device=dev)
     d_len_in = t.empty(n, dtype=t.int32, device=dev)
-    d_len_sh = t.empty(n, dtype=t.int32, device=dev)
-    d_off = (t.arange(n, dtype=t.int64, device=dev) * slot)
-    d_resp_sh = t.empty(n * rslot, dtype=t.uint8, device=dev)
-    p_resp = t.empty(n * rslot, dtype=t.uint8).pin_memory()
+    p_resp = t.empty(n * eng.rslot, dtype=t.uint8).pin_memory()
     p_rlen = t.empty(n, dtype=t.int32).pin_memory()
 
     times = []
@@ -156,17 +153,9 @@ def run_multi(eng, payloads, steps, warmup, rank, world):
         # ingress: H2D staging of this shard's accepted connections
         d_in.copy_(p_in, non_blocking=True)
         d_len_in.copy_(p_len, non_blocking=True)
-        # re-balance: all-to-all over xGMI (7 p2p links used concurrently;
-        # SURVEY.md §2.3 — rings are per-link-bound, all-to-all is the
-        # right shape for request scatter)
-        dist.all_to_all_single(d_sh, d_in)
-        dist.all_to_all_single(d_len_sh, d_len_in)
-        d_resp, d_rlen = eng.process_device(d_sh, d_off, d_len_sh, n)
-        # response gather: return each response slab to its ingress rank
-        dist.all_to_all_single(d_resp_sh, d_resp[:n * rslot])
-        dist.all_to_all_single(d_len_in, d_rlen[:n])  # reuse buffer
-        p_resp.copy_(d_resp_sh, non_blocking=True)
-        p_rlen.copy_(d_len_in, non_blocking=True)
+        resp_sh, rlen_sh = sharder.step(d_in, d_len_in)
+        p_resp.copy_(resp_sh, non_blocking=True)
+        p_rlen.copy_(rlen_sh, non_blocking=True)
         torch.cuda.synchronize(dev)
         times.append(time.perf_counter() - t0)
         if it == 0:
@@ -186,6 +175,10 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=32768)
     ap.add_argument("--payload", type=int, default=1024)
+    ap.add_argument("--routes", type=int, default=4,
+                    help="route-table size (config 4: 64)")
+    ap.add_argument("--middleware", default="",
+                    help="comma list: auth,gzip (config 4)")
     args = ap.parse_args()
 
     import torch
@@ -199,13 +192,29 @@ def main():
         batch = min(batch, 256)  # CPU mirror sanity mode
 
     app = build_app()
+    for i in range(4, args.routes):
+        app.GET(f"/r{i}/{{id}}", handlers.static_json({"route": i}))
+    mw = [m for m in args.middleware.split(",") if m]
+    if "auth" in mw:
+        app.enable_auth(b"bench-secret")
+    if "gzip" in mw:
+        app.enable_gzip(min_size=256)
     device = f"cuda:{local_rank}" if have_gpu else "cpu"
     if have_gpu:
         torch.cuda.set_device(local_rank)
     eng = BatchEngine(app, device=device, slot=2048, max_batch=batch,
                       pipeline=(3 if (have_gpu and world == 1) else 1))
 
+    extra = b""
+    if "auth" in mw:
+        from gofr_amd.http.middleware import hmac_token
+        tok = hmac_token(b"bench-secret", "POST", "/echo")
+        extra += b"Authorization: HMAC " + tok.encode() + b"\r\n"
+    if "gzip" in mw:
+        extra += b"Accept-Encoding: gzip\r\n"
     raw = make_echo_request(args.payload)
+    if extra:
+        raw = raw.replace(b"\r\n\r\n", b"\r\n" + extra + b"\r\n", 1)
     payloads = [raw] * batch
 
     if world > 1:
@@ -249,6 +258,8 @@ def main():
                 "global_batch": batch * max(world, 1),
                 "seq_len": args.payload,
                 "parallelism": (f"alltoall{world}" if world > 1 else "single"),
+                "routes": args.routes,
+                "middleware": args.middleware or "none",
                 "p99_step_ms": round(p99_ms, 3),
                 "engine": "gpu" if have_gpu else "cpu-mirror",
             },
