@@ -417,6 +417,60 @@ __device__ __forceinline__ int itoa10(uint8_t* dst, int v) {
 #define GZ_MIN_MATCH 3
 #define GZ_MAX_DIST 2048  /* window = slot size; responses are < 4 KB */
 
+// ---- wave-parallel CRC32 (combine method) ----------------------------------
+// Each lane CRCs a contiguous chunk (init 0), advances its remainder by the
+// suffix length via GF(2) multiply mod the reflected poly, and the wave
+// XOR-reduces. Identical to zlib's crc32_combine math; constants are
+// x^(8*2^j) mod P (reflected domain, x^0 = 0x80000000).
+__constant__ uint32_t XPOW8[13] = {
+    0x00800000u, 0x00008000u, 0xedb88320u, 0xb1e6b092u, 0xa06a2517u,
+    0xed627daeu, 0x88d14467u, 0xd7bbfe6au, 0xec447f11u, 0x8e7ea170u,
+    0x6427800eu, 0x4d47bae0u, 0x09fe548fu};
+
+__device__ __forceinline__ uint32_t gf2_multmodp(uint32_t a, uint32_t b) {
+    uint32_t m = 1u << 31, p = 0;
+    while (true) {
+        if (a & m) {
+            p ^= b;
+            if ((a & (m - 1)) == 0) break;
+        }
+        m >>= 1;
+        b = (b & 1) ? (b >> 1) ^ 0xEDB88320u : (b >> 1);
+    }
+    return p;
+}
+
+__device__ __forceinline__ uint32_t xnmodp8(int nbytes) {
+    uint32_t r = 0x80000000u;  // x^0
+    int j = 0;
+    while (nbytes) {
+        if (nbytes & 1) r = gf2_multmodp(XPOW8[j], r);
+        nbytes >>= 1;
+        ++j;
+    }
+    return r;
+}
+
+// standard CRC32 (init 0xFFFFFFFF, final xor) of src[0..len), whole wave
+__device__ uint32_t crc32_wave(const uint8_t* src, int len,
+                               const uint32_t* crc_tab, int lane) {
+    const int C = (len + WAVE - 1) / WAVE;
+    const int s = lane * C;
+    int e = s + C;
+    if (e > len) e = len;
+    uint32_t contrib = 0;
+    if (s < len) {
+        uint32_t r = 0;
+        for (int i = s; i < e; ++i)
+            r = (r >> 8) ^ crc_tab[(r ^ src[i]) & 0xFF];
+        contrib = gf2_multmodp(xnmodp8(len - e), r);
+    }
+    if (lane == 0) contrib ^= gf2_multmodp(xnmodp8(len), 0xFFFFFFFFu);
+    for (int off = 32; off; off >>= 1)
+        contrib ^= __shfl_xor(contrib, off);
+    return contrib ^ 0xFFFFFFFFu;
+}
+
 struct BitWriter {
     uint8_t* out;
     int bytepos;
@@ -507,11 +561,16 @@ __device__ __forceinline__ void dist_code(int dist, int* sym, int* ebits,
 }
 
 // whole-wave gzip of src[0..len) in LDS -> dst (LDS); returns total gzip
-// bytes (negative on overflow of cap). hash: per-wave LDS u16 table.
+// bytes (negative on overflow of cap). hash: per-wave LDS u32 table
+// (u32 so interior-of-match inserts can be one wave-parallel atomicMax —
+// the serial model inserts ascending positions, so last-write == max).
 // crc_tab: block-shared 256-entry LDS CRC32 table.
 __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
-                                 int cap, uint16_t* hash,
+                                 int cap, uint32_t* hash,
                                  const uint32_t* crc_tab, int lane) {
+    // CRC of the whole plain body up front, wave-parallel (the LZ77 loop
+    // below no longer touches it)
+    const uint32_t crc_final = crc32_wave(src, len, crc_tab, lane);
     // cooperative hash clear
     for (int i = lane; i < GZ_HASH_SIZE; i += WAVE) hash[i] = 0;
     // gzip header (lane 0)
@@ -526,7 +585,6 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
         bw_put(&w, 1, 2);   // BTYPE=01 static
     }
     int pos = 0;
-    uint32_t crc = 0xFFFFFFFFu;
     // whole-wave greedy LZ77 loop: lane 0 decides; the wave measures
     while (true) {
         const int p0 = __shfl(pos, 0);
@@ -540,7 +598,7 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
             const int stored = (int)hash[h] - 1;
             if (stored >= 0 && p0 - stored <= GZ_MAX_DIST && stored < p0)
                 cand = stored;
-            hash[h] = (uint16_t)(p0 + 1);
+            hash[h] = (uint32_t)(p0 + 1);
         }
         cand = __shfl(cand, 0);
         int mlen = 0;
@@ -559,6 +617,22 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
             }
             mlen = agree > 258 ? 258 : agree;
         }
+        const int mlen0 = __shfl(mlen, 0);
+        if (mlen0 >= GZ_MIN_MATCH) {
+            // wave-parallel hash inserts over the matched interior. The
+            // serial model writes ascending positions (last write wins),
+            // so atomicMax of q+1 yields the identical final table.
+            for (int i = 1 + lane; i < mlen0; i += WAVE) {
+                const int q = p0 + i;
+                if (q + GZ_MIN_MATCH <= len) {
+                    const uint32_t h2 = ((src[q] | (src[q+1] << 8) |
+                                          (src[q+2] << 16))
+                                         * 0x9E3779B1u)
+                                        >> (32 - GZ_HASH_BITS);
+                    atomicMax(&hash[h2], (uint32_t)(q + 1));
+                }
+            }
+        }
         if (lane == 0) {
             if (mlen >= GZ_MIN_MATCH) {
                 int sym, ebits, eval;
@@ -570,30 +644,18 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
                 dist_code(p0 - cand, &sym, &ebits, &eval);
                 bw_put(&w, bitrev((uint32_t)sym, 5), 5);
                 if (ebits) bw_put(&w, (uint32_t)eval, ebits);
-                // crc + hash inserts over the matched span
-                for (int i = 0; i < mlen; ++i) {
-                    const uint8_t b = src[p0 + i];
-                    crc = (crc >> 8) ^ crc_tab[(crc ^ b) & 0xFF];
-                    if (i > 0 && p0 + i + GZ_MIN_MATCH <= len) {
-                        const int q = p0 + i;
-                        const uint32_t h2 = ((src[q] | (src[q+1] << 8) |
-                                              (src[q+2] << 16))
-                                             * 0x9E3779B1u)
-                                            >> (32 - GZ_HASH_BITS);
-                        hash[h2] = (uint16_t)(q + 1);
-                    }
-                }
                 pos = p0 + mlen;
             } else {
                 const uint8_t b = src[p0];
                 uint32_t code; int clen;
                 fixed_lit_code(b, &code, &clen);
                 bw_put(&w, code, clen);
-                crc = (crc >> 8) ^ crc_tab[(crc ^ b) & 0xFF];
                 pos = p0 + 1;
             }
             if (w.bytepos > cap - 24) pos = len + 1;  // overflow guard
         }
+        // the next iteration's lane-0 lookup must see this round's atomics
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         pos = __shfl(pos, 0);
         if (pos > len) return -1;
     }
@@ -603,7 +665,7 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
         bw_put(&w, code, clen);
         bw_flush(&w);
         uint8_t* p = dst + 10 + w.bytepos;
-        const uint32_t c = crc ^ 0xFFFFFFFFu;
+        const uint32_t c = crc_final;
         p[0]=(uint8_t)c; p[1]=(uint8_t)(c>>8);
         p[2]=(uint8_t)(c>>16); p[3]=(uint8_t)(c>>24);
         p[4]=(uint8_t)len; p[5]=(uint8_t)(len>>8);
@@ -636,7 +698,7 @@ __device__ __forceinline__ void respond_impl(
         const uint8_t* __restrict__ host_blob,
         const int32_t* __restrict__ host_tab,
         uint64_t seed, int auth_env_off, int auth_env_len,
-        int gzip_min, uint8_t* obuf, uint8_t* plainbuf, uint16_t* hash,
+        int gzip_min, uint8_t* obuf, uint8_t* plainbuf, uint32_t* hash,
         const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
@@ -913,12 +975,12 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
              int auth_env_off, int auth_env_len, int gzip_min) {
     // single __shared__ block (cdna guide §5 trap 4a)
     __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
-                           WAVES_PER_BLOCK * GZ_HASH_SIZE * 2 + 256 * 4];
+                           WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4];
     uint8_t* obuf_all = lds;
     uint8_t* plain_all = lds + WAVES_PER_BLOCK * MAX_SLOT;
-    uint16_t* hash_all = (uint16_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2);
+    uint32_t* hash_all = (uint32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2);
     uint32_t* crc_tab = (uint32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2 +
-                                    WAVES_PER_BLOCK * GZ_HASH_SIZE * 2);
+                                    WAVES_PER_BLOCK * GZ_HASH_SIZE * 4);
     // build the CRC32 table cooperatively BEFORE any thread can exit
     for (int i = threadIdx.x; i < 256; i += BLOCK_THREADS) {
         uint32_t c = (uint32_t)i;
