@@ -121,6 +121,152 @@ def run_single(eng, payloads, steps, warmup):
     return elapsed, lat
 
 
+def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
+    """BASELINE config 5 (single-GPU shard): mixed HTTP+gRPC batches with
+    device-resident conn state. Per lane-step: hipGraph replay of the
+    HTTP pipeline, then (same stream) H2D of the gRPC sub-batch,
+    k_varint_spans + k_grpc_echo, conn-state scatter update
+    (ConnStateTable.record_batch), D2H of the gRPC frames. Every request
+    is attributed to one of `conns` open connections."""
+    import torch
+    from gofr_amd.grpc.codec import HELLO_REQUEST, encode_message
+    from gofr_amd import ops
+    from gofr_amd.engine import pack_batch as _pack
+    from gofr_amd.engine.connstate import (ConnStateTable, PROTO_GRPC,
+                                           PROTO_HTTP)
+    t = torch
+    dev = eng.device
+    n_http = len(payloads)
+    # conn table: config-5 sizing (100k conns ≈ 0.6% of 288 GB HBM)
+    tab = ConnStateTable(conns, device=dev)
+    http_ids = tab.open(min(conns * 3 // 4, conns - 1), PROTO_HTTP)
+    grpc_ids = tab.open(conns - len(http_ids), PROTO_GRPC)
+
+    # gRPC sub-batch (HelloRequest frames, varied name lengths)
+    import random
+    rng = random.Random(5)
+    gpay = [encode_message({"name": "client-%d-%s" %
+                            (i, "x" * rng.randrange(0, 40))}, HELLO_REQUEST)
+            for i in range(n_grpc)]
+    gbuf, goffs, glens = _pack(gpay)
+    GR = 256  # gRPC response slot
+
+    class GLane:
+        def __init__(self):
+            self.p_gbuf = t.from_numpy(gbuf).pin_memory()
+            self.d_gbuf = t.empty(len(gbuf), dtype=t.uint8, device=dev)
+            self.d_goff = t.from_numpy(goffs).to(dev)
+            self.d_glen = t.from_numpy(glens).to(dev)
+            self.d_spans = t.zeros(n_grpc * ops.MAX_PB_FIELDS * 4,
+                                   dtype=t.int32, device=dev)
+            self.d_span_n = t.zeros(n_grpc, dtype=t.int32, device=dev)
+            self.d_gout = t.empty(n_grpc * GR, dtype=t.uint8, device=dev)
+            self.d_gout_len = t.empty(n_grpc, dtype=t.int32, device=dev)
+            self.p_gout = t.empty(n_grpc * GR, dtype=t.uint8).pin_memory()
+            self.p_gout_len = t.empty(n_grpc, dtype=t.int32).pin_memory()
+            # conn attribution for this lane's batches (device-resident)
+            self.d_http_ids = t.from_numpy(
+                http_ids[np.arange(n_http) % len(http_ids)]).to(dev)
+            self.d_grpc_ids = t.from_numpy(
+                grpc_ids[np.arange(n_grpc) % len(grpc_ids)]).to(dev)
+            self.d_http_bin = t.empty(n_http, dtype=t.int64, device=dev)
+            self.d_grpc_bin = t.from_numpy(
+                glens.astype(np.int64)).to(dev)
+
+    # HTTP staging + graph capture per lane (same as run_single)
+    buf, offs, lens = pack_batch(payloads)
+    nbytes = int(offs[-1] + lens[-1])
+    for ln in eng.lanes:
+        ln.p_reqs[:nbytes] = t.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n_http] = t.from_numpy(offs)
+        ln.p_req_len[:n_http] = t.from_numpy(lens)
+    P = len(eng.lanes)
+    glanes = [GLane() for _ in range(P)]
+    for li in range(P):
+        eng.capture_graph(n_http, nbytes, li)
+        gl = glanes[li]
+        gl.d_http_bin.copy_(eng.lanes[li].d_req_len[:n_http].to(t.int64))
+
+    def submit_mixed(i, li):
+        ln = eng.lanes[li]
+        gl = glanes[li]
+        eng.submit(n_http, nbytes, li)
+        with t.cuda.stream(ln.stream):
+            cs = ln.stream.cuda_stream
+            gl.d_gbuf.copy_(gl.p_gbuf, non_blocking=True)
+            eng.hip.varint_spans(cs, gl.d_gbuf, gl.d_goff, gl.d_glen,
+                                 gl.d_spans, gl.d_span_n, n_grpc)
+            eng.hip.grpc_echo(cs, gl.d_gbuf, gl.d_spans, gl.d_span_n,
+                              gl.d_gout, gl.d_gout_len, n_grpc, GR)
+            # conn-state accounting for both protocol halves
+            tab.record_batch(gl.d_http_ids, gl.d_http_bin,
+                             ln.d_resp_len[:n_http].to(t.int64), i)
+            tab.record_batch(gl.d_grpc_ids, gl.d_grpc_bin,
+                             gl.d_gout_len.to(t.int64), i)
+            gl.p_gout.copy_(gl.d_gout, non_blocking=True)
+            gl.p_gout_len.copy_(gl.d_gout_len, non_blocking=True)
+            ln.event.record(ln.stream)
+
+    # warmup serial
+    for w in range(max(1, warmup)):
+        submit_mixed(w, 0)
+        out_t, roff_t, rlen_t = eng.complete(0)
+    first = out_t[:int(rlen_t[0])].numpy().tobytes()
+    assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+    gfirst = glanes[0].p_gout[:int(glanes[0].p_gout_len[0])].numpy()
+    assert gfirst.tobytes().startswith(b"\x00"), gfirst[:16]
+    t.cuda.synchronize(dev)
+
+    submit_at = [0.0] * steps
+    lat = []
+    t_start = time.perf_counter()
+    for i in range(steps):
+        lane = i % P
+        if i >= P:
+            eng.complete(lane)
+            lat.append(time.perf_counter() - submit_at[i - P])
+        submit_at[i] = time.perf_counter()
+        submit_mixed(i, lane)
+    for i in range(max(0, steps - P), steps):
+        eng.complete(i % P)
+        lat.append(time.perf_counter() - submit_at[i])
+    t.cuda.synchronize(dev)
+    elapsed = time.perf_counter() - t_start
+    # spot-check conn accounting ran
+    s = tab.stats(http_ids[:1])
+    assert s[0][2] > 0, "conn-state REQS not updated"
+    return elapsed, lat, tab
+
+
+def run_config5_cpu(eng, payloads, steps, warmup, n_grpc, conns):
+    """CPU-mirror sanity mode of config 5 (no GPU on this box)."""
+    from gofr_amd import ops
+    from gofr_amd.engine import pack_batch as _pack
+    from gofr_amd.engine.connstate import ConnStateTable, PROTO_HTTP
+    from gofr_amd.grpc.codec import HELLO_REQUEST, encode_message
+    tab = ConnStateTable(conns)
+    ids = tab.open(min(256, conns))
+    gpay = [encode_message({"name": f"c{i}"}, HELLO_REQUEST)
+            for i in range(n_grpc)]
+    gbuf, goffs, glens = _pack(gpay)
+    buf, offs, lens = pack_batch(payloads)
+    times = []
+    for it in range(warmup + steps):
+        if it == warmup:
+            t_start = time.perf_counter()
+        t0 = time.perf_counter()
+        out, roffs, rlens = eng.process_packed(buf, offs, lens)
+        spans, span_n = ops.cpu_varint_spans(gbuf, goffs, glens)
+        gout, gout_len = ops.cpu_grpc_echo(gbuf, spans, span_n, 256)
+        nb = np.zeros(len(ids), np.int64)
+        tab.record_batch(ids, nb, nb, it)
+        times.append(time.perf_counter() - t0)
+        if it == 0:
+            assert out[:17].tobytes() == b"HTTP/1.1 200 OK\r\n"
+            assert int(gout_len[0]) > 0
+    return time.perf_counter() - t_start, times[warmup:]
+
+
 def run_multi(eng, payloads, steps, warmup, rank, world):
     """RCCL all-to-all sharding (AllToAllSharder — the same class the
     gloo multi-process tests cover): scatter request slots to owner
@@ -179,6 +325,12 @@ def main():
                     help="route-table size (config 4: 64)")
     ap.add_argument("--middleware", default="",
                     help="comma list: auth,gzip (config 4)")
+    ap.add_argument("--grpc-frac", type=float, default=0.0,
+                    help="fraction of the batch that is gRPC unary echo "
+                         "(config 5 mixed mode)")
+    ap.add_argument("--conns", type=int, default=100_000,
+                    help="open connections in the device conn-state "
+                         "table (config 5)")
     args = ap.parse_args()
 
     import torch
@@ -216,6 +368,47 @@ def main():
     if extra:
         raw = raw.replace(b"\r\n\r\n", b"\r\n" + extra + b"\r\n", 1)
     payloads = [raw] * batch
+
+    n_grpc = int(batch * args.grpc_frac)
+    if n_grpc and world == 1:
+        if have_gpu:
+            elapsed, times, _tab = run_config5(
+                eng, payloads[:batch - n_grpc], args.steps, args.warmup,
+                n_grpc, args.conns)
+        else:
+            elapsed, times = run_config5_cpu(
+                eng, payloads[:batch - n_grpc], args.steps, args.warmup,
+                n_grpc, min(args.conns, 4096))
+        p99_ms = float(np.percentile(times, 99) * 1000)
+        out = {
+            "metric": "mixed HTTP+gRPC requests/sec (whole node)",
+            "value": round(batch * args.steps / elapsed, 1),
+            "unit": "req/s",
+            "n_gpus": 1 if have_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8-exact",
+            "data": "synthetic",
+            "config": {
+                "model": "request-batch engine, mixed HTTP+gRPC, "
+                         "conn-state in HBM",
+                "global_batch": batch,
+                "seq_len": args.payload,
+                "parallelism": "single",
+                "grpc_frac": args.grpc_frac,
+                "conns": args.conns,
+                "conn_state_mb": round(
+                    args.conns * 16432 / 1e6, 1),
+                "p99_step_ms": round(p99_ms, 3),
+                "engine": "gpu" if have_gpu else "cpu-mirror",
+            },
+        }
+        print(json.dumps(out))
+        return
 
     if world > 1:
         import torch.distributed as dist

@@ -1243,6 +1243,62 @@ k_varint_spans(const uint8_t* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
+// k_grpc_echo — batched unary SayHello responder (BASELINE config 5's
+// gRPC half). One message per lane: read the name span from the
+// k_varint_spans table, emit the gRPC length-prefixed HelloResponse
+// frame {message: "Hello <name>!"} ("World" when the name is empty —
+// examples/grpc-server semantics; reference examples/grpc-server/grpc/
+// server.go:12-21). Host golden model: gofr_amd/ops cpu_grpc_echo.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_grpc_echo(const uint8_t* __restrict__ buf,
+            const int32_t* __restrict__ spans,   // [n, MAX_PB_FIELDS, 4]
+            const int32_t* __restrict__ span_n,  // [n]
+            uint8_t* __restrict__ out,           // [n, rslot]
+            int32_t* __restrict__ out_len,
+            int n, int rslot) {
+    const int m = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+    if (m >= n) return;
+    const int32_t* row = spans + (size_t)m * MAX_PB_FIELDS * 4;
+    const int nf = span_n[m];
+    int name_off = 0, name_len = 0;
+    for (int i = 0; i < nf; ++i) {
+        if (row[i*4+0] == 1 && row[i*4+1] == 2) {
+            name_off = row[i*4+2];
+            name_len = row[i*4+3];
+        }
+    }
+    const uint8_t* nm = buf + name_off;
+    static const char WORLD[5] = {'W', 'o', 'r', 'l', 'd'};
+    if (name_len == 0) { nm = (const uint8_t*)WORLD; name_len = 5; }
+    // HelloResponse { 1: "Hello <name>!" }: tag 0x0A + varint len + payload
+    const int payload_len = 6 + name_len + 1;
+    const int vlen = payload_len < 128 ? 1 : 2;
+    const int msg_len = 1 + vlen + payload_len;
+    if (nf < 0 || 5 + msg_len > rslot) {
+        out_len[m] = -1;  // malformed/oversized: host error path
+        return;
+    }
+    uint8_t* o = out + (size_t)m * rslot;
+    o[0] = 0;  // uncompressed
+    o[1] = (uint8_t)(msg_len >> 24); o[2] = (uint8_t)(msg_len >> 16);
+    o[3] = (uint8_t)(msg_len >> 8);  o[4] = (uint8_t)msg_len;
+    int p = 5;
+    o[p++] = 0x0A;
+    if (vlen == 1) {
+        o[p++] = (uint8_t)payload_len;
+    } else {
+        o[p++] = (uint8_t)(payload_len | 0x80);
+        o[p++] = (uint8_t)(payload_len >> 7);
+    }
+    o[p++]='H'; o[p++]='e'; o[p++]='l'; o[p++]='l'; o[p++]='o'; o[p++]=' ';
+    for (int i = 0; i < name_len; ++i) o[p + i] = nm[i];
+    p += name_len;
+    o[p++] = '!';
+    out_len[m] = p;
+}
+
+// ---------------------------------------------------------------------------
 // k_compact — gather response slots into a contiguous 16B-aligned stream
 // (halves the D2H bytes vs slot-strided responses; offsets are the
 // exclusive cumsum of round16(resp_len), computed host-side via torch)
@@ -1356,6 +1412,19 @@ int gofr_launch_varint_spans(
                        (const uint8_t*)buf, (const int64_t*)msg_off,
                        (const int32_t*)msg_len,
                        (int32_t*)out, (int32_t*)out_n, n);
+    return (int)hipGetLastError();
+}
+
+int gofr_launch_grpc_echo(
+        void* stream,
+        const void* buf, const void* spans, const void* span_n,
+        void* out, void* out_len, int n, int rslot) {
+    const int blocks = (n + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    hipLaunchKernelGGL(k_grpc_echo, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       (hipStream_t)stream,
+                       (const uint8_t*)buf, (const int32_t*)spans,
+                       (const int32_t*)span_n,
+                       (uint8_t*)out, (int32_t*)out_len, n, rslot);
     return (int)hipGetLastError();
 }
 

@@ -101,6 +101,9 @@ class HipOps:
         self.lib.gofr_launch_varint_spans.restype = ctypes.c_int
         self.lib.gofr_launch_varint_spans.argtypes = \
             [ctypes.c_void_p] * 6 + [ctypes.c_int]
+        self.lib.gofr_launch_grpc_echo.restype = ctypes.c_int
+        self.lib.gofr_launch_grpc_echo.argtypes = \
+            [ctypes.c_void_p] * 6 + [ctypes.c_int, ctypes.c_int]
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,
@@ -171,6 +174,18 @@ class HipOps:
         if rc != 0:
             raise RuntimeError(
                 f"k_varint_spans launch failed: hipError {rc}")
+
+    def grpc_echo(self, stream, buf_t, spans_t, span_n_t, out_t, out_len_t,
+                  n, rslot):
+        rc = self.lib.gofr_launch_grpc_echo(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(buf_t.data_ptr()),
+            ctypes.c_void_p(spans_t.data_ptr()),
+            ctypes.c_void_p(span_n_t.data_ptr()),
+            ctypes.c_void_p(out_t.data_ptr()),
+            ctypes.c_void_p(out_len_t.data_ptr()), n, rslot)
+        if rc != 0:
+            raise RuntimeError(f"k_grpc_echo launch failed: hipError {rc}")
 
     def compact(self, stream, resp_slots_t, resp_len_t, resp_off_t, out_t,
                 n, rslot):
@@ -731,3 +746,43 @@ def cpu_varint_spans(buf: np.ndarray, msg_off: np.ndarray,
             nf += 1
         out_n[m] = -1 if bad else nf
     return out, out_n
+
+
+def cpu_grpc_echo(buf: np.ndarray, spans: np.ndarray, span_n: np.ndarray,
+                  rslot: int):
+    """Mirror of k_grpc_echo: gRPC length-prefixed HelloResponse frames
+    {message: "Hello <name>!"} from the varint span tables ("World" when
+    the name field is empty — examples/grpc-server semantics)."""
+    n = len(span_n)
+    out = np.zeros(n * rslot, np.uint8)
+    out_len = np.zeros(n, np.int32)
+    data = buf.tobytes()
+    for m in range(n):
+        nf = int(span_n[m])
+        name = b""
+        for i in range(max(0, nf)):
+            if spans[m, i, 0] == 1 and spans[m, i, 1] == 2:
+                off, ln = int(spans[m, i, 2]), int(spans[m, i, 3])
+                name = data[off:off + ln]
+        if not name:
+            name = b"World"
+        payload = b"Hello " + name + b"!"
+        vlen = 1 if len(payload) < 128 else 2
+        msg_len = 1 + vlen + len(payload)
+        if nf < 0 or 5 + msg_len > rslot:
+            out_len[m] = -1
+            continue
+        frame = bytearray()
+        frame.append(0)
+        frame += msg_len.to_bytes(4, "big")
+        frame.append(0x0A)
+        if vlen == 1:
+            frame.append(len(payload))
+        else:
+            frame.append((len(payload) & 0x7F) | 0x80)
+            frame.append(len(payload) >> 7)
+        frame += payload
+        out[m * rslot:m * rslot + len(frame)] = np.frombuffer(
+            bytes(frame), np.uint8)
+        out_len[m] = len(frame)
+    return out, out_len

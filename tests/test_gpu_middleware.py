@@ -132,3 +132,54 @@ def test_k_respond_gz_matches_mirror_bytes():
             n_gz += 1
             _gz.decompress(body)  # must be a valid gzip stream
     assert n_gz > 100
+
+
+def test_k_grpc_echo_matches_mirror():
+    from gofr_amd.grpc.codec import HELLO_REQUEST, encode_message
+
+    hip = ops.HipOps()
+    dev = torch.device("cuda:0")
+    names = ["alice", "", "x" * 130, "bob"] + [f"c{i}" for i in range(252)]
+    payloads = [encode_message({"name": n}, HELLO_REQUEST) for n in names]
+    buf, offs, lens = pack_batch(payloads)
+    n = len(lens)
+    rslot = 512
+    spans_c, span_n_c = ops.cpu_varint_spans(buf, offs, lens)
+    out_c, out_len_c = ops.cpu_grpc_echo(buf, spans_c, span_n_c, rslot)
+
+    d_buf = torch.from_numpy(buf).to(dev)
+    d_off = torch.from_numpy(offs).to(dev)
+    d_len = torch.from_numpy(lens).to(dev)
+    d_spans = torch.zeros(n * ops.MAX_PB_FIELDS * 4, dtype=torch.int32,
+                          device=dev)
+    d_span_n = torch.zeros(n, dtype=torch.int32, device=dev)
+    d_out = torch.zeros(n * rslot, dtype=torch.uint8, device=dev)
+    d_out_len = torch.zeros(n, dtype=torch.int32, device=dev)
+    stream = torch.cuda.current_stream().cuda_stream
+    hip.varint_spans(stream, d_buf, d_off, d_len, d_spans, d_span_n, n)
+    hip.grpc_echo(stream, d_buf, d_spans, d_span_n, d_out, d_out_len, n,
+                  rslot)
+    torch.cuda.synchronize()
+    assert (d_out_len.cpu().numpy() == out_len_c).all()
+    g = d_out.cpu().numpy()
+    for i in range(n):
+        ln = int(out_len_c[i])
+        assert g[i*rslot:i*rslot+ln].tobytes() == \
+            out_c[i*rslot:i*rslot+ln].tobytes(), f"msg {i}"
+
+
+def test_connstate_device_matches_cpu():
+    from gofr_amd.engine.connstate import ConnStateTable
+
+    gpu = ConnStateTable(capacity=1024, device="cuda")
+    cpu = ConnStateTable(capacity=1024)
+    ids = gpu.open(16)
+    ids_c = cpu.open(16)
+    assert (ids == ids_c).all()
+    bin_ = np.arange(16, dtype=np.int64) * 100
+    bout = np.arange(16, dtype=np.int64) * 7
+    d_ids = torch.from_numpy(ids).to("cuda")
+    gpu.record_batch(d_ids, torch.from_numpy(bin_).to("cuda"),
+                     torch.from_numpy(bout).to("cuda"), batch_no=3)
+    cpu.record_batch(ids_c, bin_, bout, batch_no=3)
+    assert (gpu.stats(ids) == cpu.stats(ids_c)).all()

@@ -169,3 +169,43 @@ def test_grpc_rpc_logged():
     time.sleep(0.05)
     assert "/hello.Hello/SayHello" in app.container.logger.stdout
     server.stop()
+
+
+def test_cpu_grpc_echo_mirror_matches_codec():
+    """cpu_grpc_echo (golden model of k_grpc_echo) emits exactly the
+    frame GRPCServer._dispatch would send for SayHello."""
+    import struct
+
+    import numpy as np
+
+    from gofr_amd import ops
+    from gofr_amd.engine import pack_batch
+
+    names = ["alice", "", "x" * 130, "bob-123"]
+    payloads = [encode_message({"name": n}, HELLO_REQUEST) for n in names]
+    buf, offs, lens = pack_batch(payloads)
+    spans, span_n = ops.cpu_varint_spans(buf, offs, lens)
+    rslot = 512
+    out, out_len = ops.cpu_grpc_echo(buf, spans, span_n, rslot)
+    for i, n in enumerate(names):
+        expect_msg = encode_message(
+            {"message": f"Hello {n or 'World'}!"}, HELLO_RESPONSE)
+        expect = bytes([0]) + struct.pack(">I", len(expect_msg)) + expect_msg
+        got = out[i * rslot:i * rslot + int(out_len[i])].tobytes()
+        assert got == expect, f"msg {i}: {got!r} != {expect!r}"
+        # and the frame decodes back to the expected dict
+        decoded = decode_message(got[5:], HELLO_RESPONSE)
+        assert decoded["message"] == f"Hello {n or 'World'}!"
+
+
+def test_cpu_grpc_echo_malformed():
+    import numpy as np
+
+    from gofr_amd import ops
+
+    buf = np.frombuffer(b"\xff\xff\xff\xff", np.uint8)
+    spans, span_n = ops.cpu_varint_spans(
+        buf, np.array([0], np.int64), np.array([4], np.int32))
+    assert span_n[0] == -1
+    out, out_len = ops.cpu_grpc_echo(buf, spans, span_n, 512)
+    assert out_len[0] == -1
