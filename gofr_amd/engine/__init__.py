@@ -273,9 +273,13 @@ class BatchEngine:
         if ln.graph is not None and ln.graph_key == (n, nbytes):
             # hipGraph replay: one launch for the whole batch pipeline
             # (the captured H2D copies re-read the pinned rings, so fresh
-            # request bytes and the fresh seed flow through the replay)
-            ln.graph.replay()
-            ln.event.record(ln.stream)
+            # request bytes and the fresh seed flow through the replay).
+            # Replay on the LANE stream: torch replays on the current
+            # stream, and the legacy default stream would both serialize
+            # the lanes against each other and escape ln.event.
+            with t.cuda.stream(ln.stream):
+                ln.graph.replay()
+                ln.event.record(ln.stream)
             return
         with t.cuda.stream(ln.stream):
             self._submit_body(ln, n, nbytes)
