@@ -172,6 +172,7 @@ def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
             self.d_http_bin = t.empty(n_http, dtype=t.int64, device=dev)
             self.d_grpc_bin = t.from_numpy(
                 glens.astype(np.int64)).to(dev)
+            self.ev = t.cuda.Event()
 
     # HTTP staging + graph capture per lane (same as run_single)
     buf, offs, lens = pack_batch(payloads)
@@ -198,19 +199,26 @@ def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
                                  gl.d_spans, gl.d_span_n, n_grpc)
             eng.hip.grpc_echo(cs, gl.d_gbuf, gl.d_spans, gl.d_span_n,
                               gl.d_gout, gl.d_gout_len, n_grpc, GR)
-            # conn-state accounting for both protocol halves
+            # conn-state accounting for both protocol halves (the HTTP
+            # half reads d_resp_len, written by the lane's kernel stage)
+            ln.stream.wait_event(ln.e_k)
             tab.record_batch(gl.d_http_ids, gl.d_http_bin,
                              ln.d_resp_len[:n_http].to(t.int64), i)
             tab.record_batch(gl.d_grpc_ids, gl.d_grpc_bin,
                              gl.d_gout_len.to(t.int64), i)
             gl.p_gout.copy_(gl.d_gout, non_blocking=True)
             gl.p_gout_len.copy_(gl.d_gout_len, non_blocking=True)
-            ln.event.record(ln.stream)
+            gl.ev.record(ln.stream)
+
+    def complete_mixed(li):
+        out = eng.complete(li)
+        glanes[li].ev.synchronize()
+        return out
 
     # warmup serial
     for w in range(max(1, warmup)):
         submit_mixed(w, 0)
-        out_t, roff_t, rlen_t = eng.complete(0)
+        out_t, roff_t, rlen_t = complete_mixed(0)
     first = out_t[:int(rlen_t[0])].numpy().tobytes()
     assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
     gfirst = glanes[0].p_gout[:int(glanes[0].p_gout_len[0])].numpy()
@@ -223,12 +231,12 @@ def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
     for i in range(steps):
         lane = i % P
         if i >= P:
-            eng.complete(lane)
+            complete_mixed(lane)
             lat.append(time.perf_counter() - submit_at[i - P])
         submit_at[i] = time.perf_counter()
         submit_mixed(i, lane)
     for i in range(max(0, steps - P), steps):
-        eng.complete(i % P)
+        complete_mixed(i % P)
         lat.append(time.perf_counter() - submit_at[i])
     t.cuda.synchronize(dev)
     elapsed = time.perf_counter() - t_start

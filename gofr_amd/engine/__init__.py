@@ -103,11 +103,22 @@ def make_batch(payloads: list[bytes], slot: int):
 
 
 class _Lane:
-    """One pipeline stage: its own HIP stream + device/pinned buffer set."""
+    """One in-flight batch: a device/pinned buffer set + stage events.
+
+    Lanes share the engine's three stage streams (s_in H2D, s_k kernels,
+    s_out D2H) — measured on MI355X, eager SDMA copies on dedicated
+    per-direction streams run full duplex at ~96 GB/s aggregate while
+    graph-captured copies reach only ~85 and compute-driven host
+    dereference ~59 (benchmarks/overlap_probe.py) — so ingress of batch
+    i+1 overlaps egress of batch i-1 and the kernels of batch i.
+    """
 
     def __init__(self, t, dev, nb, max_bytes, rslot, host_blob_cap=4 << 20):
-        self.stream = t.cuda.Stream(device=dev)
+        self.stream = t.cuda.Stream(device=dev)  # legacy per-lane stream
         self.event = t.cuda.Event()
+        self.e_in = t.cuda.Event()
+        self.e_k = t.cuda.Event()
+        self.e_out = t.cuda.Event()
         self.d_reqs = t.empty(max_bytes, dtype=t.uint8, device=dev)
         self.d_req_off = t.empty(nb, dtype=t.int64, device=dev)
         self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
@@ -196,6 +207,11 @@ class BatchEngine:
                 np.frombuffer(secret, np.uint8).copy()).to(dev)
         else:
             self.d_secret = None
+        # stage streams shared by all lanes: one per direction so the
+        # host link runs full duplex, one for compute
+        self.s_in = t.cuda.Stream(device=dev)
+        self.s_k = t.cuda.Stream(device=dev)
+        self.s_out = t.cuda.Stream(device=dev)
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot) for _ in range(self.pipeline)]
 
@@ -262,35 +278,48 @@ class BatchEngine:
     # -- pipelined API --------------------------------------------------------
     def submit(self, n: int, nbytes: int, lane_idx: int = 0) -> None:
         """Enqueue one batch (already staged in lane.p_reqs[:nbytes] /
-        p_req_off / p_req_len) on the lane's stream: H2D -> parse ->
-        respond (optimistic) -> pad16 cumsum -> compact -> D2H of
-        lens/offs/total. Never blocks; complete() finishes it."""
+        p_req_off / p_req_len). With a captured kernel graph the batch
+        flows through the three stage streams — eager H2D on s_in,
+        kernel-graph replay on s_k, eager egress D2H on s_out — chained
+        by events, so concurrent lanes overlap ingress, compute and
+        egress (full-duplex host link). Never blocks; complete()
+        finishes it. A lane must be complete()d before it is resubmitted.
+        """
         t = self.torch
         ln = self.lanes[lane_idx]
         ln.n, ln.nbytes = n, nbytes
         seed = self._next_seed()
         ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
         if ln.graph is not None and ln.graph_key == (n, nbytes):
-            # hipGraph replay: one launch for the whole batch pipeline
-            # (the captured H2D copies re-read the pinned rings, so fresh
-            # request bytes and the fresh seed flow through the replay).
-            # Replay on the LANE stream: torch replays on the current
-            # stream, and the legacy default stream would both serialize
-            # the lanes against each other and escape ln.event.
-            with t.cuda.stream(ln.stream):
+            with t.cuda.stream(self.s_in):
+                self._ingress_body(ln, n, nbytes)
+                ln.e_in.record(self.s_in)
+            with t.cuda.stream(self.s_k):
+                self.s_k.wait_event(ln.e_in)
                 ln.graph.replay()
-                ln.event.record(ln.stream)
+                ln.e_k.record(self.s_k)
+            with t.cuda.stream(self.s_out):
+                self.s_out.wait_event(ln.e_k)
+                if ln.egress_budget:
+                    ln.p_out[:ln.egress_budget].copy_(
+                        ln.d_out[:ln.egress_budget], non_blocking=True)
+                ln.event.record(self.s_out)
             return
         with t.cuda.stream(ln.stream):
             self._submit_body(ln, n, nbytes)
             ln.event.record(ln.stream)
 
-    def _submit_body(self, ln, n, nbytes):
-        cs = ln.stream.cuda_stream
+    def _ingress_body(self, ln, n, nbytes):
+        """H2D staging copies (runs on the caller's current stream)."""
         ln.d_seed.copy_(ln.p_seed, non_blocking=True)
         ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
         ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
         ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
+
+    def _kernel_body(self, ln, n):
+        """parse -> auth -> respond -> compact + small D2H (current
+        stream; this is the section captured as the lane's hipGraph)."""
+        cs = self.torch.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
         self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
                              ln.d_fields, n, self.d_trie,
@@ -304,19 +333,23 @@ class BatchEngine:
         # fixup pass overwrites before the responses are released
         self._respond_compact(ln, n)
 
+    def _submit_body(self, ln, n, nbytes):
+        self._ingress_body(ln, n, nbytes)
+        self._kernel_body(ln, n)
+
     def capture_graph(self, n: int, nbytes: int, lane_idx: int = 0) -> bool:
-        """Capture the lane's batch pipeline into a hipGraph for (n,
-        nbytes)-shaped batches; submit() then replays it (one host call
-        instead of ~10 launches/copies per batch). Returns False when the
-        torch build lacks graph support."""
+        """Capture the lane's KERNEL section into a hipGraph for (n,
+        nbytes)-shaped batches; submit() then replays it on s_k between
+        the eager stage copies. Returns False when the torch build lacks
+        graph support."""
         t = self.torch
         ln = self.lanes[lane_idx]
         if not hasattr(t.cuda, "CUDAGraph"):
             return False
         # warmup pass (allocations settle) then capture
-        with t.cuda.stream(ln.stream):
+        with t.cuda.stream(self.s_k):
             self._submit_body(ln, n, nbytes)
-        ln.stream.synchronize()
+        self.s_k.synchronize()
         # size the in-pipeline egress copy from the warmup batch's actual
         # egress bytes (+ slack for batch-to-batch response variation);
         # complete() falls back to an explicit tail copy if a later batch
@@ -325,17 +358,17 @@ class BatchEngine:
         ln.egress_budget = min(len(ln.p_out),
                                max(4096, int(total * 1.25) + 4096))
         g = t.cuda.CUDAGraph()
-        with t.cuda.graph(g, stream=ln.stream):
-            self._submit_body(ln, n, nbytes)
+        with t.cuda.graph(g, stream=self.s_k):
+            self._kernel_body(ln, n)
         ln.graph = g
         ln.graph_key = (n, nbytes)
         return True
 
     def _respond_compact(self, ln, n):
         """respond + pad16 cumsum + compact + D2H of lens/offs/total on
-        the lane's stream (caller holds the stream context)."""
+        the caller's current stream."""
         t = self.torch
-        cs = ln.stream.cuda_stream
+        cs = t.cuda.current_stream(self.device).cuda_stream
         self.hip.respond(cs, ln.d_reqs, ln.d_req_off, ln.d_fields,
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
@@ -350,9 +383,6 @@ class BatchEngine:
         ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
         ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
         ln.p_total.copy_(csum[-1:], non_blocking=True)
-        if ln.egress_budget:
-            ln.p_out[:ln.egress_budget].copy_(ln.d_out[:ln.egress_budget],
-                                              non_blocking=True)
 
     def complete(self, lane_idx: int = 0):
         """Wait for the lane's in-flight batch; run the host fixup pass if
@@ -362,7 +392,9 @@ class BatchEngine:
         ln = self.lanes[lane_idx]
         n = ln.n
         ln.event.synchronize()
+        fixed_up = False
         if int(ln.p_host_needed[0]):
+            fixed_up = True
             # fixup: run Python handlers for HK_HOST rows, re-serialize
             with t.cuda.stream(ln.stream):
                 ln.p_fields[:n * ops.NF].copy_(ln.d_fields[:n * ops.NF],
@@ -383,9 +415,11 @@ class BatchEngine:
                 self._respond_compact(ln, n)
             ln.stream.synchronize()
         total = int(ln.p_total[0])
-        if not (ln.egress_budget and total <= ln.egress_budget):
-            # no in-pipeline egress copy (or the budget overflowed):
-            # explicit D2H of the compact stream
+        if fixed_up or not (ln.egress_budget
+                            and total <= ln.egress_budget):
+            # no in-pipeline egress copy (budget overflow / no graph /
+            # fixup rewrote responses): explicit D2H of the compact
+            # stream
             with t.cuda.stream(ln.stream):
                 ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
             ln.stream.synchronize()
