@@ -141,6 +141,7 @@ class _Lane:
         self.p_host_needed = t.zeros(1, dtype=t.int32).pin_memory()
         self.d_seed = t.zeros(1, dtype=t.int64, device=dev)
         self.p_seed = t.zeros(1, dtype=t.int64).pin_memory()
+        self.d_total = t.zeros(1, dtype=t.int32, device=dev)
         self.n = 0
         self.nbytes = 0
         self.graph = None
@@ -300,6 +301,7 @@ class BatchEngine:
                 ln.e_k.record(self.s_k)
             with t.cuda.stream(self.s_out):
                 self.s_out.wait_event(ln.e_k)
+                self._emit_host_tables(ln, n)
                 if ln.egress_budget:
                     ln.p_out[:ln.egress_budget].copy_(
                         ln.d_out[:ln.egress_budget], non_blocking=True)
@@ -316,22 +318,34 @@ class BatchEngine:
         ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
         ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
 
-    def _kernel_body(self, ln, n):
-        """parse -> auth -> respond -> compact + small D2H (current
-        stream; this is the section captured as the lane's hipGraph)."""
+    def _kernel_body(self, ln, n, emit_host=True):
+        """parse -> auth -> respond -> compact (current stream; this is
+        the section captured as the lane's hipGraph). With
+        emit_host=False no D2H copy is enqueued at all — the captured
+        graph stays pure compute, so no s_k memcpy node can queue behind
+        another lane's 40 MB egress on a shared SDMA engine; submit()
+        emits the small D2Hs on the egress stream instead."""
         cs = self.torch.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
         self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
                              ln.d_fields, n, self.d_trie,
                              self.d_handler_tab, self.program.n_routes,
                              ln.d_host_needed)
-        ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
+        if emit_host:
+            ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
         if self.d_secret is not None:
             self.hip.auth(cs, ln.d_reqs, ln.d_req_off, ln.d_fields, n,
                           self.d_secret, len(self.app.auth_secret))
         # optimistic respond: host rows render a 500 fallback that the
         # fixup pass overwrites before the responses are released
-        self._respond_compact(ln, n)
+        self._respond_compact(ln, n, emit_host=emit_host)
+
+    def _emit_host_tables(self, ln, n):
+        """Small D2Hs of the per-batch result tables (current stream)."""
+        ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
+        ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
+        ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
+        ln.p_total.copy_(ln.d_total, non_blocking=True)
 
     def _submit_body(self, ln, n, nbytes):
         self._ingress_body(ln, n, nbytes)
@@ -359,14 +373,14 @@ class BatchEngine:
                                max(4096, int(total * 1.25) + 4096))
         g = t.cuda.CUDAGraph()
         with t.cuda.graph(g, stream=self.s_k):
-            self._kernel_body(ln, n)
+            self._kernel_body(ln, n, emit_host=False)
         ln.graph = g
         ln.graph_key = (n, nbytes)
         return True
 
-    def _respond_compact(self, ln, n):
-        """respond + pad16 cumsum + compact + D2H of lens/offs/total on
-        the caller's current stream."""
+    def _respond_compact(self, ln, n, emit_host=True):
+        """respond + pad16 cumsum + compact (+ D2H of lens/offs/total
+        unless emit_host=False) on the caller's current stream."""
         t = self.torch
         cs = t.cuda.current_stream(self.device).cuda_stream
         self.hip.respond(cs, ln.d_reqs, ln.d_req_off, ln.d_fields,
@@ -380,9 +394,11 @@ class BatchEngine:
         ln.d_resp_off[:n].copy_(csum - pads)
         self.hip.compact(cs, ln.d_resp, ln.d_resp_len, ln.d_resp_off,
                          ln.d_out, n, self.rslot)
-        ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
-        ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
-        ln.p_total.copy_(csum[-1:], non_blocking=True)
+        ln.d_total.copy_(csum[-1:])
+        if emit_host:
+            ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
+            ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
+            ln.p_total.copy_(ln.d_total, non_blocking=True)
 
     def complete(self, lane_idx: int = 0):
         """Wait for the lane's in-flight batch; run the host fixup pass if
