@@ -146,11 +146,12 @@ class _Lane:
         self.nbytes = 0
         self.graph = None
         self.graph_key = None
-        # egress budget: when >0, _respond_compact enqueues the pinned
-        # egress copy of this many bytes on the lane stream, so the
-        # host never blocks the next H2D on the previous D2H — the
-        # full-duplex host link runs both directions concurrently.
-        self.egress_budget = 0
+        # True once the lane runs in staged-graph mode: k_compact on the
+        # egress stream writes responses STRAIGHT into the pinned p_out
+        # ring (kernel-driven host writes, ~54 GB/s — faster than the
+        # runtime's blit and off the compute queues), so complete() has
+        # no explicit D2H to do.
+        self.egress_inline = False
 
 
 class BatchEngine:
@@ -301,10 +302,12 @@ class BatchEngine:
                 ln.e_k.record(self.s_k)
             with t.cuda.stream(self.s_out):
                 self.s_out.wait_event(ln.e_k)
+                # egress: k_compact streams the responses directly into
+                # the pinned ring (out = host pointer)
+                self.hip.compact(self.s_out.cuda_stream, ln.d_resp,
+                                 ln.d_resp_len, ln.d_resp_off, ln.p_out,
+                                 n, self.rslot)
                 self._emit_host_tables(ln, n)
-                if ln.egress_budget:
-                    ln.p_out[:ln.egress_budget].copy_(
-                        ln.d_out[:ln.egress_budget], non_blocking=True)
                 ln.event.record(self.s_out)
             return
         with t.cuda.stream(ln.stream):
@@ -338,7 +341,8 @@ class BatchEngine:
                           self.d_secret, len(self.app.auth_secret))
         # optimistic respond: host rows render a 500 fallback that the
         # fixup pass overwrites before the responses are released
-        self._respond_compact(ln, n, emit_host=emit_host)
+        self._respond_compact(ln, n, emit_host=emit_host,
+                              compact=emit_host)
 
     def _emit_host_tables(self, ln, n):
         """Small D2Hs of the per-batch result tables (current stream)."""
@@ -364,13 +368,7 @@ class BatchEngine:
         with t.cuda.stream(self.s_k):
             self._submit_body(ln, n, nbytes)
         self.s_k.synchronize()
-        # size the in-pipeline egress copy from the warmup batch's actual
-        # egress bytes (+ slack for batch-to-batch response variation);
-        # complete() falls back to an explicit tail copy if a later batch
-        # overflows the budget.
-        total = int(ln.p_total[0])
-        ln.egress_budget = min(len(ln.p_out),
-                               max(4096, int(total * 1.25) + 4096))
+        ln.egress_inline = True
         g = t.cuda.CUDAGraph()
         with t.cuda.graph(g, stream=self.s_k):
             self._kernel_body(ln, n, emit_host=False)
@@ -378,9 +376,11 @@ class BatchEngine:
         ln.graph_key = (n, nbytes)
         return True
 
-    def _respond_compact(self, ln, n, emit_host=True):
-        """respond + pad16 cumsum + compact (+ D2H of lens/offs/total
-        unless emit_host=False) on the caller's current stream."""
+    def _respond_compact(self, ln, n, emit_host=True, compact=True):
+        """respond + pad16 cumsum (+ compact into d_out and D2H of
+        lens/offs/total unless running as the captured pure-compute
+        graph, where submit() compacts straight to the pinned ring on
+        the egress stream instead) on the caller's current stream."""
         t = self.torch
         cs = t.cuda.current_stream(self.device).cuda_stream
         self.hip.respond(cs, ln.d_reqs, ln.d_req_off, ln.d_fields,
@@ -392,8 +392,9 @@ class BatchEngine:
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
-        self.hip.compact(cs, ln.d_resp, ln.d_resp_len, ln.d_resp_off,
-                         ln.d_out, n, self.rslot)
+        if compact:
+            self.hip.compact(cs, ln.d_resp, ln.d_resp_len, ln.d_resp_off,
+                             ln.d_out, n, self.rslot)
         ln.d_total.copy_(csum[-1:])
         if emit_host:
             ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
@@ -431,11 +432,9 @@ class BatchEngine:
                 self._respond_compact(ln, n)
             ln.stream.synchronize()
         total = int(ln.p_total[0])
-        if fixed_up or not (ln.egress_budget
-                            and total <= ln.egress_budget):
-            # no in-pipeline egress copy (budget overflow / no graph /
-            # fixup rewrote responses): explicit D2H of the compact
-            # stream
+        if fixed_up or not ln.egress_inline:
+            # responses are in d_out (no inline egress, or the fixup
+            # pass rewrote them): explicit D2H of the compact stream
             with t.cuda.stream(ln.stream):
                 ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
             ln.stream.synchronize()

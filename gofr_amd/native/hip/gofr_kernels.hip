@@ -1299,9 +1299,15 @@ k_grpc_echo(const uint8_t* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
-// k_compact — gather response slots into a contiguous 16B-aligned stream
-// (halves the D2H bytes vs slot-strided responses; offsets are the
-// exclusive cumsum of round16(resp_len), computed host-side via torch)
+// k_compact — gather response slots into a contiguous 16B-aligned stream.
+// `out` may be DEVICE memory (classic compact; offsets are the exclusive
+// cumsum of round16(resp_len)) or PINNED HOST memory: the kernel then IS
+// the egress path, streaming responses over the host link directly.
+// Measured on MI355X, kernel writes to pinned host sustain ~54 GB/s vs
+// ~44 GB/s for the runtime's copyBuffer blit (ROCm never uses SDMA for
+// D2H-to-pinned here — see benchmarks/overlap_probe.py), and fusing the
+// egress removes one queue hop. Grid-strided with a capped grid so the
+// link-bound sweep leaves CUs free for the next batch's parse/respond.
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
 k_compact(const uint8_t* __restrict__ resp_slots,
@@ -1309,15 +1315,16 @@ k_compact(const uint8_t* __restrict__ resp_slots,
           const int32_t* __restrict__ resp_off,
           uint8_t* __restrict__ out,
           int n, int rslot) {
-    const int wv = threadIdx.x / WAVE;
-    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
-    if (req >= n) return;
     const int lane = lane_id();
-    const int len = resp_len[req];
-    const int nv = (len + 15) >> 4;
-    const uint4* src = (const uint4*)(resp_slots + (size_t)req * rslot);
-    uint4* dst = (uint4*)(out + (size_t)resp_off[req]);
-    for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+    const int wv = threadIdx.x / WAVE;
+    for (int req = blockIdx.x * WAVES_PER_BLOCK + wv; req < n;
+         req += gridDim.x * WAVES_PER_BLOCK) {
+        const int len = resp_len[req];
+        const int nv = (len + 15) >> 4;
+        const uint4* src = (const uint4*)(resp_slots + (size_t)req * rslot);
+        uint4* dst = (uint4*)(out + (size_t)resp_off[req]);
+        for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -1432,7 +1439,8 @@ int gofr_launch_compact(
         void* stream,
         const void* resp_slots, const void* resp_len, const void* resp_off,
         void* out, int n, int rslot) {
-    const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    if (blocks > 768) blocks = 768;  // link-bound: don't hog CUs
     hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
                        (const uint8_t*)resp_slots, (const int32_t*)resp_len,
