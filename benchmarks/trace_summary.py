@@ -31,14 +31,21 @@ def main():
         return
     t1 = max(x['end'] for x in copies)
     t0 = t1 - window_ms * 1e6
+    strs = {}
+    for t in tabs:
+        if 'rocpd_string' in t:
+            for row in c.execute(f"SELECT * FROM {t}"):
+                rd = dict(row)
+                strs[rd.get('id')] = rd.get('string', rd.get('value', ''))
     evs = []
     for x in copies:
         if x['end'] >= t0:
             sz = x.get('size', 0) or 0
-            q = x.get('queue_id', x.get('stream_id', '?'))
-            name = x.get('name', '') or ''
+            q = x.get('queue_id', '?')
+            st = x.get('stream_id', '?')
+            name = strs.get(x.get('name_id'), '')
             evs.append((x['start'], x['end'],
-                        f"COPY {sz>>20:3d}MB q={q} {name[:40]}"))
+                        f"COPY {sz>>20:3d}MB q={q} s={st} {str(name)[:44]}"))
     ksyms = {}
     for t in tabs:
         if 'kernel_symbol' in t:
@@ -48,9 +55,11 @@ def main():
                                              rd.get('kernel_name', ''))
     for x in kerns:
         if x['end'] >= t0:
-            q = x.get('queue_id', x.get('stream_id', '?'))
+            q = x.get('queue_id', '?')
+            st = x.get('stream_id', '?')
             nm = ksyms.get(x.get('kernel_id'), '')
-            evs.append((x['start'], x['end'], f"KERN q={q} {str(nm)[:40]}"))
+            evs.append((x['start'], x['end'],
+                        f"KERN q={q} s={st} {str(nm)[:44]}"))
     evs.sort()
     base = evs[0][0]
     for s, e, dsc in evs[:120]:

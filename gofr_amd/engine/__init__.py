@@ -20,6 +20,7 @@ so every test of this module runs on the CPU box.
 
 from __future__ import annotations
 
+import os
 import time
 
 import numpy as np
@@ -292,13 +293,16 @@ class BatchEngine:
         ln.n, ln.nbytes = n, nbytes
         seed = self._next_seed()
         ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
-        if ln.graph is not None and ln.graph_key == (n, nbytes):
+        if ln.egress_inline and ln.graph_key == (n, nbytes):
             with t.cuda.stream(self.s_in):
                 self._ingress_body(ln, n, nbytes)
                 ln.e_in.record(self.s_in)
             with t.cuda.stream(self.s_k):
                 self.s_k.wait_event(ln.e_in)
-                ln.graph.replay()
+                if ln.graph is not None:
+                    ln.graph.replay()
+                else:
+                    self._kernel_body(ln, n, emit_host=False)
                 ln.e_k.record(self.s_k)
             with t.cuda.stream(self.s_out):
                 self.s_out.wait_event(ln.e_k)
@@ -362,18 +366,23 @@ class BatchEngine:
         graph support."""
         t = self.torch
         ln = self.lanes[lane_idx]
-        if not hasattr(t.cuda, "CUDAGraph"):
-            return False
         # warmup pass (allocations settle) then capture
         with t.cuda.stream(self.s_k):
             self._submit_body(ln, n, nbytes)
         self.s_k.synchronize()
         ln.egress_inline = True
+        ln.graph_key = (n, nbytes)
+        if not hasattr(t.cuda, "CUDAGraph") or \
+                os.environ.get("GOFR_EAGER_KERNELS"):
+            # staged mode without a graph: submit() launches the kernel
+            # section eagerly on s_k (10 launches vs 1 replay — worth it
+            # if replay setup latency exceeds the launch overhead)
+            ln.graph = None
+            return False
         g = t.cuda.CUDAGraph()
         with t.cuda.graph(g, stream=self.s_k):
             self._kernel_body(ln, n, emit_host=False)
         ln.graph = g
-        ln.graph_key = (n, nbytes)
         return True
 
     def _respond_compact(self, ln, n, emit_host=True, compact=True):
