@@ -121,7 +121,9 @@ class _Lane:
         self.e_k = t.cuda.Event()
         self.e_out = t.cuda.Event()
         self.d_reqs = t.empty(max_bytes, dtype=t.uint8, device=dev)
-        self.d_req_off = t.empty(nb, dtype=t.int64, device=dev)
+        # +1: the batch seed rides in the tail slot of the offsets copy
+        # (an 8-byte pinned H2D on its own costs a ~60 us blit kernel)
+        self.d_req_off = t.empty(nb + 1, dtype=t.int64, device=dev)
         self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
         self.d_fields = t.zeros(nb * ops.NF, dtype=t.int32, device=dev)
         self.d_resp = t.empty(nb * rslot, dtype=t.uint8, device=dev)
@@ -132,7 +134,7 @@ class _Lane:
         self.d_host_tab = t.zeros(nb * 4, dtype=t.int32, device=dev)
         self.d_host_blob = t.zeros(host_blob_cap, dtype=t.uint8, device=dev)
         self.p_reqs = t.empty(max_bytes, dtype=t.uint8).pin_memory()
-        self.p_req_off = t.empty(nb, dtype=t.int64).pin_memory()
+        self.p_req_off = t.empty(nb + 1, dtype=t.int64).pin_memory()
         self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
         self.p_out = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
         self.p_resp_len = t.empty(nb, dtype=t.int32).pin_memory()
@@ -211,9 +213,11 @@ class BatchEngine:
         else:
             self.d_secret = None
         # stage streams shared by all lanes: one per direction so the
-        # host link runs full duplex, one for compute
+        # host link runs full duplex, one for compute. s_k runs at high
+        # priority so the parse/respond chain preempts the (link-bound,
+        # grid-capped) egress sweep for CUs.
         self.s_in = t.cuda.Stream(device=dev)
-        self.s_k = t.cuda.Stream(device=dev)
+        self.s_k = t.cuda.Stream(device=dev, priority=-1)
         self.s_out = t.cuda.Stream(device=dev)
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot) for _ in range(self.pipeline)]
@@ -319,10 +323,11 @@ class BatchEngine:
             ln.event.record(ln.stream)
 
     def _ingress_body(self, ln, n, nbytes):
-        """H2D staging copies (runs on the caller's current stream)."""
-        ln.d_seed.copy_(ln.p_seed, non_blocking=True)
+        """H2D staging copies (runs on the caller's current stream).
+        The batch seed travels as element n of the offsets copy."""
+        ln.p_req_off[n] = ln.p_seed[0]
         ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
-        ln.d_req_off[:n].copy_(ln.p_req_off[:n], non_blocking=True)
+        ln.d_req_off[:n + 1].copy_(ln.p_req_off[:n + 1], non_blocking=True)
         ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
 
     def _kernel_body(self, ln, n, emit_host=True):
@@ -396,7 +401,8 @@ class BatchEngine:
                          ln.d_resp, ln.d_resp_len, n, self.rslot,
                          self.d_handler_tab, self.program.n_routes,
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
-                         ln.d_seed, auth_env=self.program.auth_env,
+                         ln.d_req_off[n:n + 1],  # seed (offsets tail)
+                         auth_env=self.program.auth_env,
                          gzip_min=self.app.gzip_min_size or 0)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)

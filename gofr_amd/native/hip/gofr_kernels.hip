@@ -23,6 +23,7 @@
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #define WAVE 64
 #define WAVES_PER_BLOCK 4
@@ -1439,8 +1440,17 @@ int gofr_launch_compact(
         void* stream,
         const void* resp_slots, const void* resp_len, const void* resp_off,
         void* out, int n, int rslot) {
+    // Link-bound when out is pinned host memory: cap the grid so the
+    // egress sweep leaves CUs free for the next batch's parse/respond
+    // (tunable for the bandwidth/occupancy sweep in benchmarks/).
+    static int cap = 0;
+    if (cap == 0) {
+        const char* e = getenv("GOFR_COMPACT_BLOCKS");
+        cap = e ? atoi(e) : 128;
+        if (cap <= 0) cap = 128;
+    }
     int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
-    if (blocks > 768) blocks = 768;  // link-bound: don't hog CUs
+    if (blocks > cap) blocks = cap;
     hipLaunchKernelGGL(k_compact, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
                        (const uint8_t*)resp_slots, (const int32_t*)resp_len,
