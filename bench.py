@@ -363,7 +363,8 @@ def main():
     if have_gpu:
         torch.cuda.set_device(local_rank)
     eng = BatchEngine(app, device=device, slot=2048, max_batch=batch,
-                      pipeline=(3 if (have_gpu and world == 1) else 1))
+                      pipeline=(int(os.environ.get("GOFR_PIPELINE", "5"))
+                                if (have_gpu and world == 1) else 1))
 
     extra = b""
     if "auth" in mw:
