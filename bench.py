@@ -105,19 +105,27 @@ def run_single(eng, payloads, steps, warmup):
 
     submit_at = [0.0] * steps
     lat = []
+    t_sub = t_comp = 0.0
     t_start = time.perf_counter()
     for i in range(steps):
         lane = i % P
         if i >= P:
+            t0 = time.perf_counter()
             eng.complete(lane)
+            t_comp += time.perf_counter() - t0
             lat.append(time.perf_counter() - submit_at[i - P])
         submit_at[i] = time.perf_counter()
         eng.submit(n, nbytes, lane)
+        t_sub += time.perf_counter() - submit_at[i]
     for i in range(max(0, steps - P), steps):
         eng.complete(i % P)
         lat.append(time.perf_counter() - submit_at[i])
     torch.cuda.synchronize(eng.device)
     elapsed = time.perf_counter() - t_start
+    if os.environ.get("GOFR_TIMING"):
+        print(f"[timing] submit {t_sub/steps*1000:.3f} ms/step, "
+              f"complete-wait {t_comp/max(1,steps-P)*1000:.3f} ms/step",
+              file=sys.stderr)
     return elapsed, lat
 
 

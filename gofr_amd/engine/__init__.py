@@ -145,16 +145,18 @@ class _Lane:
         self.d_seed = t.zeros(1, dtype=t.int64, device=dev)
         self.p_seed = t.zeros(1, dtype=t.int64).pin_memory()
         self.d_total = t.zeros(1, dtype=t.int32, device=dev)
+        # native staged-submit result tables (int32 [2n+2] for batch n:
+        # [0:n] resp_len, [n:2n] resp_off, [2n] total, [2n+1]
+        # host_needed) — one D2H moves the whole per-batch result set
+        self.d_tables = t.zeros(2 * nb + 2, dtype=t.int32, device=dev)
+        self.p_tables = t.zeros(2 * nb + 2, dtype=t.int32).pin_memory()
+        self.p_tables_np = self.p_tables.numpy()
+        self.p_req_off_np = self.p_req_off.numpy()
         self.n = 0
         self.nbytes = 0
-        self.graph = None
         self.graph_key = None
-        # True once the lane runs in staged-graph mode: k_compact on the
-        # egress stream writes responses STRAIGHT into the pinned p_out
-        # ring (kernel-driven host writes, ~54 GB/s — faster than the
-        # runtime's blit and off the compute queues), so complete() has
-        # no explicit D2H to do.
-        self.egress_inline = False
+        self.c_args = None   # ops.GofrSubmitArgs once armed
+        self.mode = "torch"  # which path produced the in-flight batch
 
 
 class BatchEngine:
@@ -285,39 +287,26 @@ class BatchEngine:
     # -- pipelined API --------------------------------------------------------
     def submit(self, n: int, nbytes: int, lane_idx: int = 0) -> None:
         """Enqueue one batch (already staged in lane.p_reqs[:nbytes] /
-        p_req_off / p_req_len). With a captured kernel graph the batch
-        flows through the three stage streams — eager H2D on s_in,
-        kernel-graph replay on s_k, eager egress D2H on s_out — chained
-        by events, so concurrent lanes overlap ingress, compute and
-        egress (full-duplex host link). Never blocks; complete()
-        finishes it. A lane must be complete()d before it is resubmitted.
-        """
+        p_req_off / p_req_len). Once the lane is armed (arm_lane), the
+        whole staged pipeline — eager H2D on s_in, kernel chain on s_k,
+        k_compact straight into the pinned egress ring + one result-table
+        D2H on s_out, chained by events — is enqueued by ONE native call
+        (gofr_submit_staged): measured, the ~14 framework dispatches the
+        Python version cost paced the serving loop at ~0.3 ms/step of
+        host time. Never blocks; complete() finishes it. A lane must be
+        complete()d before it is resubmitted."""
         t = self.torch
         ln = self.lanes[lane_idx]
         ln.n, ln.nbytes = n, nbytes
         seed = self._next_seed()
-        ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
-        if ln.egress_inline and ln.graph_key == (n, nbytes):
-            with t.cuda.stream(self.s_in):
-                self._ingress_body(ln, n, nbytes)
-                ln.e_in.record(self.s_in)
-            with t.cuda.stream(self.s_k):
-                self.s_k.wait_event(ln.e_in)
-                if ln.graph is not None:
-                    ln.graph.replay()
-                else:
-                    self._kernel_body(ln, n, emit_host=False)
-                ln.e_k.record(self.s_k)
-            with t.cuda.stream(self.s_out):
-                self.s_out.wait_event(ln.e_k)
-                # egress: k_compact streams the responses directly into
-                # the pinned ring (out = host pointer)
-                self.hip.compact(self.s_out.cuda_stream, ln.d_resp,
-                                 ln.d_resp_len, ln.d_resp_off, ln.p_out,
-                                 n, self.rslot)
-                self._emit_host_tables(ln, n)
-                ln.event.record(self.s_out)
+        signed = seed - (1 << 64) if seed >= (1 << 63) else seed
+        ln.p_seed[0] = signed
+        if ln.c_args is not None and ln.graph_key == (n, nbytes):
+            ln.mode = "c"
+            ln.p_req_off_np[n] = signed  # seed rides the offsets tail
+            self.hip.submit_staged(ln.c_args)
             return
+        ln.mode = "torch"
         with t.cuda.stream(ln.stream):
             self._submit_body(ln, n, nbytes)
             ln.event.record(ln.stream)
@@ -331,12 +320,8 @@ class BatchEngine:
         ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
 
     def _kernel_body(self, ln, n, emit_host=True):
-        """parse -> auth -> respond -> compact (current stream; this is
-        the section captured as the lane's hipGraph). With
-        emit_host=False no D2H copy is enqueued at all — the captured
-        graph stays pure compute, so no s_k memcpy node can queue behind
-        another lane's 40 MB egress on a shared SDMA engine; submit()
-        emits the small D2Hs on the egress stream instead."""
+        """parse -> auth -> respond -> compact (current stream; the
+        torch fallback path)."""
         cs = self.torch.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
         self.hip.parse_route(cs, ln.d_reqs, ln.d_req_off, ln.d_req_len,
@@ -350,44 +335,72 @@ class BatchEngine:
                           self.d_secret, len(self.app.auth_secret))
         # optimistic respond: host rows render a 500 fallback that the
         # fixup pass overwrites before the responses are released
-        self._respond_compact(ln, n, emit_host=emit_host,
-                              compact=emit_host)
-
-    def _emit_host_tables(self, ln, n):
-        """Small D2Hs of the per-batch result tables (current stream)."""
-        ln.p_host_needed.copy_(ln.d_host_needed, non_blocking=True)
-        ln.p_resp_len[:n].copy_(ln.d_resp_len[:n], non_blocking=True)
-        ln.p_resp_off[:n].copy_(ln.d_resp_off[:n], non_blocking=True)
-        ln.p_total.copy_(ln.d_total, non_blocking=True)
+        self._respond_compact(ln, n, emit_host=emit_host)
 
     def _submit_body(self, ln, n, nbytes):
         self._ingress_body(ln, n, nbytes)
         self._kernel_body(ln, n)
 
     def capture_graph(self, n: int, nbytes: int, lane_idx: int = 0) -> bool:
-        """Capture the lane's KERNEL section into a hipGraph for (n,
-        nbytes)-shaped batches; submit() then replays it on s_k between
-        the eager stage copies. Returns False when the torch build lacks
-        graph support."""
+        """Arm the lane for (n, nbytes)-shaped batches: run one warmup
+        pass (allocations settle, event handles materialize) and build
+        the native GofrSubmitArgs block that gofr_submit_staged replays
+        per batch. (Name kept from the hipGraph era — the native driver
+        replaced graph replay, which cost more in setup latency than it
+        saved in launches.)"""
+        import ctypes
         t = self.torch
         ln = self.lanes[lane_idx]
-        # warmup pass (allocations settle) then capture
+        # warmup pass (allocations settle) then arm
         with t.cuda.stream(self.s_k):
             self._submit_body(ln, n, nbytes)
         self.s_k.synchronize()
-        ln.egress_inline = True
+        # materialize torch event handles for HIP interop
+        for ev in (ln.e_in, ln.e_k, ln.event):
+            ev.record(self.s_k)
+        self.s_k.synchronize()
+        a = ops.GofrSubmitArgs()
+        a.s_in = self.s_in.cuda_stream
+        a.s_k = self.s_k.cuda_stream
+        a.s_out = self.s_out.cuda_stream
+        a.ev_in = ln.e_in.cuda_event
+        a.ev_k = ln.e_k.cuda_event
+        a.ev_done = ln.event.cuda_event
+        a.p_reqs = ln.p_reqs.data_ptr()
+        a.d_reqs = ln.d_reqs.data_ptr()
+        a.nbytes = nbytes
+        a.p_off = ln.p_req_off.data_ptr()
+        a.d_off = ln.d_req_off.data_ptr()
+        a.p_len = ln.p_req_len.data_ptr()
+        a.d_len = ln.d_req_len.data_ptr()
+        a.d_fields = ln.d_fields.data_ptr()
+        tr = self.d_trie
+        a.trie = (ctypes.c_void_p * 9)(
+            tr["seg_blob"].data_ptr(), tr["node_child_first"].data_ptr(),
+            tr["node_child_count"].data_ptr(),
+            tr["child_seg_off"].data_ptr(), tr["child_seg_len"].data_ptr(),
+            tr["child_node"].data_ptr(), tr["node_param"].data_ptr(),
+            tr["node_prefix"].data_ptr(), tr["node_route"].data_ptr())
+        a.handler_tab = self.d_handler_tab.data_ptr()
+        a.n_routes = self.program.n_routes
+        a.d_host_needed = ln.d_host_needed.data_ptr()
+        a.secret = self.d_secret.data_ptr() if self.d_secret is not None \
+            else 0
+        a.secret_len = len(self.app.auth_secret) \
+            if self.d_secret is not None else 0
+        a.d_resp = ln.d_resp.data_ptr()
+        a.d_tables = ln.d_tables.data_ptr()
+        a.p_tables = ln.p_tables.data_ptr()
+        a.blob = self.d_blob.data_ptr()
+        a.host_blob = ln.d_host_blob.data_ptr()
+        a.host_tab = ln.d_host_tab.data_ptr()
+        a.auth_env_off, a.auth_env_len = self.program.auth_env
+        a.gzip_min = self.app.gzip_min_size or 0
+        a.p_out = ln.p_out.data_ptr()
+        a.n = n
+        a.rslot = self.rslot
+        ln.c_args = a
         ln.graph_key = (n, nbytes)
-        if not hasattr(t.cuda, "CUDAGraph") or \
-                os.environ.get("GOFR_EAGER_KERNELS"):
-            # staged mode without a graph: submit() launches the kernel
-            # section eagerly on s_k (10 launches vs 1 replay — worth it
-            # if replay setup latency exceeds the launch overhead)
-            ln.graph = None
-            return False
-        g = t.cuda.CUDAGraph()
-        with t.cuda.graph(g, stream=self.s_k):
-            self._kernel_body(ln, n, emit_host=False)
-        ln.graph = g
         return True
 
     def _respond_compact(self, ln, n, emit_host=True, compact=True):
@@ -424,8 +437,12 @@ class BatchEngine:
         ln = self.lanes[lane_idx]
         n = ln.n
         ln.event.synchronize()
+        if ln.mode == "c":
+            host_needed = int(ln.p_tables_np[2 * n + 1])
+        else:
+            host_needed = int(ln.p_host_needed[0])
         fixed_up = False
-        if int(ln.p_host_needed[0]):
+        if host_needed:
             fixed_up = True
             # fixup: run Python handlers for HK_HOST rows, re-serialize
             with t.cuda.stream(ln.stream):
@@ -446,13 +463,19 @@ class BatchEngine:
                     non_blocking=True)
                 self._respond_compact(ln, n)
             ln.stream.synchronize()
+        if ln.mode == "c" and not fixed_up:
+            # native staged path: k_compact already streamed the
+            # responses into the pinned ring; tables came down in one
+            # D2H (layout: [0:n] len, [n:2n] off, [2n] total)
+            total = int(ln.p_tables_np[2 * n])
+            return (ln.p_out[:total], ln.p_tables[n:2 * n],
+                    ln.p_tables[:n])
         total = int(ln.p_total[0])
-        if fixed_up or not ln.egress_inline:
-            # responses are in d_out (no inline egress, or the fixup
-            # pass rewrote them): explicit D2H of the compact stream
-            with t.cuda.stream(ln.stream):
-                ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
-            ln.stream.synchronize()
+        # responses are in d_out (torch fallback, or the fixup pass
+        # rewrote them): explicit D2H of the compact stream
+        with t.cuda.stream(ln.stream):
+            ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
+        ln.stream.synchronize()
         return ln.p_out[:total], ln.p_resp_off[:n], ln.p_resp_len[:n]
 
     def process_device(self, d_reqs, d_req_off, d_req_len, n):

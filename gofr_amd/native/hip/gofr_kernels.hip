@@ -1244,6 +1244,48 @@ k_varint_spans(const uint8_t* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
+// k_padscan — exclusive cumsum of round16(resp_len) (the compact-stream
+// offsets) + total + host_needed mirror, in ONE single-workgroup kernel.
+// Replaces the host-side torch chain (elementwise pad + rocprim scan +
+// copies), whose ~6 dispatches of host overhead paced the whole serving
+// loop. tables layout (int32): [0:n] resp_len (written by k_respond),
+// [n:2n] resp_off (written here), [2n] total, [2n+1] host_needed.
+// ---------------------------------------------------------------------------
+#define SCAN_THREADS 1024
+
+extern "C" __global__ void __launch_bounds__(SCAN_THREADS)
+k_padscan(int32_t* __restrict__ tables,
+          const int32_t* __restrict__ host_needed,
+          int n) {
+    __shared__ int32_t partials[SCAN_THREADS];
+    const int tid = threadIdx.x;
+    const int chunk = (n + SCAN_THREADS - 1) / SCAN_THREADS;
+    const int s = tid * chunk;
+    int e = s + chunk;
+    if (e > n) e = n;
+    // per-thread serial sum of padded lengths
+    int32_t sum = 0;
+    for (int i = s; i < e; ++i) sum += (tables[i] + 15) & ~15;
+    partials[tid] = sum;
+    __syncthreads();
+    // Hillis-Steele inclusive scan over the 1024 partials
+    for (int off = 1; off < SCAN_THREADS; off <<= 1) {
+        int32_t v = (tid >= off) ? partials[tid - off] : 0;
+        __syncthreads();
+        partials[tid] += v;
+        __syncthreads();
+    }
+    // exclusive offsets for this thread's chunk
+    int32_t run = (tid > 0) ? partials[tid - 1] : 0;
+    for (int i = s; i < e; ++i) {
+        tables[n + i] = run;
+        run += (tables[i] + 15) & ~15;
+    }
+    if (tid == SCAN_THREADS - 1) tables[2 * n] = partials[tid];
+    if (tid == 0) tables[2 * n + 1] = *host_needed;
+}
+
+// ---------------------------------------------------------------------------
 // k_grpc_echo — batched unary SayHello responder (BASELINE config 5's
 // gRPC half). One message per lane: read the name span from the
 // k_varint_spans table, emit the gRPC length-prefixed HelloResponse
@@ -1331,6 +1373,142 @@ k_compact(const uint8_t* __restrict__ resp_slots,
 // ---------------------------------------------------------------------------
 // C API (ctypes-driven; raw device pointers from torch tensors)
 // ---------------------------------------------------------------------------
+
+// One-call staged submit (the native serving-loop driver). The Python
+// loop was spending ~0.3 ms/step in framework dispatch enqueueing the
+// same ~14 stream operations — at 32k requests/batch that overhead,
+// not the GPU, paced the engine. This enqueues the whole batch
+// pipeline with raw HIP calls:
+//   s_in : H2D reqs + offsets(+seed tail) + lens, record ev_in
+//   s_k  : wait ev_in; memset host_needed; k_parse_route; [k_auth];
+//          k_respond; k_padscan  -> record ev_k
+//   s_out: wait ev_k; k_compact straight into the pinned egress ring;
+//          one D2H of the result tables; record ev_done
+// Event handles come from torch (interop: same HIP runtime), so the
+// Python side still waits with event.synchronize().
+typedef struct {
+    void* s_in; void* s_k; void* s_out;
+    void* ev_in; void* ev_k; void* ev_done;
+    const void* p_reqs; void* d_reqs; long long nbytes;
+    const void* p_off; void* d_off;        // (n+1) int64, seed in tail
+    const void* p_len; void* d_len;        // n int32
+    void* d_fields;
+    const void* trie[9];
+    const void* handler_tab; int n_routes;
+    void* d_host_needed;
+    const void* secret; int secret_len;
+    void* d_resp;
+    void* d_tables;                        // int32[2n+2]
+    void* p_tables;                        // pinned int32[2n+2]
+    const void* blob; const void* host_blob; const void* host_tab;
+    int auth_env_off; int auth_env_len; int gzip_min;
+    void* p_out;                           // pinned egress ring
+    int n; int rslot;
+} GofrSubmitArgs;
+
+extern "C" int gofr_submit_staged(const GofrSubmitArgs* a) {
+    hipStream_t s_in = (hipStream_t)a->s_in;
+    hipStream_t s_k = (hipStream_t)a->s_k;
+    hipStream_t s_out = (hipStream_t)a->s_out;
+    const int n = a->n;
+    hipError_t rc;
+    // ---- ingress ----------------------------------------------------------
+    rc = hipMemcpyAsync(a->d_reqs, a->p_reqs, (size_t)a->nbytes,
+                        hipMemcpyHostToDevice, s_in);
+    if (rc) return (int)rc;
+    rc = hipMemcpyAsync(a->d_off, a->p_off, (size_t)(n + 1) * 8,
+                        hipMemcpyHostToDevice, s_in);
+    if (rc) return (int)rc;
+    rc = hipMemcpyAsync(a->d_len, a->p_len, (size_t)n * 4,
+                        hipMemcpyHostToDevice, s_in);
+    if (rc) return (int)rc;
+    rc = hipEventRecord((hipEvent_t)a->ev_in, s_in);
+    if (rc) return (int)rc;
+    // ---- kernel stage -----------------------------------------------------
+    rc = hipStreamWaitEvent(s_k, (hipEvent_t)a->ev_in, 0);
+    if (rc) return (int)rc;
+    rc = hipMemsetAsync(a->d_host_needed, 0, 4, s_k);
+    if (rc) return (int)rc;
+    TrieDev trie{(const uint8_t*)a->trie[0], (const int32_t*)a->trie[1],
+                 (const int32_t*)a->trie[2], (const int32_t*)a->trie[3],
+                 (const int32_t*)a->trie[4], (const int32_t*)a->trie[5],
+                 (const int32_t*)a->trie[6], (const int32_t*)a->trie[7],
+                 (const int32_t*)a->trie[8]};
+    const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    hipLaunchKernelGGL(k_parse_route, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       s_k,
+                       (const uint8_t*)a->d_reqs, (const int64_t*)a->d_off,
+                       (const int32_t*)a->d_len, (int32_t*)a->d_fields, n,
+                       trie, (const int32_t*)a->handler_tab, a->n_routes,
+                       (int32_t*)a->d_host_needed);
+    if ((rc = hipGetLastError())) return (int)rc;
+    if (a->secret_len > 0) {
+        const int ablocks = (n + BLOCK_THREADS - 1) / BLOCK_THREADS;
+        hipLaunchKernelGGL(k_auth, dim3(ablocks), dim3(BLOCK_THREADS), 0,
+                           s_k,
+                           (const uint8_t*)a->d_reqs,
+                           (const int64_t*)a->d_off,
+                           (int32_t*)a->d_fields, n,
+                           (const uint8_t*)a->secret, a->secret_len);
+        if ((rc = hipGetLastError())) return (int)rc;
+    }
+    int32_t* tables = (int32_t*)a->d_tables;
+    const uint64_t* seed_ptr = (const uint64_t*)a->d_off + n;
+    if (a->gzip_min > 0) {
+        hipLaunchKernelGGL(k_respond_gz, dim3(blocks), dim3(BLOCK_THREADS),
+                           0, s_k,
+                           (const uint8_t*)a->d_reqs,
+                           (const int64_t*)a->d_off,
+                           (int32_t*)a->d_fields,
+                           (uint8_t*)a->d_resp, tables, n, a->rslot,
+                           (const int32_t*)a->handler_tab, a->n_routes,
+                           (const uint8_t*)a->blob,
+                           (const uint8_t*)a->host_blob,
+                           (const int32_t*)a->host_tab, seed_ptr,
+                           a->auth_env_off, a->auth_env_len, a->gzip_min);
+    } else {
+        hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS),
+                           0, s_k,
+                           (const uint8_t*)a->d_reqs,
+                           (const int64_t*)a->d_off,
+                           (int32_t*)a->d_fields,
+                           (uint8_t*)a->d_resp, tables, n, a->rslot,
+                           (const int32_t*)a->handler_tab, a->n_routes,
+                           (const uint8_t*)a->blob,
+                           (const uint8_t*)a->host_blob,
+                           (const int32_t*)a->host_tab, seed_ptr,
+                           a->auth_env_off, a->auth_env_len);
+    }
+    if ((rc = hipGetLastError())) return (int)rc;
+    hipLaunchKernelGGL(k_padscan, dim3(1), dim3(SCAN_THREADS), 0, s_k,
+                       tables, (const int32_t*)a->d_host_needed, n);
+    if ((rc = hipGetLastError())) return (int)rc;
+    rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);
+    if (rc) return (int)rc;
+    // ---- egress -----------------------------------------------------------
+    rc = hipStreamWaitEvent(s_out, (hipEvent_t)a->ev_k, 0);
+    if (rc) return (int)rc;
+    int cblocks = blocks;
+    {
+        static int cap = 0;
+        if (cap == 0) {
+            const char* e = getenv("GOFR_COMPACT_BLOCKS");
+            cap = e ? atoi(e) : 128;
+            if (cap <= 0) cap = 128;
+        }
+        if (cblocks > cap) cblocks = cap;
+    }
+    hipLaunchKernelGGL(k_compact, dim3(cblocks), dim3(BLOCK_THREADS), 0,
+                       s_out,
+                       (const uint8_t*)a->d_resp, tables, tables + n,
+                       (uint8_t*)a->p_out, n, a->rslot);
+    if ((rc = hipGetLastError())) return (int)rc;
+    rc = hipMemcpyAsync(a->p_tables, a->d_tables, (size_t)(2 * n + 2) * 4,
+                        hipMemcpyDeviceToHost, s_out);
+    if (rc) return (int)rc;
+    return (int)hipEventRecord((hipEvent_t)a->ev_done, s_out);
+}
+
 extern "C" {
 
 int gofr_launch_parse_route(

@@ -61,6 +61,35 @@ _SO_PATH = os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "_gofr_hip.so")
 
 
+class GofrSubmitArgs(ctypes.Structure):
+    """Mirror of GofrSubmitArgs in gofr_kernels.hip (the one-call staged
+    submit: ingress H2D -> kernel chain -> egress, all raw HIP)."""
+
+    _fields_ = [
+        ("s_in", ctypes.c_void_p), ("s_k", ctypes.c_void_p),
+        ("s_out", ctypes.c_void_p),
+        ("ev_in", ctypes.c_void_p), ("ev_k", ctypes.c_void_p),
+        ("ev_done", ctypes.c_void_p),
+        ("p_reqs", ctypes.c_void_p), ("d_reqs", ctypes.c_void_p),
+        ("nbytes", ctypes.c_longlong),
+        ("p_off", ctypes.c_void_p), ("d_off", ctypes.c_void_p),
+        ("p_len", ctypes.c_void_p), ("d_len", ctypes.c_void_p),
+        ("d_fields", ctypes.c_void_p),
+        ("trie", ctypes.c_void_p * 9),
+        ("handler_tab", ctypes.c_void_p), ("n_routes", ctypes.c_int),
+        ("d_host_needed", ctypes.c_void_p),
+        ("secret", ctypes.c_void_p), ("secret_len", ctypes.c_int),
+        ("d_resp", ctypes.c_void_p),
+        ("d_tables", ctypes.c_void_p), ("p_tables", ctypes.c_void_p),
+        ("blob", ctypes.c_void_p), ("host_blob", ctypes.c_void_p),
+        ("host_tab", ctypes.c_void_p),
+        ("auth_env_off", ctypes.c_int), ("auth_env_len", ctypes.c_int),
+        ("gzip_min", ctypes.c_int),
+        ("p_out", ctypes.c_void_p),
+        ("n", ctypes.c_int), ("rslot", ctypes.c_int),
+    ]
+
+
 class HipOps:
     """ctypes driver for the device kernels."""
 
@@ -104,6 +133,9 @@ class HipOps:
         self.lib.gofr_launch_grpc_echo.restype = ctypes.c_int
         self.lib.gofr_launch_grpc_echo.argtypes = \
             [ctypes.c_void_p] * 6 + [ctypes.c_int, ctypes.c_int]
+        self.lib.gofr_submit_staged.restype = ctypes.c_int
+        self.lib.gofr_submit_staged.argtypes = \
+            [ctypes.POINTER(GofrSubmitArgs)]
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,
@@ -174,6 +206,11 @@ class HipOps:
         if rc != 0:
             raise RuntimeError(
                 f"k_varint_spans launch failed: hipError {rc}")
+
+    def submit_staged(self, args: GofrSubmitArgs):
+        rc = self.lib.gofr_submit_staged(ctypes.byref(args))
+        if rc != 0:
+            raise RuntimeError(f"gofr_submit_staged failed: hipError {rc}")
 
     def grpc_echo(self, stream, buf_t, spans_t, span_n_t, out_t, out_len_t,
                   n, rslot):
