@@ -126,6 +126,17 @@ def run_single(eng, payloads, steps, warmup):
         print(f"[timing] submit {t_sub/steps*1000:.3f} ms/step, "
               f"complete-wait {t_comp/max(1,steps-P)*1000:.3f} ms/step",
               file=sys.stderr)
+        try:
+            import ctypes
+            buf = (ctypes.c_double * 5)()
+            eng.hip.lib.gofr_submit_stats(buf)
+            if buf[4]:
+                print(f"[timing-c] per-call us: bigH2D {buf[0]/buf[4]:.0f} "
+                      f"ingress-rest {buf[1]/buf[4]:.0f} "
+                      f"kernels {buf[2]/buf[4]:.0f} egress {buf[3]/buf[4]:.0f}"
+                      f" calls {int(buf[4])}", file=sys.stderr)
+        except (AttributeError, OSError):
+            pass
     return elapsed, lat
 
 

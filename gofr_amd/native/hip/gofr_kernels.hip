@@ -1406,16 +1406,32 @@ typedef struct {
     int n; int rslot;
 } GofrSubmitArgs;
 
+// enqueue-time breakdown (µs, cumulative): [0] big H2D, [1] rest of
+// ingress, [2] kernel stage, [3] egress; [4] = calls
+static double g_submit_us[5] = {0, 0, 0, 0, 0};
+static inline double now_us() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return ts.tv_sec * 1e6 + ts.tv_nsec * 1e-3;
+}
+extern "C" void gofr_submit_stats(double* out) {
+    for (int i = 0; i < 5; ++i) { out[i] = g_submit_us[i];
+                                  g_submit_us[i] = 0; }
+}
+
 extern "C" int gofr_submit_staged(const GofrSubmitArgs* a) {
     hipStream_t s_in = (hipStream_t)a->s_in;
     hipStream_t s_k = (hipStream_t)a->s_k;
     hipStream_t s_out = (hipStream_t)a->s_out;
     const int n = a->n;
     hipError_t rc;
+    double t0 = now_us();
+    g_submit_us[4] += 1;
     // ---- ingress ----------------------------------------------------------
     rc = hipMemcpyAsync(a->d_reqs, a->p_reqs, (size_t)a->nbytes,
                         hipMemcpyHostToDevice, s_in);
     if (rc) return (int)rc;
+    g_submit_us[0] += now_us() - t0; t0 = now_us();
     rc = hipMemcpyAsync(a->d_off, a->p_off, (size_t)(n + 1) * 8,
                         hipMemcpyHostToDevice, s_in);
     if (rc) return (int)rc;
@@ -1424,6 +1440,7 @@ extern "C" int gofr_submit_staged(const GofrSubmitArgs* a) {
     if (rc) return (int)rc;
     rc = hipEventRecord((hipEvent_t)a->ev_in, s_in);
     if (rc) return (int)rc;
+    g_submit_us[1] += now_us() - t0; t0 = now_us();
     // ---- kernel stage -----------------------------------------------------
     rc = hipStreamWaitEvent(s_k, (hipEvent_t)a->ev_in, 0);
     if (rc) return (int)rc;
@@ -1485,6 +1502,7 @@ extern "C" int gofr_submit_staged(const GofrSubmitArgs* a) {
     if ((rc = hipGetLastError())) return (int)rc;
     rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);
     if (rc) return (int)rc;
+    g_submit_us[2] += now_us() - t0; t0 = now_us();
     // ---- egress -----------------------------------------------------------
     rc = hipStreamWaitEvent(s_out, (hipEvent_t)a->ev_k, 0);
     if (rc) return (int)rc;
@@ -1506,7 +1524,9 @@ extern "C" int gofr_submit_staged(const GofrSubmitArgs* a) {
     rc = hipMemcpyAsync(a->p_tables, a->d_tables, (size_t)(2 * n + 2) * 4,
                         hipMemcpyDeviceToHost, s_out);
     if (rc) return (int)rc;
-    return (int)hipEventRecord((hipEvent_t)a->ev_done, s_out);
+    rc = hipEventRecord((hipEvent_t)a->ev_done, s_out);
+    g_submit_us[3] += now_us() - t0;
+    return (int)rc;
 }
 
 extern "C" {
