@@ -447,7 +447,14 @@ class BatchEngine:
         t = self.torch
         ln = self.lanes[lane_idx]
         n = ln.n
-        ln.event.synchronize()
+        if ln.mode == "c":
+            # spin on the event: hipEventSynchronize wakes ~0.3 ms late
+            # (interrupt path); query() polls the signal directly
+            q = ln.event.query
+            while not q():
+                pass
+        else:
+            ln.event.synchronize()
         if ln.mode == "c":
             host_needed = int(ln.p_tables_np[2 * n + 1])
         else:
