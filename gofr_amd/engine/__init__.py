@@ -20,6 +20,7 @@ so every test of this module runs on the CPU box.
 
 from __future__ import annotations
 
+import ctypes
 import os
 import time
 
@@ -188,6 +189,7 @@ class BatchEngine:
                 self.device = torch.device(device)
         except ImportError:
             pass
+        self._pump_done = None
         if self.device is not None:
             self.hip = ops.HipOps()  # raises if extension missing
             self._alloc_device()
@@ -310,7 +312,10 @@ class BatchEngine:
         if ln.c_args is not None and ln.graph_key == (n, nbytes):
             ln.mode = "c"
             ln.p_req_off_np[n] = signed  # seed rides the offsets tail
-            self.hip.submit_staged(ln.c_args)
+            serial = self.hip.lib.gofr_pump_submit(ctypes.byref(ln.c_args))
+            if serial == 0:
+                raise RuntimeError("pump ring full (lane protocol bug)")
+            ln.serial = serial
             return
         ln.mode = "torch"
         with t.cuda.stream(ln.stream):
@@ -354,7 +359,6 @@ class BatchEngine:
         per batch. (Name kept from the hipGraph era — the native driver
         replaced graph replay, which cost more in setup latency than it
         saved in launches.)"""
-        import ctypes
         t = self.torch
         ln = self.lanes[lane_idx]
         # warmup pass (allocations settle) then arm
@@ -410,6 +414,12 @@ class BatchEngine:
         a.p_out = ln.p_out.data_ptr()
         a.n = n
         a.rslot = self.rslot
+        if self._pump_done is None:
+            if self.hip.lib.gofr_pump_start() != 0:
+                raise RuntimeError("gofr_pump_start failed")
+            addr = self.hip.lib.gofr_pump_done_ptr()
+            self._pump_done = np.frombuffer(
+                (ctypes.c_uint64 * 1).from_address(addr), dtype=np.uint64)
         ln.c_args = a
         ln.graph_key = (n, nbytes)
         return True
@@ -448,11 +458,17 @@ class BatchEngine:
         ln = self.lanes[lane_idx]
         n = ln.n
         if ln.mode == "c":
-            # spin on the event: hipEventSynchronize wakes ~0.3 ms late
-            # (interrupt path); query() polls the signal directly
-            q = ln.event.query
-            while not q():
+            # spin on the pump's published serial: a plain host-memory
+            # read, so the serving thread never touches runtime locks
+            # (the pump's completion thread absorbs the event-wake
+            # latency off the critical path)
+            done = self._pump_done
+            serial = ln.serial
+            while done[0] < serial:
                 pass
+            if self.hip.lib.gofr_pump_err():
+                raise RuntimeError(
+                    f"pump error: hipError {self.hip.lib.gofr_pump_err()}")
         else:
             ln.event.synchronize()
         if ln.mode == "c":

@@ -136,6 +136,12 @@ class HipOps:
         self.lib.gofr_submit_staged.restype = ctypes.c_int
         self.lib.gofr_submit_staged.argtypes = \
             [ctypes.POINTER(GofrSubmitArgs)]
+        self.lib.gofr_pump_start.restype = ctypes.c_int
+        self.lib.gofr_pump_submit.restype = ctypes.c_uint64
+        self.lib.gofr_pump_submit.argtypes = \
+            [ctypes.POINTER(GofrSubmitArgs)]
+        self.lib.gofr_pump_done_ptr.restype = ctypes.c_void_p
+        self.lib.gofr_pump_err.restype = ctypes.c_int
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,
