@@ -1404,6 +1404,12 @@ typedef struct {
     int auth_env_off; int auth_env_len; int gzip_min;
     void* p_out;                           // pinned egress ring
     int n; int rslot;
+    // egress mode: budget > 0 -> k_compact into d_out (HBM) + one
+    // budget-sized hipMemcpyAsync D2H (runtime blit; fastest duplex
+    // partner for the SDMA ingress). budget == 0 -> k_compact writes
+    // the pinned ring directly (kernel-driven host writes).
+    void* d_out;
+    long long egress_budget;
 } GofrSubmitArgs;
 
 // enqueue-time breakdown (µs, cumulative): [0] big H2D, [1] rest of
@@ -1522,11 +1528,22 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
         }
         if (cblocks > cap) cblocks = cap;
     }
-    hipLaunchKernelGGL(k_compact, dim3(cblocks), dim3(BLOCK_THREADS), 0,
-                       s_out,
-                       (const uint8_t*)a->d_resp, tables, tables + n,
-                       (uint8_t*)a->p_out, n, a->rslot);
-    if ((rc = hipGetLastError())) return (int)rc;
+    if (a->egress_budget > 0) {
+        hipLaunchKernelGGL(k_compact, dim3(cblocks), dim3(BLOCK_THREADS),
+                           0, s_out,
+                           (const uint8_t*)a->d_resp, tables, tables + n,
+                           (uint8_t*)a->d_out, n, a->rslot);
+        if ((rc = hipGetLastError())) return (int)rc;
+        rc = hipMemcpyAsync(a->p_out, a->d_out, (size_t)a->egress_budget,
+                            hipMemcpyDeviceToHost, s_out);
+        if (rc) return (int)rc;
+    } else {
+        hipLaunchKernelGGL(k_compact, dim3(cblocks), dim3(BLOCK_THREADS),
+                           0, s_out,
+                           (const uint8_t*)a->d_resp, tables, tables + n,
+                           (uint8_t*)a->p_out, n, a->rslot);
+        if ((rc = hipGetLastError())) return (int)rc;
+    }
     rc = hipMemcpyAsync(a->p_tables, a->d_tables, (size_t)(2 * n + 2) * 4,
                         hipMemcpyDeviceToHost, s_out);
     if (rc) return (int)rc;

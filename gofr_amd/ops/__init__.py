@@ -87,6 +87,8 @@ class GofrSubmitArgs(ctypes.Structure):
         ("gzip_min", ctypes.c_int),
         ("p_out", ctypes.c_void_p),
         ("n", ctypes.c_int), ("rslot", ctypes.c_int),
+        ("d_out", ctypes.c_void_p),
+        ("egress_budget", ctypes.c_longlong),
     ]
 
 
