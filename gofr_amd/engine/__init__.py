@@ -231,7 +231,7 @@ class BatchEngine:
         self.s_k = t.cuda.Stream(device=dev, priority=-1)
         self.s_k2 = t.cuda.Stream(device=dev, priority=-1)
         self.s_out = t.cuda.Stream(device=dev)
-        self.n_channels = int(os.environ.get("GOFR_CHANNELS", "2"))
+        self.n_channels = int(os.environ.get("GOFR_CHANNELS", "1"))
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot) for _ in range(self.pipeline)]
 
