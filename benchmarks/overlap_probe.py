@@ -131,5 +131,66 @@ def main():
           f"({gbs(ms_sgb, 2*N):.1f} GB/s agg)")
 
 
+
+
+def extra_mix_tests():
+    """SDMA-H2D + kernel-write mix; non-coherent-host D2H."""
+    import ctypes
+    dev = torch.device("cuda:0")
+    from gofr_amd import ops as O
+    hip = O.HipOps()
+    hip.lib.gofr_host_alloc.restype = ctypes.c_void_p
+    hip.lib.gofr_host_alloc.argtypes = [ctypes.c_longlong, ctypes.c_uint]
+    hip.lib.gofr_memcpy_async.restype = ctypes.c_int
+    hip.lib.gofr_memcpy_async.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p]
+    p_a = torch.zeros(N, dtype=torch.uint8).pin_memory()
+    p_b = torch.zeros(N, dtype=torch.uint8).pin_memory()
+    d_a = torch.zeros(N, dtype=torch.uint8, device=dev)
+    d_b = torch.zeros(N, dtype=torch.uint8, device=dev)
+    s1 = torch.cuda.Stream()
+    s2 = torch.cuda.Stream()
+    full = torch.full((NSLOTS,), RSLOT, dtype=torch.int32, device=dev)
+    offs = (torch.arange(NSLOTS, dtype=torch.int32, device=dev) * RSLOT)
+
+    def h2d():
+        with torch.cuda.stream(s1):
+            d_a.copy_(p_a, non_blocking=True)
+
+    def kwrite():
+        with torch.cuda.stream(s2):
+            hip.compact(s2.cuda_stream, d_b, full, offs, p_b, NSLOTS, RSLOT)
+
+    ms = t_ms(lambda: (h2d(), kwrite()))
+    print(f"mix SDMA-H2D + kernel-write-D2H duplex: {ms:.3f} ms "
+          f"({gbs(ms, 2*N):.1f} GB/s agg)")
+
+    # non-coherent host egress buffer
+    for name, flags in [("noncoherent", 0x80000000),
+                        ("coherent", 0x40000000),
+                        ("default", 0x0)]:
+        nc = hip.lib.gofr_host_alloc(N, flags)
+        if not nc:
+            print(f"hostalloc {name}: failed")
+            continue
+
+        def d2h_nc():
+            with torch.cuda.stream(s2):
+                rc = hip.lib.gofr_memcpy_async(
+                    nc, d_b.data_ptr(), N, 2, s2.cuda_stream)
+                assert rc == 0, rc
+
+        ms1 = t_ms(d2h_nc)
+        ms2 = t_ms(lambda: (h2d(), d2h_nc()))
+        print(f"D2H into {name:12s}: alone {ms1:.3f} ms "
+              f"({gbs(ms1):.1f} GB/s), +SDMA-H2D duplex {ms2:.3f} ms "
+              f"({gbs(ms2, 2*N):.1f} GB/s agg)")
+
+
 if __name__ == "__main__":
-    main()
+    import sys as _sys
+    if "--mix" in _sys.argv:
+        extra_mix_tests()
+    else:
+        main()

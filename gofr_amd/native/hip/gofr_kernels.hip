@@ -1675,6 +1675,20 @@ extern "C" const volatile unsigned long long* gofr_pump_done_ptr() {
 
 extern "C" int gofr_pump_err() { return g_pump.err; }
 
+// host-memory helpers (egress-ring allocation experiments: coherence
+// flags decide whether the runtime will use SDMA for D2H)
+extern "C" void* gofr_host_alloc(long long size, unsigned flags) {
+    void* p = nullptr;
+    if (hipHostMalloc(&p, (size_t)size, flags)) return nullptr;
+    return p;
+}
+
+extern "C" int gofr_memcpy_async(void* dst, const void* src, long long n,
+                                 int kind, void* stream) {
+    return (int)hipMemcpyAsync(dst, src, (size_t)n, (hipMemcpyKind)kind,
+                               (hipStream_t)stream);
+}
+
 extern "C" {
 
 int gofr_launch_parse_route(
