@@ -115,7 +115,8 @@ class _Lane:
     i+1 overlaps egress of batch i-1 and the kernels of batch i.
     """
 
-    def __init__(self, t, dev, nb, max_bytes, rslot, host_blob_cap=4 << 20):
+    def __init__(self, t, dev, nb, max_bytes, rslot, hip=None,
+                 host_blob_cap=4 << 20):
         self.stream = t.cuda.Stream(device=dev)  # legacy per-lane stream
         self.event = t.cuda.Event()
         self.e_in = t.cuda.Event()
@@ -137,7 +138,13 @@ class _Lane:
         self.p_reqs = t.empty(max_bytes, dtype=t.uint8).pin_memory()
         self.p_req_off = t.empty(nb + 1, dtype=t.int64).pin_memory()
         self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
-        self.p_out = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
+        # egress ring via hipHostMalloc: D2H SDMA works into it (torch
+        # pin_memory is hipHostRegister'd, which the runtime serves with
+        # a blit kernel instead — see ops.HipOps.host_alloc)
+        if hip is not None:
+            self.p_out = hip.host_alloc(nb * rslot)
+        else:
+            self.p_out = t.empty(nb * rslot, dtype=t.uint8).pin_memory()
         self.p_resp_len = t.empty(nb, dtype=t.int32).pin_memory()
         self.p_resp_off = t.empty(nb, dtype=t.int32).pin_memory()
         self.p_total = t.empty(1, dtype=t.int32).pin_memory()
@@ -150,7 +157,11 @@ class _Lane:
         # [0:n] resp_len, [n:2n] resp_off, [2n] total, [2n+1]
         # host_needed) — one D2H moves the whole per-batch result set
         self.d_tables = t.zeros(2 * nb + 2, dtype=t.int32, device=dev)
-        self.p_tables = t.zeros(2 * nb + 2, dtype=t.int32).pin_memory()
+        if hip is not None:
+            self.p_tables = hip.host_alloc((2 * nb + 2) * 4,
+                                           dtype=np.int32)
+        else:
+            self.p_tables = t.zeros(2 * nb + 2, dtype=t.int32).pin_memory()
         self.p_tables_np = self.p_tables.numpy()
         self.p_req_off_np = self.p_req_off.numpy()
         self.n = 0
@@ -233,7 +244,8 @@ class BatchEngine:
         self.s_out = t.cuda.Stream(device=dev)
         self.n_channels = int(os.environ.get("GOFR_CHANNELS", "1"))
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
-                            self.rslot) for _ in range(self.pipeline)]
+                            self.rslot, hip=self.hip)
+                      for _ in range(self.pipeline)]
 
     def _next_seed(self) -> int:
         self._seed = ops.splitmix64(self._seed)

@@ -144,6 +144,22 @@ class HipOps:
             [ctypes.POINTER(GofrSubmitArgs)]
         self.lib.gofr_pump_done_ptr.restype = ctypes.c_void_p
         self.lib.gofr_pump_err.restype = ctypes.c_int
+        self.lib.gofr_host_alloc.restype = ctypes.c_void_p
+        self.lib.gofr_host_alloc.argtypes = [ctypes.c_longlong,
+                                             ctypes.c_uint]
+
+    def host_alloc(self, nbytes: int, dtype=np.uint8):
+        """hipHostMalloc'd host memory as a torch CPU tensor. Unlike
+        torch's pin_memory (hipHostRegister'd), the runtime serves D2H
+        SDMA into this memory — registered destinations fall back to a
+        CU-hungry blit kernel (measured: benchmarks/overlap_probe.py
+        --mix: 96.6 GB/s duplex vs ~65 engine-observed with blit)."""
+        import torch
+        addr = self.lib.gofr_host_alloc(int(nbytes), 0)
+        if not addr:
+            raise MemoryError(f"gofr_host_alloc({nbytes}) failed")
+        buf = (ctypes.c_ubyte * int(nbytes)).from_address(addr)
+        return torch.from_numpy(np.frombuffer(buf, dtype=dtype))
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,
