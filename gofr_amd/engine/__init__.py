@@ -390,14 +390,24 @@ class BatchEngine:
             ev.record(self.s_k)
         self.s_k.synchronize()
         a = ops.GofrSubmitArgs()
-        if self.n_channels >= 2:
+        kmode = os.environ.get("GOFR_KMODE", "staged")
+        if kmode == "chan2":
+            # shared SDMA ingress stream; kernels AND link-bound egress
+            # sweep fused on one of two channel streams per lane parity
+            # (channels overlap each other; no egress-stage handoff)
+            chan = (self.s_k, self.s_k2)[lane_idx % 2]
+            a.s_in = self.s_in.cuda_stream
+            a.s_k = chan.cuda_stream
+            a.s_out = chan.cuda_stream
+        elif self.n_channels >= 2:
             chan = (self.s_k, self.s_k2)[lane_idx % 2]
             a.s_in = chan.cuda_stream
             a.s_k = chan.cuda_stream
+            a.s_out = self.s_out.cuda_stream
         else:
             a.s_in = self.s_in.cuda_stream
             a.s_k = self.s_k.cuda_stream
-        a.s_out = self.s_out.cuda_stream
+            a.s_out = self.s_out.cuda_stream
         a.ev_in = ln.e_in.cuda_event
         a.ev_k = ln.e_k.cuda_event
         a.ev_done = ln.event.cuda_event
