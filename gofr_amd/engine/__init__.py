@@ -445,7 +445,7 @@ class BatchEngine:
         a.n = n
         a.rslot = self.rslot
         a.d_out = ln.d_out.data_ptr()
-        if os.environ.get("GOFR_EGRESS", "sdma") == "kernel":
+        if os.environ.get("GOFR_EGRESS", "kernel") == "kernel":
             a.egress_budget = 0
         else:
             # budget-sized D2H via the runtime blit (measured the
