@@ -1523,8 +1523,8 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
         static int cap = 0;
         if (cap == 0) {
             const char* e = getenv("GOFR_COMPACT_BLOCKS");
-            cap = e ? atoi(e) : 128;
-            if (cap <= 0) cap = 128;
+            cap = e ? atoi(e) : 64;
+            if (cap <= 0) cap = 64;
         }
         if (cblocks > cap) cblocks = cap;
     }
@@ -1804,8 +1804,8 @@ int gofr_launch_compact(
     static int cap = 0;
     if (cap == 0) {
         const char* e = getenv("GOFR_COMPACT_BLOCKS");
-        cap = e ? atoi(e) : 128;
-        if (cap <= 0) cap = 128;
+        cap = e ? atoi(e) : 64;
+        if (cap <= 0) cap = 64;
     }
     int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     if (blocks > cap) blocks = cap;
