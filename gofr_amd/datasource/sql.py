@@ -56,7 +56,19 @@ class Tx:
         return self._db.Query(query, *args)
 
     def Exec(self, query: str, *args):
-        return self._db.Exec(query, *args)
+        # unlike DB.Exec, no autocommit — the tx owns commit/rollback
+        # (reference: db.go:70-117 Tx ops run inside the open tx)
+        t0 = time.perf_counter_ns()
+        db = self._db
+        with db._lock:
+            cur = db._conn.cursor()
+            try:
+                cur.execute(db._fix_params(query), args)
+                result = (cur.rowcount, getattr(cur, "lastrowid", None))
+            finally:
+                cur.close()
+        db._log("Exec", query, t0)
+        return result
 
     def Commit(self):
         t0 = time.perf_counter_ns()
