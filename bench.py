@@ -378,9 +378,12 @@ def main():
         app.enable_auth(b"bench-secret")
     if "gzip" in mw:
         app.enable_gzip(min_size=256)
-    device = f"cuda:{local_rank}" if have_gpu else "cpu"
     if have_gpu:
+        # map ranks onto the devices that exist (lets a world-2 smoke
+        # run on a 1-GPU box; on the 8-GPU node it is the identity)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
     eng = BatchEngine(app, device=device, slot=2048, max_batch=batch,
                       pipeline=(int(os.environ.get("GOFR_PIPELINE", "5"))
                                 if (have_gpu and world == 1) else 1))
