@@ -384,10 +384,6 @@ def main():
         local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     device = f"cuda:{local_rank}" if have_gpu else "cpu"
-    eng = BatchEngine(app, device=device, slot=2048, max_batch=batch,
-                      pipeline=(int(os.environ.get("GOFR_PIPELINE", "5"))
-                                if (have_gpu and world == 1) else 1))
-
     extra = b""
     if "auth" in mw:
         from gofr_amd.http.middleware import hmac_token
@@ -399,6 +395,13 @@ def main():
     if extra:
         raw = raw.replace(b"\r\n\r\n", b"\r\n" + extra + b"\r\n", 1)
     payloads = [raw] * batch
+    # size the request slot to the workload (the slot bounds the
+    # largest request; for N>1 it is also the fixed all-to-all exchange
+    # granularity, so tighter slots mean fewer xGMI + host-link bytes)
+    slot = max(1024, ((len(raw) + 255) // 256) * 256 + 256)
+    eng = BatchEngine(app, device=device, slot=slot, max_batch=batch,
+                      pipeline=(int(os.environ.get("GOFR_PIPELINE", "5"))
+                                if (have_gpu and world == 1) else 1))
 
     n_grpc = int(batch * args.grpc_frac)
     if n_grpc and world == 1:
