@@ -1244,6 +1244,38 @@ k_varint_spans(const uint8_t* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
+// MFMA i8 probe — one v_mfma_i32_16x16x64_i8 with both operands loaded
+// from global using the assumed CDNA4 fragment mapping (A: lane l holds
+// A[row=l&15][k=(l>>4)*16 .. +15]; B: lane l holds the matching
+// B[k][col=l&15] K-slice; D: col=lane&15, row=(lane>>4)*4+reg —
+// cdna_hip_programming.md §"Fragment layout"). tests/test_gpu_engine.py
+// asserts D == numpy A@B, pinning the layout the ETag hash builds on.
+// ---------------------------------------------------------------------------
+typedef int v4i_t __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ void __launch_bounds__(64)
+k_mfma_i8_probe(const int8_t* __restrict__ A,   // [16][64]
+                const int8_t* __restrict__ B,   // [64][16]
+                int32_t* __restrict__ D) {      // [16][16]
+    const int lane = threadIdx.x & 63;
+    const int row = lane & 15;
+    const int k0 = (lane >> 4) * 16;
+    int8_t a_bytes[16], b_bytes[16];
+    for (int j = 0; j < 16; ++j) {
+        a_bytes[j] = A[row * 64 + (k0 + j)];
+        b_bytes[j] = B[(k0 + j) * 16 + row];  // col == lane&15 == row var
+    }
+    v4i_t av, bv, cv = {0, 0, 0, 0};
+    __builtin_memcpy(&av, a_bytes, 16);
+    __builtin_memcpy(&bv, b_bytes, 16);
+    v4i_t dv = __builtin_amdgcn_mfma_i32_16x16x64_i8(av, bv, cv, 0, 0, 0);
+    const int col = lane & 15;
+    const int row_d = (lane >> 4) * 4;
+    for (int r = 0; r < 4; ++r)
+        D[(row_d + r) * 16 + col] = dv[r];
+}
+
+// ---------------------------------------------------------------------------
 // k_padscan — exclusive cumsum of round16(resp_len) (the compact-stream
 // offsets) + total + host_needed mirror, in ONE single-workgroup kernel.
 // Replaces the host-side torch chain (elementwise pad + rocprim scan +
@@ -1817,3 +1849,11 @@ int gofr_launch_compact(
 }
 
 }  // extern "C"
+
+extern "C" int gofr_launch_mfma_probe(void* stream, const void* A,
+                                      const void* B, void* D) {
+    hipLaunchKernelGGL(k_mfma_i8_probe, dim3(1), dim3(64), 0,
+                       (hipStream_t)stream,
+                       (const int8_t*)A, (const int8_t*)B, (int32_t*)D);
+    return (int)hipGetLastError();
+}
