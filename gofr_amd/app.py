@@ -139,6 +139,13 @@ class App:
         GPU engine: k_auth kernel; CPU transport: checked in dispatch."""
         self.auth_secret = bytes(secret)
 
+    def enable_etag(self):
+        """ETag middleware: every response carries a strong validator
+        computed by the MFMA batched body hash (k_respond's
+        mfma_etag_wave; model: ops.etag_u32)."""
+        self.etag_on = True
+        return self
+
     def enable_gzip(self, min_size: int = 256):
         """gzip-compress JSON responses when the request advertises
         Accept-Encoding: gzip and the body is >= min_size bytes."""

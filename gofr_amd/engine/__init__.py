@@ -293,7 +293,8 @@ class BatchEngine:
             buf, offs, fields, self.rslot, self.program.handler_tab,
             self.program.blob, host_blob, host_tab, seed,
             auth_env=self.program.auth_env,
-            gzip_min=self.app.gzip_min_size or 0)
+            gzip_min=self.app.gzip_min_size or 0,
+            etag_on=getattr(self.app, "etag_on", False))
         # compaction mirror (same round16 layout as k_compact)
         n = len(lens)
         pads = (resp_len + 15) & ~15
@@ -441,6 +442,7 @@ class BatchEngine:
         a.host_tab = ln.d_host_tab.data_ptr()
         a.auth_env_off, a.auth_env_len = self.program.auth_env
         a.gzip_min = self.app.gzip_min_size or 0
+        a.etag_on = 1 if getattr(self.app, "etag_on", False) else 0
         a.p_out = ln.p_out.data_ptr()
         a.n = n
         a.rslot = self.rslot
@@ -478,7 +480,9 @@ class BatchEngine:
                          self.d_blob, ln.d_host_blob, ln.d_host_tab,
                          ln.d_req_off[n:n + 1],  # seed (offsets tail)
                          auth_env=self.program.auth_env,
-                         gzip_min=self.app.gzip_min_size or 0)
+                         gzip_min=self.app.gzip_min_size or 0,
+                         etag_on=1 if getattr(self.app, "etag_on", False)
+                         else 0)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)

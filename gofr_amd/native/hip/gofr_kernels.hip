@@ -79,6 +79,8 @@
 #define N_METHODS_PAD 8
 #define MAX_PARAMS 4
 
+typedef int v4i_t __attribute__((ext_vector_type(4)));
+
 struct TrieDev {
     const uint8_t* seg_blob;
     const int32_t* node_child_first;
@@ -677,6 +679,55 @@ __device__ int deflate_gzip_wave(const uint8_t* src, int len, uint8_t* dst,
 }
 
 __constant__ char HDR_CE[] = "Content-Encoding: gzip\r\n";
+__constant__ char HDR_ETAG[] = "ETag: \"";
+#define ETAG_HDR_LEN 18  /* ETag: "xxxxxxxx"\r\n */
+
+// ---------------------------------------------------------------------------
+// MFMA batched body hash (the ETag middleware). The response body is
+// treated as a [16 x 64]-byte tile stream and each tile is one
+// v_mfma_i32_16x16x64_i8 against a fixed i8 coefficient matrix — the
+// matrix cores do the batched byte-dot work (BASELINE north star:
+// MFMA for the batched byte-compare/hash; fragment layout validated by
+// k_mfma_i8_probe + tests). All arithmetic is uint32 wraparound; the
+// byte-exact model is gofr_amd/ops etag_u32().
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int8_t etag_bcoef(int k, int c) {
+    return (int8_t)(((k * 17 + c * 29 + 3) % 251) - 125);
+}
+
+__device__ uint32_t mfma_etag_wave(const uint8_t* body, int len, int lane) {
+    const int row = lane & 15;
+    const int k0 = (lane >> 4) * 16;
+    int8_t bb[16];
+    for (int j = 0; j < 16; ++j) bb[j] = etag_bcoef(k0 + j, row);
+    v4i_t bv;
+    __builtin_memcpy(&bv, bb, 16);
+    uint32_t state[4] = {0, 0, 0, 0};
+    const int ntiles = (len + 1023) / 1024;
+    for (int t = 0; t < ntiles; ++t) {
+        int8_t ab[16];
+        const int base = t * 1024 + row * 64 + k0;
+        for (int j = 0; j < 16; ++j) {
+            const int idx = base + j;
+            ab[j] = (idx < len) ? (int8_t)body[idx] : 0;
+        }
+        v4i_t av;
+        __builtin_memcpy(&av, ab, 16);
+        v4i_t z = {0, 0, 0, 0};
+        v4i_t dv = __builtin_amdgcn_mfma_i32_16x16x64_i8(av, bv, z,
+                                                         0, 0, 0);
+        for (int r2 = 0; r2 < 4; ++r2)
+            state[r2] = state[r2] * 33u + (uint32_t)dv[r2];
+    }
+    const int col = lane & 15;
+    const int rowd = (lane >> 4) * 4;
+    uint32_t h = 0;
+    for (int r2 = 0; r2 < 4; ++r2)
+        h ^= state[r2] * (uint32_t)(((rowd + r2) * 16 + col) * 2 + 1);
+    for (int off = 32; off; off >>= 1)
+        h ^= (uint32_t)__shfl_xor((int)h, off);
+    return h ^ (uint32_t)len;
+}
 
 // host result table row: [off, len, status, ct_id]
 //
@@ -699,8 +750,8 @@ __device__ __forceinline__ void respond_impl(
         const uint8_t* __restrict__ host_blob,
         const int32_t* __restrict__ host_tab,
         uint64_t seed, int auth_env_off, int auth_env_len,
-        int gzip_min, uint8_t* obuf, uint8_t* plainbuf, uint32_t* hash,
-        const uint32_t* crc_tab, int req, int lane) {
+        int gzip_min, int etag_on, uint8_t* obuf, uint8_t* plainbuf,
+        uint32_t* hash, const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
     const uint8_t* rbase = reqs + req_off[req];
@@ -855,7 +906,9 @@ __device__ __forceinline__ void respond_impl(
     const int o_reason = o_status + 3 + 1;        // "xxx "
     const int o_ct = o_reason + rlen_reason + 2;  // reason \r\n
     const int o_ce = o_ct + ct_len;
-    const int o_cors = o_ce + ce_len;
+    const int et_len = etag_on ? ETAG_HDR_LEN : 0;
+    const int o_etag = o_ce + ce_len;
+    const int o_cors = o_etag + et_len;
     const int o_corr = o_cors + (int)sizeof(HDR_CORS) - 1;
     const int o_corrhex = o_corr + (int)sizeof(HDR_CORR) - 1;
     const int o_cl = o_corrhex + 32 + 2;
@@ -868,6 +921,13 @@ __device__ __forceinline__ void respond_impl(
     if (lane < (unsigned)rlen_reason) obuf[o_reason + lane] = reason[lane];
     if (lane < (unsigned)ct_len) obuf[o_ct + lane] = ct_str[lane];
     if (content_enc && lane < ce_len) obuf[o_ce + lane] = HDR_CE[lane];
+    if (etag_on && lane < 7) obuf[o_etag + lane] = HDR_ETAG[lane];
+    if (etag_on && lane == 0) {
+        obuf[o_etag + 15] = '"';
+        obuf[o_etag + 16] = '\r';
+        obuf[o_etag + 17] = '\n';
+        // hex digits patched after the body lands in LDS
+    }
     {
         const int cors_len = (int)sizeof(HDR_CORS) - 1;
         for (int i = lane; i < cors_len; i += WAVE)
@@ -920,6 +980,14 @@ __device__ __forceinline__ void respond_impl(
             obuf[body_start + i] = body_src[i];
     }
 
+    if (etag_on) {
+        // MFMA hash of the final body bytes -> patch the reserved hex
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        const uint32_t h = mfma_etag_wave(obuf + hl, body_total, lane);
+        if (lane < 8)
+            obuf[o_etag + 7 + lane] = HEXD[(h >> (28 - 4 * lane)) & 0xF];
+    }
+
     const int total = hl + body_total;
     // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -949,14 +1017,14 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ host_blob,
           const int32_t* __restrict__ host_tab,
           const uint64_t* __restrict__ seed_ptr,
-          int auth_env_off, int auth_env_len) {
+          int auth_env_off, int auth_env_len, int etag_on) {
     __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
     const int wv = threadIdx.x / WAVE;
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
     if (req >= n) return;
     respond_impl<false>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                         handler_tab, n_routes, blob, host_blob, host_tab,
-                        *seed_ptr, auth_env_off, auth_env_len, 0,
+                        *seed_ptr, auth_env_off, auth_env_len, 0, etag_on,
                         obuf_all + wv * MAX_SLOT, nullptr, nullptr, nullptr,
                         req, lane_id());
 }
@@ -973,7 +1041,8 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
              const uint8_t* __restrict__ host_blob,
              const int32_t* __restrict__ host_tab,
              const uint64_t* __restrict__ seed_ptr,
-             int auth_env_off, int auth_env_len, int gzip_min) {
+             int auth_env_off, int auth_env_len, int gzip_min,
+             int etag_on) {
     // single __shared__ block (cdna guide §5 trap 4a)
     __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
                            WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4];
@@ -996,6 +1065,7 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
     respond_impl<true>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                        handler_tab, n_routes, blob, host_blob, host_tab,
                        *seed_ptr, auth_env_off, auth_env_len, gzip_min,
+                       etag_on,
                        obuf_all + wv * MAX_SLOT, plain_all + wv * MAX_SLOT,
                        hash_all + wv * GZ_HASH_SIZE, crc_tab,
                        req, lane_id());
@@ -1251,7 +1321,6 @@ k_varint_spans(const uint8_t* __restrict__ buf,
 // cdna_hip_programming.md §"Fragment layout"). tests/test_gpu_engine.py
 // asserts D == numpy A@B, pinning the layout the ETag hash builds on.
 // ---------------------------------------------------------------------------
-typedef int v4i_t __attribute__((ext_vector_type(4)));
 
 extern "C" __global__ void __launch_bounds__(64)
 k_mfma_i8_probe(const int8_t* __restrict__ A,   // [16][64]
@@ -1433,7 +1502,7 @@ typedef struct {
     void* d_tables;                        // int32[2n+2]
     void* p_tables;                        // pinned int32[2n+2]
     const void* blob; const void* host_blob; const void* host_tab;
-    int auth_env_off; int auth_env_len; int gzip_min;
+    int auth_env_off; int auth_env_len; int gzip_min; int etag_on;
     void* p_out;                           // pinned egress ring
     int n; int rslot;
     // egress mode: budget > 0 -> k_compact into d_out (HBM) + one
@@ -1526,7 +1595,8 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->blob,
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
-                           a->auth_env_off, a->auth_env_len, a->gzip_min);
+                           a->auth_env_off, a->auth_env_len, a->gzip_min,
+                           a->etag_on);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS),
                            0, s_k,
@@ -1538,7 +1608,7 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->blob,
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
-                           a->auth_env_off, a->auth_env_len);
+                           a->auth_env_off, a->auth_env_len, a->etag_on);
     }
     if ((rc = hipGetLastError())) return (int)rc;
     hipLaunchKernelGGL(k_padscan, dim3(1), dim3(SCAN_THREADS), 0, s_k,
@@ -1757,7 +1827,7 @@ int gofr_launch_respond(
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
         const void* seed_ptr, int auth_env_off, int auth_env_len,
-        int gzip_min) {
+        int gzip_min, int etag_on) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     if (gzip_min > 0) {
         hipLaunchKernelGGL(k_respond_gz, dim3(blocks), dim3(BLOCK_THREADS),
@@ -1770,7 +1840,7 @@ int gofr_launch_respond(
                            (const uint8_t*)blob, (const uint8_t*)host_blob,
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
-                           auth_env_off, auth_env_len, gzip_min);
+                           auth_env_off, auth_env_len, gzip_min, etag_on);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                            (hipStream_t)stream,
@@ -1782,7 +1852,7 @@ int gofr_launch_respond(
                            (const uint8_t*)blob, (const uint8_t*)host_blob,
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
-                           auth_env_off, auth_env_len);
+                           auth_env_off, auth_env_len, etag_on);
     }
     return (int)hipGetLastError();
 }

@@ -84,7 +84,7 @@ class GofrSubmitArgs(ctypes.Structure):
         ("blob", ctypes.c_void_p), ("host_blob", ctypes.c_void_p),
         ("host_tab", ctypes.c_void_p),
         ("auth_env_off", ctypes.c_int), ("auth_env_len", ctypes.c_int),
-        ("gzip_min", ctypes.c_int),
+        ("gzip_min", ctypes.c_int), ("etag_on", ctypes.c_int),
         ("p_out", ctypes.c_void_p),
         ("n", ctypes.c_int), ("rslot", ctypes.c_int),
         ("d_out", ctypes.c_void_p),
@@ -121,7 +121,8 @@ class HipOps:
             ctypes.c_int, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-            ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int]
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int]
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
@@ -188,7 +189,7 @@ class HipOps:
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
                 host_blob_t, host_tab_t, seed_t, auth_env=(0, 0),
-                gzip_min=0):
+                gzip_min=0, etag_on=0):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -202,7 +203,7 @@ class HipOps:
             ctypes.c_void_p(host_blob_t.data_ptr()),
             ctypes.c_void_p(host_tab_t.data_ptr()),
             ctypes.c_void_p(seed_t.data_ptr()),
-            auth_env[0], auth_env[1], gzip_min)
+            auth_env[0], auth_env[1], gzip_min, etag_on)
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
 
@@ -491,11 +492,34 @@ def _json_body_valid(body: bytes) -> bool:
     return depth == 0 and not in_str and len(body) > 0
 
 
+_ETAG_B = ((np.arange(64)[:, None] * 17 + np.arange(16)[None, :] * 29
+            + 3) % 251 - 125).astype(np.int8)
+_ETAG_SALT = (np.arange(256, dtype=np.uint32) * 2 + 1).reshape(16, 16)
+
+
+def etag_u32(body: bytes) -> int:
+    """Mirror of mfma_etag_wave (the MFMA batched body hash): tiles of
+    16x64 i8 against the fixed coefficient matrix, uint32 fold."""
+    n = len(body)
+    pad = (-n) % 1024
+    a = np.frombuffer(body + b"\x00" * pad, np.uint8).astype(np.int8)
+    tiles = a.reshape(-1, 16, 64)
+    state = np.zeros((16, 16), np.uint32)
+    for t in range(tiles.shape[0]):
+        d = (tiles[t].astype(np.int32) @ _ETAG_B.astype(np.int32))
+        state = state * np.uint32(33) + d.astype(np.uint32)
+    h = np.uint32(0)
+    for v in (state * _ETAG_SALT).reshape(-1):
+        h ^= v
+    return int(h ^ np.uint32(n))
+
+
 def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 fields: np.ndarray, rslot: int,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
-                seed: int, auth_env=(0, 0), gzip_min: int = 0):
+                seed: int, auth_env=(0, 0), gzip_min: int = 0,
+                etag_on: bool = False):
     """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
     n = len(fields)
     host_tab = np.asarray(host_tab, np.int32).reshape(-1)
@@ -560,9 +584,15 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
         h1 = splitmix64(seed ^ r)
         h2 = splitmix64(h1 ^ 0xD1B54A32D192ED03)
         corr = f"{h1:016x}{h2:016x}".encode()
+        final_body = (b'{"data":' + body_src + b"}") if env else body_src
+        etag_hdr = b""
+        if etag_on:
+            etag_hdr = b'ETag: "' + f"{etag_u32(final_body):08x}".encode() \
+                + b'"\r\n'
         head = (b"HTTP/1.1 " + f"{status:03d}".encode() + b" " + reason +
                 b"\r\n" + _CT_STRS[ct_id] +
                 (b"Content-Encoding: gzip\r\n" if content_enc else b"") +
+                etag_hdr +
                 _CORS +
                 b"X-Correlation-ID: " + corr + b"\r\n" +
                 b"Content-Length: " + str(body_total).encode() + b"\r\n" +

@@ -183,3 +183,47 @@ def test_connstate_device_matches_cpu():
                      torch.from_numpy(bout).to("cuda"), batch_no=3)
     cpu.record_batch(ids_c, bin_, bout, batch_no=3)
     assert (gpu.stats(ids) == cpu.stats(ids_c)).all()
+
+
+def test_etag_gpu_matches_mirror_bytes():
+    """k_respond's MFMA ETag vs the numpy model, byte-for-byte, across
+    echo/static/404 and multi-tile bodies."""
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.enable_etag()
+    gpu = BatchEngine(app, device="cuda", max_batch=2048)
+    cpu = BatchEngine(app, device="cpu", max_batch=2048)
+    cpu._seed = gpu._seed
+    raws = []
+    for i in range(512):
+        if i % 3 == 0:
+            body = b'{"k":"' + b"x" * (i % 1900) + b'"}'
+            raws.append(http_req("POST", "/echo", body=body))
+        elif i % 3 == 1:
+            raws.append(http_req("GET", "/greet"))
+        else:
+            raws.append(http_req("GET", "/nope"))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, f"req {i}\nGPU {go[:200]!r}\nCPU {co[:200]!r}"
+    assert b'ETag: "' in g[0]
+
+
+def test_etag_with_gzip_gpu_matches_mirror():
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.enable_etag()
+    app.enable_gzip(min_size=64)
+    gpu = BatchEngine(app, device="cuda", max_batch=512)
+    cpu = BatchEngine(app, device="cpu", max_batch=512)
+    cpu._seed = gpu._seed
+    body = b'{"payload":"' + b"a" * 900 + b'"}'
+    raws = [http_req("POST", "/echo", body=body,
+                     headers={"Accept-Encoding": "gzip"})] * 128
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, f"req {i}\nGPU {go[:200]!r}\nCPU {co[:200]!r}"
+    assert b"Content-Encoding: gzip" in g[0] and b'ETag: "' in g[0]

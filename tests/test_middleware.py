@@ -232,3 +232,50 @@ def test_engine_gzip_fused():
     head3, _, rbody3 = outs[2].partition(b"\r\n\r\n")
     assert b"Content-Encoding: gzip" in head3
     assert json.loads(gzip.decompress(rbody3)) == {"data": {"pad": "y" * 500}}
+
+
+def test_etag_middleware_cpu_mirror():
+    """MFMA ETag middleware: header present, hash matches the model,
+    stable across identical bodies, distinct across different ones."""
+    import gofr_amd
+    from gofr_amd import handlers, ops
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.enable_etag()
+    eng = BatchEngine(app, device="cpu", max_batch=64)
+
+    def req(body):
+        return (b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+                b"Content-Type: application/json\r\n"
+                b"Content-Length: " + str(len(body)).encode() +
+                b"\r\n\r\n" + body)
+
+    b1 = b'{"k":"v1"}'
+    b2 = b'{"k":"v2"}'
+    outs = eng.process([req(b1), req(b1), req(b2)])
+    tags = []
+    for out in outs:
+        head = out.split(b"\r\n\r\n", 1)[0].decode()
+        line = [h for h in head.split("\r\n")
+                if h.startswith("ETag: ")][0]
+        tags.append(line.split('"')[1])
+    assert tags[0] == tags[1] and tags[0] != tags[2]
+    # the header value IS the model hash of the final body
+    final = b'{"data":' + b1 + b"}"
+    assert tags[0] == f"{ops.etag_u32(final):08x}"
+    # 404 (catch-all static) also carries an ETag
+    out404 = eng.process([b"GET /nope HTTP/1.1\r\nHost: h\r\n\r\n"])[0]
+    assert b"ETag: \"" in out404
+
+
+def test_etag_u32_properties():
+    from gofr_amd import ops
+    assert ops.etag_u32(b"") != ops.etag_u32(b"\x00")  # length folded in
+    assert ops.etag_u32(b"abc") == ops.etag_u32(b"abc")
+    long = bytes(range(256)) * 17  # multi-tile, bytes >= 128 (signed i8)
+    assert isinstance(ops.etag_u32(long), int)
+    assert ops.etag_u32(long) != ops.etag_u32(long[:-1])
