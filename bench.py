@@ -378,6 +378,8 @@ def main():
         app.enable_auth(b"bench-secret")
     if "gzip" in mw:
         app.enable_gzip(min_size=256)
+    if "etag" in mw:
+        app.enable_etag()
     if have_gpu:
         # map ranks onto the devices that exist (lets a world-2 smoke
         # run on a 1-GPU box; on the 8-GPU node it is the identity)
