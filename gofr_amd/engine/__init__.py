@@ -164,6 +164,8 @@ class _Lane:
             self.p_tables = t.zeros(2 * nb + 2, dtype=t.int32).pin_memory()
         self.p_tables_np = self.p_tables.numpy()
         self.p_req_off_np = self.p_req_off.numpy()
+        self.p_serial = t.zeros(1, dtype=t.int64).pin_memory()
+        self.p_serial_np = self.p_serial.numpy()
         self.n = 0
         self.nbytes = 0
         self.graph_key = None
@@ -242,6 +244,10 @@ class BatchEngine:
         self.s_k = t.cuda.Stream(device=dev, priority=-1)
         self.s_k2 = t.cuda.Stream(device=dev, priority=-1)
         self.s_out = t.cuda.Stream(device=dev)
+        # SDMA-flag gate state (GOFR_GATE=1): monotonic batch serial
+        self.d_flag = t.zeros(1, dtype=t.int64, device=dev)
+        self._gate_serial = 0
+        self._use_gate = os.environ.get("GOFR_GATE", "0") == "1"
         self.n_channels = int(os.environ.get("GOFR_CHANNELS", "1"))
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot, hip=self.hip)
@@ -328,6 +334,10 @@ class BatchEngine:
         if ln.c_args is not None and ln.graph_key == (n, nbytes):
             ln.mode = "c"
             ln.p_req_off_np[n] = signed  # seed rides the offsets tail
+            if self._use_gate:
+                self._gate_serial += 1
+                ln.p_serial_np[0] = self._gate_serial
+                ln.c_args.serial = self._gate_serial
             if self._use_pump:
                 serial = self.hip.lib.gofr_pump_submit(
                     ctypes.byref(ln.c_args))
@@ -443,6 +453,13 @@ class BatchEngine:
         a.auth_env_off, a.auth_env_len = self.program.auth_env
         a.gzip_min = self.app.gzip_min_size or 0
         a.etag_on = 1 if getattr(self.app, "etag_on", False) else 0
+        if self._use_gate:
+            a.d_flag = self.d_flag.data_ptr()
+            a.p_serial = ln.p_serial.data_ptr()
+        else:
+            a.d_flag = 0
+            a.p_serial = 0
+        a.serial = 0
         a.p_out = ln.p_out.data_ptr()
         a.n = n
         a.rslot = self.rslot

@@ -1511,7 +1511,31 @@ typedef struct {
     // the pinned ring directly (kernel-driven host writes).
     void* d_out;
     long long egress_budget;
+    // SDMA-flag gate (optional): when d_flag != 0, the kernel stage is
+    // released by k_gate spinning on d_flag instead of an event wait
+    void* d_flag;
+    const void* p_serial;        // pinned cell holding `serial`
+    unsigned long long serial;
 } GofrSubmitArgs;
+
+// k_gate — single-wave stream gate: spins (system-scope acquire
+// loads) until the ingress SDMA writes the batch serial to d_flag
+// (enqueued AFTER the payload copies, so FIFO order implies the
+// payload landed). Replaces hipStreamWaitEvent on an SDMA-recorded
+// event, whose signal delivery costs ~0.3 ms serialized in the kernel
+// stream's FIFO; the gate releases within the SDMA's own completion.
+// Timeout guard: gives up after ~0.5 s so a lost flag can never hang
+// the stream (the serving loop surfaces the stall as a bad batch).
+extern "C" __global__ void k_gate(const unsigned long long* flag,
+                                  unsigned long long serial) {
+    if (threadIdx.x != 0) return;
+    for (long i = 0; i < (1L << 19); ++i) {
+        const unsigned long long v = __hip_atomic_load(
+            flag, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+        if (v >= serial) return;
+        __builtin_amdgcn_s_sleep(32);
+    }
+}
 
 // enqueue-time breakdown (µs, cumulative): [0] big H2D, [1] rest of
 // ingress, [2] kernel stage, [3] egress; [4] = calls
@@ -1551,12 +1575,26 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     rc = hipMemcpyAsync(a->d_len, a->p_len, (size_t)n * 4,
                         hipMemcpyHostToDevice, s_in);
     if (rc) return (int)rc;
-    rc = hipEventRecord((hipEvent_t)a->ev_in, s_in);
-    if (rc) return (int)rc;
+    if (a->d_flag) {
+        // serial lands AFTER the payload (same SDMA FIFO)
+        rc = hipMemcpyAsync(a->d_flag, a->p_serial, 8,
+                            hipMemcpyHostToDevice, s_in);
+        if (rc) return (int)rc;
+    } else {
+        rc = hipEventRecord((hipEvent_t)a->ev_in, s_in);
+        if (rc) return (int)rc;
+    }
     g_submit_us[1] += now_us() - t0; t0 = now_us();
     // ---- kernel stage -----------------------------------------------------
-    rc = hipStreamWaitEvent(s_k, (hipEvent_t)a->ev_in, 0);
-    if (rc) return (int)rc;
+    if (a->d_flag) {
+        hipLaunchKernelGGL(k_gate, dim3(1), dim3(64), 0, s_k,
+                           (const unsigned long long*)a->d_flag,
+                           a->serial);
+        if ((rc = hipGetLastError())) return (int)rc;
+    } else {
+        rc = hipStreamWaitEvent(s_k, (hipEvent_t)a->ev_in, 0);
+        if (rc) return (int)rc;
+    }
     rc = hipMemsetAsync(a->d_host_needed, 0, 4, s_k);
     if (rc) return (int)rc;
     TrieDev trie{(const uint8_t*)a->trie[0], (const int32_t*)a->trie[1],

@@ -89,6 +89,9 @@ class GofrSubmitArgs(ctypes.Structure):
         ("n", ctypes.c_int), ("rslot", ctypes.c_int),
         ("d_out", ctypes.c_void_p),
         ("egress_budget", ctypes.c_longlong),
+        ("d_flag", ctypes.c_void_p),
+        ("p_serial", ctypes.c_void_p),
+        ("serial", ctypes.c_uint64),
     ]
 
 
