@@ -49,7 +49,9 @@
 #define FI_AUTH_LEN 14
 #define FI_RESP_OFF 15
 #define FI_PARAM0 16 /* 4 x (off, len) pairs: 16..23 */
-#define NF 24
+#define FI_INM_OFF 24  /* If-None-Match header value span */
+#define FI_INM_LEN 25
+#define NF 26
 
 // flags
 #define FL_ERR_PARSE 1
@@ -232,6 +234,7 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     int clen = 0;
     bool keep_alive = true;  // HTTP/1.1 default
     int body_off = len, auth_off = 0, auth_len = 0;
+    int inm_off = 0, inm_len = 0;
     int prev_lf = lf1;
     for (int guard = 0; guard < 64; ++guard) {
         const int ls = prev_lf + 1;           // line start
@@ -263,6 +266,8 @@ k_parse_route(const uint8_t* __restrict__ reqs,
                 flags |= FL_JSON_CT;
         } else if (nlen == 13 && ieq(nm, "authorization", 13)) {
             auth_off = vs; auth_len = vlen;
+        } else if (nlen == 13 && ieq(nm, "if-none-match", 13)) {
+            inm_off = vs; inm_len = vlen;
         } else if (nlen == 17 && ieq(nm, "transfer-encoding", 17)) {
             flags |= FL_NEEDS_HOST;  // chunked -> host slow path
         } else if (nlen == 15 && ieq(nm, "accept-encoding", 15)) {
@@ -287,6 +292,8 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     F[FI_CLEN] = clen;
     F[FI_AUTH_OFF] = auth_off;
     F[FI_AUTH_LEN] = auth_len;
+    F[FI_INM_OFF] = inm_off;
+    F[FI_INM_LEN] = inm_len;
 
     // ---- route match: trie walk (gofr_amd/http/router.py compile layout) --
     int node = 0;
@@ -980,15 +987,41 @@ __device__ __forceinline__ void respond_impl(
             obuf[body_start + i] = body_src[i];
     }
 
+    int send_body = body_total;
     if (etag_on) {
         // MFMA hash of the final body bytes -> patch the reserved hex
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         const uint32_t h = mfma_etag_wave(obuf + hl, body_total, lane);
         if (lane < 8)
             obuf[o_etag + 7 + lane] = HEXD[(h >> (28 - 4 * lane)) & 0xF];
+        // If-None-Match -> 304: in-place rewrite (status digits + CL
+        // digits zeroed, body dropped) so no header offset moves.
+        // Conscious quirks (documented): the reason phrase stays the
+        // 200 one (clients must ignore it, RFC 9112 §4) and
+        // Content-Length keeps its width as all-zero DIGITs.
+        if (status == 200) {
+            const int io = F[FI_INM_OFF], il = F[FI_INM_LEN];
+            if (il == 10 && rbase[io] == '"' && rbase[io + 9] == '"') {
+                bool m = true;
+                for (int i = 0; i < 8; ++i) {
+                    const uint8_t hx = HEXD[(h >> (28 - 4 * i)) & 0xF];
+                    if (rbase[io + 1 + i] != hx) { m = false; break; }
+                }
+                if (m) {
+                    if (lane == 0) {
+                        obuf[o_status] = '3';
+                        obuf[o_status + 1] = '0';
+                        obuf[o_status + 2] = '4';
+                        for (int i = 0; i < cl_digits; ++i)
+                            obuf[o_cld + i] = '0';
+                    }
+                    send_body = 0;
+                }
+            }
+        }
     }
 
-    const int total = hl + body_total;
+    const int total = hl + send_body;
     // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     if (total <= rslot) {

@@ -37,7 +37,9 @@ FI_AUTH_OFF = 13
 FI_AUTH_LEN = 14
 FI_RESP_OFF = 15
 FI_PARAM0 = 16
-NF = 24
+FI_INM_OFF = 24
+FI_INM_LEN = 25
+NF = 26
 
 FL_ERR_PARSE = 1
 FL_NEEDS_HOST = 2
@@ -338,6 +340,7 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         keep_alive = True
         body_off = ln
         auth_off = auth_len = 0
+        inm_off = inm_len = 0
         prev_lf = lf1
         for _ in range(64):
             ls = prev_lf + 1
@@ -373,6 +376,8 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
                     flags |= FL_JSON_CT
             elif nlen == 13 and _ieq(buf, ls, b"authorization"):
                 auth_off, auth_len = vs, vlen
+            elif nlen == 13 and _ieq(buf, ls, b"if-none-match"):
+                inm_off, inm_len = vs, vlen
             elif nlen == 17 and _ieq(buf, ls, b"transfer-encoding"):
                 flags |= FL_NEEDS_HOST
             elif nlen == 15 and _ieq(buf, ls, b"accept-encoding"):
@@ -389,6 +394,8 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         F[FI_CLEN] = clen
         F[FI_AUTH_OFF] = auth_off
         F[FI_AUTH_LEN] = auth_len
+        F[FI_INM_OFF] = inm_off
+        F[FI_INM_LEN] = inm_len
 
         # trie walk
         node = 0
@@ -589,9 +596,15 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
         corr = f"{h1:016x}{h2:016x}".encode()
         final_body = (b'{"data":' + body_src + b"}") if env else body_src
         etag_hdr = b""
+        not_modified = False
         if etag_on:
-            etag_hdr = b'ETag: "' + f"{etag_u32(final_body):08x}".encode() \
-                + b'"\r\n'
+            tag = f"{etag_u32(final_body):08x}"
+            etag_hdr = b'ETag: "' + tag.encode() + b'"\r\n'
+            io, il = int(F[FI_INM_OFF]), int(F[FI_INM_LEN])
+            if status == 200 and il == 10:
+                inm = reqs[base + io:base + io + il].tobytes()
+                if inm == b'"' + tag.encode() + b'"':
+                    not_modified = True
         head = (b"HTTP/1.1 " + f"{status:03d}".encode() + b" " + reason +
                 b"\r\n" + _CT_STRS[ct_id] +
                 (b"Content-Encoding: gzip\r\n" if content_enc else b"") +
@@ -601,7 +614,15 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 b"Content-Length: " + str(body_total).encode() + b"\r\n" +
                 (b"Connection: keep-alive\r\n\r\n" if keep
                  else b"Connection: close\r\n\r\n"))
-        if env:
+        if not_modified:
+            # in-place 304 rewrite quirks mirrored from the kernel:
+            # status digits patched, reason kept, CL zero-padded
+            cl = b"Content-Length: " + str(body_total).encode()
+            head = head.replace(
+                cl, b"Content-Length: " + b"0" * len(str(body_total)), 1)
+            head = head[:9] + b"304" + head[12:]
+            payload = head
+        elif env:
             payload = head + b'{"data":' + body_src + b"}"
         else:
             payload = head + body_src

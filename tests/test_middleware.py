@@ -279,3 +279,33 @@ def test_etag_u32_properties():
     long = bytes(range(256)) * 17  # multi-tile, bytes >= 128 (signed i8)
     assert isinstance(ops.etag_u32(long), int)
     assert ops.etag_u32(long) != ops.etag_u32(long[:-1])
+
+
+def test_if_none_match_304_cpu():
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.enable_etag()
+    eng = BatchEngine(app, device="cpu", max_batch=16)
+    first = eng.process([b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n"])[0]
+    head = first.split(b"\r\n\r\n", 1)[0].decode()
+    tag = [h for h in head.split("\r\n") if h.startswith("ETag")][0]
+    tag = tag.split(" ", 1)[1]  # '"xxxxxxxx"'
+    # conditional revalidation -> 304, empty body, zero-padded CL
+    second = eng.process([
+        ("GET /greet HTTP/1.1\r\nHost: h\r\n"
+         f"If-None-Match: {tag}\r\n\r\n").encode()])[0]
+    assert second.startswith(b"HTTP/1.1 304 ")
+    h2, _, body = second.partition(b"\r\n\r\n")
+    assert body == b""
+    assert b"Content-Length: 00" in h2  # width-preserving zero CL
+    assert tag.encode() in h2           # ETag still present
+    # stale validator -> normal 200 with body
+    third = eng.process([
+        b"GET /greet HTTP/1.1\r\nHost: h\r\n"
+        b'If-None-Match: "deadbeef"\r\n\r\n'])[0]
+    assert third.startswith(b"HTTP/1.1 200 OK")
