@@ -297,7 +297,14 @@ def run_config5_cpu(eng, payloads, steps, warmup, n_grpc, conns):
 def run_multi(eng, payloads, steps, warmup, rank, world):
     """RCCL all-to-all sharding (AllToAllSharder — the same class the
     gloo multi-process tests cover): scatter request slots to owner
-    shards, process on the local engine, gather responses back."""
+    shards, process on the local engine, gather responses back.
+
+    On GPU the loop is pipelined over 2 lanes, each with its own stream
+    + sharder buffer set: every stage (H2D, a2a scatter, kernels, a2a
+    gather, D2H) is enqueued under the lane stream (torch's NCCL calls
+    chain onto the calling stream), so lane A's collectives overlap
+    lane B's kernels/copies. Host-side collective call order is the
+    lane order, identical on every rank, so NCCL matching is safe."""
     import torch
     from gofr_amd.engine.shard import AllToAllSharder
     import torch.distributed as dist
@@ -306,39 +313,85 @@ def run_multi(eng, payloads, steps, warmup, rank, world):
     reqs, lens = make_batch(payloads, eng.slot)
     n = len(lens)
     assert n % world == 0
-    sharder = AllToAllSharder(eng, world)
-    sharder.alloc(n)
-    p_in = t.from_numpy(reqs).pin_memory()
-    p_len = t.from_numpy(lens.astype(np.int32)).pin_memory()
-    d_in = t.empty(n * eng.slot, dtype=t.uint8, device=dev)
-    d_len_in = t.empty(n, dtype=t.int32, device=dev)
-    p_resp = t.empty(n * eng.rslot, dtype=t.uint8).pin_memory()
-    p_rlen = t.empty(n, dtype=t.int32).pin_memory()
+    P = min(2, len(eng.lanes)) if dev is not None else 1
 
-    times = []
-    for it in range(warmup + steps):
-        if it == warmup:
-            torch.cuda.synchronize(dev)
-            dist.barrier()
-            torch.cuda.synchronize(dev)
-            t_start = time.perf_counter()
-        t0 = time.perf_counter()
-        # ingress: H2D staging of this shard's accepted connections
-        d_in.copy_(p_in, non_blocking=True)
-        d_len_in.copy_(p_len, non_blocking=True)
-        resp_sh, rlen_sh = sharder.step(d_in, d_len_in)
-        p_resp.copy_(resp_sh, non_blocking=True)
-        p_rlen.copy_(rlen_sh, non_blocking=True)
-        torch.cuda.synchronize(dev)
-        times.append(time.perf_counter() - t0)
-        if it == 0:
-            first = p_resp[:int(p_rlen[0])].numpy().tobytes()
-            assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+    class MLane:
+        def __init__(self, li):
+            self.sh = AllToAllSharder(eng, world, lane=li,
+                                      sync_host=(dev is None))
+            self.sh.alloc(n)
+            if dev is not None:
+                self.stream = t.cuda.Stream(device=dev)
+                self.ev = t.cuda.Event()
+                self.p_in = t.from_numpy(reqs).pin_memory()
+                self.p_len = t.from_numpy(
+                    lens.astype(np.int32)).pin_memory()
+                self.d_in = t.empty(n * eng.slot, dtype=t.uint8,
+                                    device=dev)
+                self.d_len = t.empty(n, dtype=t.int32, device=dev)
+                self.p_resp = t.empty(n * eng.rslot,
+                                      dtype=t.uint8).pin_memory()
+                self.p_rlen = t.empty(n, dtype=t.int32).pin_memory()
+            else:
+                self.d_in = t.from_numpy(reqs)
+                self.d_len = t.from_numpy(lens.astype(np.int32))
+
+    mlanes = [MLane(li) for li in range(P)]
+
+    if dev is None:
+        ml = mlanes[0]
+        times = []
+        for it in range(warmup + steps):
+            if it == warmup:
+                dist.barrier()
+                t_start = time.perf_counter()
+            t0 = time.perf_counter()
+            resp_sh, rlen_sh = ml.sh.step(ml.d_in, ml.d_len)
+            times.append(time.perf_counter() - t0)
+            if it == 0:
+                first = resp_sh.numpy()[:int(rlen_sh[0])].tobytes()
+                assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+        dist.barrier()
+        return time.perf_counter() - t_start, times[warmup:]
+
+    def submit(ml):
+        with t.cuda.stream(ml.stream):
+            ml.d_in.copy_(ml.p_in, non_blocking=True)
+            ml.d_len.copy_(ml.p_len, non_blocking=True)
+            resp_sh, rlen_sh = ml.sh.step(ml.d_in, ml.d_len)
+            ml.p_resp.copy_(resp_sh, non_blocking=True)
+            ml.p_rlen.copy_(rlen_sh, non_blocking=True)
+            ml.ev.record(ml.stream)
+
+    # warmup (serial, both lanes so NCCL per-lane state initializes;
+    # identical order on every rank)
+    for w in range(max(P, warmup)):
+        submit(mlanes[w % P])
+        mlanes[w % P].ev.synchronize()
+    first = mlanes[0].p_resp[:int(mlanes[0].p_rlen[0])].numpy().tobytes()
+    assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
+    torch.cuda.synchronize(dev)
+    dist.barrier()
+    torch.cuda.synchronize(dev)
+
+    submit_at = [0.0] * steps
+    lat = []
+    t_start = time.perf_counter()
+    for i in range(steps):
+        li = i % P
+        if i >= P:
+            mlanes[li].ev.synchronize()
+            lat.append(time.perf_counter() - submit_at[i - P])
+        submit_at[i] = time.perf_counter()
+        submit(mlanes[li])
+    for i in range(max(0, steps - P), steps):
+        mlanes[i % P].ev.synchronize()
+        lat.append(time.perf_counter() - submit_at[i])
     torch.cuda.synchronize(dev)
     dist.barrier()
     torch.cuda.synchronize(dev)
     elapsed = time.perf_counter() - t_start
-    return elapsed, times[warmup:]
+    return elapsed, lat
 
 
 def main():
@@ -401,9 +454,13 @@ def main():
     # largest request; for N>1 it is also the fixed all-to-all exchange
     # granularity, so tighter slots mean fewer xGMI + host-link bytes)
     slot = max(1024, ((len(raw) + 255) // 256) * 256 + 256)
+    if have_gpu:
+        pipeline = int(os.environ.get("GOFR_PIPELINE", "5")) \
+            if world == 1 else 2
+    else:
+        pipeline = 1
     eng = BatchEngine(app, device=device, slot=slot, max_batch=batch,
-                      pipeline=(int(os.environ.get("GOFR_PIPELINE", "5"))
-                                if (have_gpu and world == 1) else 1))
+                      pipeline=pipeline)
 
     n_grpc = int(batch * args.grpc_frac)
     if n_grpc and world == 1:
@@ -446,9 +503,14 @@ def main():
         print(json.dumps(out))
         return
 
-    if world > 1:
+    if world > 1 or os.environ.get("GOFR_FORCE_MULTI"):
         import torch.distributed as dist
-        dist.init_process_group("nccl")
+        if world == 1:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29531")
+            os.environ.setdefault("RANK", "0")
+            os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("nccl" if have_gpu else "gloo")
         elapsed, times = run_multi(eng, payloads, args.steps, args.warmup,
                                    rank, world)
         # whole-job aggregate: max elapsed over ranks

@@ -22,13 +22,16 @@ from .. import ops
 class AllToAllSharder:
     """One rank's view of the sharded serving step."""
 
-    def __init__(self, engine, world: int):
+    def __init__(self, engine, world: int, lane: int = 0,
+                 sync_host: bool = True):
         import torch
         import torch.distributed as dist
         self.t = torch
         self.dist = dist
         self.engine = engine
         self.world = world
+        self.lane = lane
+        self.sync_host = sync_host
         self.device = engine.device if engine.device is not None \
             else torch.device("cpu")
         self.slot = engine.slot
@@ -60,7 +63,8 @@ class AllToAllSharder:
         dist.all_to_all_single(self.d_len_sh, d_len_in)
         if self.engine.device is not None:
             d_resp, d_rlen = self.engine.process_device(
-                self.d_sh, self.d_off, self.d_len_sh, n)
+                self.d_sh, self.d_off, self.d_len_sh, n,
+                lane_idx=self.lane, sync_host=self.sync_host)
             dist.all_to_all_single(self.d_resp_sh,
                                    d_resp[:n * self.rslot].contiguous())
             dist.all_to_all_single(self.d_rlen_sh,
