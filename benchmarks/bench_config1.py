@@ -64,11 +64,32 @@ def client_loop(port, depth, stop, counts, lats, idx):
     s.close()
 
 
+def run_native_client(port, conns, depth, seconds, threads=4):
+    """Compile + run the C++ epoll load generator (the Python client
+    GIL-caps around ~0.5M req/s)."""
+    import subprocess
+    src = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                       "loadgen.cpp")
+    exe = "/tmp/gofr_loadgen"
+    if (not os.path.exists(exe)
+            or os.path.getmtime(exe) < os.path.getmtime(src)):
+        subprocess.run(["g++", "-O2", "-pthread", src, "-o", exe],
+                       check=True)
+    out = subprocess.run(
+        [exe, "127.0.0.1", str(port), str(conns), str(depth),
+         str(seconds), str(threads)],
+        check=True, capture_output=True, text=True)
+    return json.loads(out.stdout.strip())
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--conns", type=int, default=16)
     ap.add_argument("--pipeline", type=int, default=64)
     ap.add_argument("--seconds", type=float, default=5.0)
+    ap.add_argument("--client", choices=["python", "native"],
+                    default="python")
+    ap.add_argument("--threads", type=int, default=4)
     args = ap.parse_args()
 
     app = gofr_amd.New(config=MapConfig({"APP_NAME": "hello",
@@ -76,6 +97,25 @@ def main():
     app.GET("/greet", handlers.static_json("Hello World!"))
     srv = GPUServer(app, 0, batch_window_us=200, max_batch=8192)
     srv.start()
+    if args.client == "native":
+        try:
+            r = run_native_client(srv.port, args.conns, args.pipeline,
+                                  args.seconds, args.threads)
+            print(json.dumps({
+                "metric": "HTTP req/s, hello-world on CPU listener "
+                          "(config 1, native client)",
+                "value": r["req_per_s"],
+                "unit": "req/s",
+                "n_gpus": 0,
+                "conns": r["conns"], "pipeline_depth": r["depth"],
+                "client_threads": r["threads"],
+                "seconds": r["seconds"],
+                "p50_us": r["p50_us"], "p99_us": r["p99_us"],
+                "higher_is_better": True, "data": "synthetic",
+            }))
+        finally:
+            srv.stop()
+        return
     try:
         stop = threading.Event()
         counts = [0] * args.conns
