@@ -29,6 +29,14 @@ REQ = (b"GET /greet HTTP/1.1\r\n"
        b"Host: localhost\r\n\r\n")
 
 
+def _have_gpu():
+    try:
+        import torch
+        return torch.cuda.is_available()
+    except ImportError:
+        return False
+
+
 def client_loop(port, depth, stop, counts, lats, idx):
     s = socket.create_connection(("127.0.0.1", port))
     s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
@@ -90,12 +98,15 @@ def main():
     ap.add_argument("--client", choices=["python", "native"],
                     default="python")
     ap.add_argument("--threads", type=int, default=4)
+    ap.add_argument("--max-batch", type=int, default=8192)
+    ap.add_argument("--window-us", type=int, default=200)
     args = ap.parse_args()
 
     app = gofr_amd.New(config=MapConfig({"APP_NAME": "hello",
                                          "LOG_LEVEL": "FATAL"}))
     app.GET("/greet", handlers.static_json("Hello World!"))
-    srv = GPUServer(app, 0, batch_window_us=200, max_batch=8192)
+    srv = GPUServer(app, 0, batch_window_us=args.window_us,
+                max_batch=args.max_batch)
     srv.start()
     if args.client == "native":
         try:
@@ -106,7 +117,7 @@ def main():
                           "(config 1, native client)",
                 "value": r["req_per_s"],
                 "unit": "req/s",
-                "n_gpus": 0,
+                "n_gpus": 1 if _have_gpu() else 0,
                 "conns": r["conns"], "pipeline_depth": r["depth"],
                 "client_threads": r["threads"],
                 "seconds": r["seconds"],
