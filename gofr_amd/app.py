@@ -192,10 +192,25 @@ class App:
             self.container.logger.Infof("gRPC server listening on :%d",
                                         self.grpc_port)
         if block:
+            # Graceful shutdown on SIGTERM/SIGINT — a conscious fix of
+            # the reference's block-forever Run (gofr.go:125 wg.Wait()
+            # with no signal handling; SURVEY.md §5 "no graceful
+            # shutdown"): in-flight batches complete, sockets close,
+            # datasources disconnect.
+            import signal
             ev = threading.Event()
+            prev = None
+            if threading.current_thread() is threading.main_thread():
+                prev = signal.signal(signal.SIGTERM,
+                                     lambda *_: ev.set())
             try:
-                ev.wait()  # block forever — reference: gofr.go:125 wg.Wait()
+                ev.wait()  # blocks until SIGTERM (or KeyboardInterrupt)
             except KeyboardInterrupt:
+                pass
+            finally:
+                if prev is not None:
+                    signal.signal(signal.SIGTERM, prev)
+                self.container.logger.Infof("shutting down")
                 self.shutdown()
 
     run = Run

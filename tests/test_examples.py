@@ -14,6 +14,7 @@ import pytest
 
 _EXAMPLES = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                          "..", "examples")
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def _load_example(name):
@@ -149,3 +150,42 @@ def test_grpc_example_unary():
     server.stop()
     app.shutdown()
     os.environ.pop("LOG_LEVEL", None)
+
+
+def test_graceful_shutdown_on_sigterm():
+    """App.Run blocks until SIGTERM, then stops servers cleanly (a
+    conscious improvement over the reference's wg.Wait-forever)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import gofr_amd\n"
+        "from gofr_amd.config import MapConfig\n"
+        "app = gofr_amd.New(config=MapConfig({'LOG_LEVEL': 'INFO',\n"
+        "                                     'HTTP_PORT': '0'}))\n"
+        "app.GET('/x', lambda ctx: 'ok')\n"
+        "print('READY', flush=True)\n"
+        "app.Run()\n"
+        "print('CLEAN-EXIT', flush=True)\n" % REPO)
+    p = subprocess.Popen([sys.executable, "-c", code],
+                         stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                         text=True)
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 20:
+            line = p.stdout.readline()
+            if "READY" in line:
+                break
+        time.sleep(0.5)
+        p.send_signal(signal.SIGTERM)
+        out = p.stdout.read()
+        rc = p.wait(timeout=20)
+    finally:
+        if p.poll() is None:
+            p.kill()
+    assert rc == 0, out
+    assert "CLEAN-EXIT" in out and "shutting down" in out
