@@ -72,7 +72,8 @@ def client_loop(port, depth, stop, counts, lats, idx):
     s.close()
 
 
-def run_native_client(port, conns, depth, seconds, threads=4):
+def run_native_client(port, conns, depth, seconds, threads=4,
+                      n_addrs=1):
     """Compile + run the C++ epoll load generator (the Python client
     GIL-caps around ~0.5M req/s)."""
     import subprocess
@@ -85,7 +86,7 @@ def run_native_client(port, conns, depth, seconds, threads=4):
                        check=True)
     out = subprocess.run(
         [exe, "127.0.0.1", str(port), str(conns), str(depth),
-         str(seconds), str(threads)],
+         str(seconds), str(threads), str(n_addrs)],
         check=True, capture_output=True, text=True)
     return json.loads(out.stdout.strip())
 
@@ -100,6 +101,9 @@ def main():
     ap.add_argument("--threads", type=int, default=4)
     ap.add_argument("--max-batch", type=int, default=8192)
     ap.add_argument("--window-us", type=int, default=200)
+    ap.add_argument("--n-addrs", type=int, default=1,
+                    help="spread client conns over 127.0.0.{1..N} "
+                         "(needed above ~50k conns)")
     args = ap.parse_args()
 
     app = gofr_amd.New(config=MapConfig({"APP_NAME": "hello",
@@ -111,7 +115,8 @@ def main():
     if args.client == "native":
         try:
             r = run_native_client(srv.port, args.conns, args.pipeline,
-                                  args.seconds, args.threads)
+                                  args.seconds, args.threads,
+                                  args.n_addrs)
             print(json.dumps({
                 "metric": "HTTP req/s, hello-world on CPU listener "
                           "(config 1, native client)",

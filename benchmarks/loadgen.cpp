@@ -46,7 +46,7 @@ struct Stats {
 };
 
 static void reactor(const char* host, int port, int nconns, int depth,
-                    double seconds, Stats* st) {
+                    double seconds, Stats* st, int n_addrs) {
     int ep = epoll_create1(0);
     std::vector<Conn> conns(nconns);
     for (int i = 0; i < nconns; ++i) {
@@ -54,7 +54,15 @@ static void reactor(const char* host, int port, int nconns, int depth,
         sockaddr_in a{};
         a.sin_family = AF_INET;
         a.sin_port = htons(port);
-        inet_pton(AF_INET, host, &a.sin_addr);
+        if (n_addrs > 1) {
+            // spread over loopback addresses so >64k conns fit inside
+            // the ephemeral-port space (config 5: 100k real conns)
+            char ab[32];
+            snprintf(ab, sizeof(ab), "127.0.0.%d", 1 + (i % n_addrs));
+            inet_pton(AF_INET, ab, &a.sin_addr);
+        } else {
+            inet_pton(AF_INET, host, &a.sin_addr);
+        }
         if (connect(fd, (sockaddr*)&a, sizeof(a))) { perror("connect"); _exit(2); }
         int one = 1;
         setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
@@ -133,20 +141,23 @@ static void reactor(const char* host, int port, int nconns, int depth,
 
 int main(int argc, char** argv) {
     if (argc < 7) {
-        fprintf(stderr, "usage: %s host port conns depth seconds threads\n",
-                argv[0]);
+        fprintf(stderr,
+                "usage: %s host port conns depth seconds threads "
+                "[n_addrs]\n", argv[0]);
         return 1;
     }
     const char* host = argv[1];
     int port = atoi(argv[2]), conns = atoi(argv[3]), depth = atoi(argv[4]);
     double seconds = atof(argv[5]);
     int threads = atoi(argv[6]);
+    int n_addrs = argc > 7 ? atoi(argv[7]) : 1;
     std::vector<Stats> st(threads);
     std::vector<std::thread> ts;
     const double t0 = now_s();
     int per = conns / threads;
     for (int t = 0; t < threads; ++t)
-        ts.emplace_back(reactor, host, port, per, depth, seconds, &st[t]);
+        ts.emplace_back(reactor, host, port, per, depth, seconds, &st[t],
+                        n_addrs);
     for (auto& t : ts) t.join();
     const double elapsed = now_s() - t0;
     long total = 0;
