@@ -91,6 +91,22 @@ def run_native_client(port, conns, depth, seconds, threads=4,
     return json.loads(out.stdout.strip())
 
 
+def _raise_nofile(target=1 << 20):
+    """Raise RLIMIT_NOFILE (100k-conn runs need ~2x conns in fds;
+    containers often cap soft at 20k — as root the hard limit can go
+    up)."""
+    import resource
+    try:
+        resource.setrlimit(resource.RLIMIT_NOFILE, (target, target))
+    except (ValueError, OSError):
+        soft, hard = resource.getrlimit(resource.RLIMIT_NOFILE)
+        try:
+            resource.setrlimit(resource.RLIMIT_NOFILE, (hard, hard))
+        except (ValueError, OSError):
+            pass
+    return resource.getrlimit(resource.RLIMIT_NOFILE)[0]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--conns", type=int, default=16)
@@ -106,6 +122,10 @@ def main():
                          "(needed above ~50k conns)")
     args = ap.parse_args()
 
+    nofile = _raise_nofile()
+    if args.conns * 2 + 256 > nofile:
+        print(f"warning: RLIMIT_NOFILE {nofile} < 2x conns",
+              file=sys.stderr)
     app = gofr_amd.New(config=MapConfig({"APP_NAME": "hello",
                                          "LOG_LEVEL": "FATAL"}))
     app.GET("/greet", handlers.static_json("Hello World!"))
