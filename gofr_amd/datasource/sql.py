@@ -90,11 +90,14 @@ class DB:
     """database/sql-style wrapper over a DB-API connection."""
 
     def __init__(self, conn, logger=None, dialect: str = "sqlite",
-                 paramstyle: str = "qmark"):
+                 paramstyle: str = ""):
         self._conn = conn
         self.logger = logger
         self.dialect = dialect
-        self.paramstyle = paramstyle
+        # MySQL's DB-API surface uses the `format` (%s) paramstyle;
+        # sqlite uses qmark — derive from the dialect unless overridden
+        self.paramstyle = paramstyle or ("format" if dialect == "mysql"
+                                         else "qmark")
         self._lock = threading.RLock()
         self._stats = {"queries": 0, "execs": 0}
 
