@@ -145,7 +145,13 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     const int lane = lane_id();
     const uint8_t* base = reqs + req_off[req];
     int len = req_len[req];
-    if (len > MAX_SLOT) len = MAX_SLOT;
+    bool oversized = false;
+    if (len > MAX_SLOT) {
+        // larger than the LDS working set: the host trampoline parses
+        // the full bytes from the ring (k_respond serves its result)
+        len = MAX_SLOT;
+        oversized = true;
+    }
     int32_t* F = fields + (size_t)req * NF;
 
     const int nchunks = (len + WAVE - 1) / WAVE;
@@ -186,7 +192,7 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     };
 
     int32_t flags = 0;
-    if (any_percent) flags |= FL_NEEDS_HOST;
+    if (any_percent || oversized) flags |= FL_NEEDS_HOST;
 
     // request line: METHOD SP target SP version CRLF
     const int sp1 = next_bit(CLS_SP, 0);
@@ -359,6 +365,11 @@ k_parse_route(const uint8_t* __restrict__ reqs,
             status = handler_tab[route * 4 + 3];
         }
         if (flags & FL_IS_OPTIONS) { kind = HK_STATIC; status = 200; }
+        // echo bodies near the LDS working set would overflow the
+        // envelope assembly: host path (large responses then take
+        // k_respond's direct-to-global body route, env-free)
+        if (kind == HK_ECHO_JSON && body_len > MAX_SLOT - 600)
+            kind = HK_HOST;
         if (flags & (FL_ERR_PARSE | FL_NEEDS_HOST)) kind = HK_HOST;
         F[FI_KIND] = kind;
         F[FI_STATUS] = status;
@@ -971,8 +982,16 @@ __device__ __forceinline__ void respond_impl(
         }
     }
 
+    // responses larger than the LDS working set keep the body in
+    // global memory (headers still assemble in LDS); guaranteed
+    // env-free: envelope (echo) responses are LDS-sized by the parse
+    // kernel's HK_HOST rerouting of large echo bodies
+    const bool large = hl + body_total > MAX_SLOT;
+
     // ---- body into LDS -----------------------------------------------------
-    if (GZ && content_enc) {
+    if (large) {
+        // nothing staged: the final copy streams body_src directly
+    } else if (GZ && content_enc) {
         // shift the deflate output from obuf+512 down to obuf+hl.
         // dst < src and the wave is lockstep (loads of an iteration all
         // execute before its stores), so a forward sweep is safe.
@@ -991,7 +1010,9 @@ __device__ __forceinline__ void respond_impl(
     if (etag_on) {
         // MFMA hash of the final body bytes -> patch the reserved hex
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        const uint32_t h = mfma_etag_wave(obuf + hl, body_total, lane);
+        const uint32_t h = large
+            ? mfma_etag_wave(body_src, body_total, lane)
+            : mfma_etag_wave(obuf + hl, body_total, lane);
         if (lane < 8)
             obuf[o_etag + 7 + lane] = HEXD[(h >> (28 - 4 * lane)) & 0xF];
         // If-None-Match -> 304: in-place rewrite (status digits + CL
@@ -1025,10 +1046,17 @@ __device__ __forceinline__ void respond_impl(
     // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     if (total <= rslot) {
-        const int nv = (total + 15) >> 4;
-        const uint4* src = (const uint4*)obuf;
-        uint4* dst = (uint4*)out;
-        for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+        if (large) {
+            for (int i = lane; i < hl; i += WAVE) out[i] = obuf[i];
+            if (send_body)
+                for (int i = lane; i < body_total; i += WAVE)
+                    out[hl + i] = body_src[i];
+        } else {
+            const int nv = (total + 15) >> 4;
+            const uint4* src = (const uint4*)obuf;
+            uint4* dst = (uint4*)out;
+            for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+        }
     }
     if (lane == 0) {
         const int tl = (total <= rslot) ? total : 0;

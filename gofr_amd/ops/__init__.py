@@ -299,10 +299,13 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
     for r in range(n):
         base = int(req_off[r])
         ln = int(req_len[r])
+        oversized = ln > MAX_SLOT
+        if oversized:
+            ln = MAX_SLOT
         buf = reqs[base:base + ln].tobytes()
         F = fields[r]
         flags = 0
-        if b"%" in buf:
+        if b"%" in buf or oversized:
             flags |= FL_NEEDS_HOST
         sp1 = buf.find(b" ")
         lf1 = buf.find(b"\n")
@@ -456,6 +459,10 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
             status = int(handler_tab[route * 4 + 3])
         if flags & FL_IS_OPTIONS:
             kind, status = HK_STATIC, 200
+        # kernel parity: large echo bodies go to the host so envelope
+        # responses stay inside the LDS working set
+        if kind == HK_ECHO_JSON and int(F[FI_BODY_LEN]) > MAX_SLOT - 600:
+            kind = HK_HOST
         if flags & (FL_ERR_PARSE | FL_NEEDS_HOST):
             kind = HK_HOST
         F[FI_KIND] = kind

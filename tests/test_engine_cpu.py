@@ -162,3 +162,27 @@ def test_parse_fields_match_reference_parser(engine):
                 base + F[ops.FI_AUTH_OFF] + F[ops.FI_AUTH_LEN]].tobytes()
     assert auth == b"Bearer tok123"
     assert not (F[ops.FI_FLAGS] & ops.FL_KEEP_ALIVE)
+
+
+def test_oversized_request_routes_to_host():
+    """Requests beyond the 4 KiB kernel working set go through the host
+    trampoline (full-fidelity parse) instead of being truncated."""
+    import json
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    eng = BatchEngine(app, device="cpu", slot=16384, max_batch=8)
+    body = json.dumps({"big": "x" * 6000}).encode()
+    raw = (b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+           b"Content-Type: application/json\r\n"
+           b"Content-Length: " + str(len(body)).encode() + b"\r\n\r\n" +
+           body)
+    out = eng.process([raw])[0]
+    assert out.startswith(b"HTTP/1.1 200 OK"), out[:80]
+    _, _, rbody = out.partition(b"\r\n\r\n")
+    assert json.loads(rbody)["data"]["big"] == "x" * 6000

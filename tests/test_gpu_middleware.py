@@ -255,3 +255,35 @@ def test_if_none_match_304_gpu_matches_mirror():
         assert go == co, f"req {i}\nGPU {go[:160]!r}\nCPU {co[:160]!r}"
     assert g[0].startswith(b"HTTP/1.1 304 ")
     assert g[1].startswith(b"HTTP/1.1 200 OK")
+
+
+def test_large_and_oversized_bodies_gpu_matches_mirror():
+    """Host-path responses larger than the LDS working set (direct
+    global body route) and >4 KiB requests (host trampoline) — GPU
+    bytes == mirror bytes."""
+    import json
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/big", lambda ctx: {"blob": "z" * 6000})
+    app.enable_etag()
+    gpu = BatchEngine(app, device="cuda", slot=16384, max_batch=64)
+    cpu = BatchEngine(app, device="cpu", slot=16384, max_batch=64)
+    cpu._seed = gpu._seed
+    big_body = json.dumps({"big": "x" * 6000}).encode()
+    raws = []
+    for i in range(32):
+        if i % 3 == 0:
+            raws.append(http_req("GET", "/big"))
+        elif i % 3 == 1:
+            raws.append(http_req("POST", "/echo", body=big_body))
+        else:
+            raws.append(http_req("POST", "/echo",
+                                 body=b'{"small":"ok"}'))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, f"req {i}\nGPU {go[:160]!r}\nCPU {co[:160]!r}"
+    assert b'"z' in g[0] and len(g[0]) > 6000      # big host body served
+    _, _, rb = g[1].partition(b"\r\n\r\n")
+    assert json.loads(rb)["data"]["big"] == "x" * 6000
