@@ -192,5 +192,36 @@ if __name__ == "__main__":
     import sys as _sys
     if "--mix" in _sys.argv:
         extra_mix_tests()
+    elif "--enq" in _sys.argv:
+        enqueue_block_test()
     else:
         main()
+
+
+def enqueue_block_test():
+    """Does enqueueing a 2nd/3rd H2D on a busy stream block the host?"""
+    dev = torch.device("cuda:0")
+    p = torch.zeros(N, dtype=torch.uint8).pin_memory()
+    d = [torch.zeros(N, dtype=torch.uint8, device=dev) for _ in range(4)]
+    s = torch.cuda.Stream()
+    torch.cuda.synchronize()
+    t = []
+    for i in range(4):
+        t0 = time.perf_counter()
+        with torch.cuda.stream(s):
+            d[i].copy_(p, non_blocking=True)
+        t.append((time.perf_counter() - t0) * 1000)
+    torch.cuda.synchronize()
+    print("same-stream H2D enqueue times (ms):",
+          " ".join(f"{x:.3f}" for x in t))
+    s2 = [torch.cuda.Stream() for _ in range(4)]
+    torch.cuda.synchronize()
+    t = []
+    for i in range(4):
+        t0 = time.perf_counter()
+        with torch.cuda.stream(s2[i]):
+            d[i].copy_(p, non_blocking=True)
+        t.append((time.perf_counter() - t0) * 1000)
+    torch.cuda.synchronize()
+    print("per-stream  H2D enqueue times (ms):",
+          " ".join(f"{x:.3f}" for x in t))

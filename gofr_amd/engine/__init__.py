@@ -402,6 +402,7 @@ class BatchEngine:
         self.s_k.synchronize()
         a = ops.GofrSubmitArgs()
         kmode = os.environ.get("GOFR_KMODE", "staged")
+        lane_ingress = os.environ.get("GOFR_LANE_INGRESS", "0") == "1"
         if kmode == "chan2":
             # shared SDMA ingress stream; kernels AND link-bound egress
             # sweep fused on one of two channel streams per lane parity
@@ -416,7 +417,11 @@ class BatchEngine:
             a.s_k = chan.cuda_stream
             a.s_out = self.s_out.cuda_stream
         else:
-            a.s_in = self.s_in.cuda_stream
+            # per-lane ingress stream option: hipMemcpyAsync enqueue can
+            # block while the stream's previous SDMA is in flight, which
+            # would pace the pump's enqueue thread at the H2D cadence
+            a.s_in = (ln.stream.cuda_stream if lane_ingress
+                      else self.s_in.cuda_stream)
             a.s_k = self.s_k.cuda_stream
             a.s_out = self.s_out.cuda_stream
         a.ev_in = ln.e_in.cuda_event
