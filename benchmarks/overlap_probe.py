@@ -188,14 +188,6 @@ def extra_mix_tests():
               f"({gbs(ms2, 2*N):.1f} GB/s agg)")
 
 
-if __name__ == "__main__":
-    import sys as _sys
-    if "--mix" in _sys.argv:
-        extra_mix_tests()
-    elif "--enq" in _sys.argv:
-        enqueue_block_test()
-    else:
-        main()
 
 
 def enqueue_block_test():
@@ -225,3 +217,13 @@ def enqueue_block_test():
     torch.cuda.synchronize()
     print("per-stream  H2D enqueue times (ms):",
           " ".join(f"{x:.3f}" for x in t))
+
+
+if __name__ == "__main__":
+    import sys as _sys
+    if "--mix" in _sys.argv:
+        extra_mix_tests()
+    elif "--enq" in _sys.argv:
+        enqueue_block_test()
+    else:
+        main()
