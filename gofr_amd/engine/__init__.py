@@ -122,11 +122,18 @@ class _Lane:
         self.e_in = t.cuda.Event()
         self.e_k = t.cuda.Event()
         self.e_out = t.cuda.Event()
-        self.d_reqs = t.empty(max_bytes, dtype=t.uint8, device=dev)
-        # +1: the batch seed rides in the tail slot of the offsets copy
-        # (an 8-byte pinned H2D on its own costs a ~60 us blit kernel)
-        self.d_req_off = t.empty(nb + 1, dtype=t.int64, device=dev)
-        self.d_req_len = t.empty(nb, dtype=t.int32, device=dev)
+        # ingress block: [off (nb+1)*int64 | len nb*int32 | serial u64 |
+        # pad16 | request bytes] — ONE contiguous region so the whole
+        # batch ingress is ONE SDMA copy (each queued SDMA op costs a
+        # scheduling gap; 4 ops/batch paced the pipeline). The seed
+        # rides in the offsets tail (slot n).
+        self.hdr_bytes = (((nb + 1) * 8 + nb * 4 + 8 + 15) // 16) * 16
+        self.d_ingress = t.empty(self.hdr_bytes + max_bytes,
+                                 dtype=t.uint8, device=dev)
+        self.d_req_off = self.d_ingress[:(nb + 1) * 8].view(t.int64)
+        self.d_req_len = self.d_ingress[
+            (nb + 1) * 8:(nb + 1) * 8 + nb * 4].view(t.int32)
+        self.d_reqs = self.d_ingress[self.hdr_bytes:]
         self.d_fields = t.zeros(nb * ops.NF, dtype=t.int32, device=dev)
         self.d_resp = t.empty(nb * rslot, dtype=t.uint8, device=dev)
         self.d_resp_len = t.empty(nb, dtype=t.int32, device=dev)
@@ -135,9 +142,12 @@ class _Lane:
         self.d_host_needed = t.zeros(1, dtype=t.int32, device=dev)
         self.d_host_tab = t.zeros(nb * 4, dtype=t.int32, device=dev)
         self.d_host_blob = t.zeros(host_blob_cap, dtype=t.uint8, device=dev)
-        self.p_reqs = t.empty(max_bytes, dtype=t.uint8).pin_memory()
-        self.p_req_off = t.empty(nb + 1, dtype=t.int64).pin_memory()
-        self.p_req_len = t.empty(nb, dtype=t.int32).pin_memory()
+        self.p_ingress = t.empty(self.hdr_bytes + max_bytes,
+                                 dtype=t.uint8).pin_memory()
+        self.p_req_off = self.p_ingress[:(nb + 1) * 8].view(t.int64)
+        self.p_req_len = self.p_ingress[
+            (nb + 1) * 8:(nb + 1) * 8 + nb * 4].view(t.int32)
+        self.p_reqs = self.p_ingress[self.hdr_bytes:]
         # egress ring via hipHostMalloc: D2H SDMA works into it (torch
         # pin_memory is hipHostRegister'd, which the runtime serves with
         # a blit kernel instead — see ops.HipOps.host_alloc)
@@ -354,12 +364,12 @@ class BatchEngine:
             ln.event.record(ln.stream)
 
     def _ingress_body(self, ln, n, nbytes):
-        """H2D staging copies (runs on the caller's current stream).
-        The batch seed travels as element n of the offsets copy."""
+        """ONE H2D staging copy: header (offsets+lens+seed) + payload
+        travel together (runs on the caller's current stream)."""
         ln.p_req_off[n] = ln.p_seed[0]
-        ln.d_reqs[:nbytes].copy_(ln.p_reqs[:nbytes], non_blocking=True)
-        ln.d_req_off[:n + 1].copy_(ln.p_req_off[:n + 1], non_blocking=True)
-        ln.d_req_len[:n].copy_(ln.p_req_len[:n], non_blocking=True)
+        total = ln.hdr_bytes + nbytes
+        ln.d_ingress[:total].copy_(ln.p_ingress[:total],
+                                   non_blocking=True)
 
     def _kernel_body(self, ln, n, emit_host=True):
         """parse -> auth -> respond -> compact (current stream; the
@@ -429,7 +439,9 @@ class BatchEngine:
         a.ev_done = ln.event.cuda_event
         a.p_reqs = ln.p_reqs.data_ptr()
         a.d_reqs = ln.d_reqs.data_ptr()
-        a.nbytes = nbytes
+        # combined one-copy ingress: base = the ingress block, size =
+        # header + payload (gofr_submit_impl copies p_off -> d_off once)
+        a.nbytes = ln.hdr_bytes + nbytes
         a.p_off = ln.p_req_off.data_ptr()
         a.d_off = ln.d_req_off.data_ptr()
         a.p_len = ln.p_req_len.data_ptr()

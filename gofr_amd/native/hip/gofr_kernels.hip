@@ -1625,17 +1625,12 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     hipError_t rc;
     double t0 = now_us();
     g_submit_us[4] += 1;
-    // ---- ingress ----------------------------------------------------------
-    rc = hipMemcpyAsync(a->d_reqs, a->p_reqs, (size_t)a->nbytes,
+    // ---- ingress: ONE copy (header: offsets+lens+seed, then payload;
+    // a->nbytes = header + payload bytes, a->p_off/d_off = block base)
+    rc = hipMemcpyAsync(a->d_off, a->p_off, (size_t)a->nbytes,
                         hipMemcpyHostToDevice, s_in);
     if (rc) return (int)rc;
     g_submit_us[0] += now_us() - t0; t0 = now_us();
-    rc = hipMemcpyAsync(a->d_off, a->p_off, (size_t)(n + 1) * 8,
-                        hipMemcpyHostToDevice, s_in);
-    if (rc) return (int)rc;
-    rc = hipMemcpyAsync(a->d_len, a->p_len, (size_t)n * 4,
-                        hipMemcpyHostToDevice, s_in);
-    if (rc) return (int)rc;
     if (a->d_flag) {
         // serial lands AFTER the payload (same SDMA FIFO)
         rc = hipMemcpyAsync(a->d_flag, a->p_serial, 8,
