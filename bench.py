@@ -399,7 +399,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=32768)
+    ap.add_argument("--batch", type=int, default=65536)
     ap.add_argument("--payload", type=int, default=1024)
     ap.add_argument("--routes", type=int, default=4,
                     help="route-table size (config 4: 64)")
@@ -455,7 +455,7 @@ def main():
     # granularity, so tighter slots mean fewer xGMI + host-link bytes)
     slot = max(1024, ((len(raw) + 255) // 256) * 256 + 256)
     if have_gpu:
-        pipeline = int(os.environ.get("GOFR_PIPELINE", "5")) \
+        pipeline = int(os.environ.get("GOFR_PIPELINE", "4")) \
             if world == 1 else 2
     else:
         pipeline = 1
