@@ -433,6 +433,15 @@ def main():
         app.enable_gzip(min_size=256)
     if "etag" in mw:
         app.enable_etag()
+    if "log" in mw:
+        # config-4's log middleware: per-batch aggregate + sampled
+        # request records at INFO (sink = stderr-less devnull so the
+        # bench measures emission cost, not terminal IO)
+        import gofr_amd.logging as glog
+        devnull = open(os.devnull, "w")
+        app.container.logger = glog.Logger(level=glog.INFO, out=devnull,
+                                           err=devnull, force_json=True)
+        app.enable_request_log(sample_every=4096)
     if have_gpu:
         # map ranks onto the devices that exist (lets a world-2 smoke
         # run on a 1-GPU box; on the 8-GPU node it is the identity)

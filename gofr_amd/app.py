@@ -146,6 +146,16 @@ class App:
         self.etag_on = True
         return self
 
+    def enable_request_log(self, sample_every: int = 4096):
+        """Batch-aware access-log middleware: the engine emits one
+        BatchLog aggregate per processed batch plus every Nth request
+        as a full RequestLog record (parsing 22M+ full per-request log
+        lines/s is neither writable nor readable; sampling preserves
+        the reference's observability signal — middleware/logger.go:
+        24-63 — at batch rates)."""
+        self.request_log_every = max(1, int(sample_every))
+        return self
+
     def enable_gzip(self, min_size: int = 256):
         """gzip-compress JSON responses when the request advertises
         Accept-Encoding: gzip and the body is >= min_size bytes."""

@@ -185,6 +185,28 @@ class _Lane:
         self.egress_budget = 0
 
 
+class BatchLog:
+    """Per-batch access-log aggregate (the log middleware's unit of
+    emission under batching)."""
+
+    __slots__ = ("n", "bytes_in", "bytes_out", "batch_ms")
+
+    def __init__(self, n, bytes_in, bytes_out, batch_ms):
+        self.n = n
+        self.bytes_in = bytes_in
+        self.bytes_out = bytes_out
+        self.batch_ms = batch_ms
+
+    def to_dict(self):
+        return {"batch": self.n, "bytesIn": self.bytes_in,
+                "bytesOut": self.bytes_out,
+                "batchMs": round(self.batch_ms, 3)}
+
+    def pretty(self):
+        return (f"BATCH  {self.n:6d} reqs  {self.batch_ms:7.3f} ms  "
+                f"in {self.bytes_in} out {self.bytes_out}")
+
+
 class BatchEngine:
     """Processes request batches through the GPU kernels (or CPU mirrors)."""
 
@@ -322,6 +344,10 @@ class BatchEngine:
             o = int(roffs[i])
             out[o:o + int(resp_len[i])] = resp_slots[
                 i * self.rslot:i * self.rslot + int(resp_len[i])]
+        if getattr(self.app, "request_log_every", 0):
+            self._emit_logs(n, int(offs[-1] + lens[-1]) if n else 0,
+                            int(out.size), buf, offs, out, roffs,
+                            resp_len, 0.0)
         return out, roffs, resp_len
 
     # -- pipelined API --------------------------------------------------------
@@ -341,6 +367,7 @@ class BatchEngine:
         seed = self._next_seed()
         signed = seed - (1 << 64) if seed >= (1 << 63) else seed
         ln.p_seed[0] = signed
+        ln.submit_t = time.perf_counter()
         if ln.c_args is not None and ln.graph_key == (n, nbytes):
             ln.mode = "c"
             ln.p_req_off_np[n] = signed  # seed rides the offsets tail
@@ -588,6 +615,8 @@ class BatchEngine:
                     ln.p_out[:total].copy_(ln.d_out[:total],
                                            non_blocking=True)
                 ln.stream.synchronize()
+            if getattr(self.app, "request_log_every", 0):
+                self._emit_request_logs(ln, n)
             return (ln.p_out[:total], ln.p_tables[n:2 * n],
                     ln.p_tables[:n])
         total = int(ln.p_total[0])
@@ -596,7 +625,61 @@ class BatchEngine:
         with t.cuda.stream(ln.stream):
             ln.p_out[:total].copy_(ln.d_out[:total], non_blocking=True)
         ln.stream.synchronize()
+        if getattr(self.app, "request_log_every", 0):
+            self._emit_request_logs(ln, n)
         return ln.p_out[:total], ln.p_resp_off[:n], ln.p_resp_len[:n]
+
+    def _emit_request_logs(self, ln, n):
+        """Batch-aware log middleware: emit from a completed lane."""
+        dur_ms = (time.perf_counter() - getattr(ln, "submit_t",
+                                                time.perf_counter())) * 1e3
+        if ln.mode == "c":
+            roffs = ln.p_tables_np[n:2 * n]
+            rlens = ln.p_tables_np[:n]
+            total_out = int(ln.p_tables_np[2 * n])
+        else:
+            roffs = ln.p_resp_off[:n].numpy()
+            rlens = ln.p_resp_len[:n].numpy()
+            total_out = int(ln.p_total[0])
+        out_np = ln.p_out.numpy() if hasattr(ln.p_out, "numpy") \
+            else ln.p_out
+        self._emit_logs(n, ln.nbytes, total_out, ln.p_reqs.numpy(),
+                        ln.p_req_off[:n].numpy(), out_np, roffs, rlens,
+                        dur_ms)
+
+    def _emit_logs(self, n, bytes_in, bytes_out, req_np, offs, out_np,
+                   roffs, rlens, dur_ms):
+        """See App.enable_request_log: one BatchLog aggregate + every
+        Nth request as a RequestLog (method/URI from the request bytes,
+        status + correlation id from the response bytes)."""
+        logger = self.app.container.logger
+        if logger is None:
+            return
+        from ..http.middleware import RequestLog, rfc3339nano
+        every = self.app.request_log_every
+        logger.log_record(1, BatchLog(n, bytes_in, bytes_out, dur_ms))
+        for r in range(0, n, every):
+            raw = req_np[int(offs[r]):int(offs[r]) + 256].tobytes()
+            line = raw.split(b"\r\n", 1)[0].decode("latin1")
+            parts = line.split(" ")
+            method = parts[0] if parts else "?"
+            uri = parts[1] if len(parts) > 1 else "?"
+            resp = out_np[int(roffs[r]):int(roffs[r]) +
+                          min(256, int(rlens[r]))].tobytes()
+            status = 0
+            corr = ""
+            head = resp.split(b"\r\n\r\n", 1)[0].decode("latin1")
+            first = head.split("\r\n", 1)[0].split(" ")
+            if len(first) > 1 and first[1].isdigit():
+                status = int(first[1])
+            for h in head.split("\r\n"):
+                if h.startswith("X-Correlation-ID: "):
+                    corr = h[18:]
+                    break
+            logger.log_record(1, RequestLog(
+                trace_id=corr, start_time=rfc3339nano(),
+                response_time_us=dur_ms * 1000, method=method,
+                user_agent="", ip="", uri=uri, response=status))
 
     def process_device(self, d_reqs, d_req_off, d_req_len, n,
                        lane_idx=0, sync_host=True):

@@ -309,3 +309,29 @@ def test_if_none_match_304_cpu():
         b"GET /greet HTTP/1.1\r\nHost: h\r\n"
         b'If-None-Match: "deadbeef"\r\n\r\n'])[0]
     assert third.startswith(b"HTTP/1.1 200 OK")
+
+
+def test_batch_request_log_middleware():
+    """enable_request_log: one BatchLog aggregate + sampled RequestLog
+    records per processed batch, parsed from the rings."""
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+    from gofr_amd.testutil import MockLogger
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.enable_request_log(sample_every=4)
+    log = MockLogger()
+    app.container.logger = log
+    eng = BatchEngine(app, device="cpu", max_batch=32)
+    raws = [(b"POST /echo?q=1 HTTP/1.1\r\nHost: h\r\n"
+             b"Content-Type: application/json\r\n"
+             b"Content-Length: 9\r\n\r\n" + b'{"a":"b"}')] * 8
+    eng.process(raws)
+    out = log.stdout
+    assert '"batch": 8' in out           # BatchLog aggregate
+    assert '"uri": "/echo?q=1"' in out   # sampled RequestLog
+    assert '"response": 200' in out
+    assert out.count('"method": "POST"') == 2  # every 4th of 8
