@@ -1,0 +1,88 @@
+#!/usr/bin/env python3
+"""Persistent serving-cycle probe (round-2 architecture derisk).
+
+Measures the control-loop cost of a persistent kernel driven by
+SDMA-written flags against today's per-batch staged submission:
+per cycle the host enqueues an ingress H2D + a go-flag copy, the
+resident kernel wakes, streams the batch to the pinned egress ring,
+and publishes a done-flag the host spins on (plain memory reads).
+
+cycle(work=0) = pure control latency (flag in, barriers, flag out)
+cycle(work=1) - ideal(copy times) = residual overhead of the design
+"""
+
+import ctypes
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                ".."))
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+from gofr_amd import ops  # noqa: E402
+
+MB = 1 << 20
+NBYTES = 38 * MB          # config-2-shaped batch
+NBLOCKS = 64
+NBATCH = 40
+
+
+def run(hip, do_work):
+    dev = torch.device("cuda:0")
+    d_in = torch.zeros(NBYTES, dtype=torch.uint8, device=dev)
+    p_in = torch.zeros(NBYTES, dtype=torch.uint8).pin_memory()
+    p_out = hip.host_alloc(NBYTES)
+    d_go = torch.zeros(1, dtype=torch.int64, device=dev)
+    d_barrier = torch.zeros(2, dtype=torch.int32, device=dev)
+    p_done = hip.host_alloc(8, dtype=np.uint64)
+    p_done_np = p_done.numpy()
+    p_serial = torch.zeros(1, dtype=torch.int64).pin_memory()
+
+    s_k = torch.cuda.Stream()
+    s_in = torch.cuda.Stream()
+    hip.lib.gofr_launch_persist_cycle.restype = ctypes.c_int
+    hip.lib.gofr_launch_persist_cycle.argtypes = \
+        [ctypes.c_void_p] * 5 + [ctypes.c_longlong, ctypes.c_int,
+                                 ctypes.c_int, ctypes.c_void_p,
+                                 ctypes.c_int]
+    rc = hip.lib.gofr_launch_persist_cycle(
+        ctypes.c_void_p(s_k.cuda_stream),
+        ctypes.c_void_p(d_go.data_ptr()),
+        ctypes.c_void_p(p_done.data_ptr()),
+        ctypes.c_void_p(d_in.data_ptr()),
+        ctypes.c_void_p(p_out.data_ptr()),
+        NBYTES, NBATCH, do_work,
+        ctypes.c_void_p(d_barrier.data_ptr()), NBLOCKS)
+    assert rc == 0, rc
+
+    cyc = []
+    for b in range(1, NBATCH + 1):
+        t0 = time.perf_counter()
+        with torch.cuda.stream(s_in):
+            d_in.copy_(p_in, non_blocking=True)       # ingress payload
+            p_serial[0] = b
+            d_go.copy_(p_serial, non_blocking=True)   # go-flag (after)
+        while p_done_np[0] < b:
+            pass
+        cyc.append((time.perf_counter() - t0) * 1e3)
+    torch.cuda.synchronize()
+    return cyc
+
+
+def main():
+    hip = ops.HipOps()
+    for do_work in (0, 1):
+        cyc = run(hip, do_work)
+        steady = np.asarray(cyc[5:])
+        label = "work(38MB in + 38MB egress)" if do_work else "control-only"
+        print(f"persistent cycle [{label}]: "
+              f"mean {steady.mean():.3f} ms  p50 {np.percentile(steady,50):.3f}"
+              f"  p99 {np.percentile(steady,99):.3f}  min {steady.min():.3f}")
+    print("today's staged pipeline step (same bytes): ~1.27 ms; "
+          "ingress SDMA ~0.67 + kernel egress ~0.70 at 54 GB/s")
+
+
+if __name__ == "__main__":
+    main()

@@ -1504,6 +1504,78 @@ k_grpc_echo(const uint8_t* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
+// k_persist_cycle — persistent serving-cycle probe (the round-2
+// architecture candidate: one resident kernel replaces per-batch
+// launches and cross-stream dependencies). G co-resident blocks loop
+// over nbatch batches; per batch:
+//   block 0 spins (bounded) on d_go until the host's SDMA writes the
+//   batch serial (enqueued after the payload copy, same FIFO), a grid
+//   barrier releases all blocks, every block streams its share of the
+//   batch to the pinned egress ring (the link-bound work), a second
+//   barrier, then block 0 publishes the serial to pinned p_done.
+// All spins are iteration-bounded: the kernel always terminates.
+// Driver: benchmarks/persistent_probe.py.
+// ---------------------------------------------------------------------------
+__device__ void grid_barrier(int* counter, int* gen_ptr, int nblocks) {
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        const int my_gen = __hip_atomic_load(gen_ptr, __ATOMIC_ACQUIRE,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+        if (atomicAdd(counter, 1) == nblocks - 1) {
+            __hip_atomic_store(counter, 0, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            __hip_atomic_store(gen_ptr, my_gen + 1, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_AGENT);
+        } else {
+            for (long i = 0; i < (1L << 22); ++i) {
+                if (__hip_atomic_load(gen_ptr, __ATOMIC_ACQUIRE,
+                                      __HIP_MEMORY_SCOPE_AGENT) != my_gen)
+                    break;
+                __builtin_amdgcn_s_sleep(8);
+            }
+        }
+    }
+    __syncthreads();
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_persist_cycle(const unsigned long long* __restrict__ d_go,
+                volatile unsigned long long* p_done,
+                const uint8_t* __restrict__ d_in,
+                uint8_t* __restrict__ p_out,
+                long nbytes, int nbatch, int do_work,
+                int* __restrict__ d_barrier) {
+    const int nblocks = gridDim.x;
+    for (int b = 1; b <= nbatch; ++b) {
+        if (blockIdx.x == 0 && threadIdx.x == 0) {
+            for (long i = 0; i < (1L << 20); ++i) {
+                if (__hip_atomic_load(d_go, __ATOMIC_ACQUIRE,
+                                      __HIP_MEMORY_SCOPE_SYSTEM) >=
+                    (unsigned long long)b)
+                    break;
+                __builtin_amdgcn_s_sleep(16);
+            }
+        }
+        grid_barrier(d_barrier, d_barrier + 1, nblocks);
+        if (do_work) {
+            // link-bound egress: stream the batch to the pinned ring
+            const long nv = nbytes >> 4;
+            const uint4* src = (const uint4*)d_in;
+            uint4* dst = (uint4*)p_out;
+            const long stride = (long)nblocks * BLOCK_THREADS;
+            for (long i = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+                 i < nv; i += stride)
+                dst[i] = src[i];
+        }
+        grid_barrier(d_barrier, d_barrier + 1, nblocks);
+        if (blockIdx.x == 0 && threadIdx.x == 0)
+            __hip_atomic_store(p_done, (unsigned long long)b,
+                               __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // k_compact — gather response slots into a contiguous 16B-aligned stream.
 // `out` may be DEVICE memory (classic compact; offsets are the exclusive
 // cumsum of round16(resp_len)) or PINNED HOST memory: the kernel then IS
@@ -2019,5 +2091,18 @@ extern "C" int gofr_launch_mfma_probe(void* stream, const void* A,
     hipLaunchKernelGGL(k_mfma_i8_probe, dim3(1), dim3(64), 0,
                        (hipStream_t)stream,
                        (const int8_t*)A, (const int8_t*)B, (int32_t*)D);
+    return (int)hipGetLastError();
+}
+
+extern "C" int gofr_launch_persist_cycle(
+        void* stream, const void* d_go, void* p_done, const void* d_in,
+        void* p_out, long long nbytes, int nbatch, int do_work,
+        void* d_barrier, int nblocks) {
+    hipLaunchKernelGGL(k_persist_cycle, dim3(nblocks), dim3(BLOCK_THREADS),
+                       0, (hipStream_t)stream,
+                       (const unsigned long long*)d_go,
+                       (volatile unsigned long long*)p_done,
+                       (const uint8_t*)d_in, (uint8_t*)p_out,
+                       (long)nbytes, nbatch, do_work, (int*)d_barrier);
     return (int)hipGetLastError();
 }
