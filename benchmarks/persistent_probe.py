@@ -29,11 +29,12 @@ NBLOCKS = 64
 NBATCH = 40
 
 
-def run(hip, do_work):
+def run(hip, do_work, dst="host", nbytes=NBYTES):
     dev = torch.device("cuda:0")
-    d_in = torch.zeros(NBYTES, dtype=torch.uint8, device=dev)
-    p_in = torch.zeros(NBYTES, dtype=torch.uint8).pin_memory()
-    p_out = hip.host_alloc(NBYTES)
+    d_in = torch.zeros(nbytes, dtype=torch.uint8, device=dev)
+    p_in = torch.zeros(nbytes, dtype=torch.uint8).pin_memory()
+    p_out = (hip.host_alloc(nbytes) if dst == "host"
+             else torch.zeros(nbytes, dtype=torch.uint8, device=dev))
     d_go = torch.zeros(1, dtype=torch.int64, device=dev)
     d_barrier = torch.zeros(2, dtype=torch.int32, device=dev)
     p_done = hip.host_alloc(8, dtype=np.uint64)
@@ -53,7 +54,7 @@ def run(hip, do_work):
         ctypes.c_void_p(p_done.data_ptr()),
         ctypes.c_void_p(d_in.data_ptr()),
         ctypes.c_void_p(p_out.data_ptr()),
-        NBYTES, NBATCH, do_work,
+        nbytes, NBATCH, do_work,
         ctypes.c_void_p(d_barrier.data_ptr()), NBLOCKS)
     assert rc == 0, rc
 
@@ -73,10 +74,13 @@ def run(hip, do_work):
 
 def main():
     hip = ops.HipOps()
-    for do_work in (0, 1):
-        cyc = run(hip, do_work)
+    cases = [(0, "host", NBYTES, "control-only"),
+             (1, "dev", NBYTES, "work 38MB -> DEVICE dst"),
+             (1, "host", 4 * MB, "work 4MB -> host dst"),
+             (1, "host", NBYTES, "work 38MB -> host dst")]
+    for do_work, dst, nb, label in cases:
+        cyc = run(hip, do_work, dst, nb)
         steady = np.asarray(cyc[5:])
-        label = "work(38MB in + 38MB egress)" if do_work else "control-only"
         print(f"persistent cycle [{label}]: "
               f"mean {steady.mean():.3f} ms  p50 {np.percentile(steady,50):.3f}"
               f"  p99 {np.percentile(steady,99):.3f}  min {steady.min():.3f}")
