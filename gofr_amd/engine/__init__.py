@@ -168,10 +168,11 @@ class _Lane:
         # host_needed) — one D2H moves the whole per-batch result set
         self.d_tables = t.zeros(2 * nb + 2, dtype=t.int32, device=dev)
         if hip is not None:
-            self.p_tables = hip.host_alloc((2 * nb + 2) * 4,
+            # +4: [2n+2] carries the flagged pipeline's done serial
+            self.p_tables = hip.host_alloc((2 * nb + 4) * 4,
                                            dtype=np.int32)
         else:
-            self.p_tables = t.zeros(2 * nb + 2, dtype=t.int32).pin_memory()
+            self.p_tables = t.zeros(2 * nb + 4, dtype=t.int32).pin_memory()
         self.p_tables_np = self.p_tables.numpy()
         self.p_req_off_np = self.p_req_off.numpy()
         self.p_serial = t.zeros(1, dtype=t.int64).pin_memory()
@@ -238,6 +239,7 @@ class BatchEngine:
             pass
         self._pump_done = None
         self._use_pump = os.environ.get("GOFR_PUMP", "1") != "0"
+        self._flagged = os.environ.get("GOFR_FLAGGED", "0") == "1"
         if self.device is not None:
             self.hip = ops.HipOps()  # raises if extension missing
             self._alloc_device()
@@ -371,10 +373,12 @@ class BatchEngine:
         if ln.c_args is not None and ln.graph_key == (n, nbytes):
             ln.mode = "c"
             ln.p_req_off_np[n] = signed  # seed rides the offsets tail
-            if self._use_gate:
+            if self._use_gate or self._flagged:
                 self._gate_serial += 1
                 ln.p_serial_np[0] = self._gate_serial
                 ln.c_args.serial = self._gate_serial
+                if self._flagged:
+                    ln.serial_flag = self._gate_serial & 0x7fffffff
             if self._use_pump:
                 serial = self.hip.lib.gofr_pump_submit(
                     ctypes.byref(ln.c_args))
@@ -497,13 +501,22 @@ class BatchEngine:
         a.auth_env_off, a.auth_env_len = self.program.auth_env
         a.gzip_min = self.app.gzip_min_size or 0
         a.etag_on = 1 if getattr(self.app, "etag_on", False) else 0
-        if self._use_gate:
+        if self._use_gate or self._flagged:
             a.d_flag = self.d_flag.data_ptr()
             a.p_serial = ln.p_serial.data_ptr()
         else:
             a.d_flag = 0
             a.p_serial = 0
         a.serial = 0
+        a.flagged = 1 if self._flagged else 0
+        if self._flagged:
+            # event-free channel pipeline: the whole chain (gate ->
+            # kernels -> egress -> k_done) rides one of two channel
+            # streams per lane parity; ingress stays on the shared
+            # SDMA stream
+            chan = (self.s_k, self.s_k2)[lane_idx % 2]
+            a.s_k = chan.cuda_stream
+            a.s_out = chan.cuda_stream
         a.p_out = ln.p_out.data_ptr()
         a.n = n
         a.rslot = self.rslot
@@ -563,7 +576,18 @@ class BatchEngine:
         t = self.torch
         ln = self.lanes[lane_idx]
         n = ln.n
-        if ln.mode == "c" and ln.serial is not None:
+        if ln.mode == "c" and self._flagged:
+            # event-free completion: k_done published the batch serial
+            # into the pinned tables; spin on plain memory
+            cell = 2 * n + 2
+            want = ln.serial_flag
+            tbl = ln.p_tables_np
+            while tbl[cell] != want:
+                pass
+            if self._use_pump and self.hip.lib.gofr_pump_err():
+                raise RuntimeError(
+                    f"pump error: hipError {self.hip.lib.gofr_pump_err()}")
+        elif ln.mode == "c" and ln.serial is not None:
             # spin on the pump's published serial: a plain host-memory
             # read, so the serving thread never touches runtime locks
             # (the pump's completion thread absorbs the event-wake

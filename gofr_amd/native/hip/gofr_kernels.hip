@@ -1418,7 +1418,7 @@ k_mfma_i8_probe(const int8_t* __restrict__ A,   // [16][64]
 extern "C" __global__ void __launch_bounds__(SCAN_THREADS)
 k_padscan(int32_t* __restrict__ tables,
           const int32_t* __restrict__ host_needed,
-          int n) {
+          int n, int32_t* __restrict__ host_mirror) {
     __shared__ int32_t partials[SCAN_THREADS];
     const int tid = threadIdx.x;
     const int chunk = (n + SCAN_THREADS - 1) / SCAN_THREADS;
@@ -1440,11 +1440,33 @@ k_padscan(int32_t* __restrict__ tables,
     // exclusive offsets for this thread's chunk
     int32_t run = (tid > 0) ? partials[tid - 1] : 0;
     for (int i = s; i < e; ++i) {
+        const int32_t len_i = tables[i];
         tables[n + i] = run;
-        run += (tables[i] + 15) & ~15;
+        if (host_mirror) {
+            host_mirror[i] = len_i;
+            host_mirror[n + i] = run;
+        }
+        run += (len_i + 15) & ~15;
     }
-    if (tid == SCAN_THREADS - 1) tables[2 * n] = partials[tid];
-    if (tid == 0) tables[2 * n + 1] = *host_needed;
+    if (tid == SCAN_THREADS - 1) {
+        tables[2 * n] = partials[tid];
+        if (host_mirror) host_mirror[2 * n] = partials[tid];
+    }
+    if (tid == 0) {
+        tables[2 * n + 1] = *host_needed;
+        if (host_mirror) host_mirror[2 * n + 1] = *host_needed;
+    }
+}
+
+// k_done — publish the batch serial to pinned host memory (system
+// release; flagged pipeline's completion signal — replaces the event
+// machinery entirely). Launched after k_compact on the same stream,
+// so kernel-boundary coherence makes all prior host writes (compact's
+// egress, padscan's table mirror) visible before the serial lands.
+extern "C" __global__ void k_done(int32_t* cell, int serial) {
+    if (threadIdx.x == 0)
+        __hip_atomic_store(cell, serial, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
 }
 
 // ---------------------------------------------------------------------------
@@ -1649,6 +1671,12 @@ typedef struct {
     void* d_flag;
     const void* p_serial;        // pinned cell holding `serial`
     unsigned long long serial;
+    // flagged pipeline: no events — k_gate releases the kernel chain,
+    // k_padscan mirrors the tables to pinned host, k_done publishes
+    // the serial into p_tables[2n+2]. s_k carries the WHOLE chain
+    // (kernels + egress), so batches on alternating channel streams
+    // overlap each other and the SDMA ingress.
+    int flagged;
 } GofrSubmitArgs;
 
 // k_gate — single-wave stream gate: spins (system-scope acquire
@@ -1778,14 +1806,23 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     }
     if ((rc = hipGetLastError())) return (int)rc;
     hipLaunchKernelGGL(k_padscan, dim3(1), dim3(SCAN_THREADS), 0, s_k,
-                       tables, (const int32_t*)a->d_host_needed, n);
+                       tables, (const int32_t*)a->d_host_needed, n,
+                       a->flagged ? (int32_t*)a->p_tables : nullptr);
     if ((rc = hipGetLastError())) return (int)rc;
-    rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);
-    if (rc) return (int)rc;
+    if (a->flagged) {
+        // egress continues ON THE CHANNEL STREAM (caller set s_out ==
+        // s_k); no event handoff
+        s_out = s_k;
+    } else {
+        rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);
+        if (rc) return (int)rc;
+    }
     g_submit_us[2] += now_us() - t0; t0 = now_us();
     // ---- egress -----------------------------------------------------------
-    rc = hipStreamWaitEvent(s_out, (hipEvent_t)a->ev_k, 0);
-    if (rc) return (int)rc;
+    if (!a->flagged) {
+        rc = hipStreamWaitEvent(s_out, (hipEvent_t)a->ev_k, 0);
+        if (rc) return (int)rc;
+    }
     int cblocks = blocks;
     {
         static int cap = 0;
@@ -1811,6 +1848,14 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->d_resp, tables, tables + n,
                            (uint8_t*)a->p_out, n, a->rslot);
         if ((rc = hipGetLastError())) return (int)rc;
+    }
+    if (a->flagged) {
+        hipLaunchKernelGGL(k_done, dim3(1), dim3(64), 0, s_out,
+                           (int32_t*)a->p_tables + 2 * n + 2,
+                           (int)(a->serial & 0x7fffffff));
+        rc = hipGetLastError();
+        g_submit_us[3] += now_us() - t0;
+        return (int)rc;
     }
     rc = hipMemcpyAsync(a->p_tables, a->d_tables, (size_t)(2 * n + 2) * 4,
                         hipMemcpyDeviceToHost, s_out);

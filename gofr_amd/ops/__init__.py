@@ -94,6 +94,7 @@ class GofrSubmitArgs(ctypes.Structure):
         ("d_flag", ctypes.c_void_p),
         ("p_serial", ctypes.c_void_p),
         ("serial", ctypes.c_uint64),
+        ("flagged", ctypes.c_int),
     ]
 
 
