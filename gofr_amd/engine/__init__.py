@@ -239,7 +239,7 @@ class BatchEngine:
             pass
         self._pump_done = None
         self._use_pump = os.environ.get("GOFR_PUMP", "1") != "0"
-        self._flagged = os.environ.get("GOFR_FLAGGED", "0") == "1"
+        self._flagged = os.environ.get("GOFR_FLAGGED", "1") == "1"
         if self.device is not None:
             self.hip = ops.HipOps()  # raises if extension missing
             self._alloc_device()
