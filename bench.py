@@ -472,10 +472,6 @@ def main():
                       pipeline=pipeline)
 
     n_grpc = int(batch * args.grpc_frac)
-    if n_grpc and world == 1:
-        # the mixed loop chains its gRPC stage off ln.e_k, which the
-        # event-free flagged pipeline does not record
-        eng._flagged = False
         if have_gpu:
             elapsed, times, _tab = run_config5(
                 eng, payloads[:batch - n_grpc], args.steps, args.warmup,

@@ -1811,7 +1811,11 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     if ((rc = hipGetLastError())) return (int)rc;
     if (a->flagged) {
         // egress continues ON THE CHANNEL STREAM (caller set s_out ==
-        // s_k); no event handoff
+        // s_k); no event handoff — but still RECORD e_k (cheap
+        // kernel-signal) so auxiliary consumers (the mixed
+        // HTTP+gRPC loop) can chain off the kernel stage
+        rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);
+        if (rc) return (int)rc;
         s_out = s_k;
     } else {
         rc = hipEventRecord((hipEvent_t)a->ev_k, s_k);

@@ -118,3 +118,58 @@ def test_native_extension_is_loaded():
     assert ops.lib is not None
     # the so is the in-tree artifact
     assert _SO_PATH.endswith("gofr_amd/_gofr_hip.so")
+
+
+def test_armed_flagged_pipeline_matches_mirror():
+    """The production serving path (armed lanes, event-free flagged
+    pipeline — the default) byte-matches the CPU mirrors across
+    repeated submits."""
+    import numpy as np
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine, pack_batch
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    gpu = BatchEngine(app, device="cuda", max_batch=512, pipeline=2)
+    cpu = BatchEngine(app, device="cpu", max_batch=512)
+    cpu._seed = gpu._seed
+
+    raws = []
+    for i in range(512):
+        if i % 2 == 0:
+            body = b'{"i":' + str(i).encode() + b"}"
+            raws.append(b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+                        b"Content-Type: application/json\r\n"
+                        b"Content-Length: " + str(len(body)).encode() +
+                        b"\r\n\r\n" + body)
+        else:
+            raws.append(b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n")
+    buf, offs, lens = pack_batch(raws)
+    n = len(lens)
+    nbytes = int(offs[-1] + lens[-1])
+    import torch
+    for ln in gpu.lanes:
+        ln.p_reqs[:nbytes] = torch.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n] = torch.from_numpy(offs)
+        ln.p_req_len[:n] = torch.from_numpy(lens)
+    for li in range(len(gpu.lanes)):
+        assert gpu.capture_graph(n, nbytes, li)
+    assert gpu._flagged  # the default path under test
+
+    for step in range(4):  # alternate lanes, repeated submits
+        lane = step % len(gpu.lanes)
+        gpu.submit(n, nbytes, lane)
+        out_t, roff_t, rlen_t = gpu.complete(lane)
+        c_out, c_roffs, c_rlens = cpu.process_packed(buf, offs, lens)
+        g_out = out_t.numpy()
+        roffs = roff_t.numpy()
+        rlens = rlen_t.numpy()
+        assert (rlens == c_rlens).all()
+        for i in range(n):
+            g = g_out[int(roffs[i]):int(roffs[i]) + int(rlens[i])]
+            c = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
+            assert (g == c).all(), f"step {step} req {i}"
