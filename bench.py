@@ -472,6 +472,7 @@ def main():
                       pipeline=pipeline)
 
     n_grpc = int(batch * args.grpc_frac)
+    if n_grpc and world == 1:
         if have_gpu:
             elapsed, times, _tab = run_config5(
                 eng, payloads[:batch - n_grpc], args.steps, args.warmup,
