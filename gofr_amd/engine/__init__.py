@@ -105,14 +105,17 @@ def make_batch(payloads: list[bytes], slot: int):
 
 
 class _Lane:
-    """One in-flight batch: a device/pinned buffer set + stage events.
+    """One in-flight batch: a device/pinned buffer set + events.
 
-    Lanes share the engine's three stage streams (s_in H2D, s_k kernels,
-    s_out D2H) — measured on MI355X, eager SDMA copies on dedicated
-    per-direction streams run full duplex at ~96 GB/s aggregate while
-    graph-captured copies reach only ~85 and compute-driven host
-    dereference ~59 (benchmarks/overlap_probe.py) — so ingress of batch
-    i+1 overlaps egress of batch i-1 and the kernels of batch i.
+    In the default flagged pipeline a lane's batch is ONE SDMA ingress
+    copy on the shared s_in (header+payload, serial flag trailing in
+    the same FIFO) followed by the whole kernel chain + egress sweep on
+    one of two channel streams, gated by k_gate and completed by
+    k_done's serial store into the pinned tables — no events on the
+    hot path (every measured event/stream handoff cost 0.1-0.5 ms;
+    see profiles/SUMMARY.md "host-link story"). The event-based staged
+    path (GOFR_FLAGGED=0) and the eager per-lane-stream fallback remain
+    for A/B and for unarmed batches.
     """
 
     def __init__(self, t, dev, nb, max_bytes, rslot, hip=None,
@@ -121,7 +124,6 @@ class _Lane:
         self.event = t.cuda.Event()
         self.e_in = t.cuda.Event()
         self.e_k = t.cuda.Event()
-        self.e_out = t.cuda.Event()
         # ingress block: [off (nb+1)*int64 | len nb*int32 | serial u64 |
         # pad16 | request bytes] — ONE contiguous region so the whole
         # batch ingress is ONE SDMA copy (each queued SDMA op costs a
