@@ -801,11 +801,13 @@ class GPUServer:
     """
 
     def __init__(self, app, port: int, batch_window_us: int = 200,
-                 max_batch: int = 8192):
+                 max_batch: int = 8192, arm_chunk: int = 1024):
         self.app = app
         self.port = port
-        self.engine = BatchEngine(app, max_batch=max_batch)
+        self.engine = BatchEngine(app, max_batch=max_batch,
+                                  pipeline=4)
         self.batch_window_us = batch_window_us
+        self.arm_chunk = arm_chunk
         self._core = None
         self._thread = None
         self._stop = None
@@ -823,40 +825,77 @@ class GPUServer:
 
     def _serve_loop(self):
         eng = self.engine
+        if eng.device is None:
+            self._serve_loop_cpu()
+            return
+        # Armed fixed-shape serving: harvests are padded (len-0 slots
+        # emit nothing — FL_EMPTY) to a fixed chunk so every batch runs
+        # the native flagged pipeline; chunks round-robin over lanes so
+        # ingress/kernels/egress of consecutive chunks overlap.
+        from collections import deque
+        P = len(eng.lanes)
+        CH = min(self.arm_chunk, eng.max_batch)
+        cap = CH * eng.slot
+        armed = True
+        try:
+            for li in range(P):
+                eng.capture_graph(CH, cap, li)
+        except Exception:  # noqa: BLE001 — fall back to dynamic batches
+            armed = False
+        conn_ids = [np.zeros(CH if armed else eng.max_batch, np.uint64)
+                    for _ in range(P)]
+        len_views = [ln.p_req_len.numpy() for ln in eng.lanes]
+        off_views = [ln.p_req_off.numpy() for ln in eng.lanes]
+        free = deque(range(P))
+        inflight = deque()
+        while not self._stop.is_set():
+            progressed = False
+            if free:
+                L = free.popleft()
+                ln = eng.lanes[L]
+                maxn = CH if armed else eng.max_batch
+                bufcap = cap if armed else eng.max_bytes
+                n, nbytes = self._core.harvest(
+                    ln.p_reqs.data_ptr(), bufcap,
+                    ln.p_req_off.data_ptr(), ln.p_req_len.data_ptr(),
+                    conn_ids[L].ctypes.data, maxn, self.batch_window_us)
+                if n:
+                    progressed = True
+                    if armed:
+                        len_views[L][n:CH] = 0
+                        off_views[L][n:CH] = 0
+                        eng.submit(CH, cap, L)
+                    else:
+                        eng.submit(n, nbytes, L)
+                    inflight.append((L, n))
+                else:
+                    free.appendleft(L)
+            if inflight and (not free or not progressed):
+                L, n = inflight.popleft()
+                out_t, roff_t, rlen_t = eng.complete(L)
+                self._core.send(conn_ids[L].ctypes.data, n,
+                                out_t.data_ptr(), roff_t.data_ptr(),
+                                rlen_t.data_ptr())
+                free.append(L)
+
+    def _serve_loop_cpu(self):
+        eng = self.engine
         nb = eng.max_batch
         conn_ids = np.zeros(nb, np.uint64)
         conn_ptr = conn_ids.ctypes.data
-        gpu = eng.device is not None
-        if gpu:
-            ln = eng.lanes[0]
-            buf_ptr = ln.p_reqs.data_ptr()
-            cap = eng.max_bytes
-            off_ptr = ln.p_req_off.data_ptr()
-            len_ptr = ln.p_req_len.data_ptr()
-        else:
-            buf_np = np.zeros(eng.max_bytes, np.uint8)
-            off_np = np.zeros(nb, np.int64)
-            len_np = np.zeros(nb, np.int32)
-            buf_ptr = buf_np.ctypes.data
-            cap = eng.max_bytes
-            off_ptr = off_np.ctypes.data
-            len_ptr = len_np.ctypes.data
+        buf_np = np.zeros(eng.max_bytes, np.uint8)
+        off_np = np.zeros(nb, np.int64)
+        len_np = np.zeros(nb, np.int32)
         while not self._stop.is_set():
             n, nbytes = self._core.harvest(
-                buf_ptr, cap, off_ptr, len_ptr, conn_ptr, nb,
-                self.batch_window_us)
+                buf_np.ctypes.data, eng.max_bytes, off_np.ctypes.data,
+                len_np.ctypes.data, conn_ptr, nb, self.batch_window_us)
             if n == 0:
                 continue
-            if gpu:
-                eng.submit(n, nbytes, 0)
-                out_t, roff_t, rlen_t = eng.complete(0)
-                self._core.send(conn_ptr, n, out_t.data_ptr(),
-                                roff_t.data_ptr(), rlen_t.data_ptr())
-            else:
-                out, roffs, rlens = eng.process_packed(
-                    buf_np, off_np[:n], len_np[:n])
-                self._core.send(conn_ptr, n, out.ctypes.data,
-                                roffs.ctypes.data, rlens.ctypes.data)
+            out, roffs, rlens = eng.process_packed(
+                buf_np, off_np[:n], len_np[:n])
+            self._core.send(conn_ptr, n, out.ctypes.data,
+                            roffs.ctypes.data, rlens.ctypes.data)
 
     def stop(self):
         if self._stop is not None:

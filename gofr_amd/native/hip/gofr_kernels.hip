@@ -62,6 +62,7 @@
 #define FL_BODY_INVALID 32
 #define FL_ACCEPT_GZIP 64
 #define FL_AUTH_FAIL 128
+#define FL_EMPTY 256  /* padding slot (len 0): emit no response bytes */
 
 // handler kinds (handler table rows: [kind, arg_off, arg_len, status])
 #define HK_HOST 0
@@ -145,6 +146,17 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     const int lane = lane_id();
     const uint8_t* base = reqs + req_off[req];
     int len = req_len[req];
+    if (len == 0) {
+        // padding slot (fixed-shape armed batches): no response
+        if (lane_id() == 0) {
+            int32_t* Fe = fields + (size_t)req * NF;
+            for (int i = 0; i < NF; ++i) Fe[i] = 0;
+            Fe[FI_FLAGS] = FL_EMPTY;
+            Fe[FI_KIND] = HK_STATIC;
+            Fe[FI_ROUTE] = -1;
+        }
+        return;
+    }
     bool oversized = false;
     if (len > MAX_SLOT) {
         // larger than the LDS working set: the host trampoline parses
@@ -777,6 +789,14 @@ __device__ __forceinline__ void respond_impl(
     const int kind = F[FI_KIND];
     int status = F[FI_STATUS];
     const int flags = F[FI_FLAGS];
+    if (flags & FL_EMPTY) {
+        if (lane == 0) {
+            F[FI_RESP_LEN] = 0;
+            F[FI_RESP_OFF] = (int)((size_t)req * rslot);
+            resp_len_out[req] = 0;
+        }
+        return;
+    }
     const bool keep = (flags & FL_KEEP_ALIVE) != 0;
     const bool is_options = (flags & FL_IS_OPTIONS) != 0;
 
@@ -1246,7 +1266,7 @@ k_auth(const uint8_t* __restrict__ reqs,
     if (req >= n) return;
     int32_t* F = fields + (size_t)req * NF;
     const int flags = F[FI_FLAGS];
-    if (flags & (FL_ERR_PARSE | FL_IS_OPTIONS)) return;  // OPTIONS exempt
+    if (flags & (FL_ERR_PARSE | FL_IS_OPTIONS | FL_EMPTY)) return;
     const uint8_t* base = reqs + req_off[req];
     const int aoff = F[FI_AUTH_OFF], alen = F[FI_AUTH_LEN];
     bool ok = false;

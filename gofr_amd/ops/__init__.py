@@ -49,6 +49,7 @@ FL_IS_OPTIONS = 16
 FL_BODY_INVALID = 32
 FL_ACCEPT_GZIP = 64
 FL_AUTH_FAIL = 128
+FL_EMPTY = 256  # padding slot (len 0): emit no response bytes
 
 HK_HOST = 0
 HK_ECHO_JSON = 1
@@ -300,6 +301,12 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
     for r in range(n):
         base = int(req_off[r])
         ln = int(req_len[r])
+        if ln == 0:
+            F = fields[r]
+            F[FI_FLAGS] = FL_EMPTY
+            F[FI_KIND] = HK_STATIC
+            F[FI_ROUTE] = -1
+            continue
         oversized = ln > MAX_SLOT
         if oversized:
             ln = MAX_SLOT
@@ -550,6 +557,11 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
         kind = int(F[FI_KIND])
         status = int(F[FI_STATUS])
         flags = int(F[FI_FLAGS])
+        if flags & FL_EMPTY:
+            resp_len[r] = 0
+            fields[r][FI_RESP_LEN] = 0
+            fields[r][FI_RESP_OFF] = r * rslot
+            continue
         keep = bool(flags & FL_KEEP_ALIVE)
         is_options = bool(flags & FL_IS_OPTIONS)
         body_src = b""
@@ -763,7 +775,7 @@ def cpu_auth(reqs: np.ndarray, req_off: np.ndarray, fields: np.ndarray,
     for r in range(len(fields)):
         F = fields[r]
         flags = int(F[FI_FLAGS])
-        if flags & (FL_ERR_PARSE | FL_IS_OPTIONS):
+        if flags & (FL_ERR_PARSE | FL_IS_OPTIONS | FL_EMPTY):
             continue
         base = int(req_off[r])
         aoff, alen = int(F[FI_AUTH_OFF]), int(F[FI_AUTH_LEN])
