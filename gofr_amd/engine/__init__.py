@@ -839,6 +839,12 @@ class GPUServer:
         armed = True
         try:
             for li in range(P):
+                ln = eng.lanes[li]
+                # arm over a fully-padded batch: the warmup must not
+                # read uninitialized ring contents (garbage offsets
+                # would fault the parse kernel)
+                ln.p_req_len.numpy()[:] = 0
+                ln.p_req_off.numpy()[:] = 0
                 eng.capture_graph(CH, cap, li)
         except Exception:  # noqa: BLE001 — fall back to dynamic batches
             armed = False
