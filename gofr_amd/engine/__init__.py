@@ -584,8 +584,15 @@ class BatchEngine:
             cell = 2 * n + 2
             want = ln.serial_flag
             tbl = ln.p_tables_np
+            spins = 0
+            t0 = time.perf_counter()
             while tbl[cell] != want:
-                pass
+                spins += 1
+                if (spins & 0xFFFFF) == 0 and \
+                        time.perf_counter() - t0 > 30:
+                    raise RuntimeError(
+                        "flagged completion timeout (lane serial "
+                        f"{want}, cell {int(tbl[cell])})")
             if self._use_pump and self.hip.lib.gofr_pump_err():
                 raise RuntimeError(
                     f"pump error: hipError {self.hip.lib.gofr_pump_err()}")

@@ -1969,6 +1969,9 @@ static void* pump_cmp_main(void* _) {
 }
 
 extern "C" int gofr_pump_start() {
+    // idempotent: several engines in one process share the pump (a
+    // re-init would memset over live worker threads)
+    if (g_pump.running) return 0;
     memset((void*)&g_pump, 0, sizeof(g_pump));
     pthread_mutex_init(&g_pump.mu, nullptr);
     pthread_cond_init(&g_pump.cv, nullptr);
