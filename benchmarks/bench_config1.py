@@ -87,7 +87,10 @@ def run_native_client(port, conns, depth, seconds, threads=4,
     out = subprocess.run(
         [exe, "127.0.0.1", str(port), str(conns), str(depth),
          str(seconds), str(threads), str(n_addrs)],
-        check=True, capture_output=True, text=True)
+        capture_output=True, text=True)
+    if out.returncode != 0:
+        raise RuntimeError(f"loadgen rc={out.returncode}: "
+                           f"{out.stderr[-400:]}")
     return json.loads(out.stdout.strip())
 
 
