@@ -106,7 +106,7 @@ public:
         if (!running_) return;
         running_ = false;
         if (loop_.joinable()) loop_.join();
-        for (auto& kv : conns_) ::close(kv.second.fd);
+        for (auto& kv : conns_) ::close(kv.second->fd);
         conns_.clear();
         if (listen_fd_ >= 0) ::close(listen_fd_);
         if (ep_ >= 0) ::close(ep_);
@@ -137,7 +137,7 @@ public:
     void queue_writes(const uint64_t* cids, const int* rows, int nrows,
                       const uint8_t* out, const int32_t* roffs,
                       const int32_t* rlens) {
-        std::lock_guard<std::mutex> lk(wmu_);
+        std::lock_guard<std::recursive_mutex> lk(wmu_);
         for (int k = 0; k < nrows; ++k) {
             const int i = rows[k];
             auto it = conn_index_.find(cids[i]);
@@ -171,7 +171,7 @@ private:
                 }
                 auto it = conns_.find(id);
                 if (it == conns_.end()) continue;
-                Conn& c = it->second;
+                Conn& c = *it->second;
                 if (events[i].events & (EPOLLHUP | EPOLLERR)) {
                     close_conn(c);
                     continue;
@@ -196,11 +196,13 @@ private:
             int one = 1;
             setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
             const uint64_t id = ((uint64_t)idx_ << 48) | next_id_++;
-            Conn& c = conns_[id];
+            auto up = std::make_unique<Conn>();
+            Conn& c = *up;
+            conns_[id] = std::move(up);
             c.fd = fd;
             c.id = id;
             {
-                std::lock_guard<std::mutex> lk(wmu_);
+                std::lock_guard<std::recursive_mutex> lk(wmu_);
                 conn_index_[id] = &c;
             }
             epoll_event ev{};
@@ -259,16 +261,17 @@ private:
     void drain_pending_writes() {
         std::vector<uint64_t> ids;
         {
-            std::lock_guard<std::mutex> lk(wmu_);
+            std::lock_guard<std::recursive_mutex> lk(wmu_);
             ids.swap(pending_writes_);
         }
         for (uint64_t id : ids) {
             auto it = conns_.find(id);
-            if (it != conns_.end()) flush_conn(it->second);
+            if (it != conns_.end()) flush_conn(*it->second);
         }
     }
 
     void flush_conn(Conn& c) {
+        std::lock_guard<std::recursive_mutex> lk(wmu_);
         while (!c.wbuf.empty()) {
             const ssize_t w = ::send(c.fd, c.wbuf.data(), c.wbuf.size(),
                                      MSG_NOSIGNAL);
@@ -297,13 +300,13 @@ private:
         c.dead = true;
         epoll_ctl(ep_, EPOLL_CTL_DEL, c.fd, nullptr);
         ::close(c.fd);
-        std::lock_guard<std::mutex> lk(wmu_);
+        std::lock_guard<std::recursive_mutex> lk(wmu_);
         conn_index_.erase(c.id);
     }
 
     void gc_dead() {
         for (auto it = conns_.begin(); it != conns_.end();) {
-            if (it->second.dead) it = conns_.erase(it);
+            if (it->second->dead) it = conns_.erase(it);
             else ++it;
         }
     }
@@ -316,10 +319,16 @@ private:
     std::atomic<bool> running_{false};
     std::thread loop_;
     uint64_t next_id_ = 1;
-    std::unordered_map<uint64_t, Conn> conns_;
+    // unique_ptr: Conn addresses must survive map rehash (conn_index_
+    // and in-flight harvests hold raw pointers/ids)
+    std::unordered_map<uint64_t, std::unique_ptr<Conn>> conns_;
     std::unordered_map<uint64_t, Conn*> conn_index_;
     std::mutex mu_;    // guards ready_
-    std::mutex wmu_;   // guards conn_index_ + pending_writes_
+    // recursive: flush_conn (called under wmu_) may close_conn, which
+    // locks wmu_ again. Guards conn_index_, pending_writes_ AND every
+    // wbuf access (queue_writes appends from the engine thread while
+    // the reactor flushes).
+    std::recursive_mutex wmu_;
     std::deque<PendingReq> ready_;
     std::vector<uint64_t> pending_writes_;
 };
