@@ -173,3 +173,45 @@ def test_armed_flagged_pipeline_matches_mirror():
             g = g_out[int(roffs[i]):int(roffs[i]) + int(rlens[i])]
             c = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
             assert (g == c).all(), f"step {step} req {i}"
+
+
+def test_armed_staged_event_path_matches_mirror():
+    """The event-based staged fallback (GOFR_FLAGGED=0) stays correct
+    alongside the default flagged path."""
+    import numpy as np
+    import torch
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine, pack_batch
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    gpu = BatchEngine(app, device="cuda", max_batch=256, pipeline=2)
+    gpu._flagged = False  # force the staged event path
+    cpu = BatchEngine(app, device="cpu", max_batch=256)
+    cpu._seed = gpu._seed
+    body = b'{"x":"yz"}'
+    raws = [b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+            b"Content-Type: application/json\r\n"
+            b"Content-Length: " + str(len(body)).encode() +
+            b"\r\n\r\n" + body for _ in range(256)]
+    buf, offs, lens = pack_batch(raws)
+    n, nbytes = len(lens), int(offs[-1] + lens[-1])
+    for ln in gpu.lanes:
+        ln.p_reqs[:nbytes] = torch.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n] = torch.from_numpy(offs)
+        ln.p_req_len[:n] = torch.from_numpy(lens)
+    for li in range(len(gpu.lanes)):
+        gpu.capture_graph(n, nbytes, li)
+    for step in range(2):
+        gpu.submit(n, nbytes, step % 2)
+        out_t, roff_t, rlen_t = gpu.complete(step % 2)
+        c_out, c_roffs, c_rlens = cpu.process_packed(buf, offs, lens)
+        assert (rlen_t.numpy() == c_rlens).all()
+        g = out_t.numpy()
+        for i in range(n):
+            go = g[int(roff_t[i]):int(roff_t[i]) + int(rlen_t[i])]
+            co = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
+            assert (go == co).all(), f"step {step} req {i}"
