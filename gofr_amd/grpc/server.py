@@ -43,22 +43,144 @@ class RPCLog:
 
 
 class ServiceDesc:
-    """methods: {name: (request MessageDesc, response MessageDesc)}."""
+    """methods: {name: (request MessageDesc, response MessageDesc)}.
 
-    def __init__(self, name: str, methods: dict):
+    gpu_methods marks unary methods the batched GPU codec can serve
+    end-to-end (currently the "hello_echo" shape: HelloRequest in,
+    "Hello <name>!" HelloResponse out — k_varint_spans + k_grpc_echo).
+    Marked methods are decoded/responded in whole batches; everything
+    else takes the host path.
+    """
+
+    def __init__(self, name: str, methods: dict,
+                 gpu_methods: dict | None = None):
         self.name = name
         self.methods = methods
+        self.gpu_methods = gpu_methods or {}
 
 
 class GRPCServer:
-    def __init__(self, app, port: int):
+    def __init__(self, app, port: int, batch_window_us: int = 200,
+                 max_codec_batch: int = 8192):
         self.app = app
         self.port = port
         self._services: dict[str, tuple[ServiceDesc, object]] = {}
+        self._gpu_methods: dict[tuple[str, str], str] = {}
         for service, impl in app._grpc_services:
             self._services[service.name] = (service, impl)
+            for meth, kind in getattr(service, "gpu_methods", {}).items():
+                self._gpu_methods[(service.name, meth)] = kind
         self._stop = threading.Event()
         self._sock = None
+        # batched GPU codec worker (config-3 data plane in production):
+        # marked unary methods are queued, decoded and responded in
+        # whole batches by k_varint_spans + k_grpc_echo (CPU mirrors on
+        # a CPU box)
+        self.batch_window_us = batch_window_us
+        self.max_codec_batch = max_codec_batch
+        self.codec_batches = 0
+        self.codec_msgs = 0
+        self._codec_q = None
+        if self._gpu_methods:
+            import queue as _queue
+            self._codec_q = _queue.Queue()
+            threading.Thread(target=self._codec_worker,
+                             daemon=True).start()
+
+    def _codec_worker(self) -> None:
+        import queue as _queue
+
+        import numpy as np
+
+        from .. import ops
+        from ..engine import pack_batch
+        gpu = None
+        try:
+            import torch
+            if torch.cuda.is_available():
+                hip = ops.HipOps()
+                dev = torch.device("cuda")
+                nb = self.max_codec_batch
+                GR = 256
+                gpu = {
+                    "hip": hip, "t": torch, "dev": dev, "GR": GR,
+                    "p_buf": torch.empty(nb * 512,
+                                         dtype=torch.uint8).pin_memory(),
+                    "d_buf": torch.empty(nb * 512, dtype=torch.uint8,
+                                         device=dev),
+                    "d_off": torch.empty(nb, dtype=torch.int64,
+                                         device=dev),
+                    "d_len": torch.empty(nb, dtype=torch.int32,
+                                         device=dev),
+                    "d_spans": torch.zeros(nb * ops.MAX_PB_FIELDS * 4,
+                                           dtype=torch.int32, device=dev),
+                    "d_span_n": torch.zeros(nb, dtype=torch.int32,
+                                            device=dev),
+                    "d_out": torch.empty(nb * GR, dtype=torch.uint8,
+                                         device=dev),
+                    "d_out_len": torch.empty(nb, dtype=torch.int32,
+                                             device=dev),
+                    "p_out": hip.host_alloc(nb * GR),
+                    "p_out_len": torch.empty(
+                        nb, dtype=torch.int32).pin_memory(),
+                }
+        except (ImportError, FileNotFoundError):
+            gpu = None
+        while not self._stop.is_set():
+            try:
+                first = self._codec_q.get(timeout=0.2)
+            except _queue.Empty:
+                continue
+            batch = [first]
+            deadline = time.perf_counter() + self.batch_window_us / 1e6
+            while len(batch) < self.max_codec_batch and \
+                    time.perf_counter() < deadline:
+                try:
+                    batch.append(self._codec_q.get_nowait())
+                except _queue.Empty:
+                    time.sleep(0)
+            payloads = [b[0] for b in batch]
+            buf, offs, lens = pack_batch(payloads)
+            n = len(lens)
+            if gpu is not None and \
+                    all(len(p) <= 512 for p in payloads):
+                t = gpu["t"]
+                nbytes = len(buf)
+                gpu["p_buf"][:nbytes] = t.from_numpy(buf)
+                stream = t.cuda.current_stream().cuda_stream
+                gpu["d_buf"][:nbytes].copy_(gpu["p_buf"][:nbytes],
+                                            non_blocking=True)
+                gpu["d_off"][:n].copy_(t.from_numpy(offs),
+                                       non_blocking=True)
+                gpu["d_len"][:n].copy_(t.from_numpy(lens),
+                                       non_blocking=True)
+                gpu["hip"].varint_spans(stream, gpu["d_buf"],
+                                        gpu["d_off"], gpu["d_len"],
+                                        gpu["d_spans"], gpu["d_span_n"],
+                                        n)
+                gpu["hip"].grpc_echo(stream, gpu["d_buf"], gpu["d_spans"],
+                                     gpu["d_span_n"], gpu["d_out"],
+                                     gpu["d_out_len"], n, gpu["GR"])
+                gpu["p_out"][:n * gpu["GR"]].copy_(
+                    gpu["d_out"][:n * gpu["GR"]], non_blocking=True)
+                gpu["p_out_len"][:n].copy_(gpu["d_out_len"][:n],
+                                           non_blocking=True)
+                t.cuda.synchronize()
+                out = gpu["p_out"].numpy()
+                out_len = gpu["p_out_len"].numpy()
+                GR = gpu["GR"]
+            else:
+                spans, span_n = ops.cpu_varint_spans(buf, offs, lens)
+                out, out_len = ops.cpu_grpc_echo(buf, spans, span_n, 256)
+                GR = 256
+            self.codec_batches += 1
+            self.codec_msgs += n
+            for i, (_, rq) in enumerate(batch):
+                ln = int(out_len[i])
+                if ln < 0:
+                    rq.put(None)  # malformed: host error path
+                else:
+                    rq.put(out[i * GR:i * GR + ln].tobytes())
 
     def start(self) -> None:
         sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
@@ -171,6 +293,18 @@ class GRPCServer:
                 mlen = struct.unpack(">I", data[1:5])[0]
                 if compressed:
                     grpc_status, err_msg = 12, "compression not supported"
+                elif (self._codec_q is not None and
+                      (service_name, method_name) in self._gpu_methods):
+                    # batched GPU codec path (k_varint_spans +
+                    # k_grpc_echo decode/respond the whole batch)
+                    import queue as _queue
+                    rq = _queue.Queue(1)
+                    self._codec_q.put((data[5:5 + mlen], rq))
+                    frame = rq.get(timeout=10)
+                    if frame is None:
+                        raise ValueError("malformed request message")
+                    # frame already carries the 5-byte gRPC prefix
+                    msg_bytes = frame[5:]
                 else:
                     req = decode_message(data[5:5 + mlen], req_desc)
                     fn = getattr(impl, method_name)

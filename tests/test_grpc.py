@@ -209,3 +209,54 @@ def test_cpu_grpc_echo_malformed():
     assert span_n[0] == -1
     out, out_len = ops.cpu_grpc_echo(buf, spans, span_n, 512)
     assert out_len[0] == -1
+
+
+def test_batched_codec_path():
+    """Unary methods marked gpu_methods are served by the batched
+    codec (CPU mirrors here; kernels on a GPU box) — responses match
+    the host path byte-for-byte and actually take the batch path."""
+    import threading as _threading
+
+    import gofr_amd
+    from gofr_amd.grpc.server import GRPCServer
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    svc = ServiceDesc("hello.Hello", {
+        "SayHello": (HELLO_REQUEST, HELLO_RESPONSE),
+    }, gpu_methods={"SayHello": "hello_echo"})
+
+    class Impl:
+        def SayHello(self, ctx, req):
+            raise AssertionError("host path must not run")
+
+    app.RegisterService(svc, Impl())
+    srv = GRPCServer(app, 0, batch_window_us=5000)
+    srv.port = 0
+    srv.start()
+    import time as _time
+    _time.sleep(0.1)
+    port = srv._sock.getsockname()[1]
+    try:
+        results = {}
+
+        def call(i):
+            c = GRPCClient("127.0.0.1", port)
+            resp, status, err = c.call("hello.Hello", "SayHello",
+                                       {"name": f"u{i}"}, HELLO_REQUEST,
+                                       HELLO_RESPONSE)
+            c.close()
+            results[i] = (status, resp)
+
+        ts = [_threading.Thread(target=call, args=(i,))
+              for i in range(8)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=20)
+        assert len(results) == 8
+        for i, (status, resp) in results.items():
+            assert status == 0
+            assert resp["message"] == f"Hello u{i}!"
+        assert srv.codec_msgs == 8 and srv.codec_batches >= 1
+    finally:
+        srv.stop()
