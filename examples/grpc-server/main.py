@@ -21,7 +21,7 @@ class Server:
 
 HELLO_SERVICE = ServiceDesc("hello.Hello", {
     "SayHello": (HELLO_REQUEST, HELLO_RESPONSE),
-})
+}, gpu_methods={"SayHello": "hello_echo"})  # batched GPU codec path
 
 
 def build_app():

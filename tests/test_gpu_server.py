@@ -80,3 +80,39 @@ def test_many_pipelined_requests(srv):
         got += d
     s.close()
     assert got.count(b'{"data":"Hello World!"}') == 600
+
+
+def test_grpc_batched_codec_on_gpu():
+    """The production gRPC server's marked methods run through the
+    device codec kernels end-to-end over a real socket."""
+    import time
+
+    from gofr_amd.grpc.codec import HELLO_REQUEST, HELLO_RESPONSE
+    from gofr_amd.grpc.server import GRPCClient, GRPCServer, ServiceDesc
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    svc = ServiceDesc("hello.Hello", {
+        "SayHello": (HELLO_REQUEST, HELLO_RESPONSE),
+    }, gpu_methods={"SayHello": "hello_echo"})
+
+    class Impl:
+        def SayHello(self, ctx, req):
+            raise AssertionError("host path must not run")
+
+    app.RegisterService(svc, Impl())
+    s = GRPCServer(app, 0, batch_window_us=1000)
+    s.start()
+    time.sleep(0.2)
+    port = s._sock.getsockname()[1]
+    try:
+        c = GRPCClient("127.0.0.1", port)
+        for name in ("gpu-a", "", "x" * 200):
+            resp, status, err = c.call("hello.Hello", "SayHello",
+                                       {"name": name}, HELLO_REQUEST,
+                                       HELLO_RESPONSE)
+            assert status == 0, err
+            assert resp["message"] == f"Hello {name or 'World'}!"
+        c.close()
+        assert s.codec_msgs >= 3
+    finally:
+        s.stop()
