@@ -186,3 +186,42 @@ def test_oversized_request_routes_to_host():
     assert out.startswith(b"HTTP/1.1 200 OK"), out[:80]
     _, _, rbody = out.partition(b"\r\n\r\n")
     assert json.loads(rbody)["data"]["big"] == "x" * 6000
+
+
+def test_fuzz_mirror_never_crashes_and_frames_validly():
+    """Seeded fuzz over mutated requests: the engine must never raise
+    and every response must be a framed HTTP response (the GPU parity
+    fuzz in tests/test_gpu_engine.py byte-compares the kernels against
+    this same behavior)."""
+    import random
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("ok"))
+    app.GET("/u/{id}", lambda ctx: {"id": ctx.PathParam("id")})
+    eng = BatchEngine(app, device="cpu", max_batch=128)
+    rng = random.Random(1234)
+    base = (b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+            b"Content-Type: application/json\r\n"
+            b"Content-Length: 9\r\n\r\n" + b'{"a":"b"}')
+    raws = []
+    for i in range(96):
+        b = bytearray(base)
+        for _ in range(rng.randrange(0, 6)):
+            b[rng.randrange(len(b))] = rng.randrange(256)
+        raws.append(bytes(b))
+    raws += [b"\r\n\r\n", b"GARBAGE", b"GET  HTTP/1.1\r\n\r\n",
+             b"GET /u/%41%zz HTTP/1.1\r\nHost: h\r\n\r\n"]
+    outs = eng.process(raws)
+    for i, out in enumerate(outs):
+        assert out.startswith(b"HTTP/1.1 "), f"req {i}: {out[:40]!r}"
+        head, sep, body = out.partition(b"\r\n\r\n")
+        assert sep, f"req {i}: unterminated headers"
+        cl = [h for h in head.split(b"\r\n")
+              if h.lower().startswith(b"content-length:")]
+        assert cl and int(cl[0].split(b":")[1]) == len(body), f"req {i}"

@@ -215,3 +215,89 @@ def test_armed_staged_event_path_matches_mirror():
             go = g[int(roff_t[i]):int(roff_t[i]) + int(rlen_t[i])]
             co = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
             assert (go == co).all(), f"step {step} req {i}"
+
+
+def test_fuzz_gpu_matches_mirror():
+    """Seeded fuzz parity: mutated/malformed requests through the GPU
+    kernels byte-match the CPU mirrors."""
+    import random
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.GET("/greet", handlers.static_json("ok"))
+    app.enable_etag()
+    gpu = BatchEngine(app, device="cuda", max_batch=256)
+    cpu = BatchEngine(app, device="cpu", max_batch=256)
+    cpu._seed = gpu._seed
+    rng = random.Random(99)
+    base = (b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+            b"Content-Type: application/json\r\n"
+            b"Content-Length: 9\r\n\r\n" + b'{"a":"b"}')
+    raws = []
+    for i in range(200):
+        b = bytearray(base)
+        for _ in range(rng.randrange(0, 8)):
+            b[rng.randrange(len(b))] = rng.randrange(1, 256)
+        raws.append(bytes(b))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, (f"req {i} input {raws[i][:60]!r}\n"
+                          f"GPU {go[:120]!r}\nCPU {co[:120]!r}")
+
+
+def test_armed_with_middlewares_matches_mirror():
+    """Armed flagged pipeline with auth+gzip+etag enabled."""
+    import torch
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine, pack_batch
+    from gofr_amd.http.middleware import hmac_token
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.POST("/echo", handlers.echo_json)
+    app.enable_auth(b"s3cret")
+    app.enable_gzip(min_size=64)
+    app.enable_etag()
+    gpu = BatchEngine(app, device="cuda", max_batch=128, pipeline=2)
+    cpu = BatchEngine(app, device="cpu", max_batch=128)
+    cpu._seed = gpu._seed
+    tok = hmac_token(b"s3cret", "POST", "/echo")
+    body = b'{"pad":"' + b"q" * 300 + b'"}'
+    raws = []
+    for i in range(128):
+        auth = (f"Authorization: HMAC {tok}\r\n" if i % 2 == 0
+                else "Authorization: HMAC " + "00" * 32 + "\r\n")
+        raws.append((f"POST /echo HTTP/1.1\r\nHost: h\r\n{auth}"
+                     "Accept-Encoding: gzip\r\n"
+                     "Content-Type: application/json\r\n"
+                     f"Content-Length: {len(body)}\r\n\r\n"
+                     ).encode() + body)
+    buf, offs, lens = pack_batch(raws)
+    n, nbytes = len(lens), int(offs[-1] + lens[-1])
+    for ln in gpu.lanes:
+        ln.p_reqs[:nbytes] = torch.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n] = torch.from_numpy(offs)
+        ln.p_req_len[:n] = torch.from_numpy(lens)
+    for li in range(len(gpu.lanes)):
+        gpu.capture_graph(n, nbytes, li)
+    gpu.submit(n, nbytes, 0)
+    out_t, roff_t, rlen_t = gpu.complete(0)
+    c_out, c_roffs, c_rlens = cpu.process_packed(buf, offs, lens)
+    g = out_t.numpy()
+    assert (rlen_t.numpy() == c_rlens).all()
+    for i in range(n):
+        go = g[int(roff_t[i]):int(roff_t[i]) + int(rlen_t[i])]
+        co = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
+        assert (go == co).all(), f"req {i}"
+    first = g[int(roff_t[0]):int(roff_t[0]) + int(rlen_t[0])].tobytes()
+    assert b"Content-Encoding: gzip" in first and b'ETag: "' in first
+    second = g[int(roff_t[1]):int(roff_t[1]) + int(rlen_t[1])].tobytes()
+    assert second.startswith(b"HTTP/1.1 401 ")
