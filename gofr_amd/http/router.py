@@ -101,7 +101,10 @@ class Router:
         params: dict[str, str] = {}
         best_prefix = self.root.prefix_route
         segs = [s for s in path.strip("/").split("/") if s != ""]
-        found_node = True
+        # gorilla/mux StrictSlash(false) parity (http/router.go:17):
+        # "/path/" is NOT "/path" — a trailing slash only reaches
+        # prefix (catch-all) routes
+        found_node = not (len(path) > 1 and path.endswith("/"))
         for seg in segs:
             nxt = node.children.get(seg)
             if nxt is None and node.param_child is not None:

@@ -321,7 +321,9 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     {
         int pos = path_off;
         const int pend = path_off + F[FI_PATH_LEN];
-        bool dead = false;
+        // StrictSlash(false) parity (reference http/router.go:17):
+        // a trailing slash only reaches prefix routes
+        bool dead = (F[FI_PATH_LEN] > 1 && base[pend - 1] == '/');
         while (pos < pend && !dead) {
             while (pos < pend && base[pos] == '/') ++pos;
             if (pos >= pend) break;

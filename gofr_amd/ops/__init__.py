@@ -414,7 +414,11 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         nparams = 0
         pspans = []
         pos, pend = path_off, path_off + int(F[FI_PATH_LEN])
-        dead = False
+        # StrictSlash(false) parity: trailing slash never matches a
+        # non-prefix route (kernel does the same)
+        dead = (int(F[FI_PATH_LEN]) > 1 and
+                buf[path_off + int(F[FI_PATH_LEN]) - 1:
+                    path_off + int(F[FI_PATH_LEN])] == b"/")
         while pos < pend and not dead:
             while pos < pend and buf[pos] == ord("/"):
                 pos += 1

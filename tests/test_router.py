@@ -66,13 +66,6 @@ def test_root_route():
     assert st == 200 and route.handler(None) == "root"
 
 
-def test_trailing_slash_insensitive():
-    r = Router()
-    r.add("GET", "/x", _h("x"))
-    route, _, st = r.match("GET", "/x/")
-    assert st == 200  # StrictSlash(false) analog (http/router.go:18)
-
-
 def test_compile_layout():
     r = Router()
     r.add("GET", "/greet", _h("g"))
@@ -132,3 +125,20 @@ def test_compile_matches_python_match_on_fuzz():
         route, _, st = r.match(method, path)
         want = route.route_id if route else -1
         assert table_match(method, path) == want, (method, path)
+
+
+def test_strict_slash_false_parity():
+    """gorilla/mux StrictSlash(false): "/path/" is not "/path"
+    (reference http/router.go:17); prefix routes still match."""
+    from gofr_amd.http.router import Router
+    r = Router()
+    r.add("GET", "/path", lambda c: "x")
+    r.add_prefix("GET", "/files", lambda c: "f")
+    route, params, st = r.match("GET", "/path")
+    assert route is not None and st == 200
+    route, params, st = r.match("GET", "/path/")
+    assert route is None and st == 404
+    route, params, st = r.match("GET", "/files/a/")
+    assert route is not None  # prefix routes still catch trailing /
+    route, params, st = r.match("GET", "/")
+    assert st in (200, 404)  # root unaffected by the rule
