@@ -66,6 +66,11 @@ def test_defaults_over_sockets(srv):
     assert st == 200
     st, _, _ = _req(srv.port, "GET", "/nope")
     assert st == 404
+    # binary File response through the armed host-trampoline path
+    st, hdrs, body = _req(srv.port, "GET", "/favicon.ico")
+    assert st == 200
+    assert hdrs.get("Content-Type") == "image/x-icon"
+    assert len(body) > 100 and body[:2] in (b"\x00\x00", b"BM")
 
 
 def test_many_pipelined_requests(srv):
