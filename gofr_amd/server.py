@@ -80,7 +80,6 @@ def dispatch(app, request: Request) -> _CapturedResponse:
         _, ct, body = envelope_bytes(None, GofrError("route not found")
                                      if status == 404 else
                                      GofrError("method not allowed"))
-        out.status = status
         out.headers.append(("Content-Type", ct))
         out.body = body
         span.End()
