@@ -260,3 +260,24 @@ def test_batched_codec_path():
         assert srv.codec_msgs == 8 and srv.codec_batches >= 1
     finally:
         srv.stop()
+
+
+def test_server_survives_garbage_bytes(grpc_app):
+    """A connection speaking garbage must not take the server down
+    (per-conn isolation — the reference's grpc-go equivalent)."""
+    import socket as _socket
+
+    port = grpc_app.port
+    s = _socket.create_connection(("127.0.0.1", port), timeout=5)
+    s.sendall(b"\x00\xff" * 300 + b"NOT HTTP2 AT ALL")
+    s.close()
+    s2 = _socket.create_connection(("127.0.0.1", port), timeout=5)
+    s2.sendall(b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n" + b"\xde\xad" * 100)
+    s2.close()
+    # a real client still works afterwards
+    c = GRPCClient("127.0.0.1", port)
+    resp, status, err = c.call("hello.Hello", "SayHello",
+                               {"name": "after-garbage"}, HELLO_REQUEST,
+                               HELLO_RESPONSE)
+    c.close()
+    assert status == 0 and "after-garbage" in resp["message"]
