@@ -142,8 +142,14 @@ public:
             const int i = rows[k];
             auto it = conn_index_.find(cids[i]);
             if (it == conn_index_.end()) continue;
-            it->second->wbuf.append((const char*)out + roffs[i],
-                                    (size_t)rlens[i]);
+            const char* resp = (const char*)out + roffs[i];
+            const size_t rl = (size_t)rlens[i];
+            it->second->wbuf.append(resp, rl);
+            // honor the engine's Connection: close verdict (headers
+            // precede the body; cap the scan)
+            const size_t scan = rl < 400 ? rl : 400;
+            if (memmem(resp, scan, "Connection: close", 17) != nullptr)
+                it->second->close_after_write = true;
             pending_writes_.push_back(cids[i]);
         }
     }

@@ -250,7 +250,12 @@ k_parse_route(const uint8_t* __restrict__ reqs,
 
     // ---- headers ----------------------------------------------------------
     int clen = 0;
-    bool keep_alive = true;  // HTTP/1.1 default
+    // keep-alive default: HTTP/1.1 yes, HTTP/1.0 no (RFC 9112 §9.3);
+    // version token = bytes between the 2nd space and CR
+    bool keep_alive = true;
+    if (sp2 > 0 && lf1 - 1 - (sp2 + 1) == 8 &&
+        base[lf1 - 2] == '0' && base[lf1 - 4] == '1')  // "HTTP/1.0"
+        keep_alive = false;
     int body_off = len, auth_off = 0, auth_len = 0;
     int inm_off = 0, inm_len = 0;
     int prev_lf = lf1;
@@ -278,7 +283,10 @@ k_parse_route(const uint8_t* __restrict__ reqs,
             }
             clen = v;
         } else if (nlen == 10 && ieq(nm, "connection", 10)) {
-            if (vlen == 5 && ieq(base + vs, "close", 5)) keep_alive = false;
+            if (vlen == 5 && ieq(base + vs, "close", 5))
+                keep_alive = false;
+            else if (vlen == 10 && ieq(base + vs, "keep-alive", 10))
+                keep_alive = true;  // HTTP/1.0 opt-in
         } else if (nlen == 12 && ieq(nm, "content-type", 12)) {
             if (vlen >= 16 && ieq(base + vs, "application/json", 16))
                 flags |= FL_JSON_CT;

@@ -348,7 +348,13 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
             flags |= FL_ERR_PARSE
 
         clen = 0
+        # HTTP/1.1 defaults keep-alive; HTTP/1.0 defaults close
+        # (kernel checks the same version-token bytes)
         keep_alive = True
+        if sp2 > 0 and (lf1 - 1) - (sp2 + 1) == 8 and \
+                buf[lf1 - 2:lf1 - 1] == b"0" and \
+                buf[lf1 - 4:lf1 - 3] == b"1":
+            keep_alive = False
         body_off = ln
         auth_off = auth_len = 0
         inm_off = inm_len = 0
@@ -382,6 +388,8 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
             elif nlen == 10 and _ieq(buf, ls, b"connection"):
                 if vlen == 5 and _ieq(buf, vs, b"close"):
                     keep_alive = False
+                elif vlen == 10 and _ieq(buf, vs, b"keep-alive"):
+                    keep_alive = True  # HTTP/1.0 opt-in
             elif nlen == 12 and _ieq(buf, ls, b"content-type"):
                 if vlen >= 16 and _ieq(buf, vs, b"application/json"):
                     flags |= FL_JSON_CT

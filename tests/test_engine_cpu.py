@@ -225,3 +225,26 @@ def test_fuzz_mirror_never_crashes_and_frames_validly():
         cl = [h for h in head.split(b"\r\n")
               if h.lower().startswith(b"content-length:")]
         assert cl and int(cl[0].split(b":")[1]) == len(body), f"req {i}"
+
+
+def test_http10_keep_alive_semantics():
+    """HTTP/1.0 defaults Connection: close; opt-in keep-alive works;
+    HTTP/1.1 unchanged (RFC 9112 §9.3)."""
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/greet", handlers.static_json("hi"))
+    eng = BatchEngine(app, device="cpu", max_batch=8)
+    outs = eng.process([
+        b"GET /greet HTTP/1.0\r\nHost: h\r\n\r\n",
+        b"GET /greet HTTP/1.0\r\nHost: h\r\nConnection: keep-alive\r\n\r\n",
+        b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n",
+        b"GET /greet HTTP/1.1\r\nHost: h\r\nConnection: close\r\n\r\n",
+    ])
+    assert b"Connection: close" in outs[0]
+    assert b"Connection: keep-alive" in outs[1]
+    assert b"Connection: keep-alive" in outs[2]
+    assert b"Connection: close" in outs[3]

@@ -94,3 +94,22 @@ def test_native_concurrent_clients(native_server):
     for t in ts:
         t.join()
     assert not errors
+
+
+def test_connection_close_honored(native_server):
+    """A Connection: close request gets its response and the server
+    closes the socket (HTTP/1.0 clients depend on it)."""
+    import socket as _socket
+
+    s = _socket.create_connection(("127.0.0.1", native_server.port),
+                                  timeout=5)
+    s.sendall(b"GET /greet HTTP/1.0\r\nHost: h\r\n\r\n")
+    got = b""
+    while True:
+        d = s.recv(65536)
+        if not d:
+            break  # server closed after the response
+        got += d
+    s.close()
+    assert got.startswith(b"HTTP/1.1 200 OK")
+    assert b"Connection: close" in got
