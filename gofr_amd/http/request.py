@@ -118,8 +118,65 @@ def parse_request_bytes(raw: bytes, remote_addr: str = "") -> Request:
             value.decode("latin-1").strip())
     target_s = target.decode("latin-1")
     path, _, qs = target_s.partition("?")
-    clen = int(headers.get("content-length", "0") or "0")
+    if "chunked" in headers.get("transfer-encoding", "").lower():
+        body = decode_chunked(body)
+    else:
+        clen = int(headers.get("content-length", "0") or "0")
+        body = body[:clen]
     return Request(method=method.decode("latin-1"), path=unquote(path),
-                   query_string=qs, headers=headers, body=body[:clen],
+                   query_string=qs, headers=headers, body=body,
                    version=version.decode("latin-1"),
                    remote_addr=remote_addr)
+
+
+def chunked_frame_len(data: bytes, start: int):
+    """Length of a complete chunked body at `start` (through the
+    terminating blank line), or None if more bytes are needed.
+    Raises ValueError on malformed framing."""
+    pos = start
+    while True:
+        eol = data.find(b"\r\n", pos)
+        if eol < 0:
+            if len(data) - pos > 18:
+                raise ValueError("chunked: size line too long")
+            return None
+        try:
+            size = int(data[pos:eol].split(b";", 1)[0].strip(), 16)
+        except ValueError:
+            raise ValueError("chunked: bad size") from None
+        pos = eol + 2
+        if size == 0:
+            while True:
+                e2 = data.find(b"\r\n", pos)
+                if e2 < 0:
+                    return None
+                if e2 == pos:
+                    return pos + 2 - start
+                pos = e2 + 2
+        if pos + size + 2 > len(data):
+            return None
+        pos += size + 2
+
+
+def decode_chunked(data: bytes) -> bytes:
+    """RFC 9112 §7.1 chunked body decode (chunk-size[;ext]CRLF data
+    CRLF ... 0 CRLF [trailers] CRLF). Raises ValueError on malformed
+    framing."""
+    out = bytearray()
+    pos = 0
+    while True:
+        eol = data.find(b"\r\n", pos)
+        if eol < 0:
+            raise ValueError("chunked: missing size line")
+        size_tok = data[pos:eol].split(b";", 1)[0].strip()
+        try:
+            size = int(size_tok, 16)
+        except ValueError:
+            raise ValueError("chunked: bad size") from None
+        pos = eol + 2
+        if size == 0:
+            return bytes(out)  # trailers (if any) ignored
+        if pos + size + 2 > len(data):
+            raise ValueError("chunked: truncated chunk")
+        out += data[pos:pos + size]
+        pos += size + 2  # data CRLF

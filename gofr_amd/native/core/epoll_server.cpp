@@ -62,6 +62,34 @@ struct PendingReq {
 // conn ids: [63:48] reactor index, [47:0] per-reactor serial
 static inline int id_reactor(uint64_t id) { return (int)(id >> 48); }
 
+// RFC 9112 §7.1 chunked framing: length of the chunked body starting
+// at `start` through the end of the terminating blank line, or -1 if
+// more bytes are needed, -2 if malformed.
+static long chunked_body_len(const std::string& b, size_t start) {
+    size_t pos = start;
+    while (true) {
+        const size_t eol = b.find("\r\n", pos);
+        if (eol == std::string::npos)
+            return b.size() - pos > 18 ? -2 : -1;  // size line too long
+        char* end = nullptr;
+        const std::string tok = b.substr(pos, eol - pos);
+        const long size = strtol(tok.c_str(), &end, 16);
+        if (end == tok.c_str() || size < 0) return -2;
+        pos = eol + 2;
+        if (size == 0) {
+            // trailer section: lines until the blank line
+            while (true) {
+                const size_t e2 = b.find("\r\n", pos);
+                if (e2 == std::string::npos) return -1;
+                if (e2 == pos) return (long)(pos + 2 - start);
+                pos = e2 + 2;
+            }
+        }
+        if (pos + (size_t)size + 2 > b.size()) return -1;
+        pos += (size_t)size + 2;
+    }
+}
+
 class Reactor {
 public:
     Reactor(int idx, int port, size_t max_req)
@@ -234,7 +262,8 @@ private:
             const size_t he = c.rbuf.find("\r\n\r\n");
             if (he == std::string::npos) break;
             size_t clen = 0;
-            // find content-length (case-insensitive) in the header block
+            bool chunked = false;
+            // scan headers for content-length / transfer-encoding
             for (size_t p = 0; p < he;) {
                 size_t eol = c.rbuf.find("\r\n", p);
                 if (eol == std::string::npos || eol > he) eol = he;
@@ -250,9 +279,33 @@ private:
                         clen = strtoul(c.rbuf.c_str() + p + 15, nullptr,
                                        10);
                 }
+                if (eol - p > 18) {
+                    static const char t[] = "transfer-encoding:";
+                    bool match = true;
+                    for (int j = 0; j < 18; ++j) {
+                        char ch = c.rbuf[p + j];
+                        if (ch >= 'A' && ch <= 'Z') ch |= 0x20;
+                        if (ch != t[j]) { match = false; break; }
+                    }
+                    if (match) {
+                        std::string v = c.rbuf.substr(p + 18, eol - p - 18);
+                        for (auto& ch : v)
+                            if (ch >= 'A' && ch <= 'Z') ch |= 0x20;
+                        if (v.find("chunked") != std::string::npos)
+                            chunked = true;
+                    }
+                }
                 p = eol + 2;
             }
-            const size_t total = he + 4 + clen;
+            size_t total;
+            if (chunked) {
+                const long blen = chunked_body_len(c.rbuf, he + 4);
+                if (blen == -2) return false;  // malformed framing
+                if (blen < 0) break;           // need more bytes
+                total = he + 4 + (size_t)blen;
+            } else {
+                total = he + 4 + clen;
+            }
             if (c.rbuf.size() < total) break;
             PendingReq req;
             req.conn_id = c.id;

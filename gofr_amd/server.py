@@ -195,16 +195,32 @@ class HTTPServer:
                 head_end = buf.index(b"\r\n\r\n") + 4
                 head = buf[:head_end]
                 clen = 0
+                chunked = False
                 for line in head.split(b"\r\n")[1:]:
                     if line[:15].lower() == b"content-length:":
                         clen = int(line.split(b":", 1)[1].strip() or b"0")
-                        break
-                while len(buf) < head_end + clen:
+                    elif line[:18].lower() == b"transfer-encoding:" and \
+                            b"chunked" in line.lower():
+                        chunked = True
+                if chunked:
+                    from .http.request import chunked_frame_len
+                    while True:
+                        blen = chunked_frame_len(buf, head_end)
+                        if blen is not None:
+                            break
+                        chunk = conn.recv(65536)
+                        if not chunk:
+                            return
+                        buf += chunk
+                    total = head_end + blen
+                else:
+                    total = head_end + clen
+                while len(buf) < total:
                     chunk = conn.recv(65536)
                     if not chunk:
                         return
                     buf += chunk
-                raw, buf = buf[:head_end + clen], buf[head_end + clen:]
+                raw, buf = buf[:total], buf[total:]
                 request = parse_request_bytes(raw, remote_addr=remote)
                 resp = dispatch(self.app, request)
                 keep = request.headers.get("connection", "").lower() != "close"

@@ -166,3 +166,19 @@ def test_hostname_forwarded_proto():
     from gofr_amd.http.request import Request
     r = Request(headers={"Host": "h", "X-Forwarded-Proto": "https"})
     assert r.HostName() == "https://h"
+
+
+def test_chunked_decode_and_frame_len():
+    from gofr_amd.http.request import (chunked_frame_len, decode_chunked,
+                                       parse_request_bytes)
+    body = b"5\r\nhello\r\n6\r\n world\r\n0\r\n\r\n"
+    assert decode_chunked(body) == b"hello world"
+    assert chunked_frame_len(body, 0) == len(body)
+    assert chunked_frame_len(body[:-4], 0) is None  # incomplete
+    raw = (b"POST /x HTTP/1.1\r\nHost: h\r\n"
+           b"Transfer-Encoding: chunked\r\n\r\n" + body)
+    req = parse_request_bytes(raw)
+    assert req.body == b"hello world"
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        decode_chunked(b"zz\r\nxx\r\n0\r\n\r\n")

@@ -113,3 +113,30 @@ def test_connection_close_honored(native_server):
     s.close()
     assert got.startswith(b"HTTP/1.1 200 OK")
     assert b"Connection: close" in got
+
+
+def test_chunked_request_body(native_server):
+    """Chunked uploads are framed by the C++ ingress and decoded on the
+    host path (the parse kernel marks transfer-encoding NEEDS_HOST)."""
+    import socket as _socket
+
+    s = _socket.create_connection(("127.0.0.1", native_server.port),
+                                  timeout=5)
+    body = b'{"chunky":"yes"}'
+    chunks = (b"%x\r\n%s\r\n" % (10, body[:10]) +
+              b"%x\r\n%s\r\n" % (len(body) - 10, body[10:]) +
+              b"0\r\n\r\n")
+    s.sendall(b"POST /echo HTTP/1.1\r\nHost: h\r\n"
+              b"Content-Type: application/json\r\n"
+              b"Transfer-Encoding: chunked\r\n\r\n" + chunks)
+    got = b""
+    while b"\r\n\r\n" not in got or b"chunky" not in got:
+        d = s.recv(65536)
+        if not d:
+            break
+        got += d
+    s.close()
+    assert got.startswith(b"HTTP/1.1 200 OK"), got[:80]
+    import json as _json
+    _, _, rbody = got.partition(b"\r\n\r\n")
+    assert _json.loads(rbody)["data"] == {"chunky": "yes"}
