@@ -911,9 +911,11 @@ class GPUServer:
                             roffs.ctypes.data, rlens.ctypes.data)
 
     def stop(self):
+        # join the serving thread BEFORE tearing the ingress down: a
+        # harvest() in flight iterates the reactors the C++ stop frees
         if self._stop is not None:
             self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
         if self._core is not None:
             self._core.stop()
-        if self._thread is not None:
-            self._thread.join(timeout=2)
