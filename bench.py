@@ -8,11 +8,13 @@ with RCCL all-to-all over xGMI (ingress shard -> owner shard exchange of
 raw request slots, response gather back), one rank per GPU.
 
 One timed step = processing one batch of --batch synthetic 1 KB JSON echo
-requests end-to-end: H2D staging copy, (N>1: all-to-all request scatter),
-k_parse_route, k_respond, (N>1: all-to-all response gather), D2H of the
-response bytes. The payload is synthetic (in-memory request generator —
-no network on the box; SURVEY.md §4's fake-transport tier), byte-exact
-HTTP; there is no numeric precision to reduce (dtype=uint8-exact).
+requests end-to-end: SDMA ingress of the packed batch, (N>1: all-to-all
+request scatter), the kernel chain (parse/route, middlewares, respond,
+offset scan), the egress sweep writing the pinned ring, (N>1:
+all-to-all response gather). The payload is synthetic (in-memory
+request generator — no network on the box; SURVEY.md §4's
+fake-transport tier), byte-exact HTTP; there is no numeric precision
+to reduce (dtype=uint8-exact).
 
 Usage: python bench.py --gpus N --steps K --warmup W
 (N>1 is launched by the driver via torch.distributed.run, one rank/GPU.)

@@ -1,18 +1,32 @@
 """GPU request-batch engine.
 
-The MI355X data plane (SURVEY.md §7 phase 2): raw request bytes are
-PACKED back-to-back into a pinned ingress ring (offset + length per
-request — no slot padding on the bus), one H2D copy moves the batch onto
-the device, k_parse_route + k_respond (native/hip/gofr_kernels.hip)
-parse, route and serialize the whole batch, k_compact gathers the
-response slots into a contiguous 16B-aligned stream, and one D2H copy
-brings exactly the response bytes back. Handlers with a GPU spec
-(gofr_amd/handlers.py) never touch the host; the rest run through the
-host trampoline fixup pass (only when the device-side counter says some
-request needs it — pure-GPU batches never block on Python).
+The MI355X data plane (SURVEY.md §7 phase 2). Default ("flagged")
+serving cycle per armed batch:
 
-Multiple lanes (stream + buffer set each) let the serving loop overlap
-H2D of batch i+1 and D2H of batch i-1 with the kernels of batch i.
+  1. ONE SDMA H2D moves the ingress block — [offsets | lengths | batch
+     seed | packed request bytes] — from the pinned ring to HBM, with
+     the batch serial trailing in the same SDMA FIFO.
+  2. On one of two channel streams: k_gate (single wave) releases on
+     the serial, then k_parse_route (ballot structural parse + trie
+     route), k_auth, k_respond(_gz) (fused handler + serializer, gzip,
+     MFMA ETag), k_padscan (offset scan, result tables mirrored
+     straight to pinned host), k_compact (egress sweep writing the
+     pinned ring DIRECTLY — measured faster than any runtime copy
+     path), and k_done publishing the serial to the pinned tables.
+  3. The serving thread spins on that serial — no events, no D2H
+     copies, no runtime calls on the hot path (each one measured
+     0.1-0.5 ms of dead time; profiles/SUMMARY.md "host-link story").
+
+The whole cycle is enqueued by ONE native call (gofr_submit_staged),
+optionally via the pump worker threads so even the enqueue cost leaves
+the serving thread. Consecutive batches alternate channel streams, so
+kernels+egress of batch i overlap the SDMA ingress of batch i+1.
+
+Handlers with a GPU spec (gofr_amd/handlers.py) never touch the host;
+the rest run through the host trampoline fixup pass (only when the
+result tables say some request needs it — pure-GPU batches never block
+on Python). Event-based staged and eager fallbacks remain for A/B and
+for unarmed batch shapes.
 
 Without a GPU the engine runs the byte-exact CPU mirrors (gofr_amd/ops),
 so every test of this module runs on the CPU box.
