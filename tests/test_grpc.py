@@ -281,3 +281,33 @@ def test_server_survives_garbage_bytes(grpc_app):
                                HELLO_RESPONSE)
     c.close()
     assert status == 0 and "after-garbage" in resp["message"]
+
+
+def test_http2_frame_roundtrip_and_limits():
+    """Frame codec units: pack/read roundtrip, oversize rejection."""
+    from gofr_amd.grpc import http2 as h2
+
+    payload = b"\x01\x02\x03" * 100
+    raw = h2.pack_frame(h2.FT_DATA, h2.FLAG_END_STREAM, 7, payload)
+    buf = [raw]
+
+    def read_exact(n):
+        out, buf[0] = buf[0][:n], buf[0][n:]
+        assert len(out) == n
+        return out
+
+    ftype, flags, sid, got = h2.read_frame(read_exact)
+    assert (ftype, flags, sid) == (h2.FT_DATA, h2.FLAG_END_STREAM, 7)
+    assert got == payload
+
+
+def test_http2_hpack_many_headers():
+    from gofr_amd.grpc.http2 import HpackDecoder, HpackEncoder
+
+    headers = [(f"x-h{i}", f"v{i}" * 7) for i in range(40)]
+    headers += [(":method", "POST"), (":path", "/a/b"),
+                ("content-type", "application/grpc")]
+    blob = HpackEncoder.encode(headers)
+    dec = HpackDecoder()
+    got = dec.decode(blob)
+    assert dict(got) == dict(headers)
