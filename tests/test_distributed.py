@@ -88,15 +88,16 @@ main()
 
 
 @pytest.mark.timeout(120)
-def test_all_to_all_sharding_gloo(tmp_path):
+@pytest.mark.parametrize("world,port", [(2, 29517), (4, 29519)])
+def test_all_to_all_sharding_gloo(tmp_path, world, port):
     script = tmp_path / "worker.py"
     script.write_text(WORKER % {"repo": REPO})
     env = dict(os.environ)
     env.pop("GOFR_ENGINE", None)
     proc = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29517", str(script)],
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(port), str(script)],
         capture_output=True, text=True, timeout=110, env=env, cwd=REPO)
     out = proc.stdout + proc.stderr
     assert proc.returncode == 0, out[-3000:]
