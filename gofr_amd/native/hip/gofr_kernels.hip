@@ -1664,14 +1664,29 @@ k_compact(const uint8_t* __restrict__ resp_slots,
 // loop was spending ~0.3 ms/step in framework dispatch enqueueing the
 // same ~14 stream operations — at 32k requests/batch that overhead,
 // not the GPU, paced the engine. This enqueues the whole batch
-// pipeline with raw HIP calls:
+// pipeline with raw HIP calls. Two completion modes:
+//
+// Event mode (flagged == 0, the original design):
 //   s_in : H2D reqs + offsets(+seed tail) + lens, record ev_in
 //   s_k  : wait ev_in; memset host_needed; k_parse_route; [k_auth];
 //          k_respond; k_padscan  -> record ev_k
 //   s_out: wait ev_k; k_compact straight into the pinned egress ring;
 //          one D2H of the result tables; record ev_done
-// Event handles come from torch (interop: same HIP runtime), so the
-// Python side still waits with event.synchronize().
+// Event handles come from torch (interop: same HIP runtime); the
+// Python side waits with event.synchronize(). Measured cost: an event
+// whose producer ends in an SDMA copy releases cross-stream waiters
+// ~0.3 ms late, and hipEventSynchronize itself wakes ~0.3 ms after
+// completion — two stalls per batch at the host-visible edge.
+//
+// Flagged mode (flagged != 0, the production default): no events at
+// all. The serial flag travels in the SAME ingress copy as the batch
+// (trailing word of the offsets block), k_gate spins on it in device
+// memory and releases the chain, k_padscan mirrors the result tables
+// into pinned host memory itself (host_mirror arg), k_compact writes
+// the pinned egress ring directly, and k_done publishes the serial to
+// a pinned word (p_serial) with a system-scope release. Python spins
+// on plain pinned memory — completion latency is memory latency, not
+// the runtime's wakeup path.
 typedef struct {
     void* s_in; void* s_k; void* s_out;
     void* ev_in; void* ev_k; void* ev_done;
