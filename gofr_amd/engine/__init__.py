@@ -931,5 +931,11 @@ class GPUServer:
             self._stop.set()
         if self._thread is not None:
             self._thread.join(timeout=5)
-        if self._core is not None:
+            if self._thread.is_alive():
+                # worst case the thread is in the flagged-completion
+                # spin (30 s timeout) — freeing the reactors under it
+                # would be a use-after-free, so wait it out
+                self._thread.join(timeout=35)
+        if self._core is not None and (
+                self._thread is None or not self._thread.is_alive()):
             self._core.stop()
