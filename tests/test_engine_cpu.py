@@ -248,3 +248,45 @@ def test_http10_keep_alive_semantics():
     assert b"Connection: keep-alive" in outs[1]
     assert b"Connection: keep-alive" in outs[2]
     assert b"Connection: close" in outs[3]
+
+
+def test_deterministic_replay_across_engines(map_config):
+    """SURVEY §5 race-detection analog: the engine is a deterministic
+    function of (route table, batch sequence, seed) — two independent
+    engine instances replaying the same batch sequence must produce
+    byte-identical responses (correlation IDs included, since the seed
+    schedule is fixed). The GPU tier ties the kernels to these same
+    mirrors by byte equality, so this pins determinism end to end."""
+    batches = [
+        [http_req("GET", "/greet"),
+         http_req("POST", "/echo", body=b'{"a":1}')],
+        [http_req("GET", "/user/7"),
+         http_req("GET", "/missing"),
+         http_req("POST", "/echo", body=b'{"b":[1,2,3]}')],
+    ]
+    outs = []
+    for _ in range(2):
+        eng = BatchEngine(build_app(map_config), device="cpu")
+        outs.append([eng.process(b) for b in batches])
+    assert outs[0] == outs[1]
+
+
+def test_poisoned_slack_does_not_leak(map_config):
+    """SURVEY §5 poisoned-buffer analog: staging-buffer slack beyond
+    the packed requests must never influence responses. Process the
+    same batch from an exact-size buffer and from one whose tail is
+    filled with 0xAA garbage; the response bytes must match."""
+    eng1 = BatchEngine(build_app(map_config), device="cpu")
+    eng2 = BatchEngine(build_app(map_config), device="cpu")
+    payloads = [http_req("GET", "/greet"),
+                http_req("POST", "/echo", body=b'{"k":"v"}')]
+    buf, offs, lens = pack_batch(payloads)
+    poisoned = np.full(len(buf) + 4096, 0xAA, np.uint8)
+    poisoned[:len(buf)] = buf
+    out_a, ro_a, rl_a = eng1.process_packed(buf, offs, lens)
+    out_b, ro_b, rl_b = eng2.process_packed(poisoned, offs, lens)
+    assert np.array_equal(ro_a, ro_b) and np.array_equal(rl_a, rl_b)
+    for i in range(len(payloads)):
+        a = out_a[int(ro_a[i]):int(ro_a[i]) + int(rl_a[i])]
+        b = out_b[int(ro_b[i]):int(ro_b[i]) + int(rl_b[i])]
+        assert np.array_equal(a, b)
