@@ -187,6 +187,7 @@ class GRPCServer:
         sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
         sock.bind(("0.0.0.0", self.port))
         sock.listen(256)
+        self.port = sock.getsockname()[1]  # resolve port-0 binds
         self._sock = sock
         threading.Thread(target=self._accept_loop, daemon=True).start()
 
