@@ -1700,7 +1700,8 @@ typedef struct {
     const void* secret; int secret_len;
     void* d_resp;
     void* d_tables;                        // int32[2n+2]
-    void* p_tables;                        // pinned int32[2n+2]
+    void* p_tables;  // pinned int32[2n+4]; flagged mode publishes the
+                     // done serial into [2n+2] (k_done)
     const void* blob; const void* host_blob; const void* host_tab;
     int auth_env_off; int auth_env_len; int gzip_min; int etag_on;
     void* p_out;                           // pinned egress ring
