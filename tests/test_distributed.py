@@ -88,7 +88,8 @@ main()
 
 
 @pytest.mark.timeout(120)
-@pytest.mark.parametrize("world,port", [(2, 29517), (4, 29519)])
+@pytest.mark.parametrize("world,port", [(2, 29517), (4, 29519),
+                                        (8, 29521)])
 def test_all_to_all_sharding_gloo(tmp_path, world, port):
     script = tmp_path / "worker.py"
     script.write_text(WORKER % {"repo": REPO})
