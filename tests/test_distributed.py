@@ -1,5 +1,5 @@
-"""Multi-process all-to-all sharding tests (gloo backend, world_size 2,
-CPU) — the distributed-correctness tier for the RCCL path (the kernels
+"""Multi-process all-to-all sharding tests (gloo backend, world_size
+2/4/8, CPU) — the distributed-correctness tier for the RCCL path (the kernels
 under it are identical; on the GPU box the same AllToAllSharder runs on
 nccl/RCCL tensors — bench.py config 4)."""
 
