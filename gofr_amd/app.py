@@ -138,6 +138,7 @@ class App:
         Authorization: HMAC <hex of HMAC-SHA256(secret, "METHOD path")>.
         GPU engine: k_auth kernel; CPU transport: checked in dispatch."""
         self.auth_secret = bytes(secret)
+        return self
 
     def enable_etag(self):
         """ETag middleware: every response carries a strong validator
@@ -160,6 +161,7 @@ class App:
         """gzip-compress JSON responses when the request advertises
         Accept-Encoding: gzip and the body is >= min_size bytes."""
         self.gzip_min_size = int(min_size)
+        return self
 
     # -- CLI — reference: gofr.go:181-183 -------------------------------------
     def SubCommand(self, pattern: str, handler):
