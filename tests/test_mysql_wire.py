@@ -222,3 +222,23 @@ def test_db_wrapper_over_wire(srv):
     assert count == 3 and lastid == 42
     assert db.HealthCheck()["status"] == "UP"
     db.close()
+
+
+def test_tx_commit_and_rollback_over_wire(srv):
+    """Tx.Exec must not autocommit (documented fix); Commit/Rollback
+    reach the wire as COMMIT/ROLLBACK statements."""
+    from gofr_amd.datasource.sql import DB
+    from gofr_amd.testutil import MockLogger
+
+    conn = connect_mysql("127.0.0.1", srv.port, USER, PASSWORD, "db")
+    db = DB(conn, logger=MockLogger(), dialect="mysql")
+    tx = db.Begin()
+    tx.Exec("UPDATE t SET a = ?", 1)
+    assert "COMMIT" not in [q.strip().upper() for q in srv.queries]
+    tx.Commit()
+    assert srv.queries[-1].strip().upper() == "COMMIT"
+    tx2 = db.Begin()
+    tx2.Exec("UPDATE t SET a = ?", 2)
+    tx2.Rollback()
+    assert srv.queries[-1].strip().upper() == "ROLLBACK"
+    db.close()
