@@ -144,3 +144,13 @@ def test_with_headers_verbs():
             assert hdrs.get("X-Check") == "yes", method
     finally:
         srv.shutdown()
+
+
+def test_head_falls_to_catch_all(app):
+    """gorilla/mux registers explicit methods only, and the reference
+    installs a PathPrefix("/") catch-all in Run (gofr.go:104-107) — so
+    HEAD on a GET-only route reaches the catch-all -> 404, not 405."""
+    app.GET("/only-get", lambda ctx: "x")
+    app.install_default_routes()
+    resp = dispatch(app, _http("HEAD", "/only-get"))
+    assert resp.status == 404
