@@ -63,3 +63,15 @@ def test_protobuf_codec_roundtrip(name, count, flag, blob, ratio):
     assert out.get("flag", False) == flag
     assert out.get("blob", b"") == blob
     assert out.get("ratio", 0.0) == ratio
+
+
+@SET
+@given(st.lists(st.tuples(
+    st.text(alphabet=st.characters(min_codepoint=97, max_codepoint=122),
+            min_size=1, max_size=10),
+    st.text(max_size=30)), min_size=0, max_size=8))
+def test_hpack_roundtrip(headers):
+    from gofr_amd.grpc.http2 import HpackDecoder, HpackEncoder
+    wire = HpackEncoder.encode(headers)
+    out = HpackDecoder().decode(wire)
+    assert out == headers
