@@ -75,3 +75,67 @@ def test_hpack_roundtrip(headers):
     wire = HpackEncoder.encode(headers)
     out = HpackDecoder().decode(wire)
     assert out == headers
+
+
+SEG = st.sampled_from(["a", "b", "c", "{x}", "{y}"])
+
+
+@SET
+@given(st.lists(st.tuples(st.sampled_from(["GET", "POST"]),
+                          st.lists(SEG, min_size=1, max_size=3)),
+                min_size=1, max_size=6, unique_by=lambda r: (r[0],
+                                                            tuple(r[1]))),
+       st.booleans(),
+       st.lists(st.lists(st.sampled_from(["a", "b", "c", "d"]),
+                         max_size=4), min_size=1, max_size=8))
+def test_router_table_matches_python_match(routes, add_prefix, paths):
+    """The compiled flat table (what k_parse_route walks) must encode
+    the same decisions as Router.match for ANY route set — the
+    hypothesis generalization of the fixed fuzz in test_router."""
+    from gofr_amd.http.request import METHOD_IDS
+    from gofr_amd.http.router import Router
+
+    r = Router()
+    seen_patterns = set()
+    for method, segs in routes:
+        pattern = "/" + "/".join(segs)
+        if (method, pattern) in seen_patterns:
+            continue
+        seen_patterns.add((method, pattern))
+        try:
+            r.add(method, pattern, lambda c: None)
+        except ValueError:
+            continue  # conflicting param names at one level etc.
+    if add_prefix:
+        r.add_prefix("GET", "/", lambda c: None)
+    t = r.compile()
+
+    def table_match(method, path):
+        node = 0
+        mid = METHOD_IDS[method]
+        best_prefix = t["node_prefix"][0]
+        for seg in [s for s in path.strip("/").split("/") if s]:
+            sb = seg.encode()
+            nxt = -1
+            f, c = t["node_child_first"][node], t["node_child_count"][node]
+            for ci in range(f, f + c):
+                off, ln = t["child_seg_off"][ci], t["child_seg_len"][ci]
+                if bytes(t["seg_blob"][off:off + ln]) == sb:
+                    nxt = t["child_node"][ci]
+                    break
+            if nxt < 0 and t["node_param"][node] >= 0:
+                nxt = t["node_param"][node]
+            if nxt < 0:
+                return int(best_prefix)
+            node = nxt
+            if t["node_prefix"][node] >= 0:
+                best_prefix = t["node_prefix"][node]
+        rid = t["node_route"][node * 8 + mid]
+        return int(rid) if rid >= 0 else int(best_prefix)
+
+    for segs in paths:
+        path = "/" + "/".join(segs)
+        for method in ("GET", "POST"):
+            route, _, _ = r.match(method, path)
+            want = route.route_id if route else -1
+            assert table_match(method, path) == want, (method, path)
