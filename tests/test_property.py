@@ -139,3 +139,65 @@ def test_router_table_matches_python_match(routes, add_prefix, paths):
             route, _, _ = r.match(method, path)
             want = route.route_id if route else -1
             assert table_match(method, path) == want, (method, path)
+
+
+@SET
+@given(st.lists(st.binary(min_size=1, max_size=200), min_size=0,
+                max_size=6),
+       st.booleans())
+def test_chunked_decode_roundtrip(chunks, with_trailer):
+    # chunks are >=1 byte: a zero-size chunk IS the stream terminator
+    from gofr_amd.http.request import decode_chunked
+    wire = b""
+    for c in chunks:
+        wire += f"{len(c):x}".encode() + b"\r\n" + c + b"\r\n"
+    wire += b"0\r\n"
+    if with_trailer:
+        wire += b"X-Trailer: v\r\n"
+    wire += b"\r\n"
+    assert decode_chunked(wire) == b"".join(chunks)
+
+
+@SET
+@given(st.sampled_from(["GET", "POST", "PUT", "DELETE", "PATCH",
+                        "OPTIONS", "HEAD"]),
+       st.lists(st.sampled_from(["a", "bb", "ccc"]), max_size=3),
+       st.one_of(st.none(), st.sampled_from(["k=v", "q=a+b&x=1"])),
+       st.binary(max_size=120))
+def test_parse_mirror_agrees_with_host_parser(method, segs, query, body):
+    """cpu_parse_route (the kernel's golden model) and
+    parse_request_bytes (the host parser) must agree on method, path,
+    query and body spans for any well-formed request."""
+    import numpy as np
+
+    from gofr_amd import ops
+    from gofr_amd.http.request import METHOD_IDS, parse_request_bytes
+    from gofr_amd.http.router import Router
+
+    path = "/" + "/".join(segs)
+    target = path + (f"?{query}" if query else "")
+    head = (f"{method} {target} HTTP/1.1\r\nHost: h\r\n"
+            + (f"Content-Length: {len(body)}\r\n" if body else "")
+            + "\r\n")
+    raw = head.encode() + body
+
+    r = Router()
+    r.add("GET", "/a", lambda c: None)
+    trie = r.compile()
+    tab = np.asarray([[ops.HK_HOST, 0, 0, 200]], np.int32).reshape(-1)
+    buf = np.frombuffer(raw, np.uint8)
+    fields = ops.cpu_parse_route(buf, np.asarray([0], np.int64),
+                                 np.asarray([len(raw)], np.int32),
+                                 trie, tab)
+    f = fields[0]
+    req = parse_request_bytes(raw)
+    assert f[ops.FI_METHOD] == METHOD_IDS[method]
+    got_path = raw[f[ops.FI_PATH_OFF]:
+                   f[ops.FI_PATH_OFF] + f[ops.FI_PATH_LEN]].decode()
+    assert got_path == req.path == path
+    got_q = raw[f[ops.FI_QUERY_OFF]:
+                f[ops.FI_QUERY_OFF] + f[ops.FI_QUERY_LEN]].decode()
+    assert got_q == (query or "")
+    got_body = raw[f[ops.FI_BODY_OFF]:
+                   f[ops.FI_BODY_OFF] + f[ops.FI_BODY_LEN]]
+    assert bytes(got_body) == req.body == body
