@@ -12,7 +12,13 @@ from hypothesis import given, settings, strategies as st
 from gofr_amd.grpc.codec import MessageDesc, decode_message, encode_message
 from gofr_amd.ops import gzip_static_mirror
 
-SET = settings(max_examples=120, deadline=None)
+# GOFR_FUZZ_EXAMPLES raises the example count for deep campaigns
+# (CI/default stays fast at 120)
+import os  # noqa: E402
+
+SET = settings(max_examples=int(os.environ.get("GOFR_FUZZ_EXAMPLES",
+                                               "120")),
+               deadline=None)
 
 
 @SET
