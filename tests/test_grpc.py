@@ -311,3 +311,58 @@ def test_http2_hpack_many_headers():
     dec = HpackDecoder()
     got = dec.decode(blob)
     assert dict(got) == dict(headers)
+
+
+def test_concurrent_clients_through_codec_worker():
+    """Concurrent unary clients racing the batched codec worker: every
+    call gets ITS OWN response (no cross-wiring under batching)."""
+    import threading
+    import time
+
+    import gofr_amd
+    from gofr_amd.config import MapConfig
+    from gofr_amd.grpc.codec import HELLO_REQUEST, HELLO_RESPONSE
+    from gofr_amd.grpc.server import GRPCClient, GRPCServer, ServiceDesc
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    svc = ServiceDesc("hello.Hello",
+                      {"SayHello": (HELLO_REQUEST, HELLO_RESPONSE)},
+                      gpu_methods={"SayHello": "hello_echo"})
+
+    class Impl:
+        def SayHello(self, ctx, req):
+            raise AssertionError("host path must not run")
+
+    app.RegisterService(svc, Impl())
+    s = GRPCServer(app, 0, batch_window_us=300)
+    s.start()
+    time.sleep(0.2)
+    ok = [0]
+    err = []
+
+    def client(i, n):
+        try:
+            c = GRPCClient("127.0.0.1", s.port)
+            for k in range(n):
+                name = f"c{i}-{k}"
+                resp, status, msg = c.call(
+                    "hello.Hello", "SayHello", {"name": name},
+                    HELLO_REQUEST, HELLO_RESPONSE)
+                assert status == 0, msg
+                assert resp["message"] == f"Hello {name}!"
+                ok[0] += 1
+            c.close()
+        except Exception as e:  # noqa: BLE001 — collected for assert
+            err.append(repr(e))
+
+    try:
+        ts = [threading.Thread(target=client, args=(i, 30))
+              for i in range(8)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        assert not err, err[:3]
+        assert ok[0] == 8 * 30 and s.codec_msgs == 8 * 30
+    finally:
+        s.stop()
