@@ -154,3 +154,53 @@ def test_head_falls_to_catch_all(app):
     app.install_default_routes()
     resp = dispatch(app, _http("HEAD", "/only-get"))
     assert resp.status == 404
+
+
+def test_ctx_trace_user_span(app):
+    """ctx.Trace(name) opens a child span of the request span
+    (reference context.go:45-50); it shares the request's trace id and
+    carries a W3C traceparent for propagation."""
+    spans = {}
+    app.GET("/t", lambda ctx: (spans.setdefault("corr", None),
+                               spans.update(user=ctx.Trace("work")),
+                               spans["user"].End(),
+                               "ok")[-1])
+    resp = dispatch(app, _http("GET", "/t"))
+    assert resp.status == 200
+    corr = dict(resp.headers)["X-Correlation-ID"]
+    user = spans["user"]
+    assert user.trace_id == corr  # same trace as the request span
+    tp = user.traceparent()
+    assert tp.startswith("00-") and corr in tp
+    assert user.duration_us >= 0  # property, µs
+
+
+def test_new_logger_from_env_respects_log_level(map_config):
+    from gofr_amd.config import MapConfig
+    from gofr_amd.logging import DEBUG, ERROR, new_logger_from_env
+    lg = new_logger_from_env(MapConfig({"LOG_LEVEL": "ERROR"}))
+    assert lg.level == ERROR
+    lg = new_logger_from_env(MapConfig({"LOG_LEVEL": "DEBUG"}))
+    assert lg.level == DEBUG
+    # default when unset: INFO (reference logging/level.go GetLevel)
+    from gofr_amd.logging import INFO
+    assert new_logger_from_env(MapConfig({})).level == INFO
+
+
+def test_rfc3339nano_format():
+    import re
+
+    from gofr_amd.http.middleware import rfc3339nano
+    s = rfc3339nano()
+    assert re.match(
+        r"^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}\.\d{9}[+-]\d{2}:\d{2}$"
+        r"|^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}\.\d{9}Z$", s), s
+
+
+def test_reason_phrases():
+    from gofr_amd.http.responder import reason_phrase
+    for st_code, want in [(200, "OK"), (404, "Not Found"),
+                          (401, "Unauthorized"),
+                          (405, "Method Not Allowed"),
+                          (500, "Internal Server Error")]:
+        assert reason_phrase(st_code) == want
