@@ -58,15 +58,18 @@ class RequestLog:
 
 def rfc3339nano(t: float | None = None) -> str:
     """RFC3339Nano start-time format (middleware/logger.go:51).
-    Go's layout renders the zone as Z07:00 — colon-separated offset,
-    "Z" for UTC (strftime %z gives +0000, which is NOT RFC 3339)."""
+    Matches Go's layout exactly: zone as Z07:00 (colon-separated
+    offset, "Z" for UTC — strftime %z gives +0000, which is NOT
+    RFC 3339) and trailing fractional zeros trimmed (".999999999"
+    layout semantics)."""
     if t is None:
         t = time.time()
     ns = int((t % 1) * 1e9)
     z = time.strftime("%z", time.localtime(t))
     z = "Z" if z in ("+0000", "-0000") else z[:3] + ":" + z[3:]
+    frac = f".{ns:09d}".rstrip("0").rstrip(".")
     return time.strftime("%Y-%m-%dT%H:%M:%S", time.localtime(t)) + \
-        f".{ns:09d}" + z
+        frac + z
 
 
 def make_request_log(span, request, status: int, start: float,

@@ -191,10 +191,13 @@ def test_rfc3339nano_format():
     import re
 
     from gofr_amd.http.middleware import rfc3339nano
-    s = rfc3339nano()
-    assert re.match(
-        r"^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}\.\d{9}[+-]\d{2}:\d{2}$"
-        r"|^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}\.\d{9}Z$", s), s
+    # Go RFC3339Nano: optional trimmed fraction, zone Z or +hh:mm
+    pat = (r"^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}(\.\d{1,9})?"
+           r"(Z|[+-]\d{2}:\d{2})$")
+    assert re.match(pat, rfc3339nano()), rfc3339nano()
+    # trimming parity with Go: .5 -> ".5", exact second -> no fraction
+    assert rfc3339nano(1000000000.5).count(".5") == 1
+    assert "." not in rfc3339nano(1000000000.0).split("T")[1]
 
 
 def test_reason_phrases():
