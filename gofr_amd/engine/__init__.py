@@ -139,11 +139,13 @@ class _Lane:
         self.e_in = t.cuda.Event()
         self.e_k = t.cuda.Event()
         # ingress block: [off (nb+1)*int64 | len nb*int32 | serial u64 |
-        # pad16 | request bytes] — ONE contiguous region so the whole
-        # batch ingress is ONE SDMA copy (each queued SDMA op costs a
-        # scheduling gap; 4 ops/batch paced the pipeline). The seed
-        # rides in the offsets tail (slot n).
-        self.hdr_bytes = (((nb + 1) * 8 + nb * 4 + 8 + 15) // 16) * 16
+        # Date slot 32B | pad16 | request bytes] — ONE contiguous region
+        # so the whole batch ingress is ONE SDMA copy (each queued SDMA
+        # op costs a scheduling gap; 4 ops/batch paced the pipeline).
+        # The seed rides in the offsets tail (slot n); the Date slot
+        # carries the batch's 29-byte IMF-fixdate for k_respond.
+        self.date_off = (nb + 1) * 8 + nb * 4 + 8
+        self.hdr_bytes = ((self.date_off + 32 + 15) // 16) * 16
         self.d_ingress = t.empty(self.hdr_bytes + max_bytes,
                                  dtype=t.uint8, device=dev)
         self.d_req_off = self.d_ingress[:(nb + 1) * 8].view(t.int64)
@@ -164,6 +166,8 @@ class _Lane:
         self.p_req_len = self.p_ingress[
             (nb + 1) * 8:(nb + 1) * 8 + nb * 4].view(t.int32)
         self.p_reqs = self.p_ingress[self.hdr_bytes:]
+        self.p_date_np = self.p_ingress[
+            self.date_off:self.date_off + 29].numpy()
         # egress ring via hipHostMalloc: D2H SDMA works into it (torch
         # pin_memory is hipHostRegister'd, which the runtime serves with
         # a blit kernel instead — see ops.HipOps.host_alloc)
@@ -241,6 +245,10 @@ class BatchEngine:
         self.pipeline = max(1, pipeline)
         self.program = RouteProgram(app)
         self._seed = 0x6F667247414D4421  # advanced per batch
+        # Date header source (Go's net/http attaches Date to every
+        # response; parity). Overridable for deterministic tests.
+        self._date_fn = time.time
+        self._date_cache = (None, b"")
         self.device = None
         self.torch = None
         try:
@@ -307,6 +315,17 @@ class BatchEngine:
         self._seed = ops.splitmix64(self._seed)
         return self._seed
 
+    def _date29(self) -> bytes:
+        t = int(self._date_fn())
+        if self._date_cache[0] != t:
+            self._date_cache = (t, ops.imf_date(t))
+        return self._date_cache[1]
+
+    def _stamp_date(self, ln) -> None:
+        """Write the batch's IMF-fixdate into the lane's ingress-block
+        Date slot (travels to the GPU in the same SDMA copy)."""
+        ln.p_date_np[:] = np.frombuffer(self._date29(), np.uint8)
+
     # -- main entries ---------------------------------------------------------
     def process(self, payloads: list[bytes]) -> list[bytes]:
         """Convenience: pack, run, slice. Returns one response byte string
@@ -350,7 +369,8 @@ class BatchEngine:
             self.program.blob, host_blob, host_tab, seed,
             auth_env=self.program.auth_env,
             gzip_min=self.app.gzip_min_size or 0,
-            etag_on=getattr(self.app, "etag_on", False))
+            etag_on=getattr(self.app, "etag_on", False),
+            date29=self._date29())
         # compaction mirror (same round16 layout as k_compact)
         n = len(lens)
         pads = (resp_len + 15) & ~15
@@ -382,6 +402,7 @@ class BatchEngine:
         t = self.torch
         ln = self.lanes[lane_idx]
         ln.n, ln.nbytes = n, nbytes
+        self._stamp_date(ln)
         seed = self._next_seed()
         signed = seed - (1 << 64) if seed >= (1 << 63) else seed
         ln.p_seed[0] = signed
@@ -449,6 +470,7 @@ class BatchEngine:
         saved in launches.)"""
         t = self.torch
         ln = self.lanes[lane_idx]
+        self._stamp_date(ln)
         # warmup pass (allocations settle) then arm
         with t.cuda.stream(self.s_k):
             self._submit_body(ln, n, nbytes)
@@ -525,6 +547,7 @@ class BatchEngine:
             a.p_serial = 0
         a.serial = 0
         a.flagged = 1 if self._flagged else 0
+        a.d_date = ln.d_ingress.data_ptr() + ln.date_off
         if self._flagged:
             # event-free channel pipeline: the whole chain (gate ->
             # kernels -> egress -> k_done) rides one of two channel
@@ -572,7 +595,8 @@ class BatchEngine:
                          auth_env=self.program.auth_env,
                          gzip_min=self.app.gzip_min_size or 0,
                          etag_on=1 if getattr(self.app, "etag_on", False)
-                         else 0)
+                         else 0,
+                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
@@ -742,6 +766,10 @@ class BatchEngine:
         seed = self._next_seed()
         ln.p_seed[0] = seed - (1 << 64) if seed >= (1 << 63) else seed
         ln.d_seed.copy_(ln.p_seed, non_blocking=True)
+        # no ingress-block copy on this path: stage the Date slot alone
+        self._stamp_date(ln)
+        ln.d_ingress[ln.date_off:ln.date_off + 29].copy_(
+            ln.p_ingress[ln.date_off:ln.date_off + 29], non_blocking=True)
         stream = t.cuda.current_stream(self.device).cuda_stream
         ln.d_host_needed.zero_()
         self.hip.parse_route(stream, d_reqs, d_req_off, d_req_len,
@@ -775,7 +803,8 @@ class BatchEngine:
                          ln.d_seed, auth_env=self.program.auth_env,
                          gzip_min=self.app.gzip_min_size or 0,
                          etag_on=1 if getattr(self.app, "etag_on", False)
-                         else 0)
+                         else 0,
+                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off)
         return ln.d_resp, ln.d_resp_len
 
     # -- host trampoline ------------------------------------------------------

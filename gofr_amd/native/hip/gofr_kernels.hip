@@ -790,7 +790,8 @@ __device__ __forceinline__ void respond_impl(
         const uint8_t* __restrict__ host_blob,
         const int32_t* __restrict__ host_tab,
         uint64_t seed, int auth_env_off, int auth_env_len,
-        int gzip_min, int etag_on, uint8_t* obuf, uint8_t* plainbuf,
+        int gzip_min, int etag_on, const uint8_t* date29,
+        uint8_t* obuf, uint8_t* plainbuf,
         uint32_t* hash, const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
@@ -952,7 +953,9 @@ __device__ __forceinline__ void respond_impl(
     // segment offsets (all lanes compute identically)
     const int o_status = 9;                       // after "HTTP/1.1 "
     const int o_reason = o_status + 3 + 1;        // "xxx "
-    const int o_ct = o_reason + rlen_reason + 2;  // reason \r\n
+    const int o_date = o_reason + rlen_reason + 2;  // reason \r\n
+    const int date_len = date29 ? 37 : 0;  // "Date: " + 29 + CRLF
+    const int o_ct = o_date + date_len;
     const int o_ce = o_ct + ct_len;
     const int et_len = etag_on ? ETAG_HDR_LEN : 0;
     const int o_etag = o_ce + ce_len;
@@ -967,6 +970,15 @@ __device__ __forceinline__ void respond_impl(
     // cooperative writes into LDS: each segment <=64B -> one store per lane
     if (lane < 9) obuf[lane] = HDR_P1[lane];
     if (lane < (unsigned)rlen_reason) obuf[o_reason + lane] = reason[lane];
+    if (date29) {
+        static const char DK[6] = {'D', 'a', 't', 'e', ':', ' '};
+        if (lane < 6) obuf[o_date + lane] = DK[lane];
+        if (lane < 29) obuf[o_date + 6 + lane] = date29[lane];
+        if (lane == 0) {
+            obuf[o_date + 35] = '\r';
+            obuf[o_date + 36] = '\n';
+        }
+    }
     if (lane < (unsigned)ct_len) obuf[o_ct + lane] = ct_str[lane];
     if (content_enc && lane < ce_len) obuf[o_ce + lane] = HDR_CE[lane];
     if (etag_on && lane < 7) obuf[o_etag + lane] = HDR_ETAG[lane];
@@ -1108,7 +1120,8 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ host_blob,
           const int32_t* __restrict__ host_tab,
           const uint64_t* __restrict__ seed_ptr,
-          int auth_env_off, int auth_env_len, int etag_on) {
+          int auth_env_off, int auth_env_len, int etag_on,
+          const uint8_t* __restrict__ date29) {
     __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
     const int wv = threadIdx.x / WAVE;
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
@@ -1116,6 +1129,7 @@ k_respond(const uint8_t* __restrict__ reqs,
     respond_impl<false>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                         handler_tab, n_routes, blob, host_blob, host_tab,
                         *seed_ptr, auth_env_off, auth_env_len, 0, etag_on,
+                        date29,
                         obuf_all + wv * MAX_SLOT, nullptr, nullptr, nullptr,
                         req, lane_id());
 }
@@ -1133,7 +1147,7 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
              const int32_t* __restrict__ host_tab,
              const uint64_t* __restrict__ seed_ptr,
              int auth_env_off, int auth_env_len, int gzip_min,
-             int etag_on) {
+             int etag_on, const uint8_t* __restrict__ date29) {
     // single __shared__ block (cdna guide §5 trap 4a)
     __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
                            WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4];
@@ -1156,7 +1170,7 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
     respond_impl<true>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                        handler_tab, n_routes, blob, host_blob, host_tab,
                        *seed_ptr, auth_env_off, auth_env_len, gzip_min,
-                       etag_on,
+                       etag_on, date29,
                        obuf_all + wv * MAX_SLOT, plain_all + wv * MAX_SLOT,
                        hash_all + wv * GZ_HASH_SIZE, crc_tab,
                        req, lane_id());
@@ -1723,6 +1737,10 @@ typedef struct {
     // (kernels + egress), so batches on alternating channel streams
     // overlap each other and the SDMA ingress.
     int flagged;
+    // 29-byte IMF-fixdate in device memory (rides the ingress block's
+    // header slot; all responses of a batch share it). 0 -> no Date
+    // header (Go's net/http attaches Date to every response; parity).
+    const void* d_date;
 } GofrSubmitArgs;
 
 // k_gate — single-wave stream gate: spins (system-scope acquire
@@ -1836,7 +1854,7 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
                            a->auth_env_off, a->auth_env_len, a->gzip_min,
-                           a->etag_on);
+                           a->etag_on, (const uint8_t*)a->d_date);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS),
                            0, s_k,
@@ -1848,7 +1866,8 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->blob,
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
-                           a->auth_env_off, a->auth_env_len, a->etag_on);
+                           a->auth_env_off, a->auth_env_len, a->etag_on,
+                           (const uint8_t*)a->d_date);
     }
     if ((rc = hipGetLastError())) return (int)rc;
     hipLaunchKernelGGL(k_padscan, dim3(1), dim3(SCAN_THREADS), 0, s_k,
@@ -2091,7 +2110,7 @@ int gofr_launch_respond(
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
         const void* seed_ptr, int auth_env_off, int auth_env_len,
-        int gzip_min, int etag_on) {
+        int gzip_min, int etag_on, const void* date29) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     if (gzip_min > 0) {
         hipLaunchKernelGGL(k_respond_gz, dim3(blocks), dim3(BLOCK_THREADS),
@@ -2104,7 +2123,8 @@ int gofr_launch_respond(
                            (const uint8_t*)blob, (const uint8_t*)host_blob,
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
-                           auth_env_off, auth_env_len, gzip_min, etag_on);
+                           auth_env_off, auth_env_len, gzip_min, etag_on,
+                           (const uint8_t*)date29);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                            (hipStream_t)stream,
@@ -2116,7 +2136,8 @@ int gofr_launch_respond(
                            (const uint8_t*)blob, (const uint8_t*)host_blob,
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
-                           auth_env_off, auth_env_len, etag_on);
+                           auth_env_off, auth_env_len, etag_on,
+                           (const uint8_t*)date29);
     }
     return (int)hipGetLastError();
 }

@@ -96,6 +96,7 @@ class GofrSubmitArgs(ctypes.Structure):
         ("p_serial", ctypes.c_void_p),
         ("serial", ctypes.c_uint64),
         ("flagged", ctypes.c_int),
+        ("d_date", ctypes.c_void_p),  # 29-byte IMF-fixdate in device mem
     ]
 
 
@@ -129,7 +130,7 @@ class HipOps:
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
             ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
-            ctypes.c_int]
+            ctypes.c_int, ctypes.c_void_p]
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
@@ -196,7 +197,7 @@ class HipOps:
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
                 host_blob_t, host_tab_t, seed_t, auth_env=(0, 0),
-                gzip_min=0, etag_on=0):
+                gzip_min=0, etag_on=0, date_ptr=0):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -210,7 +211,8 @@ class HipOps:
             ctypes.c_void_p(host_blob_t.data_ptr()),
             ctypes.c_void_p(host_tab_t.data_ptr()),
             ctypes.c_void_p(seed_t.data_ptr()),
-            auth_env[0], auth_env[1], gzip_min, etag_on)
+            auth_env[0], auth_env[1], gzip_min, etag_on,
+            ctypes.c_void_p(date_ptr))
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
 
@@ -274,6 +276,17 @@ class HipOps:
 # ---------------------------------------------------------------------------
 
 M64 = (1 << 64) - 1
+
+
+def imf_date(epoch_s: int) -> bytes:
+    """IMF-fixdate (RFC 9110 §5.6.7), always 29 bytes — the Date
+    header value Go's net/http attaches to every response (the
+    reference inherits it; parity surface). Shared by the kernel
+    (host-rendered, rides the ingress block) and the CPU mirrors."""
+    import email.utils
+    s = email.utils.formatdate(epoch_s, usegmt=True).encode()
+    assert len(s) == 29, s
+    return s
 
 
 def splitmix64(x: int) -> int:
@@ -556,7 +569,7 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
                 seed: int, auth_env=(0, 0), gzip_min: int = 0,
-                etag_on: bool = False):
+                etag_on: bool = False, date29: bytes = b""):
     """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
     n = len(fields)
     host_tab = np.asarray(host_tab, np.int32).reshape(-1)
@@ -638,7 +651,9 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 if inm == b'"' + tag.encode() + b'"':
                     not_modified = True
         head = (b"HTTP/1.1 " + f"{status:03d}".encode() + b" " + reason +
-                b"\r\n" + _CT_STRS[ct_id] +
+                b"\r\n" +
+                (b"Date: " + date29 + b"\r\n" if date29 else b"") +
+                _CT_STRS[ct_id] +
                 (b"Content-Encoding: gzip\r\n" if content_enc else b"") +
                 etag_hdr +
                 _CORS +

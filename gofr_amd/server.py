@@ -136,7 +136,10 @@ def _log_request(app, span, request, status, start, t0) -> None:
 
 
 def serialize_response(resp: _CapturedResponse, keep_alive: bool) -> bytes:
-    head = [f"HTTP/1.1 {resp.status} {reason_phrase(resp.status)}"]
+    from .ops import imf_date
+    head = [f"HTTP/1.1 {resp.status} {reason_phrase(resp.status)}",
+            # Go's net/http writes Date on every response; parity
+            f"Date: {imf_date(int(time.time())).decode()}"]
     seen_ct = False
     for k, v in resp.headers:
         if k.lower() == "content-type":

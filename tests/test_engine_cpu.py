@@ -267,6 +267,7 @@ def test_deterministic_replay_across_engines(map_config):
     outs = []
     for _ in range(2):
         eng = BatchEngine(build_app(map_config), device="cpu")
+        eng._date_fn = lambda: 1789300000.0  # pin the Date header
         outs.append([eng.process(b) for b in batches])
     assert outs[0] == outs[1]
 
@@ -278,6 +279,7 @@ def test_poisoned_slack_does_not_leak(map_config):
     filled with 0xAA garbage; the response bytes must match."""
     eng1 = BatchEngine(build_app(map_config), device="cpu")
     eng2 = BatchEngine(build_app(map_config), device="cpu")
+    eng1._date_fn = eng2._date_fn = lambda: 1789300000.0
     payloads = [http_req("GET", "/greet"),
                 http_req("POST", "/echo", body=b'{"k":"v"}')]
     buf, offs, lens = pack_batch(payloads)
@@ -290,3 +292,17 @@ def test_poisoned_slack_does_not_leak(map_config):
         a = out_a[int(ro_a[i]):int(ro_a[i]) + int(rl_a[i])]
         b = out_b[int(ro_b[i]):int(ro_b[i]) + int(rl_b[i])]
         assert np.array_equal(a, b)
+
+
+def test_date_header_on_engine_responses(engine):
+    """Every engine response carries a Date header in IMF-fixdate
+    format (Go net/http parity; rendered per batch by k_respond from
+    the ingress block's Date slot — mirror: ops.imf_date)."""
+    import re
+
+    st, hdrs, _ = run_one(engine, http_req("GET", "/greet"))
+    assert st == 200
+    assert re.match(
+        r"^(Mon|Tue|Wed|Thu|Fri|Sat|Sun), \d{2} "
+        r"(Jan|Feb|Mar|Apr|May|Jun|Jul|Aug|Sep|Oct|Nov|Dec) "
+        r"\d{4} \d{2}:\d{2}:\d{2} GMT$", hdrs["date"]), hdrs["date"]

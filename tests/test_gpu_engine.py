@@ -69,6 +69,7 @@ def test_kernels_match_cpu_mirror_bytes():
     cpu = BatchEngine(app, device="cpu", max_batch=4096)
     # identical correlation-id seeds
     cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
 
     raws = mixed_payloads(512)
     g_out = gpu.process(list(raws))
@@ -137,6 +138,7 @@ def test_armed_flagged_pipeline_matches_mirror():
     gpu = BatchEngine(app, device="cuda", max_batch=512, pipeline=2)
     cpu = BatchEngine(app, device="cpu", max_batch=512)
     cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
 
     raws = []
     for i in range(512):
@@ -192,6 +194,7 @@ def test_armed_staged_event_path_matches_mirror():
     gpu._flagged = False  # force the staged event path
     cpu = BatchEngine(app, device="cpu", max_batch=256)
     cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
     body = b'{"x":"yz"}'
     raws = [b"POST /echo HTTP/1.1\r\nHost: h\r\n"
             b"Content-Type: application/json\r\n"
@@ -234,6 +237,7 @@ def test_fuzz_gpu_matches_mirror():
     gpu = BatchEngine(app, device="cuda", max_batch=256)
     cpu = BatchEngine(app, device="cpu", max_batch=256)
     cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
     rng = random.Random(99)
     base = (b"POST /echo HTTP/1.1\r\nHost: h\r\n"
             b"Content-Type: application/json\r\n"
@@ -269,6 +273,7 @@ def test_armed_with_middlewares_matches_mirror():
     gpu = BatchEngine(app, device="cuda", max_batch=128, pipeline=2)
     cpu = BatchEngine(app, device="cpu", max_batch=128)
     cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
     tok = hmac_token(b"s3cret", "POST", "/echo")
     body = b'{"pad":"' + b"q" * 300 + b'"}'
     raws = []

@@ -198,3 +198,9 @@ def test_connection_churn_and_concurrency(native_server):
         t.join()
     assert not err, err[:3]
     assert ok[0] == 8 * 25 + 10 * 50
+
+
+def test_date_header_over_sockets(native_server):
+    st, hdrs, _ = _req(native_server.port, "GET", "/greet")
+    assert st == 200
+    assert hdrs.get("Date", "").endswith("GMT")
