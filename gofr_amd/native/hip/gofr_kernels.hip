@@ -1084,6 +1084,10 @@ __device__ __forceinline__ void respond_impl(
         }
     }
 
+    // HEAD: headers (incl. real Content-Length) without the body —
+    // net/http discards handler writes for HEAD (RFC 9110 §9.3.2)
+    if (F[FI_METHOD] == M_HEAD) send_body = 0;
+
     const int total = hl + send_body;
     // ---- one coalesced 16B/lane sweep LDS -> global ------------------------
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");

@@ -669,6 +669,8 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 cl, b"Content-Length: " + b"0" * len(str(body_total)), 1)
             head = head[:9] + b"304" + head[12:]
             payload = head
+        elif int(F[FI_METHOD]) == 6:  # HEAD (mirror of M_HEAD): no body
+            payload = head
         elif env:
             payload = head + b'{"data":' + body_src + b"}"
         else:

@@ -135,7 +135,8 @@ def _log_request(app, span, request, status, start, t0) -> None:
         make_request_log(span, request, status, start, dur_us))
 
 
-def serialize_response(resp: _CapturedResponse, keep_alive: bool) -> bytes:
+def serialize_response(resp: _CapturedResponse, keep_alive: bool,
+                       head_only: bool = False) -> bytes:
     from .ops import imf_date
     head = [f"HTTP/1.1 {resp.status} {reason_phrase(resp.status)}",
             # Go's net/http writes Date on every response; parity
@@ -149,7 +150,10 @@ def serialize_response(resp: _CapturedResponse, keep_alive: bool) -> bytes:
         head.append(f"{k}: {v}")
     head.append(f"Content-Length: {len(resp.body)}")
     head.append("Connection: " + ("keep-alive" if keep_alive else "close"))
-    return ("\r\n".join(head) + "\r\n\r\n").encode("latin-1") + resp.body
+    # HEAD: headers (incl. real Content-Length) without the body —
+    # net/http discards handler writes for HEAD (RFC 9110 §9.3.2)
+    body = b"" if head_only else resp.body
+    return ("\r\n".join(head) + "\r\n\r\n").encode("latin-1") + body
 
 
 class HTTPServer:
@@ -227,7 +231,8 @@ class HTTPServer:
                 request = parse_request_bytes(raw, remote_addr=remote)
                 resp = dispatch(self.app, request)
                 keep = request.headers.get("connection", "").lower() != "close"
-                conn.sendall(serialize_response(resp, keep))
+                conn.sendall(serialize_response(
+                    resp, keep, head_only=request.method == "HEAD"))
                 if not keep:
                     return
         except (OSError, ValueError):

@@ -306,3 +306,16 @@ def test_date_header_on_engine_responses(engine):
         r"^(Mon|Tue|Wed|Thu|Fri|Sat|Sun), \d{2} "
         r"(Jan|Feb|Mar|Apr|May|Jun|Jul|Aug|Sep|Oct|Nov|Dec) "
         r"\d{4} \d{2}:\d{2}:\d{2} GMT$", hdrs["date"]), hdrs["date"]
+
+
+def test_head_response_has_headers_but_no_body(engine):
+    """HEAD responses carry the real Content-Length but no body bytes
+    (net/http discards handler writes for HEAD; RFC 9110 §9.3.2)."""
+    raw = engine.process([http_req("HEAD", "/nope"),
+                          http_req("GET", "/nope")])
+    head_resp, get_resp = raw
+    assert head_resp.endswith(b"\r\n\r\n")  # headers only
+    st, hdrs, body = parse_http_response(head_resp)
+    assert st == 404 and body == b""
+    _, _, get_body = parse_http_response(get_resp)
+    assert int(hdrs["content-length"]) == len(get_body) > 0
