@@ -56,6 +56,8 @@ def mixed_payloads(n):
             out.append(http_req("GET", "/greet"))
         elif r == 4:
             out.append(http_req("GET", f"/user/{i}"))
+        elif r == 5 and i % 2:
+            out.append(http_req("HEAD", "/greet"))  # headers, no body
         else:
             out.append(http_req("GET", f"/missing/{i}",
                                 headers={"Connection": "close"}))
