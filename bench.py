@@ -94,7 +94,9 @@ def run_single(eng, payloads, steps, warmup):
         ln.p_req_off[:n] = torch.from_numpy(offs)
         ln.p_req_len[:n] = torch.from_numpy(lens)
     P = len(eng.lanes)
-    # capture the per-lane batch pipeline as a hipGraph (replayed below)
+    # arm each lane: build the native GofrSubmitArgs block the one-call
+    # staged submit replays per batch (capture_graph keeps its name
+    # from the hipGraph era)
     for li in range(P):
         eng.capture_graph(n, nbytes, li)
     # warmup: serial batches
@@ -144,8 +146,8 @@ def run_single(eng, payloads, steps, warmup):
 
 def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
     """BASELINE config 5 (single-GPU shard): mixed HTTP+gRPC batches with
-    device-resident conn state. Per lane-step: hipGraph replay of the
-    HTTP pipeline, then (same stream) H2D of the gRPC sub-batch,
+    device-resident conn state. Per lane-step: armed native submit of the
+    HTTP pipeline, then (lane stream) H2D of the gRPC sub-batch,
     k_varint_spans + k_grpc_echo, conn-state scatter update
     (ConnStateTable.record_batch), D2H of the gRPC frames. Every request
     is attributed to one of `conns` open connections."""
@@ -195,7 +197,7 @@ def run_config5(eng, payloads, steps, warmup, n_grpc, conns):
                 glens.astype(np.int64)).to(dev)
             self.ev = t.cuda.Event()
 
-    # HTTP staging + graph capture per lane (same as run_single)
+    # HTTP staging + lane arming (same as run_single)
     buf, offs, lens = pack_batch(payloads)
     nbytes = int(offs[-1] + lens[-1])
     for ln in eng.lanes:
