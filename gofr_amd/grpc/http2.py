@@ -239,22 +239,24 @@ class HpackEncoder:
     """Emits literal-without-indexing fields only (always legal)."""
 
     @staticmethod
+    def _put_str(out: bytearray, raw: bytes) -> None:
+        # string length as an HPACK 7-bit-prefix integer, no huffman
+        if len(raw) < 0x7F:
+            out.append(len(raw))
+        else:
+            out.append(0x7F)
+            rest = len(raw) - 0x7F
+            while rest >= 0x80:
+                out.append((rest & 0x7F) | 0x80)
+                rest >>= 7
+            out.append(rest)
+        out.extend(raw)
+
+    @staticmethod
     def encode(headers: list[tuple[str, str]]) -> bytes:
         out = bytearray()
         for name, value in headers:
             out.append(0x00)  # literal without indexing, new name
-            nb = name.encode()
-            vb = value.encode()
-            out.append(len(nb))  # no huffman, len < 127 assumed
-            out.extend(nb)
-            if len(vb) < 127:
-                out.append(len(vb))
-            else:
-                out.append(0x7F)
-                rest = len(vb) - 0x7F
-                while rest >= 0x80:
-                    out.append((rest & 0x7F) | 0x80)
-                    rest >>= 7
-                out.append(rest)
-            out.extend(vb)
+            HpackEncoder._put_str(out, name.encode())
+            HpackEncoder._put_str(out, value.encode())
         return bytes(out)

@@ -74,8 +74,8 @@ def test_protobuf_codec_roundtrip(name, count, flag, blob, ratio):
 @SET
 @given(st.lists(st.tuples(
     st.text(alphabet=st.characters(min_codepoint=97, max_codepoint=122),
-            min_size=1, max_size=10),
-    st.text(max_size=30)), min_size=0, max_size=8))
+            min_size=1, max_size=300),
+    st.text(max_size=400)), min_size=0, max_size=8))
 def test_hpack_roundtrip(headers):
     from gofr_amd.grpc.http2 import HpackDecoder, HpackEncoder
     wire = HpackEncoder.encode(headers)
