@@ -207,3 +207,25 @@ def test_parse_mirror_agrees_with_host_parser(method, segs, query, body):
     got_body = raw[f[ops.FI_BODY_OFF]:
                    f[ops.FI_BODY_OFF] + f[ops.FI_BODY_LEN]]
     assert bytes(got_body) == req.body == body
+
+
+INNER = MessageDesc("Inner", {1: ("id", "int32"), 2: ("tag", "string")})
+OUTER = MessageDesc("Outer", {
+    1: ("items", "repeated_message", INNER),
+    2: ("names", "repeated_string"),
+    3: ("kind", "sint64"),
+})
+
+
+@SET
+@given(st.lists(st.tuples(st.integers(-(2 ** 31), 2 ** 31 - 1),
+                          st.text(max_size=30)), max_size=5),
+       st.lists(st.text(max_size=20), max_size=5),
+       st.integers(-(2 ** 62), 2 ** 62))
+def test_protobuf_nested_repeated_roundtrip(items, names, kind):
+    msg = {"items": [{"id": i, "tag": t} for i, t in items],
+           "names": names, "kind": kind}
+    out = decode_message(encode_message(msg, OUTER), OUTER)
+    assert [(d["id"], d["tag"]) for d in out["items"]] == items
+    assert out["names"] == names
+    assert out["kind"] == kind
