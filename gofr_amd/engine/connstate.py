@@ -18,7 +18,7 @@ table refuses to oversubscribe what the device reports free.
 Host-side id management (open/close free list) is control-plane work and
 stays on the CPU, mirroring the reference's accept-loop split; the data
 plane (`record_batch`) is pure device ops (index_add_ scatter updates on
-the current stream — hipGraph-capturable, used by bench.py --config5).
+the current stream; bench.py exercises it via --grpc-frac/--conns).
 
 Without a GPU the same table runs on numpy so every semantic is testable
 on the CPU box (tests/test_connstate.py).
