@@ -19,7 +19,7 @@ from __future__ import annotations
 import re
 import threading
 import time
-from typing import Any, Optional
+from typing import Optional
 
 
 class Log:

@@ -90,8 +90,6 @@ class GRPCServer:
     def _codec_worker(self) -> None:
         import queue as _queue
 
-        import numpy as np
-
         from .. import ops
         from ..engine import pack_batch
         gpu = None

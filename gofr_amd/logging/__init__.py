@@ -17,7 +17,6 @@ from __future__ import annotations
 
 import io
 import json
-import os
 import sys
 import threading
 import time

@@ -8,7 +8,6 @@ from __future__ import annotations
 
 import contextlib
 import io
-import sys
 
 from .. import logging as gofr_logging
 

@@ -115,7 +115,6 @@ def test_gpu_large_echo_batch_correct():
 
 def test_native_extension_is_loaded():
     """Driver policy: the HIP .so must actually be loaded on a GPU box."""
-    import ctypes
     from gofr_amd.ops import HipOps, _SO_PATH
     ops = HipOps()
     assert ops.lib is not None

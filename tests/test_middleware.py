@@ -5,7 +5,6 @@ import gzip
 import json
 
 import numpy as np
-import pytest
 
 import gofr_amd
 from gofr_amd import handlers, ops
