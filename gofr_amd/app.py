@@ -118,6 +118,24 @@ class App:
             return fn
         return deco
 
+    def put(self, pattern: str):
+        def deco(fn):
+            self.PUT(pattern, fn)
+            return fn
+        return deco
+
+    def delete(self, pattern: str):
+        def deco(fn):
+            self.DELETE(pattern, fn)
+            return fn
+        return deco
+
+    def patch(self, pattern: str):
+        def deco(fn):
+            self.PATCH(pattern, fn)
+            return fn
+        return deco
+
     # -- gRPC — reference: gofr.go:42-46 -------------------------------------
     def RegisterService(self, service, impl=None):
         """Register a gRPC service implementation (see gofr_amd.grpc)."""

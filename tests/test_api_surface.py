@@ -207,3 +207,23 @@ def test_reason_phrases():
                           (405, "Method Not Allowed"),
                           (500, "Internal Server Error")]:
         assert reason_phrase(st_code) == want
+
+
+def test_all_verb_decorators(app):
+    @app.put("/p")
+    def p(ctx):
+        return "put"
+
+    @app.delete("/d")
+    def d(ctx):
+        return "del"
+
+    @app.patch("/pa")
+    def pa(ctx):
+        return "patch"
+
+    for method, path, want in [("PUT", "/p", "put"),
+                               ("DELETE", "/d", "del"),
+                               ("PATCH", "/pa", "patch")]:
+        resp = dispatch(app, _http(method, path))
+        assert json.loads(resp.body) == {"data": want}, (method, path)
