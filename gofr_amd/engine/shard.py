@@ -87,7 +87,9 @@ class AllToAllSharder:
                 self.engine.program.handler_tab, self.engine.program.blob,
                 host_blob, host_tab, self.engine._next_seed(),
                 auth_env=self.engine.program.auth_env,
-                gzip_min=self.engine.app.gzip_min_size or 0)
+                gzip_min=self.engine.app.gzip_min_size or 0,
+                etag_on=getattr(self.engine.app, "etag_on", False),
+                date29=self.engine._date29())
             t = self.t
             dist.all_to_all_single(self.d_resp_sh,
                                    t.from_numpy(resp_slots))
