@@ -730,11 +730,16 @@ class BatchEngine:
         every = self.app.request_log_every
         logger.log_record(1, BatchLog(n, bytes_in, bytes_out, dur_ms))
         for r in range(0, n, every):
-            raw = req_np[int(offs[r]):int(offs[r]) + 256].tobytes()
+            raw = req_np[int(offs[r]):int(offs[r]) + 512].tobytes()
             line = raw.split(b"\r\n", 1)[0].decode("latin1")
             parts = line.split(" ")
             method = parts[0] if parts else "?"
             uri = parts[1] if len(parts) > 1 else "?"
+            ua = ""
+            for h in raw.split(b"\r\n\r\n", 1)[0].split(b"\r\n")[1:]:
+                if h[:11].lower() == b"user-agent:":
+                    ua = h[11:].strip().decode("latin1")
+                    break
             resp = out_np[int(roffs[r]):int(roffs[r]) +
                           min(256, int(rlens[r]))].tobytes()
             status = 0
@@ -750,7 +755,7 @@ class BatchEngine:
             logger.log_record(1, RequestLog(
                 trace_id=corr, start_time=rfc3339nano(),
                 response_time_us=dur_ms * 1000, method=method,
-                user_agent="", ip="", uri=uri, response=status))
+                user_agent=ua, ip="", uri=uri, response=status))
 
     def process_device(self, d_reqs, d_req_off, d_req_len, n,
                        lane_idx=0, sync_host=True):
