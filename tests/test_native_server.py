@@ -204,3 +204,31 @@ def test_date_header_over_sockets(native_server):
     st, hdrs, _ = _req(native_server.port, "GET", "/greet")
     assert st == 200
     assert hdrs.get("Date", "").endswith("GMT")
+
+
+def test_lifecycle_cycles_leak_free():
+    """start -> serve -> stop cycles must not leak fds or threads
+    (reactor sockets, epoll fds, serve thread all reclaimed)."""
+    import os
+
+    pytest.importorskip("gofr_amd._core")
+    from gofr_amd.engine import GPUServer
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/greet", handlers.static_json("Hello World!"))
+    app.install_default_routes()
+
+    def fds():
+        return len(os.listdir("/proc/self/fd"))
+
+    base = None
+    for cycle in range(10):
+        s = GPUServer(app, 0, batch_window_us=500, max_batch=256)
+        s.start()
+        st, _, body = _req(s.port, "GET", "/greet")
+        assert st == 200 and b"Hello World!" in body
+        s.stop()
+        if cycle == 2:
+            base = (fds(), threading.active_count())
+    assert fds() - base[0] <= 4, "fd leak across server lifecycles"
+    assert threading.active_count() - base[1] <= 2, "thread leak"
