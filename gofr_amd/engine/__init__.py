@@ -293,9 +293,11 @@ class BatchEngine:
         # stage streams shared by all lanes: one per direction so the
         # host link runs full duplex, one for compute. s_k runs at high
         # priority so the parse/respond chain preempts the (link-bound,
-        # grid-capped) egress sweep for CUs. With GOFR_CHANNELS=2 (the
-        # default), lanes alternate between two ingress+compute channel
-        # streams so a batch's kernels follow its own H2D in-stream —
+        # grid-capped) egress sweep for CUs. GOFR_CHANNELS defaults to 1
+        # (shared s_k); setting it to 2 makes non-flagged lanes
+        # alternate between two ingress+compute channel streams so a
+        # batch's kernels follow its own H2D in-stream. The flagged
+        # production pipeline always uses the two channel streams —
         # nothing ever waits on an SDMA-recorded event (whose signal
         # wake costs ~0.3 ms via interrupts) across streams.
         self.s_in = t.cuda.Stream(device=dev)
@@ -310,6 +312,24 @@ class BatchEngine:
         self.lanes = [_Lane(t, dev, self.max_batch, self.max_bytes,
                             self.rslot, hip=self.hip)
                       for _ in range(self.pipeline)]
+
+    def close(self) -> None:
+        """Release hipHostMalloc'd pinned memory (p_out / p_tables per
+        lane). torch-managed device/pinned tensors free with GC, but
+        host_alloc'd buffers leak without this. Safe to
+        call twice; the engine must not be used afterwards."""
+        if self.device is None or not hasattr(self, "lanes"):
+            return
+        self.torch.cuda.synchronize(self.device)
+        for ln in self.lanes:
+            for name in ("p_out", "p_tables"):
+                t = getattr(ln, name, None)
+                if t is not None:
+                    if name == "p_tables":
+                        ln.p_tables_np = None
+                    setattr(ln, name, None)
+                    self.hip.host_free(t)
+        self.lanes = []
 
     def _next_seed(self) -> int:
         self._seed = ops.splitmix64(self._seed)
@@ -626,11 +646,26 @@ class BatchEngine:
             t0 = time.perf_counter()
             while tbl[cell] != want:
                 spins += 1
-                if (spins & 0xFFFFF) == 0 and \
-                        time.perf_counter() - t0 > 30:
-                    raise RuntimeError(
-                        "flagged completion timeout (lane serial "
-                        f"{want}, cell {int(tbl[cell])})")
+                if (spins & 0xFFFF) == 0:
+                    if tbl[2 * n + 3] == want:
+                        # k_gate gave up waiting for this batch's SDMA
+                        # serial: the kernel chain ran on stale ingress
+                        # bytes — never release these responses
+                        raise RuntimeError(
+                            f"ingress gate timeout (serial {want}): "
+                            "SDMA flag never arrived; batch dropped")
+                    if (spins & 0xFFFFF) == 0 and \
+                            time.perf_counter() - t0 > 30:
+                        raise RuntimeError(
+                            "flagged completion timeout (lane serial "
+                            f"{want}, cell {int(tbl[cell])})")
+            if tbl[2 * n + 3] == want:
+                # k_done can land before the spin notices the marker:
+                # re-check after completion so a gate timeout is never
+                # masked by the (stale-input) batch finishing
+                raise RuntimeError(
+                    f"ingress gate timeout (serial {want}): "
+                    "SDMA flag never arrived; batch dropped")
             if self._use_pump and self.hip.lib.gofr_pump_err():
                 raise RuntimeError(
                     f"pump error: hipError {self.hip.lib.gofr_pump_err()}")
@@ -973,3 +1008,4 @@ class GPUServer:
         if self._core is not None and (
                 self._thread is None or not self._thread.is_alive()):
             self._core.stop()
+        self.engine.close()

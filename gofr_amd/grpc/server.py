@@ -179,6 +179,11 @@ class GRPCServer:
                     rq.put(None)  # malformed: host error path
                 else:
                     rq.put(out[i * GR:i * GR + ln].tobytes())
+        if gpu is not None:
+            # worker owns the hipHostMalloc'd egress buffer: free it on
+            # shutdown so server start/stop cycles don't leak pinned mem
+            p_out = gpu.pop("p_out")
+            gpu["hip"].host_free(p_out)
 
     def start(self) -> None:
         sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)

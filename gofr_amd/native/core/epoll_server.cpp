@@ -484,8 +484,15 @@ private:
 
 }  // namespace
 
+#ifndef GOFR_SRC_HASH
+#define GOFR_SRC_HASH "unhashed"
+#endif
+
 PYBIND11_MODULE(_core, m) {
     m.doc() = "gofr_amd native epoll ingress (multi-reactor)";
+    // source-hash stamp (tests/test_build_hash.py): the loaded binary
+    // must carry the hash of the source committed in the tree
+    m.attr("src_hash") = GOFR_SRC_HASH;
     py::class_<EpollServer>(m, "EpollServer")
         .def(py::init<int, size_t, int>(), py::arg("port"),
              py::arg("max_req") = (size_t)1 << 20,

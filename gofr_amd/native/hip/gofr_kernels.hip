@@ -1754,9 +1754,13 @@ typedef struct {
 // event, whose signal delivery costs ~0.3 ms serialized in the kernel
 // stream's FIFO; the gate releases within the SDMA's own completion.
 // Timeout guard: gives up after ~0.5 s so a lost flag can never hang
-// the stream (the serving loop surfaces the stall as a bad batch).
+// the stream — and REPORTS it: the batch serial is published to the
+// pinned timeout cell (system release), which BatchEngine.complete()
+// checks before releasing the batch's responses, so a lost/late SDMA
+// flag surfaces as an error instead of a batch of stale bytes.
 extern "C" __global__ void k_gate(const unsigned long long* flag,
-                                  unsigned long long serial) {
+                                  unsigned long long serial,
+                                  int32_t* timeout_cell) {
     if (threadIdx.x != 0) return;
     for (long i = 0; i < (1L << 19); ++i) {
         const unsigned long long v = __hip_atomic_load(
@@ -1764,6 +1768,9 @@ extern "C" __global__ void k_gate(const unsigned long long* flag,
         if (v >= serial) return;
         __builtin_amdgcn_s_sleep(32);
     }
+    if (timeout_cell)
+        __hip_atomic_store(timeout_cell, (int32_t)(serial & 0x7fffffff),
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
 }
 
 // enqueue-time breakdown (µs, cumulative): [0] big H2D, [1] rest of
@@ -1811,9 +1818,13 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     g_submit_us[1] += now_us() - t0; t0 = now_us();
     // ---- kernel stage -----------------------------------------------------
     if (a->d_flag) {
+        // timeout marker cell = p_tables[2n+3] (pinned; complete()
+        // checks it before releasing the batch)
+        int32_t* tmo = a->p_tables
+            ? (int32_t*)a->p_tables + 2 * (size_t)n + 3 : nullptr;
         hipLaunchKernelGGL(k_gate, dim3(1), dim3(64), 0, s_k,
                            (const unsigned long long*)a->d_flag,
-                           a->serial);
+                           a->serial, tmo);
         if ((rc = hipGetLastError())) return (int)rc;
     } else {
         rc = hipStreamWaitEvent(s_k, (hipEvent_t)a->ev_in, 0);
@@ -1926,7 +1937,12 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
         hipLaunchKernelGGL(k_done, dim3(1), dim3(64), 0, s_out,
                            (int32_t*)a->p_tables + 2 * n + 2,
                            (int)(a->serial & 0x7fffffff));
-        rc = hipGetLastError();
+        if ((rc = hipGetLastError())) return (int)rc;
+        // record ev_done AFTER k_done so the pump's completion thread
+        // (which hipEventSynchronize's this event) publishes a truthful
+        // done_serial even in flagged mode — the primary completion
+        // signal is still the k_done serial in pinned p_tables
+        rc = hipEventRecord((hipEvent_t)a->ev_done, s_out);
         g_submit_us[3] += now_us() - t0;
         return (int)rc;
     }
@@ -2071,6 +2087,19 @@ extern "C" void* gofr_host_alloc(long long size, unsigned flags) {
     if (hipHostMalloc(&p, (size_t)size, flags)) return nullptr;
     return p;
 }
+
+extern "C" int gofr_host_free(void* p) {
+    return (int)hipHostFree(p);
+}
+
+// source-hash stamp: build() compiles with -DGOFR_SRC_HASH="<sha256>"
+// of the committed .hip source, and tests/test_build_hash.py asserts
+// the loaded .so carries the hash of the source in the tree — a stale
+// or foreign binary fails loudly instead of silently serving old code.
+#ifndef GOFR_SRC_HASH
+#define GOFR_SRC_HASH "unhashed"
+#endif
+extern "C" const char* gofr_src_hash() { return GOFR_SRC_HASH; }
 
 extern "C" int gofr_memcpy_async(void* dst, const void* src, long long n,
                                  int kind, void* stream) {

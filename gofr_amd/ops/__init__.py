@@ -156,19 +156,38 @@ class HipOps:
         self.lib.gofr_host_alloc.restype = ctypes.c_void_p
         self.lib.gofr_host_alloc.argtypes = [ctypes.c_longlong,
                                              ctypes.c_uint]
+        self.lib.gofr_host_free.restype = ctypes.c_int
+        self.lib.gofr_host_free.argtypes = [ctypes.c_void_p]
+        try:
+            self.lib.gofr_src_hash.restype = ctypes.c_char_p
+        except AttributeError:
+            pass
+        # data_ptr -> hipHostMalloc base (host_free needs the base addr)
+        self._host_allocs: dict[int, int] = {}
 
     def host_alloc(self, nbytes: int, dtype=np.uint8):
         """hipHostMalloc'd host memory as a torch CPU tensor. Unlike
         torch's pin_memory (hipHostRegister'd), the runtime serves D2H
         SDMA into this memory — registered destinations fall back to a
         CU-hungry blit kernel (measured: benchmarks/overlap_probe.py
-        --mix: 96.6 GB/s duplex vs ~65 engine-observed with blit)."""
+        --mix: 96.6 GB/s duplex vs ~65 engine-observed with blit).
+        Pair with host_free (engine/lane close paths) — hipHostMalloc'd
+        pinned memory is NOT garbage-collected with the tensor."""
         import torch
         addr = self.lib.gofr_host_alloc(int(nbytes), 0)
         if not addr:
             raise MemoryError(f"gofr_host_alloc({nbytes}) failed")
         buf = (ctypes.c_ubyte * int(nbytes)).from_address(addr)
-        return torch.from_numpy(np.frombuffer(buf, dtype=dtype))
+        t = torch.from_numpy(np.frombuffer(buf, dtype=dtype))
+        self._host_allocs[t.data_ptr()] = addr
+        return t
+
+    def host_free(self, tensor) -> None:
+        """Free a host_alloc'd tensor's pinned backing store. The caller
+        must drop every reference to the tensor (and views) first."""
+        addr = self._host_allocs.pop(tensor.data_ptr(), None)
+        if addr is not None:
+            self.lib.gofr_host_free(ctypes.c_void_p(addr))
 
     def parse_route(self, stream, reqs_t, req_off_t, req_len_t, fields_t,
                     n, trie_t: dict, handler_tab_t, n_routes,

@@ -123,6 +123,19 @@ def dispatch(app, request: Request) -> _CapturedResponse:
         import gzip as _gz
         body = _gz.compress(body, compresslevel=1, mtime=0)
         out.headers.append(("Content-Encoding", "gzip"))
+    # ETag middleware (enable_etag) — GPU analog: MFMA body hash +
+    # If-None-Match 304 fused in k_respond. Same hash (ops.etag_u32 of
+    # the final, post-gzip body) so clients see one tag across
+    # transports; this transport renders a proper 304 (status + empty
+    # body) rather than the kernel's in-place digit rewrite.
+    if getattr(app, "etag_on", False):
+        from .ops import etag_u32
+        tag = f"{etag_u32(body):08x}"
+        out.headers.append(("ETag", f'"{tag}"'))
+        if (out.status == 200 and
+                request.header("if-none-match") == f'"{tag}"'):
+            out.status = 304
+            body = b""
     out.body = body
     span.End()
     _log_request(app, span, request, out.status, start, t0)
