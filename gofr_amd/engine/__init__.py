@@ -50,13 +50,19 @@ _CT_IDS = {"application/json": 0, "image/x-icon": 1,
 INVALID_BODY_ENV = b'{"error":{"message":"invalid JSON body"}}'
 NOTFOUND_ENV = b'{"error":{"message":"http: no such file"}}'
 UNAUTHORIZED_ENV = b'{"error":{"message":"unauthorized"}}'
+KV_MISS_ENV = b'{"error":{"message":"key not found"}}'
 
 
 class RouteProgram:
     """Compiled device image of an app's route table.
 
-    blob layout: [int32 len][invalid-body envelope][static args...]
-    handler_tab: int32 [n_routes, 4] = (kind, arg_off, arg_len, status)
+    blob layout: [int32 len][invalid-body envelope][int32 len][kv-miss
+    envelope][static args / template programs...] — the kernel finds the
+    kv-miss envelope at blob[8 + invalid_len].
+    handler_tab: int32 [n_routes, 4] = (kind, arg_off, arg_len, status);
+    HK_TEMPLATE: arg_off = 4-aligned template program offset in blob;
+    HK_KV: arg_off = slot base in kv_tab (rows), arg_len = n_slots,
+    keyed by path param 0.
     """
 
     def __init__(self, app):
@@ -66,6 +72,11 @@ class RouteProgram:
         blob = bytearray()
         blob += len(INVALID_BODY_ENV).to_bytes(4, "little")
         blob += INVALID_BODY_ENV
+        blob += len(KV_MISS_ENV).to_bytes(4, "little")
+        blob += KV_MISS_ENV
+        kv_blob = bytearray()
+        kv_rows = []
+        kv_slots = 0
         rows = []
         self.py_handlers = []
         for route in app.router.routes:
@@ -84,6 +95,16 @@ class RouteProgram:
                 status = spec[2] if len(spec) > 2 else 200
                 blob += body
                 rows.append((ops.HK_STATIC, off, len(body), status))
+            elif spec[0] == "template":
+                prog_off = self._compile_template(blob, spec[1])
+                status = spec[2] if len(spec) > 2 else 200
+                rows.append((ops.HK_TEMPLATE, prog_off, 0, status))
+            elif spec[0] == "kv":
+                tab, nslots = ops.build_kv_table(spec[1], kv_blob,
+                                                 kv_slots)
+                rows.append((ops.HK_KV, kv_slots, nslots, 200))
+                kv_rows.append(tab)
+                kv_slots += nslots
             else:
                 rows.append((ops.HK_HOST, 0, 0, 200))
         # auth middleware 401 envelope (k_respond auth_env_off/len)
@@ -92,6 +113,50 @@ class RouteProgram:
         self.handler_tab = np.asarray(rows, np.int32).reshape(-1)
         self.n_routes = len(rows)
         self.blob = bytes(blob)
+        self.kv_tab = (np.concatenate(kv_rows) if kv_rows
+                       else np.full(6, -1, np.int32))
+        self.kv_blob = bytes(kv_blob) if kv_blob else b"\0"
+
+    @staticmethod
+    def _compile_template(blob: bytearray, pieces) -> int:
+        """Encode a handlers.template_json piece list into the blob
+        (int32 program, 4-aligned): [n][op, a, b, mode] x n."""
+        words = []
+        for piece in pieces:
+            if isinstance(piece, (bytes, str)):
+                lit = piece.encode("utf-8") if isinstance(piece, str) \
+                    else piece
+                off = len(blob)
+                blob += lit
+                words.append((ops.TP_LIT, off, len(lit), 0))
+                continue
+            op, arg = piece[0], piece[1]
+            if op in ("path", "path_raw"):
+                mode = ops.TM_JESC if op == "path" else 0
+                words.append((ops.TP_PATH, int(arg), 0, mode))
+            elif op in ("query", "query_raw"):
+                key = arg.encode("utf-8")
+                koff = len(blob)
+                blob += key
+                mode = (ops.TM_PCT | ops.TM_JESC) if op == "query" \
+                    else ops.TM_PCT
+                words.append((ops.TP_QUERY, koff, len(key), mode))
+            elif op in ("jfield", "jfield_str"):
+                key = arg.encode("utf-8")
+                koff = len(blob)
+                blob += key
+                mode = ops.TM_JSTR if op == "jfield_str" else 0
+                words.append((ops.TP_JFIELD, koff, len(key), mode))
+            else:
+                raise ValueError(f"unknown template piece {piece!r}")
+        while len(blob) % 4:
+            blob += b"\0"
+        prog_off = len(blob)
+        blob += len(words).to_bytes(4, "little")
+        for w in words:
+            for v in w:
+                blob += int(v).to_bytes(4, "little", signed=True)
+        return prog_off
 
 
 def pack_batch(payloads: list[bytes]):
@@ -284,6 +349,9 @@ class BatchEngine:
         self.d_handler_tab = t.as_tensor(self.program.handler_tab).to(dev)
         self.d_blob = t.as_tensor(
             np.frombuffer(self.program.blob, np.uint8).copy()).to(dev)
+        self.d_kv_tab = t.as_tensor(self.program.kv_tab.copy()).to(dev)
+        self.d_kv_blob = t.as_tensor(
+            np.frombuffer(self.program.kv_blob, np.uint8).copy()).to(dev)
         secret = getattr(self.app, "auth_secret", None)
         if secret:
             self.d_secret = t.as_tensor(
@@ -390,7 +458,8 @@ class BatchEngine:
             auth_env=self.program.auth_env,
             gzip_min=self.app.gzip_min_size or 0,
             etag_on=getattr(self.app, "etag_on", False),
-            date29=self._date29())
+            date29=self._date29(),
+            kv_tab=self.program.kv_tab, kv_blob=self.program.kv_blob)
         # compaction mirror (same round16 layout as k_compact)
         n = len(lens)
         pads = (resp_len + 15) & ~15
@@ -568,6 +637,8 @@ class BatchEngine:
         a.serial = 0
         a.flagged = 1 if self._flagged else 0
         a.d_date = ln.d_ingress.data_ptr() + ln.date_off
+        a.d_kv_tab = self.d_kv_tab.data_ptr()
+        a.d_kv_blob = self.d_kv_blob.data_ptr()
         if self._flagged:
             # event-free channel pipeline: the whole chain (gate ->
             # kernels -> egress -> k_done) rides one of two channel
@@ -616,7 +687,8 @@ class BatchEngine:
                          gzip_min=self.app.gzip_min_size or 0,
                          etag_on=1 if getattr(self.app, "etag_on", False)
                          else 0,
-                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off)
+                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off,
+                         kv_tab_t=self.d_kv_tab, kv_blob_t=self.d_kv_blob)
         pads = (ln.d_resp_len[:n] + 15).bitwise_and_(-16)
         csum = t.cumsum(pads, 0, dtype=t.int32)
         ln.d_resp_off[:n].copy_(csum - pads)
@@ -844,7 +916,8 @@ class BatchEngine:
                          gzip_min=self.app.gzip_min_size or 0,
                          etag_on=1 if getattr(self.app, "etag_on", False)
                          else 0,
-                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off)
+                         date_ptr=ln.d_ingress.data_ptr() + ln.date_off,
+                         kv_tab_t=self.d_kv_tab, kv_blob_t=self.d_kv_blob)
         return ln.d_resp, ln.d_resp_len
 
     # -- host trampoline ------------------------------------------------------

@@ -89,7 +89,9 @@ class AllToAllSharder:
                 auth_env=self.engine.program.auth_env,
                 gzip_min=self.engine.app.gzip_min_size or 0,
                 etag_on=getattr(self.engine.app, "etag_on", False),
-                date29=self.engine._date29())
+                date29=self.engine._date29(),
+                kv_tab=self.engine.program.kv_tab,
+                kv_blob=self.engine.program.kv_blob)
             t = self.t
             dist.all_to_all_single(self.d_resp_sh,
                                    t.from_numpy(resp_slots))

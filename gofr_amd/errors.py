@@ -28,6 +28,16 @@ class MissingFileError(GofrError):
         super().__init__(msg)
 
 
+class KeyNotFoundError(GofrError):
+    """KV-store miss (handlers.kv_json) — maps to 404. The GPU engine
+    renders the same envelope from the kv-miss blob slot."""
+
+    status_code = 404
+
+    def __init__(self, msg: str = "key not found"):
+        super().__init__(msg)
+
+
 class CommandNotFoundError(GofrError):
     """CLI mode: no registered command matched
     (reference: pkg/gofr/cmd.go:21-25)."""

@@ -41,3 +41,87 @@ def static_raw(body: bytes, status: int = 200):
         return Raw(json.loads(body))
     handler.__gofr_gpu__ = ("static", body, status)
     return handler
+
+
+def template_json(*pieces, status: int = 200):
+    """Template-substitution handler: the response body is a compiled
+    piece sequence rendered fully on-device (HK_TEMPLATE in k_respond —
+    no host trampoline for /user/{id}-class routes). Pieces:
+
+      "literal text"          verbatim bytes (carry your own envelope)
+      ("path", i)             i-th path param, JSON-string-escaped
+      ("path_raw", i)         i-th path param verbatim
+      ("query", "key")        query param: %XX/'+'-decoded + escaped
+      ("query_raw", "key")    query param: decoded, not escaped
+      ("jfield", "key")       top-level JSON body field, raw value span
+      ("jfield_str", "key")   ... string content (quotes stripped)
+
+    Missing query/body fields splice as empty. The Python body renders
+    the identical bytes for the CPU transport (shared ops mirrors).
+    Reference handler shape: /root/reference/examples/http-server/
+    main.go:14-29 (param routes the r1 engine sent to the trampoline).
+    """
+    from . import ops
+    from .http.response import File
+
+    def handler(ctx):
+        out = bytearray()
+        jfields = None
+        body = ctx.request.body
+        for piece in pieces:
+            if isinstance(piece, (str, bytes)):
+                out += piece.encode("utf-8") if isinstance(piece, str) \
+                    else piece
+                continue
+            op, arg = piece[0], piece[1]
+            if op in ("path", "path_raw"):
+                vals = list(ctx.request.path_params.values())
+                v = vals[arg].encode("utf-8") if arg < len(vals) else b""
+                out += ops.splice_py(v, ops.TM_JESC if op == "path"
+                                     else 0)
+            elif op in ("query", "query_raw"):
+                q = ctx.request.query_string.encode("latin-1")
+                val = ops.q_find_py(q, arg.encode("utf-8"))
+                if val is not None:
+                    mode = (ops.TM_PCT | ops.TM_JESC) if op == "query" \
+                        else ops.TM_PCT
+                    out += ops.splice_py(val, mode)
+            elif op in ("jfield", "jfield_str"):
+                if jfields is None:
+                    jfields = ops.json_top_fields_py(body)
+                for key, vs, vl in jfields:
+                    if key == arg.encode("utf-8"):
+                        if op == "jfield_str" and vl >= 2 and \
+                                body[vs] == 0x22:
+                            vs, vl = vs + 1, vl - 2
+                        out += ops.splice_py(body[vs:vs + vl], 0)
+                        break
+        return File(bytes(out), "application/json")
+
+    handler.__gofr_gpu__ = ("template", list(pieces), status)
+    return handler
+
+
+def kv_json(store: dict):
+    """Device KV-store read handler keyed by the route's first path
+    param (HK_KV): the store compiles into an open-addressing table +
+    value blob resident in HBM; k_respond probes it and splices the
+    pre-wrapped {"data":...} envelope — the /user/{name} redis-get
+    analog (reference examples/http-server/main.go:31-38) with the
+    lookup on-device. Miss -> 404 {"error":{"message":"key not found"}}.
+    """
+    from .errors import KeyNotFoundError
+    from .http.response import File
+
+    def handler(ctx):
+        vals = list(ctx.request.path_params.values())
+        key = vals[0] if vals else ""
+        if key in store:
+            body = (b'{"data":' +
+                    json.dumps(store[key], separators=(",", ":"),
+                               ensure_ascii=False).encode("utf-8") + b"}")
+            return File(body, "application/json")
+        raise KeyNotFoundError()
+
+    handler.__gofr_gpu__ = ("kv", store)
+    return handler

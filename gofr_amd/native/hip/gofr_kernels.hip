@@ -68,6 +68,21 @@
 #define HK_HOST 0
 #define HK_ECHO_JSON 1
 #define HK_STATIC 2
+#define HK_TEMPLATE 3  /* precompiled body with spliced params/fields */
+#define HK_KV 4        /* device KV-store lookup keyed by path param 0 */
+
+// template piece opcodes (blob-encoded programs; 4 int32 words per
+// piece: [op, a, b, mode] — compiled by engine.RouteProgram from
+// handlers.template_json specs; mirror: ops.render_template)
+#define TP_LIT 0     /* a=abs blob off, b=len */
+#define TP_PATH 1    /* a=param idx (spans are pct-decoded by parse) */
+#define TP_QUERY 2   /* a=key blob off, b=key len */
+#define TP_JFIELD 3  /* a=key blob off, b=key len (top-level JSON field) */
+// splice modes (bitmask)
+#define TM_PCT 1     /* pct-decode + '+'->' ' (query values) */
+#define TM_JESC 2    /* JSON-string-escape the spliced bytes */
+#define TM_JSTR 4    /* TP_JFIELD: splice string CONTENT (strip quotes) */
+#define MAX_JSON_FIELDS 8
 
 // methods — must match gofr_amd/http/request.py METHOD_IDS
 // GET POST PUT DELETE PATCH OPTIONS HEAD
@@ -129,8 +144,31 @@ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
 #define CLS_QM 3
 #define N_CLS 4
 
+__device__ __forceinline__ int hexval(uint8_t c);
+
+// in-place %XX decode of s[0..len) (writes <= len bytes at s).
+// Returns decoded length, or -1 on an invalid escape (host path then
+// serves Python-unquote leniency). '+' is NOT decoded here (only
+// query values use '+'-as-space; paths keep it literal, like Go).
+__device__ __forceinline__ int pct_decode_inplace(uint8_t* s, int len) {
+    int w = 0;
+    for (int i = 0; i < len; ++i) {
+        uint8_t c = s[i];
+        if (c == '%') {
+            if (i + 2 >= len) return -1;
+            const int hi = hexval(s[i + 1]);
+            const int lo = hexval(s[i + 2]);
+            if (hi < 0 || lo < 0) return -1;
+            c = (uint8_t)((hi << 4) | lo);
+            i += 2;
+        }
+        s[w++] = c;
+    }
+    return w;
+}
+
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
-k_parse_route(const uint8_t* __restrict__ reqs,
+k_parse_route(uint8_t* __restrict__ reqs,
               const int64_t* __restrict__ req_off,
               const int32_t* __restrict__ req_len,
               int32_t* __restrict__ fields,
@@ -144,7 +182,7 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
     if (req >= n) return;
     const int lane = lane_id();
-    const uint8_t* base = reqs + req_off[req];
+    uint8_t* base = reqs + req_off[req];
     int len = req_len[req];
     if (len == 0) {
         // padding slot (fixed-shape armed batches): no response
@@ -204,7 +242,7 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     };
 
     int32_t flags = 0;
-    if (any_percent || oversized) flags |= FL_NEEDS_HOST;
+    if (oversized) flags |= FL_NEEDS_HOST;
 
     // request line: METHOD SP target SP version CRLF
     const int sp1 = next_bit(CLS_SP, 0);
@@ -240,9 +278,25 @@ k_parse_route(const uint8_t* __restrict__ reqs,
     int query_off = (qm >= 0 && qm < target_end) ? qm + 1 : target_end;
     int query_len = (qm >= 0 && qm < target_end) ? target_end - query_off : 0;
 
+    int path_len = path_end - path_off;
+    // %XX in the path: decode IN PLACE before the trie walk (Go parity:
+    // net/http routes on the decoded URL.Path, incl. %2F -> '/'), so
+    // param spans hand handlers decoded bytes and encoded URLs stay on
+    // the GPU (r1 sent every '%' to the host). Invalid escapes only
+    // fall back (the host parser's unquote leniency applies there).
+    if (any_percent) {
+        bool in_path = false;
+        for (int i = path_off; i < path_end; ++i)
+            if (base[i] == '%') { in_path = true; break; }
+        if (in_path) {
+            const int dl = pct_decode_inplace(base + path_off, path_len);
+            if (dl < 0) flags |= FL_NEEDS_HOST;
+            else path_len = dl;
+        }
+    }
     F[FI_METHOD] = method;
     F[FI_PATH_OFF] = path_off;
-    F[FI_PATH_LEN] = path_end - path_off;
+    F[FI_PATH_LEN] = path_len;
     F[FI_QUERY_OFF] = query_off;
     F[FI_QUERY_LEN] = query_len;
     if (method == M_OPTIONS) flags |= FL_IS_OPTIONS;
@@ -769,6 +823,222 @@ __device__ uint32_t mfma_etag_wave(const uint8_t* body, int len, int lane) {
     return h ^ (uint32_t)len;
 }
 
+// ---------------------------------------------------------------------------
+// Template / KV / JSON-bind helpers (HK_TEMPLATE, HK_KV — VERDICT r1
+// items 3-5: GPU handler kinds beyond echo/static, query-param
+// extraction, batched JSON field binding). CPU golden mirrors:
+// gofr_amd/ops render_template / kv_lookup_mirror / json_top_fields.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint64_t fnv1a64(const uint8_t* p, int n) {
+    uint64_t h = 0xcbf29ce484222325ull;
+    for (int i = 0; i < n; ++i) {
+        h ^= p[i];
+        h *= 0x100000001b3ull;
+    }
+    return h;
+}
+
+// locate `key=` in the raw query string; value span (raw, undecoded)
+__device__ __forceinline__ bool q_find(const uint8_t* q, int qlen,
+                                       const uint8_t* key, int klen,
+                                       int* voff, int* vlen) {
+    int i = 0;
+    while (i < qlen) {
+        // key span of this pair
+        int ke = i;
+        while (ke < qlen && q[ke] != '=' && q[ke] != '&') ++ke;
+        if (ke - i == klen) {
+            bool eq = true;
+            for (int k = 0; k < klen; ++k)
+                if (q[i + k] != key[k]) { eq = false; break; }
+            if (eq) {
+                if (ke < qlen && q[ke] == '=') {
+                    int ve = ke + 1;
+                    while (ve < qlen && q[ve] != '&') ++ve;
+                    *voff = ke + 1;
+                    *vlen = ve - (ke + 1);
+                } else {
+                    *voff = ke;
+                    *vlen = 0;  // bare key: empty value
+                }
+                return true;
+            }
+        }
+        while (ke < qlen && q[ke] != '&') ++ke;
+        i = ke + 1;
+    }
+    return false;
+}
+
+// splice src[0..len) through the mode transforms into dst (or just
+// measure when dst == nullptr). TM_PCT: %XX + '+'->' ' decode (invalid
+// escapes copied verbatim — Python-unquote leniency, host parity).
+// TM_JESC: JSON string escaping of the (decoded) bytes.
+__device__ int splice_bytes(uint8_t* dst, const uint8_t* src, int len,
+                            int mode) {
+    int w = 0;
+    for (int i = 0; i < len; ++i) {
+        uint8_t c = src[i];
+        if (mode & TM_PCT) {
+            if (c == '+') c = ' ';
+            else if (c == '%' && i + 2 < len) {
+                const int hi = hexval(src[i + 1]);
+                const int lo = hexval(src[i + 2]);
+                if (hi >= 0 && lo >= 0) {
+                    c = (uint8_t)((hi << 4) | lo);
+                    i += 2;
+                }
+            }
+        }
+        if (mode & TM_JESC) {
+            if (c == '"' || c == '\\') {
+                if (dst) { dst[w] = '\\'; dst[w + 1] = c; }
+                w += 2;
+                continue;
+            }
+            if (c < 0x20) {
+                if (dst) {
+                    dst[w] = '\\'; dst[w+1] = 'u'; dst[w+2] = '0';
+                    dst[w+3] = '0';
+                    dst[w+4] = HEXD[(c >> 4) & 0xF];
+                    dst[w+5] = HEXD[c & 0xF];
+                }
+                w += 6;
+                continue;
+            }
+        }
+        if (dst) dst[w] = c;
+        ++w;
+    }
+    return w;
+}
+
+// top-level field scan of a JSON object body: writes up to
+// MAX_JSON_FIELDS rows of [key_off, key_len, val_off, val_len] (offsets
+// relative to `body`; key span excludes quotes; value span is the raw
+// JSON value with surrounding whitespace trimmed). Returns the field
+// count, or 0 when the body is not an object / malformed (callers
+// splice empty). Serial (one lane); bodies are request-sized.
+__device__ int json_top_fields(const uint8_t* body, int blen,
+                               int32_t* tab) {
+    int i = 0;
+    while (i < blen && body[i] <= ' ') ++i;
+    if (i >= blen || body[i] != '{') return 0;
+    ++i;
+    int nf = 0;
+    while (nf < MAX_JSON_FIELDS) {
+        while (i < blen && (body[i] <= ' ' || body[i] == ',')) ++i;
+        if (i >= blen) return 0;
+        if (body[i] == '}') return nf;
+        if (body[i] != '"') return 0;
+        const int ks = ++i;
+        while (i < blen && body[i] != '"') {
+            if (body[i] == '\\') ++i;
+            ++i;
+        }
+        if (i >= blen) return 0;
+        const int ke = i++;
+        while (i < blen && body[i] <= ' ') ++i;
+        if (i >= blen || body[i] != ':') return 0;
+        ++i;
+        while (i < blen && body[i] <= ' ') ++i;
+        if (i >= blen) return 0;
+        const int vs = i;
+        // value: string / container / scalar
+        if (body[i] == '"') {
+            ++i;
+            while (i < blen && body[i] != '"') {
+                if (body[i] == '\\') ++i;
+                ++i;
+            }
+            if (i >= blen) return 0;
+            ++i;
+        } else if (body[i] == '{' || body[i] == '[') {
+            int depth = 0;
+            bool in_str = false;
+            while (i < blen) {
+                const uint8_t c = body[i];
+                if (in_str) {
+                    if (c == '\\') ++i;
+                    else if (c == '"') in_str = false;
+                } else if (c == '"') in_str = true;
+                else if (c == '{' || c == '[') ++depth;
+                else if (c == '}' || c == ']') {
+                    if (--depth == 0) { ++i; break; }
+                }
+                ++i;
+            }
+            if (depth != 0) return 0;
+        } else {
+            while (i < blen && body[i] != ',' && body[i] != '}' &&
+                   body[i] > ' ')
+                ++i;
+        }
+        tab[nf * 4 + 0] = ks;
+        tab[nf * 4 + 1] = ke - ks;
+        tab[nf * 4 + 2] = vs;
+        tab[nf * 4 + 3] = i - vs;
+        ++nf;
+        while (i < blen && body[i] <= ' ') ++i;
+        if (i < blen && body[i] == ',') continue;
+        if (i < blen && body[i] == '}') return nf;
+        return 0;
+    }
+    return nf;
+}
+
+// render one template piece's source span + effective mode. Returns
+// false when the piece references a missing query/JSON field (splice
+// empty). jtab/jn: the request's json_top_fields table.
+__device__ __forceinline__ bool piece_src(
+        const int32_t* pc, const uint8_t* blob, const uint8_t* rbase,
+        const int32_t* F, const int32_t* jtab, int jn,
+        const uint8_t** src, int* slen, int* mode) {
+    const int op = pc[0], a = pc[1], b = pc[2];
+    *mode = pc[3];
+    if (op == TP_LIT) {
+        *src = blob + a;
+        *slen = b;
+        return true;
+    }
+    if (op == TP_PATH) {
+        if (a >= MAX_PARAMS) return false;
+        *src = rbase + F[FI_PARAM0 + 2 * a];
+        *slen = F[FI_PARAM0 + 2 * a + 1];
+        *mode &= ~TM_PCT;  // parse already decoded the path in place
+        return true;
+    }
+    if (op == TP_QUERY) {
+        int voff, vlen;
+        if (!q_find(rbase + F[FI_QUERY_OFF], F[FI_QUERY_LEN],
+                    blob + a, b, &voff, &vlen))
+            return false;
+        *src = rbase + F[FI_QUERY_OFF] + voff;
+        *slen = vlen;
+        return true;
+    }
+    // TP_JFIELD
+    const uint8_t* body = rbase + F[FI_BODY_OFF];
+    for (int f = 0; f < jn; ++f) {
+        if (jtab[f * 4 + 1] != b) continue;
+        bool eq = true;
+        for (int k = 0; k < b; ++k)
+            if (body[jtab[f * 4] + k] != blob[a + k]) { eq = false; break; }
+        if (!eq) continue;
+        int vs = jtab[f * 4 + 2], vl = jtab[f * 4 + 3];
+        if ((*mode & TM_JSTR) && vl >= 2 && body[vs] == '"') {
+            vs += 1;
+            vl -= 2;  // string content without quotes (escapes kept)
+        }
+        *src = body + vs;
+        *slen = vl;
+        *mode &= ~TM_PCT;  // body bytes are not URL-encoded
+        return true;
+    }
+    return false;
+}
+
 // host result table row: [off, len, status, ct_id]
 //
 // The whole response (headers + body) is assembled in LDS per wave, then
@@ -791,6 +1061,8 @@ __device__ __forceinline__ void respond_impl(
         const int32_t* __restrict__ host_tab,
         uint64_t seed, int auth_env_off, int auth_env_len,
         int gzip_min, int etag_on, const uint8_t* date29,
+        const int32_t* __restrict__ kv_tab,
+        const uint8_t* __restrict__ kv_blob,
         uint8_t* obuf, uint8_t* plainbuf,
         uint32_t* hash, const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
@@ -883,6 +1155,103 @@ __device__ __forceinline__ void respond_impl(
         if (route >= 0 && route < n_routes) {
             body_src = blob + handler_tab[route * 4 + 1];
             body_src_len = handler_tab[route * 4 + 2];
+        }
+    } else if (kind == HK_TEMPLATE && !(flags & FL_AUTH_FAIL)) {
+        // template program: blob-encoded pieces spliced with decoded
+        // path params, query params and top-level JSON body fields —
+        // the /user/{id}-class routes the r1 engine trampolined.
+        // Lane 0 scans + measures + emits; result is the raw body
+        // (templates carry their own envelope text), env stays 0.
+        const int route = F[FI_ROUTE];
+        const int32_t* prog = (const int32_t*)(blob +
+                                               handler_tab[route * 4 + 1]);
+        uint8_t* tbuf = plainbuf ? plainbuf : (obuf + 1024);
+        const int tcap = MAX_SLOT - 1024;
+        int total = -1;
+        if (lane == 0) {
+            int32_t jtab[MAX_JSON_FIELDS * 4];
+            int jn = -1;  // lazy: scan only when a piece needs it
+            const int np = prog[0];
+            total = 0;
+            for (int p = 0; p < np; ++p) {
+                const int32_t* pc = prog + 1 + p * 4;
+                if (pc[0] == TP_JFIELD && jn < 0)
+                    jn = json_top_fields(rbase + F[FI_BODY_OFF],
+                                         F[FI_BODY_LEN], jtab);
+                const uint8_t* src;
+                int slen, mode;
+                if (piece_src(pc, blob, rbase, F, jtab, jn < 0 ? 0 : jn,
+                              &src, &slen, &mode))
+                    total += splice_bytes(nullptr, src, slen, mode);
+            }
+            if (total > tcap) {
+                total = -1;  // overflow guard: render the 500 envelope
+            } else {
+                int w = 0;
+                for (int p = 0; p < np; ++p) {
+                    const int32_t* pc = prog + 1 + p * 4;
+                    if (pc[0] == TP_JFIELD && jn < 0)
+                        jn = 0;  // unreachable (measure pass scanned)
+                    const uint8_t* src;
+                    int slen, mode;
+                    if (piece_src(pc, blob, rbase, F, jtab,
+                                  jn < 0 ? 0 : jn, &src, &slen, &mode))
+                        w += splice_bytes(tbuf + w, src, slen, mode);
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        total = __shfl(total, 0);
+        if (total < 0) {
+            status = 500;
+            const int elen = *(const int32_t*)blob;
+            body_src = blob + 4;
+            body_src_len = elen;
+        } else {
+            body_src = tbuf;
+            body_src_len = total;
+        }
+    } else if (kind == HK_KV && !(flags & FL_AUTH_FAIL)) {
+        // device KV-store read keyed by path param 0 (decoded): the
+        // /user/{id} redis-get analog served from HBM. Values are
+        // pre-wrapped {"data":...} envelopes; a miss renders the 404
+        // envelope stored at blob[8+invalid_len].
+        const int route = F[FI_ROUTE];
+        const int slot0 = handler_tab[route * 4 + 1];
+        const int nslots = handler_tab[route * 4 + 2];
+        const uint8_t* key = rbase + F[FI_PARAM0];
+        const int klen = F[FI_PARAM0 + 1];
+        int voff = -1, vlen = 0;
+        if (nslots > 0 && klen > 0) {
+            const uint64_t h = fnv1a64(key, klen);
+            const int32_t h_lo = (int32_t)(h & 0xFFFFFFFF);
+            const int32_t h_hi = (int32_t)(h >> 32);
+            int idx = (int)(h % (uint64_t)nslots);
+            for (int probe = 0; probe < nslots; ++probe) {
+                const int32_t* row = kv_tab + (size_t)(slot0 + idx) * 6;
+                if (row[5] < 0) break;  // empty slot: miss
+                if (row[0] == h_lo && row[1] == h_hi &&
+                    row[3] == klen) {
+                    bool eq = true;
+                    for (int k = 0; k < klen; ++k)
+                        if (kv_blob[row[2] + k] != key[k]) {
+                            eq = false;
+                            break;
+                        }
+                    if (eq) { voff = row[4]; vlen = row[5]; break; }
+                }
+                idx = (idx + 1 == nslots) ? 0 : idx + 1;
+            }
+        }
+        if (voff >= 0) {
+            body_src = kv_blob + voff;
+            body_src_len = vlen;
+        } else {
+            status = 404;
+            const int elen = *(const int32_t*)blob;
+            const int miss_len = *(const int32_t*)(blob + 4 + elen);
+            body_src = blob + 8 + elen;
+            body_src_len = miss_len;
         }
     }
     int ct_id = 0;  // 0 json, 1 icon, 2 octet-stream, 3 text/plain
@@ -1125,7 +1494,9 @@ k_respond(const uint8_t* __restrict__ reqs,
           const int32_t* __restrict__ host_tab,
           const uint64_t* __restrict__ seed_ptr,
           int auth_env_off, int auth_env_len, int etag_on,
-          const uint8_t* __restrict__ date29) {
+          const uint8_t* __restrict__ date29,
+          const int32_t* __restrict__ kv_tab,
+          const uint8_t* __restrict__ kv_blob) {
     __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
     const int wv = threadIdx.x / WAVE;
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
@@ -1133,7 +1504,7 @@ k_respond(const uint8_t* __restrict__ reqs,
     respond_impl<false>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                         handler_tab, n_routes, blob, host_blob, host_tab,
                         *seed_ptr, auth_env_off, auth_env_len, 0, etag_on,
-                        date29,
+                        date29, kv_tab, kv_blob,
                         obuf_all + wv * MAX_SLOT, nullptr, nullptr, nullptr,
                         req, lane_id());
 }
@@ -1151,7 +1522,9 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
              const int32_t* __restrict__ host_tab,
              const uint64_t* __restrict__ seed_ptr,
              int auth_env_off, int auth_env_len, int gzip_min,
-             int etag_on, const uint8_t* __restrict__ date29) {
+             int etag_on, const uint8_t* __restrict__ date29,
+             const int32_t* __restrict__ kv_tab,
+             const uint8_t* __restrict__ kv_blob) {
     // single __shared__ block (cdna guide §5 trap 4a)
     __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
                            WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4];
@@ -1174,7 +1547,7 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
     respond_impl<true>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                        handler_tab, n_routes, blob, host_blob, host_tab,
                        *seed_ptr, auth_env_off, auth_env_len, gzip_min,
-                       etag_on, date29,
+                       etag_on, date29, kv_tab, kv_blob,
                        obuf_all + wv * MAX_SLOT, plain_all + wv * MAX_SLOT,
                        hash_all + wv * GZ_HASH_SIZE, crc_tab,
                        req, lane_id());
@@ -1745,6 +2118,9 @@ typedef struct {
     // header slot; all responses of a batch share it). 0 -> no Date
     // header (Go's net/http attaches Date to every response; parity).
     const void* d_date;
+    // device KV stores (HK_KV routes): hash table rows + value blob
+    const void* d_kv_tab;
+    const void* d_kv_blob;
 } GofrSubmitArgs;
 
 // k_gate — single-wave stream gate: spins (system-scope acquire
@@ -1840,7 +2216,7 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_parse_route, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        s_k,
-                       (const uint8_t*)a->d_reqs, (const int64_t*)a->d_off,
+                       (uint8_t*)a->d_reqs, (const int64_t*)a->d_off,
                        (const int32_t*)a->d_len, (int32_t*)a->d_fields, n,
                        trie, (const int32_t*)a->handler_tab, a->n_routes,
                        (int32_t*)a->d_host_needed);
@@ -1869,7 +2245,9 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
                            a->auth_env_off, a->auth_env_len, a->gzip_min,
-                           a->etag_on, (const uint8_t*)a->d_date);
+                           a->etag_on, (const uint8_t*)a->d_date,
+                           (const int32_t*)a->d_kv_tab,
+                           (const uint8_t*)a->d_kv_blob);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS),
                            0, s_k,
@@ -1882,7 +2260,9 @@ static int gofr_submit_impl(const GofrSubmitArgs* a) {
                            (const uint8_t*)a->host_blob,
                            (const int32_t*)a->host_tab, seed_ptr,
                            a->auth_env_off, a->auth_env_len, a->etag_on,
-                           (const uint8_t*)a->d_date);
+                           (const uint8_t*)a->d_date,
+                           (const int32_t*)a->d_kv_tab,
+                           (const uint8_t*)a->d_kv_blob);
     }
     if ((rc = hipGetLastError())) return (int)rc;
     hipLaunchKernelGGL(k_padscan, dim3(1), dim3(SCAN_THREADS), 0, s_k,
@@ -2128,7 +2508,7 @@ int gofr_launch_parse_route(
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     hipLaunchKernelGGL(k_parse_route, dim3(blocks), dim3(BLOCK_THREADS), 0,
                        (hipStream_t)stream,
-                       (const uint8_t*)reqs, (const int64_t*)req_off,
+                       (uint8_t*)reqs, (const int64_t*)req_off,
                        (const int32_t*)req_len,
                        (int32_t*)fields, n, trie,
                        (const int32_t*)handler_tab, n_routes,
@@ -2143,7 +2523,8 @@ int gofr_launch_respond(
         const void* handler_tab, int n_routes,
         const void* blob, const void* host_blob, const void* host_tab,
         const void* seed_ptr, int auth_env_off, int auth_env_len,
-        int gzip_min, int etag_on, const void* date29) {
+        int gzip_min, int etag_on, const void* date29,
+        const void* kv_tab, const void* kv_blob) {
     const int blocks = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
     if (gzip_min > 0) {
         hipLaunchKernelGGL(k_respond_gz, dim3(blocks), dim3(BLOCK_THREADS),
@@ -2157,7 +2538,9 @@ int gofr_launch_respond(
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
                            auth_env_off, auth_env_len, gzip_min, etag_on,
-                           (const uint8_t*)date29);
+                           (const uint8_t*)date29,
+                           (const int32_t*)kv_tab,
+                           (const uint8_t*)kv_blob);
     } else {
         hipLaunchKernelGGL(k_respond, dim3(blocks), dim3(BLOCK_THREADS), 0,
                            (hipStream_t)stream,
@@ -2170,7 +2553,9 @@ int gofr_launch_respond(
                            (const int32_t*)host_tab,
                            (const uint64_t*)seed_ptr,
                            auth_env_off, auth_env_len, etag_on,
-                           (const uint8_t*)date29);
+                           (const uint8_t*)date29,
+                           (const int32_t*)kv_tab,
+                           (const uint8_t*)kv_blob);
     }
     return (int)hipGetLastError();
 }

@@ -54,9 +54,22 @@ FL_EMPTY = 256  # padding slot (len 0): emit no response bytes
 HK_HOST = 0
 HK_ECHO_JSON = 1
 HK_STATIC = 2
+HK_TEMPLATE = 3
+HK_KV = 4
+
+# template piece opcodes / splice modes (mirror of gofr_kernels.hip)
+TP_LIT = 0
+TP_PATH = 1
+TP_QUERY = 2
+TP_JFIELD = 3
+TM_PCT = 1
+TM_JESC = 2
+TM_JSTR = 4
+MAX_JSON_FIELDS = 8
 
 MAX_PARAMS = 4
 N_METHODS_PAD = 8
+MAX_SLOT = 4096
 
 _METHOD_NAMES = ["GET", "POST", "PUT", "DELETE", "PATCH", "OPTIONS", "HEAD"]
 
@@ -97,6 +110,8 @@ class GofrSubmitArgs(ctypes.Structure):
         ("serial", ctypes.c_uint64),
         ("flagged", ctypes.c_int),
         ("d_date", ctypes.c_void_p),  # 29-byte IMF-fixdate in device mem
+        ("d_kv_tab", ctypes.c_void_p),   # HK_KV store rows (int32 x6)
+        ("d_kv_blob", ctypes.c_void_p),  # HK_KV key/value bytes
     ]
 
 
@@ -130,7 +145,8 @@ class HipOps:
             ctypes.c_void_p, ctypes.c_int,
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
             ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
-            ctypes.c_int, ctypes.c_void_p]
+            ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p]
         self.lib.gofr_launch_compact.restype = ctypes.c_int
         self.lib.gofr_launch_compact.argtypes = \
             [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_int]
@@ -216,7 +232,8 @@ class HipOps:
     def respond(self, stream, reqs_t, req_off_t, fields_t, resp_t,
                 resp_len_t, n, rslot, handler_tab_t, n_routes, blob_t,
                 host_blob_t, host_tab_t, seed_t, auth_env=(0, 0),
-                gzip_min=0, etag_on=0, date_ptr=0):
+                gzip_min=0, etag_on=0, date_ptr=0, kv_tab_t=None,
+                kv_blob_t=None):
         rc = self.lib.gofr_launch_respond(
             ctypes.c_void_p(stream),
             ctypes.c_void_p(reqs_t.data_ptr()),
@@ -231,7 +248,11 @@ class HipOps:
             ctypes.c_void_p(host_tab_t.data_ptr()),
             ctypes.c_void_p(seed_t.data_ptr()),
             auth_env[0], auth_env[1], gzip_min, etag_on,
-            ctypes.c_void_p(date_ptr))
+            ctypes.c_void_p(date_ptr),
+            ctypes.c_void_p(kv_tab_t.data_ptr() if kv_tab_t is not None
+                            else 0),
+            ctypes.c_void_p(kv_blob_t.data_ptr() if kv_blob_t is not None
+                            else 0))
         if rc != 0:
             raise RuntimeError(f"k_respond launch failed: hipError {rc}")
 
@@ -345,7 +366,7 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         buf = reqs[base:base + ln].tobytes()
         F = fields[r]
         flags = 0
-        if b"%" in buf or oversized:
+        if oversized:
             flags |= FL_NEEDS_HOST
         sp1 = buf.find(b" ")
         lf1 = buf.find(b"\n")
@@ -369,9 +390,22 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         path_end = qm if (0 <= qm < target_end) else target_end
         query_off = qm + 1 if (0 <= qm < target_end) else target_end
         query_len = target_end - query_off if (0 <= qm < target_end) else 0
+        path_len = path_end - path_off
+        # %XX in the path: decode IN PLACE before the trie walk (mirror
+        # of the kernel; Go routes on the decoded URL.Path). Invalid
+        # escapes fall to the host parser's unquote leniency.
+        if b"%" in buf[path_off:path_end]:
+            dec = pct_decode(buf[path_off:path_end])
+            if dec is None:
+                flags |= FL_NEEDS_HOST
+            else:
+                reqs[base + path_off:base + path_off + len(dec)] = \
+                    np.frombuffer(dec, np.uint8)
+                path_len = len(dec)
+                buf = reqs[base:base + ln].tobytes()
         F[FI_METHOD] = method
         F[FI_PATH_OFF] = path_off
-        F[FI_PATH_LEN] = path_end - path_off
+        F[FI_PATH_LEN] = path_len
         F[FI_QUERY_OFF] = query_off
         F[FI_QUERY_LEN] = query_len
         if method == 5:  # OPTIONS
@@ -561,6 +595,274 @@ def _json_body_valid(body: bytes) -> bool:
     return depth == 0 and not in_str and len(body) > 0
 
 
+# ---- template / KV / JSON-bind mirrors (gofr_kernels.hip helpers) ----------
+
+def pct_decode(s: bytes):
+    """Mirror of pct_decode_inplace: strict %XX decode ('+' untouched).
+    Returns None on an invalid escape (kernel: host fallback)."""
+    out = bytearray()
+    i = 0
+    n = len(s)
+    while i < n:
+        c = s[i]
+        if c == 0x25:  # %
+            if i + 2 >= n:
+                return None
+            try:
+                c = int(s[i + 1:i + 3].decode("latin-1"), 16)
+            except ValueError:
+                return None
+            i += 2
+        out.append(c)
+        i += 1
+    return bytes(out)
+
+
+def fnv1a64(data: bytes) -> int:
+    h = 0xcbf29ce484222325
+    for b in data:
+        h = ((h ^ b) * 0x100000001b3) & M64
+    return h
+
+
+def q_find_py(query: bytes, key: bytes):
+    """Mirror of q_find: raw value span of `key` in the query string."""
+    i = 0
+    n = len(query)
+    while i < n:
+        ke = i
+        while ke < n and query[ke] not in (0x3D, 0x26):  # '=' '&'
+            ke += 1
+        if query[i:ke] == key:
+            if ke < n and query[ke] == 0x3D:
+                ve = ke + 1
+                while ve < n and query[ve] != 0x26:
+                    ve += 1
+                return query[ke + 1:ve]
+            return b""
+        while ke < n and query[ke] != 0x26:
+            ke += 1
+        i = ke + 1
+    return None
+
+
+def splice_py(src: bytes, mode: int) -> bytes:
+    """Mirror of splice_bytes: TM_PCT decode then TM_JESC escaping."""
+    out = bytearray()
+    i = 0
+    n = len(src)
+    while i < n:
+        c = src[i]
+        if mode & TM_PCT:
+            if c == 0x2B:  # '+'
+                c = 0x20
+            elif c == 0x25 and i + 2 < n:
+                try:
+                    c = int(src[i + 1:i + 3].decode("latin-1"), 16)
+                    i += 2
+                except ValueError:
+                    pass
+        if mode & TM_JESC:
+            if c in (0x22, 0x5C):  # '"' '\\'
+                out.append(0x5C)
+                out.append(c)
+                i += 1
+                continue
+            if c < 0x20:
+                out += b"\\u00" + f"{c:02x}".encode()
+                i += 1
+                continue
+        out.append(c)
+        i += 1
+    return bytes(out)
+
+
+def json_top_fields_py(body: bytes):
+    """Mirror of json_top_fields: [(key, val_off, val_len)] of a JSON
+    object's top-level entries (value spans raw, ws-trimmed); [] when
+    not an object / malformed / > MAX_JSON_FIELDS semantics."""
+    n = len(body)
+    i = 0
+    while i < n and body[i] <= 0x20:
+        i += 1
+    if i >= n or body[i] != 0x7B:  # '{'
+        return []
+    i += 1
+    fields = []
+    while len(fields) < MAX_JSON_FIELDS:
+        while i < n and (body[i] <= 0x20 or body[i] == 0x2C):
+            i += 1
+        if i >= n:
+            return []
+        if body[i] == 0x7D:  # '}'
+            return fields
+        if body[i] != 0x22:  # '"'
+            return []
+        i += 1
+        ks = i
+        while i < n and body[i] != 0x22:
+            if body[i] == 0x5C:
+                i += 1
+            i += 1
+        if i >= n:
+            return []
+        ke = i
+        i += 1
+        while i < n and body[i] <= 0x20:
+            i += 1
+        if i >= n or body[i] != 0x3A:  # ':'
+            return []
+        i += 1
+        while i < n and body[i] <= 0x20:
+            i += 1
+        if i >= n:
+            return []
+        vs = i
+        if body[i] == 0x22:
+            i += 1
+            while i < n and body[i] != 0x22:
+                if body[i] == 0x5C:
+                    i += 1
+                i += 1
+            if i >= n:
+                return []
+            i += 1
+        elif body[i] in (0x7B, 0x5B):  # '{' '['
+            depth = 0
+            in_str = False
+            while i < n:
+                c = body[i]
+                if in_str:
+                    if c == 0x5C:
+                        i += 1
+                    elif c == 0x22:
+                        in_str = False
+                elif c == 0x22:
+                    in_str = True
+                elif c in (0x7B, 0x5B):
+                    depth += 1
+                elif c in (0x7D, 0x5D):
+                    depth -= 1
+                    if depth == 0:
+                        i += 1
+                        break
+                i += 1
+            if depth != 0:
+                return []
+        else:
+            while i < n and body[i] not in (0x2C, 0x7D) and body[i] > 0x20:
+                i += 1
+        fields.append((body[ks:ke], vs, i - vs))
+        while i < n and body[i] <= 0x20:
+            i += 1
+        if i < n and body[i] == 0x2C:
+            continue
+        if i < n and body[i] == 0x7D:
+            return fields
+        return []
+    return fields
+
+
+def render_template(blob: bytes, prog_off: int, req: bytes, F,
+                    cap: int = MAX_SLOT - 1024):
+    """Mirror of the HK_TEMPLATE branch: render the blob-encoded piece
+    program against one parsed request. Returns bytes, or None on
+    overflow (kernel renders the 500 envelope)."""
+    prog = np.frombuffer(blob, np.int32, offset=prog_off,
+                         count=1).item()
+    pieces = np.frombuffer(blob, np.int32, offset=prog_off + 4,
+                           count=prog * 4).reshape(prog, 4)
+    jfields = None
+    out = bytearray()
+    body = req[int(F[FI_BODY_OFF]):int(F[FI_BODY_OFF]) +
+               int(F[FI_BODY_LEN])]
+    for op, a, b, mode in pieces:
+        op, a, b, mode = int(op), int(a), int(b), int(mode)
+        if op == TP_LIT:
+            out += blob[a:a + b]
+        elif op == TP_PATH:
+            if a >= MAX_PARAMS:
+                continue
+            po = int(F[FI_PARAM0 + 2 * a])
+            pl = int(F[FI_PARAM0 + 2 * a + 1])
+            out += splice_py(req[po:po + pl], mode & ~TM_PCT)
+        elif op == TP_QUERY:
+            q = req[int(F[FI_QUERY_OFF]):int(F[FI_QUERY_OFF]) +
+                    int(F[FI_QUERY_LEN])]
+            val = q_find_py(q, blob[a:a + b])
+            if val is not None:
+                out += splice_py(val, mode)
+        elif op == TP_JFIELD:
+            if jfields is None:
+                jfields = json_top_fields_py(body)
+            for key, vs, vl in jfields:
+                if key == blob[a:a + b]:
+                    if mode & TM_JSTR and vl >= 2 and body[vs] == 0x22:
+                        vs, vl = vs + 1, vl - 2
+                    out += splice_py(body[vs:vs + vl],
+                                     (mode & ~TM_PCT) & ~TM_JSTR)
+                    break
+    if len(out) > cap:
+        return None
+    return bytes(out)
+
+
+def build_kv_table(store: dict, blob: bytearray, slot0: int):
+    """Compile one KV store into open-addressing rows + key/value bytes
+    appended to `blob` (shared kv_blob). Row layout (int32 x6, the
+    kernel's probe layout): [h_lo, h_hi, key_off, key_len, val_off,
+    val_len]; empty slots have val_len == -1. Values are pre-wrapped
+    {"data":...} envelopes. Returns (rows int32[nslots*6], nslots)."""
+    import json as _json
+    n = len(store)
+    nslots = 1
+    while nslots < max(4, 2 * n):
+        nslots *= 2
+    rows = np.zeros((nslots, 6), np.int32)
+    rows[:, 5] = -1
+    for key, value in store.items():
+        kb = key.encode("utf-8") if isinstance(key, str) else bytes(key)
+        vb = (b'{"data":' +
+              _json.dumps(value, separators=(",", ":"),
+                          ensure_ascii=False).encode("utf-8") + b"}")
+        h = fnv1a64(kb)
+        koff = len(blob)
+        blob += kb
+        voff = len(blob)
+        blob += vb
+        idx = h % nslots
+        while rows[idx, 5] >= 0:
+            idx = (idx + 1) % nslots
+        rows[idx] = (np.int32(h & 0xFFFFFFFF) if (h & 0xFFFFFFFF) < 2**31
+                     else np.int32((h & 0xFFFFFFFF) - 2**32),
+                     np.int32(h >> 32) if (h >> 32) < 2**31
+                     else np.int32((h >> 32) - 2**32),
+                     koff, len(kb), voff, len(vb))
+    return rows.reshape(-1), nslots
+
+
+def kv_lookup_mirror(kv_tab, kv_blob: bytes, slot0: int, nslots: int,
+                     key: bytes):
+    """Mirror of the HK_KV probe loop."""
+    if nslots <= 0 or not key:
+        return None
+    h = fnv1a64(key)
+    h_lo = np.int32(h & 0xFFFFFFFF) if (h & 0xFFFFFFFF) < 2**31 \
+        else np.int32((h & 0xFFFFFFFF) - 2**32)
+    h_hi = np.int32(h >> 32) if (h >> 32) < 2**31 \
+        else np.int32((h >> 32) - 2**32)
+    idx = h % nslots
+    for _ in range(nslots):
+        row = kv_tab[(slot0 + idx) * 6:(slot0 + idx) * 6 + 6]
+        if row[5] < 0:
+            return None
+        if row[0] == h_lo and row[1] == h_hi and row[3] == len(key) and \
+                kv_blob[int(row[2]):int(row[2]) + len(key)] == key:
+            return kv_blob[int(row[4]):int(row[4]) + int(row[5])]
+        idx = (idx + 1) % nslots
+    return None
+
+
 _ETAG_B = ((np.arange(64)[:, None] * 17 + np.arange(16)[None, :] * 29
             + 3) % 251 - 125).astype(np.int8)
 _ETAG_SALT = (np.arange(256, dtype=np.uint32) * 2 + 1).reshape(16, 16)
@@ -588,7 +890,8 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 handler_tab: np.ndarray, blob: bytes,
                 host_blob: bytes, host_tab: np.ndarray,
                 seed: int, auth_env=(0, 0), gzip_min: int = 0,
-                etag_on: bool = False, date29: bytes = b""):
+                etag_on: bool = False, date29: bytes = b"",
+                kv_tab=None, kv_blob: bytes = b""):
     """Mirror of k_respond. Returns (resp uint8 [n*rslot], resp_len int32)."""
     n = len(fields)
     host_tab = np.asarray(host_tab, np.int32).reshape(-1)
@@ -628,6 +931,32 @@ def cpu_respond(reqs: np.ndarray, req_off: np.ndarray,
                 off = int(handler_tab[route * 4 + 1])
                 ln = int(handler_tab[route * 4 + 2])
                 body_src = blob[off:off + ln]
+        elif kind == HK_TEMPLATE and not (flags & FL_AUTH_FAIL):
+            route = int(F[FI_ROUTE])
+            prog_off = int(handler_tab[route * 4 + 1])
+            raw = reqs[base:base + MAX_SLOT].tobytes()
+            rendered = render_template(blob, prog_off, raw, F)
+            if rendered is None:
+                status = 500
+                elen = int.from_bytes(blob[:4], "little")
+                body_src = blob[4:4 + elen]
+            else:
+                body_src = rendered
+        elif kind == HK_KV and not (flags & FL_AUTH_FAIL):
+            route = int(F[FI_ROUTE])
+            slot0 = int(handler_tab[route * 4 + 1])
+            nslots = int(handler_tab[route * 4 + 2])
+            po = int(F[FI_PARAM0])
+            pl = int(F[FI_PARAM0 + 1])
+            key = reqs[base + po:base + po + pl].tobytes()
+            val = kv_lookup_mirror(kv_tab, kv_blob, slot0, nslots, key)
+            if val is not None:
+                body_src = val
+            else:
+                status = 404
+                elen = int.from_bytes(blob[:4], "little")
+                mlen = int.from_bytes(blob[4 + elen:8 + elen], "little")
+                body_src = blob[8 + elen:8 + elen + mlen]
         if kind == HK_HOST and not is_options and \
                 not (flags & FL_AUTH_FAIL):
             off, ln, status, ct_id = (int(host_tab[r * 4 + i])

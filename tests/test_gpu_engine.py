@@ -307,3 +307,63 @@ def test_armed_with_middlewares_matches_mirror():
     assert b"Content-Encoding: gzip" in first and b'ETag: "' in first
     second = g[int(roff_t[1]):int(roff_t[1]) + int(rlen_t[1])].tobytes()
     assert second.startswith(b"HTTP/1.1 401 ")
+
+
+def test_template_kv_kernels_match_mirror():
+    """HK_TEMPLATE / HK_KV / pct-decode / query+JSON-field splice: GPU
+    bytes == CPU mirror bytes."""
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/user/{id}", handlers.template_json(
+        '{"data":{"id":"', ("path", 0), '"}}'))
+    app.GET("/hello", handlers.template_json(
+        '{"data":"Hello ', ("query", "name"), '!"}'))
+    app.POST("/order", handlers.template_json(
+        '{"data":{"item":', ("jfield", "item"),
+        ',"qty":', ("jfield", "qty"),
+        ',"note":"', ("jfield_str", "note"), '"}}'))
+    app.GET("/kv/{key}", handlers.kv_json(
+        {f"user{i}": {"id": i, "bio": "x" * (i % 40)} for i in range(64)}))
+    gpu = BatchEngine(app, device="cuda", max_batch=2048)
+    cpu = BatchEngine(app, device="cpu", max_batch=2048)
+    cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
+
+    import random
+    rng = random.Random(7)
+    raws = []
+    for i in range(1024):
+        r = rng.randrange(7)
+        if r == 0:
+            raws.append(http_req("GET", f"/user/u{i}"))
+        elif r == 1:
+            raws.append(http_req("GET", "/user/a%20b%2fc"))
+        elif r == 2:
+            raws.append(http_req("GET", f"/hello?name=n%22{i}+x&z=1"))
+        elif r == 3:
+            body = json.dumps({"item": ["a", i], "qty": i,
+                               "note": 'say "hi"\t'},
+                              separators=(",", ":")).encode()
+            raws.append(http_req("POST", "/order", body))
+        elif r == 4:
+            raws.append(http_req("GET", f"/kv/user{i % 80}"))  # ~20% miss
+        elif r == 5:
+            raws.append(http_req("GET", "/hello"))  # missing query param
+        else:
+            body = b'{"broken": '  # malformed JSON -> empty fields
+            raws.append(http_req("POST", "/order", body))
+    g = gpu.process(list(raws))
+    c = cpu.process(list(raws))
+    for i, (go, co) in enumerate(zip(g, c)):
+        assert go == co, (f"req {i}: {raws[i]!r}\nGPU {go!r}\nCPU {co!r}")
+    # pure-GPU batch: none of these kinds may touch the trampoline
+    from gofr_amd.engine import pack_batch
+    from gofr_amd import ops
+    buf, offs, lens = pack_batch(raws)
+    fields = ops.cpu_parse_route(buf, offs, lens, cpu.program.trie,
+                                 cpu.program.handler_tab)
+    assert not any(f[ops.FI_KIND] == ops.HK_HOST for f in fields)
