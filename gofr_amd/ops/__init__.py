@@ -424,20 +424,36 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         body_off = ln
         auth_off = auth_len = 0
         inm_off = inm_len = 0
-        prev_lf = lf1
-        for _ in range(64):
-            ls = prev_lf + 1
-            lf = buf.find(b"\n", ls)
-            if lf < 0:
-                flags |= FL_ERR_PARSE
+        # mirror of the kernel's one-line-per-lane header pass: LF
+        # position table, first blank line among lines 1..63, then the
+        # header lines before it (last occurrence of a repeated header
+        # wins — the reductions are max-by-line-start)
+        lfs = []
+        p = -1
+        while True:
+            p = buf.find(b"\n", p + 1)
+            if p < 0:
                 break
-            le = lf - 1
-            if le <= ls:
-                body_off = lf + 1
+            lfs.append(p)
+        nlf = len(lfs)
+        j_empty = -1
+        for j in range(1, min(nlf - 1, 63) + 1):
+            if lfs[j] <= lfs[j - 1] + 2:
+                j_empty = j
                 break
-            co = buf.find(b":", ls)
-            if co < 0 or co >= le:
-                prev_lf = lf
+        if j_empty > 0:
+            body_off = lfs[j_empty] + 1
+            hdr_last = j_empty - 1
+        elif nlf - 1 > 63:
+            hdr_last = 63  # >63 header lines: body falls back to end
+        else:
+            flags |= FL_ERR_PARSE  # LFs ran out before a blank line
+            hdr_last = min(nlf - 1, 63)
+        for j in range(1, hdr_last + 1):
+            ls = lfs[j - 1] + 1
+            le = lfs[j] - 1
+            co = buf.find(b":", ls, le)
+            if co < 0:
                 continue
             nlen = co - ls
             vs = co + 1
@@ -468,7 +484,6 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
             elif nlen == 15 and _ieq(buf, ls, b"accept-encoding"):
                 if b"gzip" in buf[vs:vs + vlen]:
                     flags |= FL_ACCEPT_GZIP
-            prev_lf = lf
         body_len = max(0, ln - body_off)
         if 0 < clen < body_len:
             body_len = clen
