@@ -187,6 +187,15 @@ public:
         return (long)ready_.size();
     }
 
+    // pop one ready request (sharded harvest path)
+    bool pop_ready(PendingReq* out) {
+        std::lock_guard<std::mutex> lk(mu_);
+        if (ready_.empty()) return false;
+        *out = std::move(ready_.front());
+        ready_.pop_front();
+        return true;
+    }
+
     int port() const { return port_; }
 
 private:
@@ -475,10 +484,83 @@ public:
         return t;
     }
 
+    // Sharded harvest (multi-GPU serving): drain ready requests into a
+    // SLOT-layout exchange buffer of `world` blocks x `bpr` slots x
+    // `slot` bytes, the fixed-size all-to-all granularity of
+    // engine.shard.AllToAllSharder. A request's block = its OWNER rank
+    // = splitmix-hash(conn_id) % world (stable conn->owner affinity).
+    // Overfull blocks spill to a backlog drained first next cycle.
+    // lens[] is zeroed for unfilled slots (FL_EMPTY padding on device);
+    // conn ids for unfilled slots are 0 (never allocated). Requests
+    // larger than `slot` are dropped (the sharded exchange is
+    // fixed-granularity; size `slot` for the workload).
+    int harvest_slots(uintptr_t buf_ptr, int slot, int world, int bpr,
+                      uintptr_t len_ptr, uintptr_t conn_ptr,
+                      int window_us) {
+        py::gil_scoped_release rel;
+        uint8_t* buf = (uint8_t*)buf_ptr;
+        int32_t* lens = (int32_t*)len_ptr;
+        uint64_t* cids = (uint64_t*)conn_ptr;
+        const int total = world * bpr;
+        memset(lens, 0, sizeof(int32_t) * (size_t)total);
+        memset(cids, 0, sizeof(uint64_t) * (size_t)total);
+        std::vector<int> fill(world, 0);
+        int n = 0;
+        auto place = [&](PendingReq& req) -> bool {
+            if ((long)req.bytes.size() > slot) return true;  // dropped
+            const uint64_t h = req.conn_id * 0x9E3779B97F4A7C15ull;
+            const int owner = (int)((h >> 32) % (uint64_t)world);
+            if (fill[owner] >= bpr) {
+                backlog_.push_back(std::move(req));
+                return false;  // block full: spilled
+            }
+            const int s = owner * bpr + fill[owner]++;
+            memcpy(buf + (size_t)s * slot, req.bytes.data(),
+                   req.bytes.size());
+            lens[s] = (int32_t)req.bytes.size();
+            cids[s] = req.conn_id;
+            ++n;
+            return true;
+        };
+        const auto deadline = std::chrono::steady_clock::now() +
+                              std::chrono::microseconds(window_us);
+        {
+            // backlog first (already popped from the reactors)
+            std::deque<PendingReq> keep;
+            while (!backlog_.empty()) {
+                PendingReq req = std::move(backlog_.front());
+                backlog_.pop_front();
+                const uint64_t h = req.conn_id * 0x9E3779B97F4A7C15ull;
+                const int owner = (int)((h >> 32) % (uint64_t)world);
+                if (fill[owner] >= bpr || (long)req.bytes.size() > slot) {
+                    if ((long)req.bytes.size() <= slot)
+                        keep.push_back(std::move(req));
+                    continue;
+                }
+                place(req);
+            }
+            backlog_.swap(keep);
+        }
+        while (true) {
+            for (auto& r : reactors_) {
+                while (true) {
+                    PendingReq req;
+                    if (!r->pop_ready(&req)) break;
+                    place(req);
+                }
+            }
+            if (n > 0 || std::chrono::steady_clock::now() >= deadline)
+                break;
+            std::this_thread::sleep_for(std::chrono::microseconds(50));
+        }
+        return n;
+    }
+
 private:
     int port_;
     size_t max_req_;
     int nthreads_;
+    std::deque<PendingReq> backlog_;
     std::vector<std::unique_ptr<Reactor>> reactors_;
 };
 

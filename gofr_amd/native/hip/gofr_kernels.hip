@@ -1346,9 +1346,11 @@ __device__ __forceinline__ void respond_impl(
         } else {
             status = 404;
             const int elen = *(const int32_t*)blob;
-            const int miss_len = *(const int32_t*)(blob + 4 + elen);
+            // blob+4+elen is NOT 4-aligned: byte-wise length read
+            const uint8_t* ml = blob + 4 + elen;
             body_src = blob + 8 + elen;
-            body_src_len = miss_len;
+            body_src_len = ml[0] | (ml[1] << 8) | (ml[2] << 16) |
+                           (ml[3] << 24);
         }
     }
     int ct_id = 0;  // 0 json, 1 icon, 2 octet-stream, 3 text/plain
