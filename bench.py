@@ -299,16 +299,18 @@ def run_config5_cpu(eng, payloads, steps, warmup, n_grpc, conns):
 
 
 def run_multi(eng, payloads, steps, warmup, rank, world):
-    """RCCL all-to-all sharding (AllToAllSharder — the same class the
-    gloo multi-process tests cover): scatter request slots to owner
-    shards, process on the local engine, gather responses back.
+    """RCCL all-to-all sharding: measures AllToAllSharder.serve_step —
+    the SAME method GPUServer._serve_loop_sharded runs per cycle in
+    production (H2D staging, all-to-all request scatter, local kernel
+    pipeline, all-to-all response gather, D2H), so this bench times the
+    serving dataflow unmodified; the only substitution is the synthetic
+    pre-staged ingress standing in for C++ harvest_slots.
 
     On GPU the loop is pipelined over 2 lanes, each with its own stream
-    + sharder buffer set: every stage (H2D, a2a scatter, kernels, a2a
-    gather, D2H) is enqueued under the lane stream (torch's NCCL calls
-    chain onto the calling stream), so lane A's collectives overlap
-    lane B's kernels/copies. Host-side collective call order is the
-    lane order, identical on every rank, so NCCL matching is safe."""
+    + sharder buffer set (torch's NCCL calls chain onto the calling
+    stream), so lane A's collectives overlap lane B's kernels/copies.
+    Host-side collective call order is the lane order, identical on
+    every rank, so NCCL matching is safe."""
     import torch
     from gofr_amd.engine.shard import AllToAllSharder
     import torch.distributed as dist
@@ -319,60 +321,49 @@ def run_multi(eng, payloads, steps, warmup, rank, world):
     assert n % world == 0
     P = min(2, len(eng.lanes)) if dev is not None else 1
 
-    class MLane:
-        def __init__(self, li):
-            self.sh = AllToAllSharder(eng, world, lane=li,
-                                      sync_host=(dev is None))
-            self.sh.alloc(n)
-            if dev is not None:
-                self.stream = t.cuda.Stream(device=dev)
-                self.ev = t.cuda.Event()
-                self.p_in = t.from_numpy(reqs).pin_memory()
-                self.p_len = t.from_numpy(
-                    lens.astype(np.int32)).pin_memory()
-                self.d_in = t.empty(n * eng.slot, dtype=t.uint8,
-                                    device=dev)
-                self.d_len = t.empty(n, dtype=t.int32, device=dev)
-                self.p_resp = t.empty(n * eng.rslot,
-                                      dtype=t.uint8).pin_memory()
-                self.p_rlen = t.empty(n, dtype=t.int32).pin_memory()
-            else:
-                self.d_in = t.from_numpy(reqs)
-                self.d_len = t.from_numpy(lens.astype(np.int32))
-
-    mlanes = [MLane(li) for li in range(P)]
+    shs = []
+    for li in range(P):
+        sh = AllToAllSharder(eng, world, lane=li, sync_host=(dev is None))
+        sh.alloc_serve(n)
+        # stage the synthetic ingress once (the production loop's
+        # harvest_slots refills these pinned buffers per cycle)
+        if dev is not None:
+            sh.p_in.copy_(t.from_numpy(reqs))
+            sh.p_len.copy_(t.from_numpy(lens.astype(np.int32)))
+            sh.stream = t.cuda.Stream(device=dev)
+            sh.ev = t.cuda.Event()
+        else:
+            sh.p_in[:] = reqs
+            sh.p_len[:] = lens
+        shs.append(sh)
 
     if dev is None:
-        ml = mlanes[0]
+        sh = shs[0]
         times = []
         for it in range(warmup + steps):
             if it == warmup:
                 dist.barrier()
                 t_start = time.perf_counter()
             t0 = time.perf_counter()
-            resp_sh, rlen_sh = ml.sh.step(ml.d_in, ml.d_len)
+            out, rlen = sh.serve_step()
             times.append(time.perf_counter() - t0)
             if it == 0:
-                first = resp_sh.numpy()[:int(rlen_sh[0])].tobytes()
+                first = out[:int(rlen[0])].tobytes()
                 assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
         dist.barrier()
         return time.perf_counter() - t_start, times[warmup:]
 
-    def submit(ml):
-        with t.cuda.stream(ml.stream):
-            ml.d_in.copy_(ml.p_in, non_blocking=True)
-            ml.d_len.copy_(ml.p_len, non_blocking=True)
-            resp_sh, rlen_sh = ml.sh.step(ml.d_in, ml.d_len)
-            ml.p_resp.copy_(resp_sh, non_blocking=True)
-            ml.p_rlen.copy_(rlen_sh, non_blocking=True)
-            ml.ev.record(ml.stream)
+    def submit(sh):
+        with t.cuda.stream(sh.stream):
+            sh.serve_step()
+            sh.ev.record(sh.stream)
 
     # warmup (serial, both lanes so NCCL per-lane state initializes;
     # identical order on every rank)
     for w in range(max(P, warmup)):
-        submit(mlanes[w % P])
-        mlanes[w % P].ev.synchronize()
-    first = mlanes[0].p_resp[:int(mlanes[0].p_rlen[0])].numpy().tobytes()
+        submit(shs[w % P])
+        shs[w % P].ev.synchronize()
+    first = shs[0].p_resp[:int(shs[0].p_rlen[0])].numpy().tobytes()
     assert first.startswith(b"HTTP/1.1 200 OK\r\n"), first[:80]
     torch.cuda.synchronize(dev)
     dist.barrier()
@@ -384,12 +375,12 @@ def run_multi(eng, payloads, steps, warmup, rank, world):
     for i in range(steps):
         li = i % P
         if i >= P:
-            mlanes[li].ev.synchronize()
+            shs[li].ev.synchronize()
             lat.append(time.perf_counter() - submit_at[i - P])
         submit_at[i] = time.perf_counter()
-        submit(mlanes[li])
+        submit(shs[li])
     for i in range(max(0, steps - P), steps):
-        mlanes[i % P].ev.synchronize()
+        shs[i % P].ev.synchronize()
         lat.append(time.perf_counter() - submit_at[i])
     torch.cuda.synchronize(dev)
     dist.barrier()

@@ -964,9 +964,15 @@ class GPUServer:
     """
 
     def __init__(self, app, port: int, batch_window_us: int = 200,
-                 max_batch: int = 8192, arm_chunk: int = 1024):
+                 max_batch: int = 8192, arm_chunk: int = 1024,
+                 world: int = 1, rank: int = 0, shard_chunk: int = 1024):
         self.app = app
         self.port = port
+        self.world = world
+        self.rank = rank
+        self.shard_chunk = shard_chunk  # slots per destination block
+        if world > 1:
+            max_batch = max(max_batch, world * shard_chunk)
         self.engine = BatchEngine(app, max_batch=max_batch,
                                   pipeline=4)
         self.batch_window_us = batch_window_us
@@ -976,14 +982,20 @@ class GPUServer:
         self._stop = None
 
     def start(self):
+        """world > 1: the caller must have initialized
+        torch.distributed (nccl on GPUs, gloo on CPU) with one process
+        per GPU BEFORE start(); every rank binds the same port via
+        SO_REUSEPORT, so the kernel spreads client connections across
+        ranks and the all-to-all re-balances them to their owner."""
         import threading
         from .. import _core  # built by setup.py / __graft_entry__.build()
         self._core = _core.EpollServer(self.port)
         self._core.start()
         self.port = self._core.port()
         self._stop = threading.Event()
-        self._thread = threading.Thread(target=self._serve_loop,
-                                        daemon=True)
+        target = self._serve_loop if self.world == 1 \
+            else self._serve_loop_sharded
+        self._thread = threading.Thread(target=target, daemon=True)
         self._thread.start()
 
     def _serve_loop(self):
@@ -1046,6 +1058,95 @@ class GPUServer:
                                 out_t.data_ptr(), roff_t.data_ptr(),
                                 rlen_t.data_ptr())
                 free.append(L)
+
+    def _serve_loop_sharded(self):
+        """Multi-GPU PRODUCTION serve loop: each
+        cycle stages this rank's ingress into owner-ordered request
+        slots (C++ harvest_slots: owner = hash(conn) % world, stable
+        affinity), runs AllToAllSharder.serve_step — RCCL all-to-all
+        scatter over xGMI, local kernel pipeline, all-to-all response
+        gather (the same method bench.py run_multi measures) — and
+        writes the returned responses to the local sockets. Every rank
+        runs the cycle at the same cadence (a stop-consensus
+        all-reduce leads each cycle), so collectives pair by order.
+        Pipelined over 2 lanes: lane A's exchange+kernels overlap lane
+        B's harvest and egress."""
+        from collections import deque
+
+        import torch.distributed as dist
+
+        from .shard import AllToAllSharder
+        eng = self.engine
+        world, bpr = self.world, self.shard_chunk
+        n = world * bpr
+        on_gpu = eng.device is not None
+        t = eng.torch
+        if t is None:
+            import torch as t  # CPU-mirror path without engine torch
+        P = 2 if on_gpu else 1
+        shs = []
+        for li in range(P):
+            sh = AllToAllSharder(eng, world, lane=li,
+                                 sync_host=not on_gpu)
+            sh.alloc_serve(n)
+            shs.append(sh)
+        conn_ids = [np.zeros(n, np.uint64) for _ in range(P)]
+        if on_gpu:
+            streams = [t.cuda.Stream(device=eng.device)
+                       for _ in range(P)]
+            evs = [t.cuda.Event() for _ in range(P)]
+            stop_t = t.zeros(1, device=eng.device)
+        else:
+            stop_t = t.zeros(1)
+        pending = deque()
+
+        def complete(li):
+            sh = shs[li]
+            if on_gpu:
+                evs[li].synchronize()
+                self._core.send(conn_ids[li].ctypes.data, n,
+                                sh.p_resp.data_ptr(),
+                                sh.p_roff.data_ptr(),
+                                sh.p_rlen.data_ptr())
+            else:
+                out, rlen = sh.result
+                self._core.send(conn_ids[li].ctypes.data, n,
+                                out.ctypes.data, sh.p_roff.ctypes.data,
+                                rlen.ctypes.data)
+
+        it = 0
+        while True:
+            # stop consensus (MIN: the loop runs until EVERY rank has
+            # requested stop, so early-stopping ranks keep serving
+            # their peers' exchanges): every rank leaves at the same
+            # cycle and no rank blocks in a collective its peers never
+            # issue
+            stop_t.fill_(1.0 if self._stop.is_set() else 0.0)
+            dist.all_reduce(stop_t, op=dist.ReduceOp.MIN)
+            if float(stop_t.item()) > 0:
+                break
+            li = it % P
+            it += 1
+            if len(pending) == P:
+                complete(pending.popleft())
+            sh = shs[li]
+            if on_gpu:
+                self._core.harvest_slots(
+                    sh.p_in.data_ptr(), eng.slot, world, bpr,
+                    sh.p_len.data_ptr(), conn_ids[li].ctypes.data,
+                    self.batch_window_us)
+                with t.cuda.stream(streams[li]):
+                    sh.serve_step()
+                    evs[li].record(streams[li])
+            else:
+                self._core.harvest_slots(
+                    sh.p_in.ctypes.data, eng.slot, world, bpr,
+                    sh.p_len.ctypes.data, conn_ids[li].ctypes.data,
+                    self.batch_window_us)
+                sh.result = sh.serve_step()
+            pending.append(li)
+        while pending:
+            complete(pending.popleft())
 
     def _serve_loop_cpu(self):
         eng = self.engine

@@ -584,5 +584,6 @@ PYBIND11_MODULE(_core, m) {
         .def("port", &EpollServer::port)
         .def("ready_count", &EpollServer::ready_count)
         .def("harvest", &EpollServer::harvest)
+        .def("harvest_slots", &EpollServer::harvest_slots)
         .def("send", &EpollServer::send);
 }
