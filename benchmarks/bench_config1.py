@@ -123,6 +123,8 @@ def main():
     ap.add_argument("--n-addrs", type=int, default=1,
                     help="spread client conns over 127.0.0.{1..N} "
                          "(needed above ~50k conns)")
+    ap.add_argument("--arm-chunk", type=int, default=1024,
+                    help="armed fixed-batch size of the serving loop")
     args = ap.parse_args()
 
     nofile = _raise_nofile()
@@ -133,7 +135,7 @@ def main():
                                          "LOG_LEVEL": "FATAL"}))
     app.GET("/greet", handlers.static_json("Hello World!"))
     srv = GPUServer(app, 0, batch_window_us=args.window_us,
-                max_batch=args.max_batch)
+                    max_batch=args.max_batch, arm_chunk=args.arm_chunk)
     srv.start()
     if args.client == "native":
         try:

@@ -989,7 +989,11 @@ class GPUServer:
         ranks and the all-to-all re-balances them to their owner."""
         import threading
         from .. import _core  # built by setup.py / __graft_entry__.build()
-        self._core = _core.EpollServer(self.port)
+        # reactor threads: each sustains ~1.5M req/s of socket+parse
+        # work (r1 measurement), so the ingress needs several to keep
+        # up with the engine's ~27M req/s
+        nreact = int(os.environ.get("GOFR_REACTORS", "8"))
+        self._core = _core.EpollServer(self.port, 1 << 20, nreact)
         self._core.start()
         self.port = self._core.port()
         self._stop = threading.Event()
