@@ -1033,6 +1033,9 @@ class GPUServer:
         off_views = [ln.p_req_off.numpy() for ln in eng.lanes]
         free = deque(range(P))
         inflight = deque()
+        stats = os.environ.get("GOFR_SERVE_STATS") == "1"
+        st = {"harvest": 0.0, "submit": 0.0, "complete": 0.0,
+              "send": 0.0, "cycles": 0, "reqs": 0}
         while not self._stop.is_set():
             progressed = False
             if free:
@@ -1040,28 +1043,50 @@ class GPUServer:
                 ln = eng.lanes[L]
                 maxn = CH if armed else eng.max_batch
                 bufcap = cap if armed else eng.max_bytes
+                t0 = time.perf_counter() if stats else 0.0
                 n, nbytes = self._core.harvest(
                     ln.p_reqs.data_ptr(), bufcap,
                     ln.p_req_off.data_ptr(), ln.p_req_len.data_ptr(),
                     conn_ids[L].ctypes.data, maxn, self.batch_window_us)
+                if stats:
+                    st["harvest"] += time.perf_counter() - t0
                 if n:
                     progressed = True
+                    t0 = time.perf_counter() if stats else 0.0
                     if armed:
                         len_views[L][n:CH] = 0
                         off_views[L][n:CH] = 0
                         eng.submit(CH, cap, L)
                     else:
                         eng.submit(n, nbytes, L)
+                    if stats:
+                        st["submit"] += time.perf_counter() - t0
+                        st["cycles"] += 1
+                        st["reqs"] += n
                     inflight.append((L, n))
                 else:
                     free.appendleft(L)
             if inflight and (not free or not progressed):
                 L, n = inflight.popleft()
+                t0 = time.perf_counter() if stats else 0.0
                 out_t, roff_t, rlen_t = eng.complete(L)
+                if stats:
+                    t1 = time.perf_counter()
+                    st["complete"] += t1 - t0
                 self._core.send(conn_ids[L].ctypes.data, n,
                                 out_t.data_ptr(), roff_t.data_ptr(),
                                 rlen_t.data_ptr())
+                if stats:
+                    st["send"] += time.perf_counter() - t1
                 free.append(L)
+        if stats and st["cycles"]:
+            c = st["cycles"]
+            print(f"[serve-stats] cycles {c} reqs {st['reqs']} "
+                  f"fill {st['reqs']/c:.0f} "
+                  f"us/cycle: harvest {st['harvest']/c*1e6:.0f} "
+                  f"submit {st['submit']/c*1e6:.0f} "
+                  f"complete {st['complete']/c*1e6:.0f} "
+                  f"send {st['send']/c*1e6:.0f}", flush=True)
 
     def _serve_loop_sharded(self):
         """Multi-GPU PRODUCTION serve loop: each
