@@ -26,9 +26,11 @@
 #include <atomic>
 #include <cerrno>
 #include <chrono>
+#include <condition_variable>
 #include <cstring>
 #include <deque>
 #include <fcntl.h>
+#include <functional>
 #include <memory>
 #include <mutex>
 #include <netinet/in.h>
@@ -141,27 +143,6 @@ public:
         listen_fd_ = ep_ = -1;
     }
 
-    // drain up to max_n ready requests into the ring; returns (n, bytes)
-    std::pair<int, long> drain(uint8_t* buf, long buf_cap, long pos,
-                               int64_t* offs, int32_t* lens,
-                               uint64_t* cids, int base_n, int max_n) {
-        int n = base_n;
-        std::lock_guard<std::mutex> lk(mu_);
-        while (n < max_n && !ready_.empty()) {
-            PendingReq& req = ready_.front();
-            const long sz = (long)req.bytes.size();
-            if (pos + sz > buf_cap) break;
-            memcpy(buf + pos, req.bytes.data(), sz);
-            offs[n] = pos;
-            lens[n] = (int32_t)sz;
-            cids[n] = req.conn_id;
-            pos += sz;
-            ++n;
-            ready_.pop_front();
-        }
-        return {n, pos};
-    }
-
     void queue_writes(const uint64_t* cids, const int* rows, int nrows,
                       const uint8_t* out, const int32_t* roffs,
                       const int32_t* rlens) {
@@ -191,6 +172,17 @@ public:
     bool pop_ready(PendingReq* out) {
         std::lock_guard<std::mutex> lk(mu_);
         if (ready_.empty()) return false;
+        *out = std::move(ready_.front());
+        ready_.pop_front();
+        return true;
+    }
+
+    // pop the next ready request only if it fits in `room` bytes
+    bool pop_ready_fit(PendingReq* out, long room) {
+        std::lock_guard<std::mutex> lk(mu_);
+        if (ready_.empty() ||
+            (long)ready_.front().bytes.size() > room)
+            return false;
         *out = std::move(ready_.front());
         ready_.pop_front();
         return true;
@@ -401,6 +393,91 @@ private:
     std::vector<uint64_t> pending_writes_;
 };
 
+// Persistent worker pool for the harvest/send fan-out: the serving
+// thread's serial per-request work (ready-queue drain + egress routing)
+// measured 150 + 120 ns/req — a ~3.6M req/s ceiling with the GPU at
+// ~3 us/batch. run() executes fn(0..ntasks) across the workers + the
+// calling thread and returns when all tasks finished.
+class WorkerPool {
+public:
+    void start(int w) {
+        nworkers_ = w < 1 ? 0 : w;
+        running_ = true;
+        for (int i = 0; i < nworkers_; ++i)
+            threads_.emplace_back([this] { worker(); });
+    }
+
+    void stop() {
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            running_ = false;
+            cv_.notify_all();
+        }
+        for (auto& t : threads_) t.join();
+        threads_.clear();
+    }
+
+    void run(const std::function<void(int)>& fn, int ntasks) {
+        if (nworkers_ == 0 || ntasks <= 1) {
+            for (int i = 0; i < ntasks; ++i) fn(i);
+            return;
+        }
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            fn_ = &fn;
+            ntasks_ = ntasks;
+            next_.store(0, std::memory_order_relaxed);
+            done_.store(0, std::memory_order_relaxed);
+            ++gen_;
+            cv_.notify_all();
+        }
+        // the caller works too
+        help();
+        std::unique_lock<std::mutex> lk(mu_);
+        cv_done_.wait(lk, [&] {
+            return done_.load(std::memory_order_acquire) >= ntasks_;
+        });
+        fn_ = nullptr;
+    }
+
+private:
+    void help() {
+        while (true) {
+            const int i = next_.fetch_add(1, std::memory_order_relaxed);
+            if (i >= ntasks_) return;
+            (*fn_)(i);
+            if (done_.fetch_add(1, std::memory_order_acq_rel) + 1 >=
+                ntasks_) {
+                std::lock_guard<std::mutex> lk(mu_);
+                cv_done_.notify_all();
+            }
+        }
+    }
+
+    void worker() {
+        uint64_t seen = 0;
+        while (true) {
+            {
+                std::unique_lock<std::mutex> lk(mu_);
+                cv_.wait(lk, [&] { return !running_ || gen_ != seen; });
+                if (!running_) return;
+                seen = gen_;
+            }
+            help();
+        }
+    }
+
+    int nworkers_ = 0;
+    std::vector<std::thread> threads_;
+    std::mutex mu_;
+    std::condition_variable cv_, cv_done_;
+    const std::function<void(int)>* fn_ = nullptr;
+    std::atomic<int> next_{0}, done_{0};
+    int ntasks_ = 0;
+    uint64_t gen_ = 0;
+    bool running_ = false;
+};
+
 class EpollServer {
 public:
     EpollServer(int port, size_t max_req = 1 << 20, int threads = 4)
@@ -416,9 +493,15 @@ public:
             port_ = reactors_.back()->bind_and_listen();  // 0 -> learned
         }
         for (auto& r : reactors_) r->run();
+        const char* e = getenv("GOFR_IO_WORKERS");
+        int w = e ? atoi(e) : 7;
+        if (w > nthreads_ - 1) w = nthreads_ - 1;
+        pool_.start(w);
+        scratch_.resize(reactors_.size());
     }
 
     void stop() {
+        pool_.stop();
         for (auto& r : reactors_) r->stop();
         reactors_.clear();
     }
@@ -426,7 +509,12 @@ public:
     int port() const { return port_; }
 
     // Harvest up to max_n complete requests into the caller's buffers
-    // (the engine's pinned ingress ring), round-robin over reactors.
+    // (the engine's pinned ingress ring). PARALLEL: reactor r's ready
+    // queue drains into byte segment r of the ring (quota max_n/R
+    // requests; leftovers stay queued for the next cycle) across the
+    // worker pool, then one cheap serial pass merges the metadata
+    // dense. The armed serving loop ships the whole fixed-size ring
+    // every batch, so segmented placement costs no extra bus bytes.
     // Blocks up to window_us for the FIRST request. GIL released.
     std::pair<int, long> harvest(uintptr_t buf_ptr, long buf_cap,
                                  uintptr_t off_ptr, uintptr_t len_ptr,
@@ -437,27 +525,81 @@ public:
         int64_t* offs = (int64_t*)off_ptr;
         int32_t* lens = (int32_t*)len_ptr;
         uint64_t* cids = (uint64_t*)conn_ptr;
+        const int R = (int)reactors_.size();
+        const long seg_bytes = buf_cap / R;
         const auto deadline = std::chrono::steady_clock::now() +
                               std::chrono::microseconds(window_us);
         int n = 0;
-        long pos = 0;
+        long nbytes = 0;
+        if (max_n < R) {
+            // tiny batches: serial drain (quota math needs max_n >= R)
+            while (true) {
+                for (auto& r : reactors_) {
+                    PendingReq req;
+                    while (n < max_n &&
+                           r->pop_ready_fit(&req, buf_cap - nbytes)) {
+                        memcpy(buf + nbytes, req.bytes.data(),
+                               req.bytes.size());
+                        offs[n] = nbytes;
+                        lens[n] = (int32_t)req.bytes.size();
+                        cids[n] = req.conn_id;
+                        nbytes += (long)req.bytes.size();
+                        ++n;
+                    }
+                }
+                if (n > 0 ||
+                    std::chrono::steady_clock::now() >= deadline)
+                    return {n, nbytes};
+                std::this_thread::sleep_for(
+                    std::chrono::microseconds(50));
+            }
+        }
+        const int quota = max_n / R;
         while (true) {
-            for (auto& r : reactors_) {
-                auto np = r->drain(buf, buf_cap, pos, offs, lens, cids,
-                                   n, max_n);
-                n = np.first;
-                pos = np.second;
-                if (n >= max_n) break;
+            pool_.run([&](int r) {
+                Scratch& s = scratch_[r];
+                s.offs.resize(quota);
+                s.lens.resize(quota);
+                s.cids.resize(quota);
+                s.n = 0;
+                const long base = (long)r * seg_bytes;
+                long pos = 0;
+                while (s.n < quota) {
+                    PendingReq req;
+                    if (!reactors_[r]->pop_ready_fit(
+                            &req, seg_bytes - pos))
+                        break;
+                    memcpy(buf + base + pos, req.bytes.data(),
+                           req.bytes.size());
+                    s.offs[s.n] = base + pos;
+                    s.lens[s.n] = (int32_t)req.bytes.size();
+                    s.cids[s.n] = req.conn_id;
+                    pos += (long)req.bytes.size();
+                    ++s.n;
+                }
+                s.used = pos;
+            }, R);
+            for (int r = 0; r < R; ++r) {
+                Scratch& s = scratch_[r];
+                if (!s.n) continue;
+                memcpy(offs + n, s.offs.data(), sizeof(int64_t) * s.n);
+                memcpy(lens + n, s.lens.data(), sizeof(int32_t) * s.n);
+                memcpy(cids + n, s.cids.data(), sizeof(uint64_t) * s.n);
+                n += s.n;
+                const long end = (long)r * seg_bytes + s.used;
+                if (end > nbytes) nbytes = end;
             }
             if (n > 0 || std::chrono::steady_clock::now() >= deadline)
                 break;
             std::this_thread::sleep_for(std::chrono::microseconds(50));
         }
-        return {n, pos};
+        return {n, nbytes};
     }
 
     // Write responses back (engine egress pinned buffer), routed to the
-    // owning reactor by the conn-id's reactor index.
+    // owning reactor by the conn-id's reactor index; the per-reactor
+    // egress work (index lookup, close-scan, wbuf append) fans out
+    // across the worker pool.
     void send(uintptr_t conn_ptr, int n, uintptr_t out_ptr,
               uintptr_t roff_ptr, uintptr_t rlen_ptr) {
         py::gil_scoped_release rel;
@@ -465,17 +607,19 @@ public:
         const uint8_t* out = (const uint8_t*)out_ptr;
         const int32_t* roffs = (const int32_t*)roff_ptr;
         const int32_t* rlens = (const int32_t*)rlen_ptr;
-        std::vector<std::vector<int>> by_reactor(reactors_.size());
+        const int R = (int)reactors_.size();
+        std::vector<std::vector<int>> by_reactor(R);
         for (int i = 0; i < n; ++i) {
             const int r = id_reactor(cids[i]);
-            if (r >= 0 && r < (int)reactors_.size())
+            if (r >= 0 && r < R && rlens[i] > 0)
                 by_reactor[r].push_back(i);
         }
-        for (size_t r = 0; r < reactors_.size(); ++r)
+        pool_.run([&](int r) {
             if (!by_reactor[r].empty())
                 reactors_[r]->queue_writes(cids, by_reactor[r].data(),
                                            (int)by_reactor[r].size(),
                                            out, roffs, rlens);
+        }, R);
     }
 
     long ready_count() {
@@ -557,11 +701,21 @@ public:
     }
 
 private:
+    struct Scratch {
+        std::vector<int64_t> offs;
+        std::vector<int32_t> lens;
+        std::vector<uint64_t> cids;
+        int n = 0;
+        long used = 0;
+    };
+
     int port_;
     size_t max_req_;
     int nthreads_;
     std::deque<PendingReq> backlog_;
     std::vector<std::unique_ptr<Reactor>> reactors_;
+    WorkerPool pool_;
+    std::vector<Scratch> scratch_;
 };
 
 }  // namespace
