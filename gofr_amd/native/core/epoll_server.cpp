@@ -188,6 +188,23 @@ public:
         return true;
     }
 
+    // pop up to `quota` ready requests fitting in `room` bytes under
+    // ONE lock acquisition (the per-request lock was the harvest cost)
+    int pop_ready_batch(std::vector<PendingReq>* out, int quota,
+                        long room) {
+        std::lock_guard<std::mutex> lk(mu_);
+        int k = 0;
+        while (k < quota && !ready_.empty()) {
+            const long sz = (long)ready_.front().bytes.size();
+            if (sz > room) break;
+            room -= sz;
+            out->push_back(std::move(ready_.front()));
+            ready_.pop_front();
+            ++k;
+        }
+        return k;
+    }
+
     int port() const { return port_; }
 
 private:
@@ -312,8 +329,13 @@ private:
             req.conn_id = c.id;
             req.bytes = c.rbuf.substr(0, total);
             c.rbuf.erase(0, total);
+            staged_.push_back(std::move(req));
+        }
+        if (!staged_.empty()) {
+            // one lock per readiness event, not per sliced request
             std::lock_guard<std::mutex> lk(mu_);
-            ready_.push_back(std::move(req));
+            for (auto& r : staged_) ready_.push_back(std::move(r));
+            staged_.clear();
         }
         return true;
     }
@@ -390,6 +412,7 @@ private:
     // the reactor flushes).
     std::recursive_mutex wmu_;
     std::deque<PendingReq> ready_;
+    std::vector<PendingReq> staged_;  // event-loop local slice buffer
     std::vector<uint64_t> pending_writes_;
 };
 
@@ -561,14 +584,13 @@ public:
                 s.offs.resize(quota);
                 s.lens.resize(quota);
                 s.cids.resize(quota);
+                s.reqs.clear();
+                reactors_[r]->pop_ready_batch(&s.reqs, quota,
+                                              seg_bytes);
                 s.n = 0;
                 const long base = (long)r * seg_bytes;
                 long pos = 0;
-                while (s.n < quota) {
-                    PendingReq req;
-                    if (!reactors_[r]->pop_ready_fit(
-                            &req, seg_bytes - pos))
-                        break;
+                for (PendingReq& req : s.reqs) {
                     memcpy(buf + base + pos, req.bytes.data(),
                            req.bytes.size());
                     s.offs[s.n] = base + pos;
@@ -705,6 +727,7 @@ private:
         std::vector<int64_t> offs;
         std::vector<int32_t> lens;
         std::vector<uint64_t> cids;
+        std::vector<PendingReq> reqs;
         int n = 0;
         long used = 0;
     };
