@@ -309,6 +309,8 @@ class BatchEngine:
         self.max_bytes = max_batch * slot
         self.pipeline = max(1, pipeline)
         self.program = RouteProgram(app)
+        import threading
+        self._seed_lock = threading.Lock()
         self._seed = 0x6F667247414D4421  # advanced per batch
         # Date header source (Go's net/http attaches Date to every
         # response; parity). Overridable for deterministic tests.
@@ -400,8 +402,10 @@ class BatchEngine:
         self.lanes = []
 
     def _next_seed(self) -> int:
-        self._seed = ops.splitmix64(self._seed)
-        return self._seed
+        # serving threads may submit concurrently on distinct lanes
+        with self._seed_lock:
+            self._seed = ops.splitmix64(self._seed)
+            return self._seed
 
     def _date29(self) -> bytes:
         t = int(self._date_fn())
@@ -710,34 +714,27 @@ class BatchEngine:
         n = ln.n
         if ln.mode == "c" and self._flagged:
             # event-free completion: k_done published the batch serial
-            # into the pinned tables; spin on plain memory
-            cell = 2 * n + 2
+            # into the pinned tables; the spin runs in native code with
+            # the GIL RELEASED (gofr_wait_cell), so concurrent serving
+            # threads overlap their waits
+            base_addr = ln.p_tables.data_ptr() if hasattr(
+                ln.p_tables, "data_ptr") else ln.p_tables_np.ctypes.data
             want = ln.serial_flag
-            tbl = ln.p_tables_np
-            spins = 0
-            t0 = time.perf_counter()
-            while tbl[cell] != want:
-                spins += 1
-                if (spins & 0xFFFF) == 0:
-                    if tbl[2 * n + 3] == want:
-                        # k_gate gave up waiting for this batch's SDMA
-                        # serial: the kernel chain ran on stale ingress
-                        # bytes — never release these responses
-                        raise RuntimeError(
-                            f"ingress gate timeout (serial {want}): "
-                            "SDMA flag never arrived; batch dropped")
-                    if (spins & 0xFFFFF) == 0 and \
-                            time.perf_counter() - t0 > 30:
-                        raise RuntimeError(
-                            "flagged completion timeout (lane serial "
-                            f"{want}, cell {int(tbl[cell])})")
-            if tbl[2 * n + 3] == want:
-                # k_done can land before the spin notices the marker:
-                # re-check after completion so a gate timeout is never
-                # masked by the (stale-input) batch finishing
+            rc = self.hip.lib.gofr_wait_cell(
+                ctypes.c_void_p(base_addr + (2 * n + 2) * 4), want,
+                ctypes.c_void_p(base_addr + (2 * n + 3) * 4),
+                ctypes.c_double(30.0))
+            if rc == 2:
+                # k_gate gave up waiting for this batch's SDMA serial:
+                # the kernel chain ran on stale ingress bytes — never
+                # release these responses
                 raise RuntimeError(
                     f"ingress gate timeout (serial {want}): "
                     "SDMA flag never arrived; batch dropped")
+            if rc == 1:
+                raise RuntimeError(
+                    "flagged completion timeout (lane serial "
+                    f"{want}, cell {int(ln.p_tables_np[2 * n + 2])})")
             if self._use_pump and self.hip.lib.gofr_pump_err():
                 raise RuntimeError(
                     f"pump error: hipError {self.hip.lib.gofr_pump_err()}")
@@ -1011,7 +1008,7 @@ class GPUServer:
         # emit nothing — FL_EMPTY) to a fixed chunk so every batch runs
         # the native flagged pipeline; chunks round-robin over lanes so
         # ingress/kernels/egress of consecutive chunks overlap.
-        from collections import deque
+        import threading
         P = len(eng.lanes)
         CH = min(self.arm_chunk, eng.max_batch)
         cap = CH * eng.slot
@@ -1027,11 +1024,36 @@ class GPUServer:
                 eng.capture_graph(CH, cap, li)
         except Exception:  # noqa: BLE001 — fall back to dynamic batches
             armed = False
-        conn_ids = [np.zeros(CH if armed else eng.max_batch, np.uint64)
-                    for _ in range(P)]
-        len_views = [ln.p_req_len.numpy() for ln in eng.lanes]
-        off_views = [ln.p_req_off.numpy() for ln in eng.lanes]
-        free = deque(range(P))
+        # serving threads: each owns a disjoint lane subset, so the
+        # host-side stages (harvest memcpy fan-out, egress routing,
+        # native submit) of one lane set overlap another's — the single
+        # serve thread's ~280 us/cycle of serial host work was the
+        # socket-path ceiling. All native sections
+        # (harvest/send/submit/wait) run with the GIL released.
+        T = max(1, min(int(os.environ.get("GOFR_SERVE_THREADS", "2")),
+                       P))
+        lane_sets = [list(range(P))[t::T] for t in range(T)]
+        if T == 1:
+            self._serve_lanes(lane_sets[0], armed, CH, cap)
+            return
+        extra = [threading.Thread(target=self._serve_lanes,
+                                  args=(ls, armed, CH, cap),
+                                  daemon=True)
+                 for ls in lane_sets[1:]]
+        for th in extra:
+            th.start()
+        self._serve_lanes(lane_sets[0], armed, CH, cap)
+        for th in extra:
+            th.join(timeout=35)
+
+    def _serve_lanes(self, lanes, armed, CH, cap):
+        from collections import deque
+        eng = self.engine
+        conn_ids = {L: np.zeros(CH if armed else eng.max_batch,
+                                np.uint64) for L in lanes}
+        len_views = {L: eng.lanes[L].p_req_len.numpy() for L in lanes}
+        off_views = {L: eng.lanes[L].p_req_off.numpy() for L in lanes}
+        free = deque(lanes)
         inflight = deque()
         stats = os.environ.get("GOFR_SERVE_STATS") == "1"
         st = {"harvest": 0.0, "submit": 0.0, "complete": 0.0,
@@ -1081,8 +1103,8 @@ class GPUServer:
                 free.append(L)
         if stats and st["cycles"]:
             c = st["cycles"]
-            print(f"[serve-stats] cycles {c} reqs {st['reqs']} "
-                  f"fill {st['reqs']/c:.0f} "
+            print(f"[serve-stats lanes={list(lanes)}] cycles {c} "
+                  f"reqs {st['reqs']} fill {st['reqs']/c:.0f} "
                   f"us/cycle: harvest {st['harvest']/c*1e6:.0f} "
                   f"submit {st['submit']/c*1e6:.0f} "
                   f"complete {st['complete']/c*1e6:.0f} "

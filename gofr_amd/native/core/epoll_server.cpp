@@ -440,53 +440,70 @@ public:
         threads_.clear();
     }
 
+    // Reentrant: several serving threads may run() concurrently (the
+    // two-serve-thread loop overlaps one lane set's harvest with
+    // another's egress); tasks queue globally and any participant may
+    // execute any job's tasks (work conservation, no deadlock).
     void run(const std::function<void(int)>& fn, int ntasks) {
         if (nworkers_ == 0 || ntasks <= 1) {
             for (int i = 0; i < ntasks; ++i) fn(i);
             return;
         }
+        Job job{&fn, ntasks};
         {
             std::lock_guard<std::mutex> lk(mu_);
-            fn_ = &fn;
-            ntasks_ = ntasks;
-            next_.store(0, std::memory_order_relaxed);
-            done_.store(0, std::memory_order_relaxed);
-            ++gen_;
+            for (int i = 0; i < ntasks; ++i) q_.push_back({&job, i});
             cv_.notify_all();
         }
-        // the caller works too
-        help();
+        // the caller works too (on any queued task)
+        while (true) {
+            Task t;
+            {
+                std::lock_guard<std::mutex> lk(mu_);
+                if (q_.empty()) break;
+                t = q_.front();
+                q_.pop_front();
+            }
+            exec(t);
+        }
         std::unique_lock<std::mutex> lk(mu_);
         cv_done_.wait(lk, [&] {
-            return done_.load(std::memory_order_acquire) >= ntasks_;
+            return job.done.load(std::memory_order_acquire) >=
+                   job.ntasks;
         });
-        fn_ = nullptr;
     }
 
 private:
-    void help() {
-        while (true) {
-            const int i = next_.fetch_add(1, std::memory_order_relaxed);
-            if (i >= ntasks_) return;
-            (*fn_)(i);
-            if (done_.fetch_add(1, std::memory_order_acq_rel) + 1 >=
-                ntasks_) {
-                std::lock_guard<std::mutex> lk(mu_);
-                cv_done_.notify_all();
-            }
+    struct Job {
+        const std::function<void(int)>* fn;
+        int ntasks;
+        std::atomic<int> done{0};
+    };
+    struct Task {
+        Job* job = nullptr;
+        int idx = 0;
+    };
+
+    void exec(const Task& t) {
+        (*t.job->fn)(t.idx);
+        if (t.job->done.fetch_add(1, std::memory_order_acq_rel) + 1 >=
+            t.job->ntasks) {
+            std::lock_guard<std::mutex> lk(mu_);
+            cv_done_.notify_all();
         }
     }
 
     void worker() {
-        uint64_t seen = 0;
         while (true) {
+            Task t;
             {
                 std::unique_lock<std::mutex> lk(mu_);
-                cv_.wait(lk, [&] { return !running_ || gen_ != seen; });
-                if (!running_) return;
-                seen = gen_;
+                cv_.wait(lk, [&] { return !running_ || !q_.empty(); });
+                if (!running_ && q_.empty()) return;
+                t = q_.front();
+                q_.pop_front();
             }
-            help();
+            exec(t);
         }
     }
 
@@ -494,10 +511,7 @@ private:
     std::vector<std::thread> threads_;
     std::mutex mu_;
     std::condition_variable cv_, cv_done_;
-    const std::function<void(int)>* fn_ = nullptr;
-    std::atomic<int> next_{0}, done_{0};
-    int ntasks_ = 0;
-    uint64_t gen_ = 0;
+    std::deque<Task> q_;
     bool running_ = false;
 };
 
@@ -520,7 +534,6 @@ public:
         int w = e ? atoi(e) : 7;
         if (w > nthreads_ - 1) w = nthreads_ - 1;
         pool_.start(w);
-        scratch_.resize(reactors_.size());
     }
 
     void stop() {
@@ -578,6 +591,11 @@ public:
             }
         }
         const int quota = max_n / R;
+        // per-serving-thread scratch (two serve threads harvest
+        // concurrently into their own lanes)
+        static thread_local std::vector<Scratch> scratch_tl;
+        std::vector<Scratch>& scratch_ = scratch_tl;
+        scratch_.resize(R);
         while (true) {
             pool_.run([&](int r) {
                 Scratch& s = scratch_[r];
@@ -738,7 +756,6 @@ private:
     std::deque<PendingReq> backlog_;
     std::vector<std::unique_ptr<Reactor>> reactors_;
     WorkerPool pool_;
-    std::vector<Scratch> scratch_;
 };
 
 }  // namespace

@@ -2571,6 +2571,30 @@ extern "C" int gofr_host_free(void* p) {
     return (int)hipHostFree(p);
 }
 
+// Flagged-completion wait OUTSIDE the GIL: spins on the pinned done
+// cell until k_done publishes `want` (0), the gate-timeout cell shows
+// `want` (2 — stale-ingress batch, caller must drop it), or timeout_s
+// elapses (1). ctypes releases the GIL for the call, so several
+// serving threads can wait on their lanes concurrently.
+extern "C" int gofr_wait_cell(const void* cell_p, int want,
+                              const void* tmo_p, double timeout_s) {
+    const volatile int* cell = (const volatile int*)cell_p;
+    const volatile int* tmo = (const volatile int*)tmo_p;
+    const double t0 = now_us();
+    long spins = 0;
+    while (__atomic_load_n((const int*)cell, __ATOMIC_ACQUIRE) != want) {
+        if (tmo &&
+            __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE) == want)
+            return 2;
+        if (((++spins) & 0xFFFF) == 0 &&
+            now_us() - t0 > timeout_s * 1e6)
+            return 1;
+    }
+    if (tmo && __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE) == want)
+        return 2;
+    return 0;
+}
+
 // source-hash stamp: build() compiles with -DGOFR_SRC_HASH="<sha256>"
 // of the committed .hip source, and tests/test_build_hash.py asserts
 // the loaded .so carries the hash of the source in the tree — a stale

@@ -174,6 +174,10 @@ class HipOps:
                                              ctypes.c_uint]
         self.lib.gofr_host_free.restype = ctypes.c_int
         self.lib.gofr_host_free.argtypes = [ctypes.c_void_p]
+        self.lib.gofr_wait_cell.restype = ctypes.c_int
+        self.lib.gofr_wait_cell.argtypes = [
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+            ctypes.c_double]
         try:
             self.lib.gofr_src_hash.restype = ctypes.c_char_p
         except AttributeError:
