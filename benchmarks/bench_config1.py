@@ -123,7 +123,7 @@ def main():
     ap.add_argument("--n-addrs", type=int, default=1,
                     help="spread client conns over 127.0.0.{1..N} "
                          "(needed above ~50k conns)")
-    ap.add_argument("--arm-chunk", type=int, default=1024,
+    ap.add_argument("--arm-chunk", type=int, default=2048,
                     help="armed fixed-batch size of the serving loop")
     args = ap.parse_args()
 

@@ -961,7 +961,7 @@ class GPUServer:
     """
 
     def __init__(self, app, port: int, batch_window_us: int = 200,
-                 max_batch: int = 8192, arm_chunk: int = 1024,
+                 max_batch: int = 16384, arm_chunk: int = 2048,
                  world: int = 1, rank: int = 0, shard_chunk: int = 1024):
         self.app = app
         self.port = port
@@ -989,7 +989,7 @@ class GPUServer:
         # reactor threads: each sustains ~1.5M req/s of socket+parse
         # work (r1 measurement), so the ingress needs several to keep
         # up with the engine's ~27M req/s
-        nreact = int(os.environ.get("GOFR_REACTORS", "8"))
+        nreact = int(os.environ.get("GOFR_REACTORS", "16"))
         self._core = _core.EpollServer(self.port, 1 << 20, nreact)
         self._core.start()
         self.port = self._core.port()

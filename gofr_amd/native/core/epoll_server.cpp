@@ -530,8 +530,10 @@ public:
             port_ = reactors_.back()->bind_and_listen();  // 0 -> learned
         }
         for (auto& r : reactors_) r->run();
+        // 3 workers + the calling thread measured best (more contend
+        // on the reactor ready-queues); see profiles/SUMMARY.md r2
         const char* e = getenv("GOFR_IO_WORKERS");
-        int w = e ? atoi(e) : 7;
+        int w = e ? atoi(e) : 3;
         if (w > nthreads_ - 1) w = nthreads_ - 1;
         pool_.start(w);
     }
