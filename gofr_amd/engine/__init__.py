@@ -149,6 +149,8 @@ class RouteProgram:
                 words.append((ops.TP_JFIELD, koff, len(key), mode))
             else:
                 raise ValueError(f"unknown template piece {piece!r}")
+        if len(words) > 24:  # kernel MAX_TPL_PIECES
+            raise ValueError("template has too many pieces (max 24)")
         while len(blob) % 4:
             blob += b"\0"
         prog_off = len(blob)
@@ -1040,7 +1042,8 @@ class GPUServer:
                                   args=(ls, armed, CH, cap),
                                   daemon=True)
                  for ls in lane_sets[1:]]
-        for th in extra:
+        self._extra_threads = extra  # stop() joins these too before
+        for th in extra:             # tearing the C++ reactors down
             th.start()
         self._serve_lanes(lane_sets[0], armed, CH, cap)
         for th in extra:
@@ -1230,7 +1233,12 @@ class GPUServer:
                 # spin (30 s timeout) — freeing the reactors under it
                 # would be a use-after-free, so wait it out
                 self._thread.join(timeout=35)
-        if self._core is not None and (
-                self._thread is None or not self._thread.is_alive()):
+        threads = [self._thread] + list(
+            getattr(self, "_extra_threads", []))
+        for th in threads:
+            if th is not None and th.is_alive():
+                th.join(timeout=35)
+        if self._core is not None and not any(
+                th is not None and th.is_alive() for th in threads):
             self._core.stop()
         self.engine.close()

@@ -433,8 +433,22 @@ public:
     void stop() {
         {
             std::lock_guard<std::mutex> lk(mu_);
+            if (!running_ && threads_.empty()) return;  // idempotent
             running_ = false;
             cv_.notify_all();
+        }
+        // drain: execute any queued tasks on this thread so a
+        // concurrent run() caller can never wait forever on work the
+        // exiting workers left behind
+        while (true) {
+            Task t;
+            {
+                std::lock_guard<std::mutex> lk(mu_);
+                if (q_.empty()) break;
+                t = q_.front();
+                q_.pop_front();
+            }
+            exec(t);
         }
         for (auto& t : threads_) t.join();
         threads_.clear();

@@ -83,6 +83,7 @@
 #define TM_JESC 2    /* JSON-string-escape the spliced bytes */
 #define TM_JSTR 4    /* TP_JFIELD: splice string CONTENT (strip quotes) */
 #define MAX_JSON_FIELDS 8
+#define MAX_TPL_PIECES 24  /* enforced by engine.RouteProgram */
 
 // methods — must match gofr_amd/http/request.py METHOD_IDS
 // GET POST PUT DELETE PATCH OPTIONS HEAD
@@ -1016,9 +1017,29 @@ __device__ int splice_bytes(uint8_t* dst, const uint8_t* src, int len,
 // relative to `body`; key span excludes quotes; value span is the raw
 // JSON value with surrounding whitespace trimmed). Returns the field
 // count, or 0 when the body is not an object / malformed (callers
-// splice empty). Serial (one lane); bodies are request-sized.
+// splice empty). WAVE function: every lane runs the identical serial
+// state machine in lockstep; the long string/container interiors are
+// skipped 64 bytes per step via ballots (the r1-shape serial byte walk
+// made the ctx.Bind()-class handler 4x slower than echo). `tab` is
+// LDS; writes are lane-0-guarded. Mirror: ops.json_top_fields_py.
 __device__ int json_top_fields(const uint8_t* body, int blen,
-                               int32_t* tab) {
+                               int32_t* tab, int lane) {
+    // whole-chunk skip: returns true when body[i0..i0+63] (clamped)
+    // contains none of the watched bytes, advancing i by the chunk
+    auto chunk_clear = [&](int i0, int end, uint64_t* mask,
+                           bool in_str_scan) -> uint64_t {
+        const int i = i0 + lane;
+        const uint8_t b = (i < end) ? body[i] : 0;
+        bool hit;
+        if (in_str_scan) {
+            hit = (b == '"' || b == '\\') && i < end;
+        } else {
+            hit = (b == '"' || b == '\\' || b == '{' || b == '}' ||
+                   b == '[' || b == ']') && i < end;
+        }
+        *mask = __ballot(hit);
+        return *mask;
+    };
     int i = 0;
     while (i < blen && body[i] <= ' ') ++i;
     if (i >= blen || body[i] != '{') return 0;
@@ -1031,6 +1052,13 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
         if (body[i] != '"') return 0;
         const int ks = ++i;
         while (i < blen && body[i] != '"') {
+            uint64_t m;
+            if (!chunk_clear(i, blen, &m, true)) {
+                i += WAVE;
+                if (i > blen) i = blen;
+                continue;
+            }
+            i += __builtin_ctzll(m);  // first watched byte of the chunk
             if (body[i] == '\\') ++i;
             ++i;
         }
@@ -1046,6 +1074,13 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
         if (body[i] == '"') {
             ++i;
             while (i < blen && body[i] != '"') {
+                uint64_t m;
+                if (!chunk_clear(i, blen, &m, true)) {
+                    i += WAVE;
+                    if (i > blen) i = blen;
+                    continue;
+                }
+                i += __builtin_ctzll(m);
                 if (body[i] == '\\') ++i;
                 ++i;
             }
@@ -1055,6 +1090,13 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
             int depth = 0;
             bool in_str = false;
             while (i < blen) {
+                uint64_t m;
+                if (!chunk_clear(i, blen, &m, in_str)) {
+                    i += WAVE;
+                    if (i > blen) i = blen;
+                    continue;
+                }
+                i += __builtin_ctzll(m);
                 const uint8_t c = body[i];
                 if (in_str) {
                     if (c == '\\') ++i;
@@ -1072,10 +1114,12 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
                    body[i] > ' ')
                 ++i;
         }
-        tab[nf * 4 + 0] = ks;
-        tab[nf * 4 + 1] = ke - ks;
-        tab[nf * 4 + 2] = vs;
-        tab[nf * 4 + 3] = i - vs;
+        if (lane == 0) {
+            tab[nf * 4 + 0] = ks;
+            tab[nf * 4 + 1] = ke - ks;
+            tab[nf * 4 + 2] = vs;
+            tab[nf * 4 + 3] = i - vs;
+        }
         ++nf;
         while (i < blen && body[i] <= ' ') ++i;
         if (i < blen && body[i] == ',') continue;
@@ -1160,7 +1204,7 @@ __device__ __forceinline__ void respond_impl(
         int gzip_min, int etag_on, const uint8_t* date29,
         const int32_t* __restrict__ kv_tab,
         const uint8_t* __restrict__ kv_blob,
-        uint8_t* obuf, uint8_t* plainbuf,
+        uint8_t* obuf, uint8_t* plainbuf, int32_t* jtab,
         uint32_t* hash, const uint32_t* crc_tab, int req, int lane) {
     int32_t* F = fields + (size_t)req * NF;
     uint8_t* out = resp + (size_t)req * rslot;
@@ -1256,55 +1300,85 @@ __device__ __forceinline__ void respond_impl(
     } else if (kind == HK_TEMPLATE && !(flags & FL_AUTH_FAIL)) {
         // template program: blob-encoded pieces spliced with decoded
         // path params, query params and top-level JSON body fields —
-        // the /user/{id}-class routes the r1 engine trampolined.
-        // Lane 0 scans + measures + emits; result is the raw body
-        // (templates carry their own envelope text), env stays 0.
+        // the /user/{id}-class + ctx.Bind()-class routes the r1 engine
+        // trampolined. The whole pass runs WAVE-UNIFORM (every lane
+        // computes identical resolution state in lockstep): the JSON
+        // field scan fast-skips string interiors via ballots, and
+        // splices without transformable bytes (ballot-checked) are
+        // wave-parallel copies; only the rare escape/decode splice
+        // runs serially on lane 0. Result is the raw body (templates
+        // carry their own envelope text), env stays 0.
         const int route = F[FI_ROUTE];
         const int32_t* prog = (const int32_t*)(blob +
                                                handler_tab[route * 4 + 1]);
         uint8_t* tbuf = plainbuf ? plainbuf : (obuf + 1024);
         const int tcap = MAX_SLOT - 1024;
-        int total = -1;
-        if (lane == 0) {
-            int32_t jtab[MAX_JSON_FIELDS * 4];
-            int jn = -1;  // lazy: scan only when a piece needs it
-            const int np = prog[0];
-            total = 0;
-            for (int p = 0; p < np; ++p) {
-                const int32_t* pc = prog + 1 + p * 4;
-                if (pc[0] == TP_JFIELD && jn < 0)
-                    jn = json_top_fields(rbase + F[FI_BODY_OFF],
-                                         F[FI_BODY_LEN], jtab);
-                const uint8_t* src;
-                int slen, mode;
-                if (piece_src(pc, blob, rbase, F, jtab, jn < 0 ? 0 : jn,
-                              &src, &slen, &mode))
-                    total += splice_bytes(nullptr, src, slen, mode);
+        const int np0 = prog[0];
+        const int np = np0 < MAX_TPL_PIECES ? np0 : MAX_TPL_PIECES;
+        const uint8_t* srcs[MAX_TPL_PIECES];
+        int slens[MAX_TPL_PIECES], pmodes[MAX_TPL_PIECES];
+        int olens[MAX_TPL_PIECES];
+        int jn = -1;  // lazy: scan only when a piece needs it
+        int total = 0;
+        const uint8_t* body = rbase + F[FI_BODY_OFF];
+        for (int p = 0; p < np; ++p) {
+            const int32_t* pc = prog + 1 + p * 4;
+            if (pc[0] == TP_JFIELD && jn < 0) {
+                jn = json_top_fields(body, F[FI_BODY_LEN], jtab, lane);
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             }
-            if (total > tcap) {
-                total = -1;  // overflow guard: render the 500 envelope
-            } else {
-                int w = 0;
-                for (int p = 0; p < np; ++p) {
-                    const int32_t* pc = prog + 1 + p * 4;
-                    if (pc[0] == TP_JFIELD && jn < 0)
-                        jn = 0;  // unreachable (measure pass scanned)
-                    const uint8_t* src;
-                    int slen, mode;
-                    if (piece_src(pc, blob, rbase, F, jtab,
-                                  jn < 0 ? 0 : jn, &src, &slen, &mode))
-                        w += splice_bytes(tbuf + w, src, slen, mode);
+            const uint8_t* src;
+            int slen, mode;
+            if (!piece_src(pc, blob, rbase, F, jtab, jn < 0 ? 0 : jn,
+                           &src, &slen, &mode)) {
+                srcs[p] = nullptr;
+                slens[p] = olens[p] = pmodes[p] = 0;
+                continue;
+            }
+            int ol = slen;
+            if (mode & (TM_PCT | TM_JESC)) {
+                // ballot: does any byte actually need the transform?
+                bool hot = false;
+                for (int c0 = 0; c0 < slen; c0 += WAVE) {
+                    const int i = c0 + lane;
+                    const uint8_t b = (i < slen) ? src[i] : 'a';
+                    const bool h =
+                        ((mode & TM_PCT) && (b == '%' || b == '+')) ||
+                        ((mode & TM_JESC) &&
+                         (b == '"' || b == '\\' || b < 0x20));
+                    if (__ballot(h)) { hot = true; break; }
                 }
+                if (!hot)
+                    mode &= ~(TM_PCT | TM_JESC);  // plain wave copy
+                else
+                    ol = splice_bytes(nullptr, src, slen, mode);
             }
+            srcs[p] = src;
+            slens[p] = slen;
+            pmodes[p] = mode;
+            olens[p] = ol;
+            total += ol;
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        total = __shfl(total, 0);
-        if (total < 0) {
-            status = 500;
+        if (total > tcap) {
+            status = 500;  // overflow guard: render the 500 envelope
             const int elen = *(const int32_t*)blob;
             body_src = blob + 4;
             body_src_len = elen;
         } else {
+            int w = 0;
+            for (int p = 0; p < np; ++p) {
+                if (!srcs[p]) continue;
+                const int mode = pmodes[p];
+                if (!(mode & (TM_PCT | TM_JESC))) {
+                    const uint8_t* src = srcs[p];
+                    for (int i = lane; i < slens[p]; i += WAVE)
+                        tbuf[w + i] = src[i];
+                } else if (lane == 0) {
+                    splice_bytes(tbuf + w, srcs[p], slens[p], mode);
+                }
+                w += olens[p];
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             body_src = tbuf;
             body_src_len = total;
         }
@@ -1596,16 +1670,20 @@ k_respond(const uint8_t* __restrict__ reqs,
           const uint8_t* __restrict__ date29,
           const int32_t* __restrict__ kv_tab,
           const uint8_t* __restrict__ kv_blob) {
-    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT];
+    // obuf per wave + jtab (JSON field table) per wave in ONE block
+    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT +
+                                WAVES_PER_BLOCK * MAX_JSON_FIELDS * 16];
     const int wv = threadIdx.x / WAVE;
     const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
     if (req >= n) return;
+    int32_t* jtab = (int32_t*)(obuf_all + WAVES_PER_BLOCK * MAX_SLOT) +
+                    wv * MAX_JSON_FIELDS * 4;
     respond_impl<false>(reqs, req_off, fields, resp, resp_len_out, n, rslot,
                         handler_tab, n_routes, blob, host_blob, host_tab,
                         *seed_ptr, auth_env_off, auth_env_len, 0, etag_on,
                         date29, kv_tab, kv_blob,
-                        obuf_all + wv * MAX_SLOT, nullptr, nullptr, nullptr,
-                        req, lane_id());
+                        obuf_all + wv * MAX_SLOT, nullptr, jtab, nullptr,
+                        nullptr, req, lane_id());
 }
 
 extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
@@ -1626,12 +1704,16 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
              const uint8_t* __restrict__ kv_blob) {
     // single __shared__ block (cdna guide §5 trap 4a)
     __shared__ uint8_t lds[WAVES_PER_BLOCK * MAX_SLOT * 2 +
-                           WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4];
+                           WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 + 256 * 4 +
+                           WAVES_PER_BLOCK * MAX_JSON_FIELDS * 16];
     uint8_t* obuf_all = lds;
     uint8_t* plain_all = lds + WAVES_PER_BLOCK * MAX_SLOT;
     uint32_t* hash_all = (uint32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2);
     uint32_t* crc_tab = (uint32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2 +
                                     WAVES_PER_BLOCK * GZ_HASH_SIZE * 4);
+    int32_t* jtab_all = (int32_t*)(lds + WAVES_PER_BLOCK * MAX_SLOT * 2 +
+                                   WAVES_PER_BLOCK * GZ_HASH_SIZE * 4 +
+                                   256 * 4);
     // build the CRC32 table cooperatively BEFORE any thread can exit
     for (int i = threadIdx.x; i < 256; i += BLOCK_THREADS) {
         uint32_t c = (uint32_t)i;
@@ -1648,6 +1730,7 @@ k_respond_gz(const uint8_t* __restrict__ reqs,
                        *seed_ptr, auth_env_off, auth_env_len, gzip_min,
                        etag_on, date29, kv_tab, kv_blob,
                        obuf_all + wv * MAX_SLOT, plain_all + wv * MAX_SLOT,
+                       jtab_all + wv * MAX_JSON_FIELDS * 4,
                        hash_all + wv * GZ_HASH_SIZE, crc_tab,
                        req, lane_id());
 }
