@@ -1059,6 +1059,7 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
                 continue;
             }
             i += __builtin_ctzll(m);  // first watched byte of the chunk
+            if (body[i] == '"') break;  // loop-exit char: don't consume
             if (body[i] == '\\') ++i;
             ++i;
         }
@@ -1081,6 +1082,7 @@ __device__ int json_top_fields(const uint8_t* body, int blen,
                     continue;
                 }
                 i += __builtin_ctzll(m);
+                if (body[i] == '"') break;  // closing quote
                 if (body[i] == '\\') ++i;
                 ++i;
             }
