@@ -240,6 +240,9 @@ class GRPCServer:
                 self._paths.append(
                     (desc, impl, meth,
                      (sname, meth) in self._gpu_methods))
+        import numpy as _np
+        self._gpu_mask = _np.array([p[3] for p in self._paths],
+                                   dtype=bool)
         core.set_grpc_blocks(
             h2.HpackEncoder.encode([(":status", "200"),
                                     ("content-type",
@@ -316,13 +319,13 @@ class GRPCServer:
             self.codec_batches += 1
             self.codec_msgs += n
             rlens[:n] = -1  # default: error trailers
-            gpu_rows = [i for i in range(n)
-                        if 0 <= pids[i] < len(self._paths)
-                        and self._paths[pids[i]][3]]
-            host_rows = [i for i in range(n)
-                         if 0 <= pids[i] < len(self._paths)
-                         and not self._paths[pids[i]][3]]
-            if gpu_rows:
+            pids_n = pids[:n]
+            valid = pids_n >= 0
+            gmask = np.zeros(n, bool)
+            gmask[valid] = self._gpu_mask[pids_n[valid]]
+            gpu_rows = np.nonzero(gmask)[0]
+            host_rows = np.nonzero(valid & ~gmask)[0]
+            if len(gpu_rows):
                 go = offs[gpu_rows]
                 gl = lens[gpu_rows]
                 m = len(gpu_rows)
@@ -355,13 +358,14 @@ class GRPCServer:
                     spans, span_n = ops.cpu_varint_spans(buf, go, gl)
                     gout, gout_len = ops.cpu_grpc_echo(buf, spans,
                                                        span_n, GR)
-                for k, i in enumerate(gpu_rows):
-                    ln = int(gout_len[k])
-                    if ln > 0:
-                        out[i * GR:i * GR + ln] = \
-                            gout[k * GR:k * GR + ln]
-                        roffs[i] = i * GR
-                        rlens[i] = ln
+                # vectorized row scatter (a per-row Python loop cost
+                # ~2 us x batch and capped the wire at ~300k msg/s)
+                ok = gout_len[:m] > 0
+                gi = gpu_rows[ok]
+                out[:MB * GR].reshape(MB, GR)[gi] = \
+                    gout[:m * GR].reshape(m, GR)[ok]
+                roffs[gi] = (gi * GR).astype(np.int32)
+                rlens[gi] = gout_len[:m][ok]
             host_pos = MB * GR
             for i in host_rows:
                 desc, impl, meth, _ = self._paths[int(pids[i])]

@@ -730,6 +730,17 @@ public:
         return true;
     }
 
+    int pop_grpc_batch(std::vector<PendingGrpc>* out, int quota) {
+        std::lock_guard<std::mutex> lk(mu_);
+        int k = 0;
+        while (k < quota && !gready_.empty()) {
+            out->push_back(std::move(gready_.front()));
+            gready_.pop_front();
+            ++k;
+        }
+        return k;
+    }
+
     // queue gRPC responses: HEADERS + DATA + trailers per row (rlen<0:
     // error -> headers + error trailers, no DATA)
     void queue_grpc_writes(const uint64_t* cids, const uint32_t* sids,
@@ -1152,10 +1163,12 @@ public:
                               std::chrono::microseconds(window_us);
         int n = 0;
         long pos = 0;
+        std::vector<PendingGrpc> batch;
         while (true) {
             for (auto& r : reactors_) {
-                PendingGrpc g;
-                while (n < max_n && r->pop_grpc(&g)) {
+                batch.clear();
+                r->pop_grpc_batch(&batch, max_n - n);
+                for (PendingGrpc& g : batch) {
                     if (pos + (long)g.msg.size() > buf_cap) break;
                     memcpy(buf + pos, g.msg.data(), g.msg.size());
                     offs[n] = pos;
@@ -1166,6 +1179,7 @@ public:
                     pos += (long)g.msg.size();
                     ++n;
                 }
+                if (n >= max_n) break;
             }
             if (n > 0 || std::chrono::steady_clock::now() >= deadline)
                 break;
