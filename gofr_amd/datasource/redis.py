@@ -157,6 +157,13 @@ class Redis:
     def Ping(self):
         return self.execute("PING")
 
+    def MGet(self, keys):
+        """One MGET round trip for a whole key batch (the engine's
+        batch-trampoline path issues one of these per GPU batch)."""
+        if not keys:
+            return []
+        return self.execute("MGET", *keys)
+
     def Get(self, key: str):
         return self.execute("GET", key)
 

@@ -921,10 +921,23 @@ class BatchEngine:
 
     # -- host trampoline ------------------------------------------------------
     def _run_host_rows(self, fields, reqs, offs, lens):
-        """Run Python handlers for HK_HOST rows; returns (blob, tab)."""
+        """Run Python handlers for HK_HOST rows; returns (blob, tab).
+
+        Handlers exposing a __gofr_batch__ hook (e.g. the redis-backed
+        handlers.redis_json) are grouped by route and served with ONE
+        call over the whole row set, so a datasource-touching batch
+        costs one pipelined round trip instead of a command per
+        request."""
         n = len(lens)
         host_tab = np.zeros((n, 4), np.int32)
         blob = bytearray()
+        batch_groups: dict[int, list] = {}
+
+        def emit(r, status, body, ct):
+            off = len(blob)
+            blob.extend(body)
+            host_tab[r] = (off, len(body), status, _CT_IDS.get(ct, 0))
+
         for r in range(n):
             if fields[r][ops.FI_KIND] != ops.HK_HOST:
                 continue
@@ -934,6 +947,24 @@ class BatchEngine:
             raw = np.asarray(reqs[o:o + int(lens[r])]).tobytes()
             try:
                 request = parse_request_bytes(raw)
+            except (ValueError, KeyError):
+                emit(r, 400, b'{"error":{"message":"malformed request"}}',
+                     "application/json")
+                continue
+            route_id = int(fields[r][ops.FI_ROUTE])
+            handler = (self.program.py_handlers[route_id]
+                       if 0 <= route_id < len(self.program.py_handlers)
+                       else None)
+            bfn = getattr(handler, "__gofr_batch__", None)
+            if bfn is not None:
+                route, params, status = self.app.router.match(
+                    request.method, request.path)
+                if route is not None and status == 200:
+                    request.path_params = params
+                    batch_groups.setdefault(route_id,
+                                            []).append((r, request))
+                    continue
+            try:
                 resp = dispatch(self.app, request)
                 status, body = resp.status, resp.body
                 ct = dict(resp.headers).get("Content-Type",
@@ -942,9 +973,21 @@ class BatchEngine:
                 status = 400
                 body = b'{"error":{"message":"malformed request"}}'
                 ct = "application/json"
-            off = len(blob)
-            blob += body
-            host_tab[r] = (off, len(body), status, _CT_IDS.get(ct, 0))
+            emit(r, status, body, ct)
+        for route_id, group in batch_groups.items():
+            bfn = self.program.py_handlers[route_id].__gofr_batch__
+            requests = [req for _, req in group]
+            try:
+                results = bfn(self.app, requests)
+                assert len(results) == len(group)
+            except Exception as e:  # noqa: BLE001 — recovery to 500s
+                import json as _json
+                msg = _json.dumps({"error": {"message":
+                                             str(e) or "batch failed"}},
+                                  separators=(",", ":")).encode()
+                results = [(500, msg, "application/json")] * len(group)
+            for (r, _), (status, body, ct) in zip(group, results):
+                emit(r, status, body, ct)
         return bytes(blob), host_tab
 
 
