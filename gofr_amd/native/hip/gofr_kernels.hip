@@ -214,7 +214,12 @@ k_parse_route(uint8_t* __restrict__ reqs,
     const int nchunks = (len + WAVE - 1) / WAVE;
     bool saw_percent = false;
     // structural classification: 64 bytes per wave-iteration, one ballot
-    // per class — no per-byte branching (SURVEY.md §7 hard part 2)
+    // per class — no per-byte branching (SURVEY.md §7 hard part 2).
+    // The pass STOPS at the chunk containing the header/body boundary
+    // (\r\n\r\n): for the 1 KB-echo shape that is ~2 chunks instead of
+    // 17 — the body's bytes never need classes (r1 scanned them all;
+    // k_parse_route was 536 us/batch, VERDICT target < 200).
+    int nclass = nchunks;
     for (int c = 0; c < nchunks; ++c) {
         const int i = c * WAVE + lane;
         const uint8_t b = (i < len) ? base[i] : 0;
@@ -226,6 +231,14 @@ k_parse_route(uint8_t* __restrict__ reqs,
             masks[wv][CLS_LF][c] = m_lf;
             masks[wv][CLS_SP][c] = m_sp;
             masks[wv][CLS_QM][c] = m_qm;
+        }
+        const bool hdr_end = i >= 3 && i < len && b == '\n' &&
+                             base[i - 1] == '\r' &&
+                             base[i - 2] == '\n' &&
+                             base[i - 3] == '\r';
+        if (__ballot(hdr_end)) {
+            nclass = c + 1;
+            break;
         }
     }
     const bool any_percent = __ballot(saw_percent) != 0;
@@ -240,7 +253,7 @@ k_parse_route(uint8_t* __restrict__ reqs,
     // r1 weak #3). Body LFs past MAX_LFS don't matter: header lines
     // precede the body, and only lines 1..63 are ever examined.
     {
-        uint64_t lm = (lane < nchunks) ? masks[wv][CLS_LF][lane] : 0;
+        uint64_t lm = (lane < nclass) ? masks[wv][CLS_LF][lane] : 0;
         const int cnt = __popcll(lm);
         int incl = cnt;
         for (int off = 1; off < WAVE; off <<= 1) {
@@ -255,8 +268,8 @@ k_parse_route(uint8_t* __restrict__ reqs,
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const int nlf = [&] {
-        int total = (lane < nchunks) ? __popcll(masks[wv][CLS_LF][lane])
-                                     : 0;
+        int total = (lane < nclass) ? __popcll(masks[wv][CLS_LF][lane])
+                                    : 0;
         for (int off = 32; off; off >>= 1)
             total += __shfl_xor(total, off);
         return total;
@@ -268,11 +281,11 @@ k_parse_route(uint8_t* __restrict__ reqs,
     // header pass below; memory writes stay lane-0-guarded) ----------------
     auto next_bit = [&](int cls, int from) -> int {
         int c = from / WAVE;
-        if (c >= nchunks) return -1;
+        if (c >= nclass) return -1;
         uint64_t m = masks[wv][cls][c] & (~0ull << (from & (WAVE - 1)));
         while (true) {
             if (m) return c * WAVE + __builtin_ctzll(m);
-            if (++c >= nchunks) return -1;
+            if (++c >= nclass) return -1;
             m = masks[wv][cls][c];
         }
     };

@@ -431,11 +431,15 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
         # mirror of the kernel's one-line-per-lane header pass: LF
         # position table, first blank line among lines 1..63, then the
         # header lines before it (last occurrence of a repeated header
-        # wins — the reductions are max-by-line-start)
+        # wins — the reductions are max-by-line-start). The kernel's
+        # classification stops at the 64B chunk containing the first
+        # \r\n\r\n, so the LF table is chunk-granular-truncated there.
+        bnd = buf.find(b"\r\n\r\n")
+        region_end = ln if bnd < 0 else min(ln, ((bnd + 3) // 64 + 1) * 64)
         lfs = []
         p = -1
         while True:
-            p = buf.find(b"\n", p + 1)
+            p = buf.find(b"\n", p + 1, region_end)
             if p < 0:
                 break
             lfs.append(p)
