@@ -172,23 +172,20 @@ __device__ __forceinline__ int pct_decode_inplace(uint8_t* s, int len) {
     return w;
 }
 
-extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
-k_parse_route(uint8_t* __restrict__ reqs,
-              const int64_t* __restrict__ req_off,
-              const int32_t* __restrict__ req_len,
-              int32_t* __restrict__ fields,
-              int n,
-              TrieDev trie,
-              const int32_t* __restrict__ handler_tab, int n_routes,
-              int32_t* __restrict__ host_needed) {
-    __shared__ uint64_t masks[WAVES_PER_BLOCK][N_CLS][MAX_CHUNKS];
-    __shared__ int lf_pos_all[WAVES_PER_BLOCK][MAX_LFS];
-
-    const int wv = threadIdx.x / WAVE;
-    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
-    if (req >= n) return;
-    const int lane = lane_id();
-    int* lf_pos = lf_pos_all[wv];
+// parse one request (one wave): callable from k_parse_route AND the
+// persistent serving kernel. `masks` / `lf_pos` are this wave's LDS
+// slices.
+__device__ void parse_one(uint8_t* __restrict__ reqs,
+                          const int64_t* __restrict__ req_off,
+                          const int32_t* __restrict__ req_len,
+                          int32_t* __restrict__ fields,
+                          int n,
+                          const TrieDev& trie,
+                          const int32_t* __restrict__ handler_tab,
+                          int n_routes,
+                          int32_t* __restrict__ host_needed,
+                          uint64_t (*masks)[MAX_CHUNKS], int* lf_pos,
+                          int req, int lane) {
     uint8_t* base = reqs + req_off[req];
     int len = req_len[req];
     if (len == 0) {
@@ -228,9 +225,9 @@ k_parse_route(uint8_t* __restrict__ reqs,
         const uint64_t m_qm = __ballot(b == '?');
         saw_percent |= (b == '%');
         if (lane == 0) {
-            masks[wv][CLS_LF][c] = m_lf;
-            masks[wv][CLS_SP][c] = m_sp;
-            masks[wv][CLS_QM][c] = m_qm;
+            masks[CLS_LF][c] = m_lf;
+            masks[CLS_SP][c] = m_sp;
+            masks[CLS_QM][c] = m_qm;
         }
         const bool hdr_end = i >= 3 && i < len && b == '\n' &&
                              base[i - 1] == '\r' &&
@@ -253,7 +250,7 @@ k_parse_route(uint8_t* __restrict__ reqs,
     // r1 weak #3). Body LFs past MAX_LFS don't matter: header lines
     // precede the body, and only lines 1..63 are ever examined.
     {
-        uint64_t lm = (lane < nclass) ? masks[wv][CLS_LF][lane] : 0;
+        uint64_t lm = (lane < nclass) ? masks[CLS_LF][lane] : 0;
         const int cnt = __popcll(lm);
         int incl = cnt;
         for (int off = 1; off < WAVE; off <<= 1) {
@@ -268,7 +265,7 @@ k_parse_route(uint8_t* __restrict__ reqs,
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const int nlf = [&] {
-        int total = (lane < nclass) ? __popcll(masks[wv][CLS_LF][lane])
+        int total = (lane < nclass) ? __popcll(masks[CLS_LF][lane])
                                     : 0;
         for (int off = 32; off; off >>= 1)
             total += __shfl_xor(total, off);
@@ -282,11 +279,11 @@ k_parse_route(uint8_t* __restrict__ reqs,
     auto next_bit = [&](int cls, int from) -> int {
         int c = from / WAVE;
         if (c >= nclass) return -1;
-        uint64_t m = masks[wv][cls][c] & (~0ull << (from & (WAVE - 1)));
+        uint64_t m = masks[cls][c] & (~0ull << (from & (WAVE - 1)));
         while (true) {
             if (m) return c * WAVE + __builtin_ctzll(m);
             if (++c >= nclass) return -1;
-            m = masks[wv][cls][c];
+            m = masks[cls][c];
         }
     };
 
@@ -563,6 +560,25 @@ k_parse_route(uint8_t* __restrict__ reqs,
             if (kind == HK_HOST) atomicAdd(host_needed, 1);
         }
     }
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_parse_route(uint8_t* __restrict__ reqs,
+              const int64_t* __restrict__ req_off,
+              const int32_t* __restrict__ req_len,
+              int32_t* __restrict__ fields,
+              int n,
+              TrieDev trie,
+              const int32_t* __restrict__ handler_tab, int n_routes,
+              int32_t* __restrict__ host_needed) {
+    __shared__ uint64_t masks_all[WAVES_PER_BLOCK][N_CLS][MAX_CHUNKS];
+    __shared__ int lf_pos_all[WAVES_PER_BLOCK][MAX_LFS];
+    const int wv = threadIdx.x / WAVE;
+    const int req = blockIdx.x * WAVES_PER_BLOCK + wv;
+    if (req >= n) return;
+    parse_one(reqs, req_off, req_len, fields, n, trie, handler_tab,
+              n_routes, host_needed, masks_all[wv], lf_pos_all[wv],
+              req, lane_id());
 }
 
 // ---------------------------------------------------------------------------
@@ -1854,14 +1870,9 @@ __device__ __forceinline__ int hexval(uint8_t c) {
     return -1;
 }
 
-extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
-k_auth(const uint8_t* __restrict__ reqs,
-       const int64_t* __restrict__ req_off,
-       int32_t* __restrict__ fields,
-       int n,
-       const uint8_t* __restrict__ secret, int secret_len) {
-    const int req = blockIdx.x * BLOCK_THREADS + threadIdx.x;
-    if (req >= n) return;
+__device__ void auth_one(const uint8_t* reqs, const int64_t* req_off,
+                         int32_t* fields, int req,
+                         const uint8_t* secret, int secret_len) {
     int32_t* F = fields + (size_t)req * NF;
     const int flags = F[FI_FLAGS];
     if (flags & (FL_ERR_PARSE | FL_IS_OPTIONS | FL_EMPTY)) return;
@@ -1904,6 +1915,17 @@ k_auth(const uint8_t* __restrict__ reqs,
         F[FI_FLAGS] = flags | FL_AUTH_FAIL;
         F[FI_STATUS] = 401;
     }
+}
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_auth(const uint8_t* __restrict__ reqs,
+       const int64_t* __restrict__ req_off,
+       int32_t* __restrict__ fields,
+       int n,
+       const uint8_t* __restrict__ secret, int secret_len) {
+    const int req = blockIdx.x * BLOCK_THREADS + threadIdx.x;
+    if (req >= n) return;
+    auth_one(reqs, req_off, fields, req, secret, secret_len);
 }
 
 // ---------------------------------------------------------------------------
@@ -2212,6 +2234,278 @@ k_persist_cycle(const unsigned long long* __restrict__ d_go,
             __hip_atomic_store(p_done, (unsigned long long)b,
                                __ATOMIC_RELEASE,
                                __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// k_persist_serve — the PERSISTENT serving engine (round-2 headline
+// path; derisked by benchmarks/persistent_probe.py in r1). One
+// resident kernel replaces the per-batch launch chain and its
+// stream-handoff dead time (~0.45 ms/batch measured):
+//
+//   compute crew (blocks egress_blocks..G-1): per batch — block 0
+//     spins (bounded) on the go serial the host's ingress SDMA writes
+//     after the payload (same FIFO), crew barrier, parse+respond fused
+//     per wave over a block-strided request range (auth phase between
+//     when a secret is armed), crew barrier, block 0 runs the pad16
+//     scan + pinned table mirror, then publishes compute_done.
+//   egress crew (blocks 0..egress_blocks-1): waits compute_done,
+//     streams the compact response stream STRAIGHT into the pinned
+//     egress ring (the link-bound work), then publishes the batch
+//     serial to the pinned done cell (system release).
+//
+// Batches alternate 2 slots, so the SDMA ingress of batch b+1 and the
+// compute of batch b and the egress of batch b-1 all overlap; slot
+// reuse is safe because the host never writes go[b+2] before it
+// completed b. Every spin is bounded and latches a timeout marker
+// (pinned cell [2n+3]) — the kernel ALWAYS terminates after nbatch
+// batches. Grid size must be fully co-resident: the host launcher
+// sizes it from hipOccupancyMaxActiveBlocksPerMultiprocessor.
+// ---------------------------------------------------------------------------
+
+typedef struct {
+    unsigned long long first;   // first batch serial of this launch
+    int nbatch;
+    int n, rslot;
+    long long hdr_bytes;        // ingress header size (offs/lens/seed/date)
+    int date_off;
+    int egress_blocks;
+    // per-slot device buffers
+    void* d_ingress[2];
+    void* d_fields[2];
+    void* d_resp[2];
+    void* d_tables[2];
+    void* p_tables[2];          // pinned int32[2n+4]
+    void* p_out[2];
+    void* host_blob[2];
+    void* host_tab[2];
+    // shared
+    const void* trie[9];
+    const void* handler_tab;
+    int n_routes;
+    const void* blob;
+    const void* d_kv_tab;
+    const void* d_kv_blob;
+    const void* secret;
+    int secret_len;
+    int auth_env_off, auth_env_len, etag_on;
+    void* d_state;              // u64[4]: [0] go, [1] compute_done, [3] tmo latch
+    void* d_barrier;            // int[8]: cbar cnt/gen, ebar cnt/gen, hn[2]
+} PersistKernArgs;
+
+// 256-thread pad16 exclusive scan (k_padscan adapted to one crew block)
+__device__ void scan_tables_crew(int32_t* tables, int hn, int n,
+                                 int32_t* mirror) {
+    __shared__ int32_t partials[BLOCK_THREADS];
+    const int tid = threadIdx.x;
+    const int chunk = (n + BLOCK_THREADS - 1) / BLOCK_THREADS;
+    const int s = tid * chunk;
+    int e = s + chunk;
+    if (e > n) e = n;
+    int32_t sum = 0;
+    for (int i = s; i < e; ++i) sum += (tables[i] + 15) & ~15;
+    partials[tid] = sum;
+    __syncthreads();
+    for (int off = 1; off < BLOCK_THREADS; off <<= 1) {
+        int32_t v = (tid >= off) ? partials[tid - off] : 0;
+        __syncthreads();
+        partials[tid] += v;
+        __syncthreads();
+    }
+    int32_t run = (tid > 0) ? partials[tid - 1] : 0;
+    for (int i = s; i < e; ++i) {
+        const int32_t len_i = tables[i];
+        tables[n + i] = run;
+        mirror[i] = len_i;
+        mirror[n + i] = run;
+        run += (len_i + 15) & ~15;
+    }
+    if (tid == BLOCK_THREADS - 1) {
+        tables[2 * n] = partials[tid];
+        mirror[2 * n] = partials[tid];
+    }
+    if (tid == 0) {
+        tables[2 * n + 1] = hn;
+        mirror[2 * n + 1] = hn;
+    }
+}
+
+// k_auth body as a per-thread device function (one request per lane)
+__device__ void auth_one(const uint8_t* reqs, const int64_t* req_off,
+                         int32_t* fields, int req,
+                         const uint8_t* secret, int secret_len);
+
+extern "C" __global__ void __launch_bounds__(BLOCK_THREADS)
+k_persist_serve(PersistKernArgs a) {
+    __shared__ uint64_t masks_all[WAVES_PER_BLOCK][N_CLS][MAX_CHUNKS];
+    __shared__ int lf_pos_all[WAVES_PER_BLOCK][MAX_LFS];
+    __shared__ uint8_t obuf_all[WAVES_PER_BLOCK * MAX_SLOT +
+                                WAVES_PER_BLOCK * MAX_JSON_FIELDS * 16];
+    const int lane = lane_id();
+    const int wv = threadIdx.x / WAVE;
+    int32_t* jtab = (int32_t*)(obuf_all + WAVES_PER_BLOCK * MAX_SLOT) +
+                    wv * MAX_JSON_FIELDS * 4;
+    uint8_t* obuf = obuf_all + wv * MAX_SLOT;
+    const bool is_egress = (int)blockIdx.x < a.egress_blocks;
+    const int crew_rank = is_egress ? blockIdx.x
+                                    : blockIdx.x - a.egress_blocks;
+    const int crew_n = is_egress ? a.egress_blocks
+                                 : gridDim.x - a.egress_blocks;
+    unsigned long long* go = (unsigned long long*)a.d_state;
+    unsigned long long* cdone = go + 1;
+    unsigned long long* tmo_latch = go + 3;
+    int* cbar = (int*)a.d_barrier;
+    int* ebar = cbar + 2;
+    int* hn_cells = cbar + 4;  // per-slot host_needed counters
+    TrieDev trie{(const uint8_t*)a.trie[0], (const int32_t*)a.trie[1],
+                 (const int32_t*)a.trie[2], (const int32_t*)a.trie[3],
+                 (const int32_t*)a.trie[4], (const int32_t*)a.trie[5],
+                 (const int32_t*)a.trie[6], (const int32_t*)a.trie[7],
+                 (const int32_t*)a.trie[8]};
+    const int n = a.n;
+    for (unsigned long long b = a.first; b < a.first + a.nbatch; ++b) {
+        // serial 1 lands in slot 0 (the engine's lane round-robin)
+        const int slot = (int)((b - 1) & 1);
+        uint8_t* d_ing = (uint8_t*)a.d_ingress[slot];
+        const int64_t* offs = (const int64_t*)d_ing;
+        const int32_t* lens = (const int32_t*)(d_ing + (n + 1) * 8);
+        uint8_t* reqs = d_ing + a.hdr_bytes;
+        int32_t* fields = (int32_t*)a.d_fields[slot];
+        uint8_t* resp = (uint8_t*)a.d_resp[slot];
+        int32_t* tables = (int32_t*)a.d_tables[slot];
+        int32_t* p_tab = (int32_t*)a.p_tables[slot];
+        const uint8_t* date29 = d_ing + a.date_off;
+        if (!is_egress) {
+            bool tmo = false;
+            if (crew_rank == 0 && threadIdx.x == 0) {
+                __hip_atomic_store(&hn_cells[slot], 0, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+                if (!__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
+                                       __HIP_MEMORY_SCOPE_AGENT)) {
+                    long i = 0;
+                    for (; i < (1L << 19); ++i) {
+                        if (__hip_atomic_load(
+                                go, __ATOMIC_ACQUIRE,
+                                __HIP_MEMORY_SCOPE_SYSTEM) >= b)
+                            break;
+                        __builtin_amdgcn_s_sleep(32);
+                    }
+                    tmo = i >= (1L << 19);
+                } else {
+                    tmo = true;
+                }
+                if (tmo) {
+                    // latch: the remaining batches of this launch
+                    // drain in microseconds (work skipped, markers
+                    // published) so the kernel always exits promptly
+                    __hip_atomic_store(tmo_latch, 1ull,
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(
+                        p_tab + 2 * n + 3,
+                        (int32_t)(b & 0x7fffffff), __ATOMIC_RELEASE,
+                        __HIP_MEMORY_SCOPE_SYSTEM);
+                }
+            }
+            grid_barrier(cbar, cbar + 1, crew_n);
+            if (__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
+                                  __HIP_MEMORY_SCOPE_AGENT)) {
+                grid_barrier(cbar, cbar + 1, crew_n);
+                if (crew_rank == 0 && threadIdx.x == 0)
+                    __hip_atomic_store(cdone, b, __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                continue;
+            }
+            const uint64_t seed = *(const uint64_t*)(offs + n);
+            for (int req = crew_rank * WAVES_PER_BLOCK + wv; req < n;
+                 req += crew_n * WAVES_PER_BLOCK) {
+                parse_one(reqs, offs, lens, fields, n, trie,
+                          (const int32_t*)a.handler_tab, a.n_routes,
+                          &hn_cells[slot], masks_all[wv],
+                          lf_pos_all[wv], req, lane);
+                if (a.secret_len == 0) {
+                    respond_impl<false>(
+                        reqs, offs, fields, resp, tables, n, a.rslot,
+                        (const int32_t*)a.handler_tab, a.n_routes,
+                        (const uint8_t*)a.blob,
+                        (const uint8_t*)a.host_blob[slot],
+                        (const int32_t*)a.host_tab[slot], seed,
+                        a.auth_env_off, a.auth_env_len, 0, a.etag_on,
+                        date29, (const int32_t*)a.d_kv_tab,
+                        (const uint8_t*)a.d_kv_blob, obuf, nullptr,
+                        jtab, nullptr, nullptr, req, lane);
+                }
+            }
+            if (a.secret_len > 0) {
+                __threadfence();
+                grid_barrier(cbar, cbar + 1, crew_n);
+                for (int req = crew_rank * BLOCK_THREADS + threadIdx.x;
+                     req < n; req += crew_n * BLOCK_THREADS)
+                    auth_one(reqs, offs, fields, req,
+                             (const uint8_t*)a.secret, a.secret_len);
+                __threadfence();
+                grid_barrier(cbar, cbar + 1, crew_n);
+                for (int req = crew_rank * WAVES_PER_BLOCK + wv;
+                     req < n; req += crew_n * WAVES_PER_BLOCK)
+                    respond_impl<false>(
+                        reqs, offs, fields, resp, tables, n, a.rslot,
+                        (const int32_t*)a.handler_tab, a.n_routes,
+                        (const uint8_t*)a.blob,
+                        (const uint8_t*)a.host_blob[slot],
+                        (const int32_t*)a.host_tab[slot], seed,
+                        a.auth_env_off, a.auth_env_len, 0, a.etag_on,
+                        date29, (const int32_t*)a.d_kv_tab,
+                        (const uint8_t*)a.d_kv_blob, obuf, nullptr,
+                        jtab, nullptr, nullptr, req, lane);
+            }
+            __threadfence();
+            grid_barrier(cbar, cbar + 1, crew_n);
+            if (crew_rank == 0)
+                scan_tables_crew(tables, hn_cells[slot], n, p_tab);
+            __threadfence_system();
+            grid_barrier(cbar, cbar + 1, crew_n);
+            if (crew_rank == 0 && threadIdx.x == 0)
+                __hip_atomic_store(cdone, b, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        } else {
+            if (crew_rank == 0 && threadIdx.x == 0) {
+                for (long i = 0; i < (1L << 22); ++i) {
+                    if (__hip_atomic_load(cdone, __ATOMIC_ACQUIRE,
+                                          __HIP_MEMORY_SCOPE_AGENT)
+                        >= b)
+                        break;
+                    __builtin_amdgcn_s_sleep(16);
+                }
+            }
+            grid_barrier(ebar, ebar + 1, crew_n);
+            if (__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
+                                  __HIP_MEMORY_SCOPE_AGENT)) {
+                grid_barrier(ebar, ebar + 1, crew_n);
+                if (crew_rank == 0 && threadIdx.x == 0)
+                    __hip_atomic_store(p_tab + 2 * n + 2,
+                                       (int32_t)(b & 0x7fffffff),
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_SYSTEM);
+                continue;
+            }
+            uint8_t* p_out = (uint8_t*)a.p_out[slot];
+            for (int req = crew_rank * WAVES_PER_BLOCK + wv; req < n;
+                 req += crew_n * WAVES_PER_BLOCK) {
+                const int len = tables[req];
+                const int nv = (len + 15) >> 4;
+                const uint4* src =
+                    (const uint4*)(resp + (size_t)req * a.rslot);
+                uint4* dst = (uint4*)(p_out + (size_t)tables[n + req]);
+                for (int i = lane; i < nv; i += WAVE) dst[i] = src[i];
+            }
+            __threadfence_system();
+            grid_barrier(ebar, ebar + 1, crew_n);
+            if (crew_rank == 0 && threadIdx.x == 0)
+                __hip_atomic_store(p_tab + 2 * n + 2,
+                                   (int32_t)(b & 0x7fffffff),
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
+        }
     }
 }
 
@@ -2850,6 +3144,45 @@ extern "C" int gofr_launch_mfma_probe(void* stream, const void* A,
                        (hipStream_t)stream,
                        (const int8_t*)A, (const int8_t*)B, (int32_t*)D);
     return (int)hipGetLastError();
+}
+
+// ---- persistent serving engine host API -----------------------------------
+
+// co-resident grid capacity for k_persist_serve (the crews wait on
+// each other, so the launch MUST fit on the chip at once)
+extern "C" int gofr_persist_grid(int* blocks_out) {
+    int per_cu = 0;
+    hipError_t rc = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+        &per_cu, (const void*)k_persist_serve, BLOCK_THREADS, 0);
+    if (rc) return (int)rc;
+    int cus = 0;
+    hipDeviceProp_t prop;
+    if ((rc = hipGetDeviceProperties(&prop, 0))) return (int)rc;
+    cus = prop.multiProcessorCount;
+    *blocks_out = per_cu * cus;
+    return 0;
+}
+
+extern "C" int gofr_persist_launch(const PersistKernArgs* a,
+                                   void* stream, int nblocks) {
+    hipLaunchKernelGGL(k_persist_serve, dim3(nblocks),
+                       dim3(BLOCK_THREADS), 0, (hipStream_t)stream, *a);
+    return (int)hipGetLastError();
+}
+
+// per-batch submit: ONE ingress SDMA (header+payload) then the go
+// serial in the same FIFO — no kernel launches, no events
+extern "C" int gofr_persist_submit(void* stream, void* d_ingress,
+                                   const void* p_ingress,
+                                   long long nbytes, void* d_go,
+                                   const void* p_serial) {
+    hipError_t rc = hipMemcpyAsync(d_ingress, p_ingress, (size_t)nbytes,
+                                   hipMemcpyHostToDevice,
+                                   (hipStream_t)stream);
+    if (rc) return (int)rc;
+    return (int)hipMemcpyAsync(d_go, p_serial, 8,
+                               hipMemcpyHostToDevice,
+                               (hipStream_t)stream);
 }
 
 extern "C" int gofr_launch_persist_cycle(

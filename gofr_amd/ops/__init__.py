@@ -115,6 +115,35 @@ class GofrSubmitArgs(ctypes.Structure):
     ]
 
 
+class PersistKernArgs(ctypes.Structure):
+    """Mirror of PersistKernArgs in gofr_kernels.hip (the persistent
+    serving kernel's argument block; passed by value at launch)."""
+
+    _fields_ = [
+        ("first", ctypes.c_uint64),
+        ("nbatch", ctypes.c_int),
+        ("n", ctypes.c_int), ("rslot", ctypes.c_int),
+        ("hdr_bytes", ctypes.c_longlong),
+        ("date_off", ctypes.c_int), ("egress_blocks", ctypes.c_int),
+        ("d_ingress", ctypes.c_void_p * 2),
+        ("d_fields", ctypes.c_void_p * 2),
+        ("d_resp", ctypes.c_void_p * 2),
+        ("d_tables", ctypes.c_void_p * 2),
+        ("p_tables", ctypes.c_void_p * 2),
+        ("p_out", ctypes.c_void_p * 2),
+        ("host_blob", ctypes.c_void_p * 2),
+        ("host_tab", ctypes.c_void_p * 2),
+        ("trie", ctypes.c_void_p * 9),
+        ("handler_tab", ctypes.c_void_p), ("n_routes", ctypes.c_int),
+        ("blob", ctypes.c_void_p),
+        ("d_kv_tab", ctypes.c_void_p), ("d_kv_blob", ctypes.c_void_p),
+        ("secret", ctypes.c_void_p), ("secret_len", ctypes.c_int),
+        ("auth_env_off", ctypes.c_int), ("auth_env_len", ctypes.c_int),
+        ("etag_on", ctypes.c_int),
+        ("d_state", ctypes.c_void_p), ("d_barrier", ctypes.c_void_p),
+    ]
+
+
 class HipOps:
     """ctypes driver for the device kernels."""
 
@@ -178,6 +207,17 @@ class HipOps:
         self.lib.gofr_wait_cell.argtypes = [
             ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
             ctypes.c_double]
+        self.lib.gofr_persist_grid.restype = ctypes.c_int
+        self.lib.gofr_persist_grid.argtypes = [
+            ctypes.POINTER(ctypes.c_int)]
+        self.lib.gofr_persist_launch.restype = ctypes.c_int
+        self.lib.gofr_persist_launch.argtypes = [
+            ctypes.POINTER(PersistKernArgs), ctypes.c_void_p,
+            ctypes.c_int]
+        self.lib.gofr_persist_submit.restype = ctypes.c_int
+        self.lib.gofr_persist_submit.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_longlong, ctypes.c_void_p, ctypes.c_void_p]
         try:
             self.lib.gofr_src_hash.restype = ctypes.c_char_p
         except AttributeError:

@@ -367,3 +367,40 @@ def test_template_kv_kernels_match_mirror():
     fields = ops.cpu_parse_route(buf, offs, lens, cpu.program.trie,
                                  cpu.program.handler_tab)
     assert not any(f[ops.FI_KIND] == ops.HK_HOST for f in fields)
+
+
+def test_persistent_engine_matches_mirror():
+    """k_persist_serve (the resident serving kernel) byte-matches the
+    CPU mirrors across slot reuse, relaunch windows and the host-fixup
+    path (mixed payloads include trampoline rows)."""
+    import os
+
+    import torch
+
+    os.environ.setdefault("GOFR_PERSIST_NBATCH", "8")  # force relaunches
+    app = build_app()
+    gpu = BatchEngine(app, device="cuda", max_batch=1024, pipeline=2)
+    cpu = BatchEngine(app, device="cpu", max_batch=1024)
+    cpu._seed = gpu._seed
+    cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
+    raws = mixed_payloads(512)
+    buf, offs, lens = pack_batch(raws)
+    n, nbytes = len(lens), int(offs[-1] + lens[-1])
+    for ln in gpu.lanes:
+        ln.p_reqs[:nbytes] = torch.from_numpy(buf[:nbytes])
+        ln.p_req_off[:n] = torch.from_numpy(offs)
+        ln.p_req_len[:n] = torch.from_numpy(lens)
+    gpu.arm_persistent(n, nbytes)
+    for it in range(20):  # crosses two relaunch windows (nbatch=8)
+        lane = it % 2
+        gpu.submit(n, nbytes, lane)
+        out_t, roff_t, rlen_t = gpu.complete(lane)
+        c_out, c_roffs, c_rlens = cpu.process_packed(buf, offs, lens)
+        g = out_t.numpy()
+        for i in range(n):
+            go = g[int(roff_t[i]):int(roff_t[i]) + int(rlen_t[i])]
+            co = c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])]
+            assert bytes(go) == bytes(co), \
+                (f"iter {it} req {i}: {raws[i][:60]!r}\n"
+                 f"GPU {bytes(go)[:120]!r}\nCPU {bytes(co)[:120]!r}")
+    del os.environ["GOFR_PERSIST_NBATCH"]
