@@ -2423,6 +2423,10 @@ k_persist_serve(PersistKernArgs a) {
                           (const int32_t*)a.handler_tab, a.n_routes,
                           &hn_cells[slot], masks_all[wv],
                           lf_pos_all[wv], req, lane);
+                // lane 0's field-table stores must land before the
+                // whole wave's respond reads them (the kernel boundary
+                // used to provide this)
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
                 if (a.secret_len == 0) {
                     respond_impl<false>(
                         reqs, offs, fields, resp, tables, n, a.rslot,
