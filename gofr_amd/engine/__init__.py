@@ -595,6 +595,8 @@ class BatchEngine:
         a.n = n
         a.rslot = self.rslot
         a.hdr_bytes = ln0.hdr_bytes
+        a.lens_off = (ln0.d_req_len.data_ptr() -
+                      ln0.d_ingress.data_ptr())
         a.date_off = ln0.date_off
         a.egress_blocks = max(32, G // 4)
         for i, ln in enumerate((ln0, ln1)):

@@ -2268,6 +2268,7 @@ typedef struct {
     int nbatch;
     int n, rslot;
     long long hdr_bytes;        // ingress header size (offs/lens/seed/date)
+    long long lens_off;         // byte offset of the int32 lens array
     int date_off;
     int egress_blocks;
     // per-slot device buffers
@@ -2368,7 +2369,7 @@ k_persist_serve(PersistKernArgs a) {
         const int slot = (int)((b - 1) & 1);
         uint8_t* d_ing = (uint8_t*)a.d_ingress[slot];
         const int64_t* offs = (const int64_t*)d_ing;
-        const int32_t* lens = (const int32_t*)(d_ing + (n + 1) * 8);
+        const int32_t* lens = (const int32_t*)(d_ing + a.lens_off);
         uint8_t* reqs = d_ing + a.hdr_bytes;
         int32_t* fields = (int32_t*)a.d_fields[slot];
         uint8_t* resp = (uint8_t*)a.d_resp[slot];

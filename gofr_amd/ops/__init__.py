@@ -124,6 +124,7 @@ class PersistKernArgs(ctypes.Structure):
         ("nbatch", ctypes.c_int),
         ("n", ctypes.c_int), ("rslot", ctypes.c_int),
         ("hdr_bytes", ctypes.c_longlong),
+        ("lens_off", ctypes.c_longlong),
         ("date_off", ctypes.c_int), ("egress_blocks", ctypes.c_int),
         ("d_ingress", ctypes.c_void_p * 2),
         ("d_fields", ctypes.c_void_p * 2),
