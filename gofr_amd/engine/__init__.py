@@ -255,9 +255,12 @@ class _Lane:
         # host_needed) — one D2H moves the whole per-batch result set
         self.d_tables = t.zeros(2 * nb + 2, dtype=t.int32, device=dev)
         if hip is not None:
-            # +4: [2n+2] carries the flagged pipeline's done serial
+            # +4: [2n+2] carries the flagged pipeline's done serial,
+            # [2n+3] the gate-timeout marker (nonzero => timed out;
+            # hipHostMalloc memory is NOT zero-initialized)
             self.p_tables = hip.host_alloc((2 * nb + 4) * 4,
                                            dtype=np.int32)
+            self.p_tables.zero_()
         else:
             self.p_tables = t.zeros(2 * nb + 4, dtype=t.int32).pin_memory()
         self.p_tables_np = self.p_tables.numpy()
@@ -385,6 +388,22 @@ class BatchEngine:
                             self.rslot, hip=self.hip)
                       for _ in range(self.pipeline)]
 
+    def stop_persistent(self) -> None:
+        """Shut the resident serving kernel down: write the timeout
+        latch (the kernel drains its remaining launch windows in
+        microseconds and exits)."""
+        if not getattr(self, "_persist", False):
+            return
+        t = self.torch
+        stop = t.ones(1, dtype=t.int64).pin_memory()
+        self.hip.lib.gofr_memcpy_async(
+            ctypes.c_void_p(self.d_pstate.data_ptr() + 24),
+            ctypes.c_void_p(stop.data_ptr()),
+            ctypes.c_longlong(8), 1,  # hipMemcpyHostToDevice
+            ctypes.c_void_p(self.s_in.cuda_stream))
+        self.torch.cuda.synchronize(self.device)
+        self._persist = False
+
     def close(self) -> None:
         """Release hipHostMalloc'd pinned memory (p_out / p_tables per
         lane). torch-managed device/pinned tensors free with GC, but
@@ -392,6 +411,7 @@ class BatchEngine:
         call twice; the engine must not be used afterwards."""
         if self.device is None or not hasattr(self, "lanes"):
             return
+        self.stop_persistent()
         self.torch.cuda.synchronize(self.device)
         for ln in self.lanes:
             for name in ("p_out", "p_tables"):

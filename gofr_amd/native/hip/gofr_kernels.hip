@@ -2383,15 +2383,27 @@ k_persist_serve(PersistKernArgs a) {
                                    __HIP_MEMORY_SCOPE_AGENT);
                 if (!__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
                                        __HIP_MEMORY_SCOPE_AGENT)) {
+                    // ~60 s bounded idle tolerance: a serving loop may
+                    // legitimately sit idle between batches; the host
+                    // shuts the kernel down EXPLICITLY by writing the
+                    // latch (engine.stop_persistent), which drains the
+                    // remaining windows in microseconds
                     long i = 0;
-                    for (; i < (1L << 19); ++i) {
+                    for (; i < (1L << 26); ++i) {
                         if (__hip_atomic_load(
                                 go, __ATOMIC_ACQUIRE,
                                 __HIP_MEMORY_SCOPE_SYSTEM) >= b)
                             break;
+                        if ((i & 0xFFF) == 0 &&
+                            __hip_atomic_load(tmo_latch,
+                                              __ATOMIC_ACQUIRE,
+                                              __HIP_MEMORY_SCOPE_SYSTEM))
+                            break;
                         __builtin_amdgcn_s_sleep(32);
                     }
-                    tmo = i >= (1L << 19);
+                    tmo = __hip_atomic_load(
+                              go, __ATOMIC_ACQUIRE,
+                              __HIP_MEMORY_SCOPE_SYSTEM) < b;
                 } else {
                     tmo = true;
                 }
@@ -2980,14 +2992,13 @@ extern "C" int gofr_wait_cell(const void* cell_p, int want,
     const double t0 = now_us();
     long spins = 0;
     while (__atomic_load_n((const int*)cell, __ATOMIC_ACQUIRE) != want) {
-        if (tmo &&
-            __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE) == want)
-            return 2;
+        if (tmo && __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE))
+            return 2;  // any nonzero marker: gate timeout / latched
         if (((++spins) & 0xFFFF) == 0 &&
             now_us() - t0 > timeout_s * 1e6)
             return 1;
     }
-    if (tmo && __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE) == want)
+    if (tmo && __atomic_load_n((const int*)tmo, __ATOMIC_ACQUIRE))
         return 2;
     return 0;
 }
