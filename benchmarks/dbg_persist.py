@@ -19,19 +19,22 @@ for ln in gpu.lanes:
     ln.p_req_off[:n] = torch.from_numpy(offs)
     ln.p_req_len[:n] = torch.from_numpy(lens)
 gpu.arm_persistent(n, nbytes)
-gpu.submit(n, nbytes, 0)
-out_t, roff_t, rlen_t = gpu.complete(0)
-torch.cuda.synchronize()
-gf = gpu.lanes[0].d_fields[:n*ops.NF].cpu().numpy().reshape(n, ops.NF)
-cbuf = buf.copy()
-cf = ops.cpu_parse_route(cbuf, offs, lens, cpu.program.trie, cpu.program.handler_tab)
-bad = 0
-for i in range(n):
-    if not np.array_equal(gf[i], cf[i]):
-        bad += 1
-        if bad <= 4:
-            print("REQ", i, repr(raws[i][:70]))
-            for k in range(ops.NF):
-                if gf[i][k] != cf[i][k]:
-                    print("  field", k, "gpu", gf[i][k], "cpu", cf[i][k])
-print("total field mismatches:", bad, "/", n)
+for it in range(3):
+    lane = it % 2
+    gpu.submit(n, nbytes, lane)
+    out_t, roff_t, rlen_t = gpu.complete(lane)
+    c_out, c_roffs, c_rlens = cpu.process_packed(buf.copy(), offs, lens)
+    g = out_t.numpy()
+    bad = 0
+    for i in range(n):
+        go = bytes(g[int(roff_t[i]):int(roff_t[i]) + int(rlen_t[i])])
+        co = bytes(c_out[int(c_roffs[i]):int(c_roffs[i]) + int(c_rlens[i])])
+        if go != co:
+            bad += 1
+            if bad <= 2:
+                d = next(k for k in range(min(len(go), len(co)) + 1)
+                         if k >= len(go) or k >= len(co) or go[k] != co[k])
+                print(f"iter {it} req {i} diff@{d}")
+                print("  GPU", go[max(0,d-20):d+30])
+                print("  CPU", co[max(0,d-20):d+30])
+    print(f"iter {it}: {bad} mismatched of {n}")

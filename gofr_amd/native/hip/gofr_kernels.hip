@@ -2383,13 +2383,15 @@ k_persist_serve(PersistKernArgs a) {
                                    __HIP_MEMORY_SCOPE_AGENT);
                 if (!__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
                                        __HIP_MEMORY_SCOPE_AGENT)) {
-                    // ~60 s bounded idle tolerance: a serving loop may
-                    // legitimately sit idle between batches; the host
-                    // shuts the kernel down EXPLICITLY by writing the
-                    // latch (engine.stop_persistent), which drains the
-                    // remaining windows in microseconds
+                    // effectively-unbounded idle tolerance (hours):
+                    // a serving loop may sit idle arbitrarily long
+                    // between batches. Clean shutdown is the host
+                    // writing the latch (engine.stop_persistent); an
+                    // ABANDONED kernel dies with its process (the
+                    // driver tears the process's queues down), so
+                    // this spin cannot leave the GPU wedged.
                     long i = 0;
-                    for (; i < (1L << 26); ++i) {
+                    for (; i < (1L << 33); ++i) {
                         if (__hip_atomic_load(
                                 go, __ATOMIC_ACQUIRE,
                                 __HIP_MEMORY_SCOPE_SYSTEM) >= b)
@@ -2486,7 +2488,10 @@ k_persist_serve(PersistKernArgs a) {
                                    __HIP_MEMORY_SCOPE_AGENT);
         } else {
             if (crew_rank == 0 && threadIdx.x == 0) {
-                for (long i = 0; i < (1L << 22); ++i) {
+                // compute's latched path still publishes cdone, so
+                // this wait always releases; the bound matches the
+                // go-wait's process-lifetime reasoning
+                for (long i = 0; i < (1L << 33); ++i) {
                     if (__hip_atomic_load(cdone, __ATOMIC_ACQUIRE,
                                           __HIP_MEMORY_SCOPE_AGENT)
                         >= b)
