@@ -109,7 +109,7 @@ def run_single(eng, payloads, steps, warmup):
         ln.p_reqs[:nbytes] = torch.from_numpy(buf[:nbytes])
         ln.p_req_off[:n] = torch.from_numpy(offs)
         ln.p_req_len[:n] = torch.from_numpy(lens)
-    persist = os.environ.get("GOFR_PERSIST", "1") == "1"
+    persist = os.environ.get("GOFR_PERSIST", "0") == "1"
     if persist:
         # resident serving kernel: two double-buffer slots (lanes 0/1)
         P = 2
