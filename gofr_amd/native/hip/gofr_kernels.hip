@@ -2189,7 +2189,12 @@ __device__ void grid_barrier(int* counter, int* gen_ptr, int nblocks) {
             __hip_atomic_store(gen_ptr, my_gen + 1, __ATOMIC_RELEASE,
                                __HIP_MEMORY_SCOPE_AGENT);
         } else {
-            for (long i = 0; i < (1L << 22); ++i) {
+            // process-lifetime bound: the persistent kernel's crews
+            // legitimately sit at barriers while the host is idle
+            // between batches (the 0.9 s probe-era bound let waiters
+            // ESCAPE and run ahead, smearing state across batches).
+            // An abandoned kernel dies with its process's queues.
+            for (long i = 0; i < (1L << 33); ++i) {
                 if (__hip_atomic_load(gen_ptr, __ATOMIC_ACQUIRE,
                                       __HIP_MEMORY_SCOPE_AGENT) != my_gen)
                     break;
