@@ -598,7 +598,7 @@ class BatchEngine:
         ln0, ln1 = self.lanes[0], self.lanes[1]
         for ln in (ln0, ln1):
             self._stamp_date(ln)
-        self.d_pstate = t.zeros(4, dtype=t.int64, device=self.device)
+        self.d_pstate = t.zeros(6, dtype=t.int64, device=self.device)
         self.d_pbar = t.zeros(8, dtype=t.int32, device=self.device)
         t.cuda.synchronize(self.device)
         g = ctypes.c_int(0)

@@ -2359,6 +2359,7 @@ k_persist_serve(PersistKernArgs a) {
                                  : gridDim.x - a.egress_blocks;
     unsigned long long* go = (unsigned long long*)a.d_state;
     unsigned long long* cdone = go + 1;
+    unsigned long long* cverdict = go + 2;  // per-batch latch verdict
     unsigned long long* tmo_latch = go + 3;
     int* cbar = (int*)a.d_barrier;
     int* ebar = cbar + 2;
@@ -2427,12 +2428,21 @@ k_persist_serve(PersistKernArgs a) {
                         __HIP_MEMORY_SCOPE_SYSTEM);
                 }
             }
+            if (crew_rank == 0 && threadIdx.x == 0)
+                __hip_atomic_store(cverdict, tmo ? 1ull : 0ull,
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
             grid_barrier(cbar, cbar + 1, crew_n);
-            if (__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
+            // UNIFORM latch verdict: published by crew block 0 BEFORE
+            // the barrier, so every block takes the same path (a
+            // direct racy read of the host-written latch could
+            // diverge and deadlock the crew barriers)
+            if (__hip_atomic_load(cverdict, __ATOMIC_ACQUIRE,
                                   __HIP_MEMORY_SCOPE_AGENT)) {
                 grid_barrier(cbar, cbar + 1, crew_n);
                 if (crew_rank == 0 && threadIdx.x == 0)
-                    __hip_atomic_store(cdone, b, __ATOMIC_RELEASE,
+                    __hip_atomic_store(cdone, (b << 1) | 1ull,
+                                       __ATOMIC_RELEASE,
                                        __HIP_MEMORY_SCOPE_AGENT);
                 continue;
             }
@@ -2489,23 +2499,29 @@ k_persist_serve(PersistKernArgs a) {
             __threadfence_system();
             grid_barrier(cbar, cbar + 1, crew_n);
             if (crew_rank == 0 && threadIdx.x == 0)
-                __hip_atomic_store(cdone, b, __ATOMIC_RELEASE,
+                __hip_atomic_store(cdone, b << 1, __ATOMIC_RELEASE,
                                    __HIP_MEMORY_SCOPE_AGENT);
         } else {
             if (crew_rank == 0 && threadIdx.x == 0) {
                 // compute's latched path still publishes cdone, so
                 // this wait always releases; the bound matches the
                 // go-wait's process-lifetime reasoning
+                unsigned long long cd = 0;
                 for (long i = 0; i < (1L << 33); ++i) {
-                    if (__hip_atomic_load(cdone, __ATOMIC_ACQUIRE,
-                                          __HIP_MEMORY_SCOPE_AGENT)
-                        >= b)
-                        break;
+                    cd = __hip_atomic_load(cdone, __ATOMIC_ACQUIRE,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+                    if ((cd >> 1) >= b) break;
                     __builtin_amdgcn_s_sleep(16);
                 }
+                // publish the verdict for THIS crew uniformly (the
+                // low bit of cdone; verdicts are monotonic so a
+                // later batch's bit only over-drops during shutdown)
+                __hip_atomic_store(tmo_latch + 1, cd & 1ull,
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
             }
             grid_barrier(ebar, ebar + 1, crew_n);
-            if (__hip_atomic_load(tmo_latch, __ATOMIC_ACQUIRE,
+            if (__hip_atomic_load(tmo_latch + 1, __ATOMIC_ACQUIRE,
                                   __HIP_MEMORY_SCOPE_AGENT)) {
                 grid_barrier(ebar, ebar + 1, crew_n);
                 if (crew_rank == 0 && threadIdx.x == 0)
