@@ -404,6 +404,10 @@ class BatchEngine:
         self.torch.cuda.synchronize(self.device)
         self._persist = False
 
+    # atexit registry protocol (shared with the servers)
+    def stop(self) -> None:
+        self.stop_persistent()
+
     def close(self) -> None:
         """Release hipHostMalloc'd pinned memory (p_out / p_tables per
         lane). torch-managed device/pinned tensors free with GC, but
@@ -661,6 +665,9 @@ class BatchEngine:
         if rc:
             raise RuntimeError(f"gofr_persist_launch failed: {rc}")
         self._persist = True
+        # a resident kernel must be latched down before interpreter
+        # teardown (a bare exit would hang the HIP context destroy)
+        _register_server(self)
 
     def _persist_submit(self, ln, n: int, nbytes: int,
                         lane_idx: int) -> None:

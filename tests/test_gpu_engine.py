@@ -377,13 +377,13 @@ def test_persistent_engine_matches_mirror():
 
     import torch
 
-    os.environ.setdefault("GOFR_PERSIST_NBATCH", "8")  # force relaunches
+    os.environ.setdefault("GOFR_PERSIST_NBATCH", "5")  # force relaunches
     app = build_app()
-    gpu = BatchEngine(app, device="cuda", max_batch=1024, pipeline=2)
-    cpu = BatchEngine(app, device="cpu", max_batch=1024)
+    gpu = BatchEngine(app, device="cuda", max_batch=256, pipeline=2)
+    cpu = BatchEngine(app, device="cpu", max_batch=256)
     cpu._seed = gpu._seed
     cpu._date_fn = gpu._date_fn = lambda: 1789300000.0
-    raws = mixed_payloads(512)
+    raws = mixed_payloads(128)
     buf, offs, lens = pack_batch(raws)
     n, nbytes = len(lens), int(offs[-1] + lens[-1])
     for ln in gpu.lanes:
@@ -391,7 +391,7 @@ def test_persistent_engine_matches_mirror():
         ln.p_req_off[:n] = torch.from_numpy(offs)
         ln.p_req_len[:n] = torch.from_numpy(lens)
     gpu.arm_persistent(n, nbytes)
-    for it in range(20):  # crosses two relaunch windows (nbatch=8)
+    for it in range(12):  # crosses two relaunch windows (nbatch=5)
         lane = it % 2
         gpu.submit(n, nbytes, lane)
         out_t, roff_t, rlen_t = gpu.complete(lane)
@@ -403,4 +403,5 @@ def test_persistent_engine_matches_mirror():
             assert bytes(go) == bytes(co), \
                 (f"iter {it} req {i}: {raws[i][:60]!r}\n"
                  f"GPU {bytes(go)[:120]!r}\nCPU {bytes(co)[:120]!r}")
+    gpu.close()
     del os.environ["GOFR_PERSIST_NBATCH"]
