@@ -622,7 +622,8 @@ class BatchEngine:
         a.lens_off = (ln0.d_req_len.data_ptr() -
                       ln0.d_ingress.data_ptr())
         a.date_off = ln0.date_off
-        a.egress_blocks = max(32, G // 4)
+        a.egress_blocks = int(os.environ.get("GOFR_PERSIST_EGRESS",
+                                             str(max(32, G // 4))))
         for i, ln in enumerate((ln0, ln1)):
             a.d_ingress[i] = ln.d_ingress.data_ptr()
             a.d_fields[i] = ln.d_fields.data_ptr()
