@@ -189,3 +189,54 @@ def test_graceful_shutdown_on_sigterm():
             p.kill()
     assert rc == 0, out
     assert "CLEAN-EXIT" in out and "shutting down" in out
+
+
+@pytest.fixture()
+def gpu_routes_example(monkeypatch, tmp_path):
+    """Start the gpu-routes example (device-resident handler showcase)
+    on a free port — CPU transport here; the engine serves the same
+    routes GPU-resident on an MI355X."""
+    monkeypatch.chdir(tmp_path)
+    for k in ("REDIS_HOST", "DB_HOST", "TRACER_HOST"):
+        monkeypatch.delenv(k, raising=False)
+    port = _free_port()
+    monkeypatch.setenv("HTTP_PORT", str(port))
+    monkeypatch.setenv("LOG_LEVEL", "FATAL")
+    mod = _load_example("gpu-routes")
+    app = mod.build_app()
+    app.Run(block=False)
+    time.sleep(0.1)
+    yield port
+    app.shutdown()
+    for k in ("HTTP_PORT", "LOG_LEVEL", "APP_NAME"):
+        os.environ.pop(k, None)
+
+
+def test_gpu_routes_example(gpu_routes_example):
+    port = gpu_routes_example
+    status, _, body = _get(port, "/user/42")
+    assert (status, body) == (200, b'{"data":{"id":"42"}}')
+    status, _, body = _get(port, "/greet?name=World")
+    assert (status, body) == (200, b'{"data":"Hello World!"}')
+    status, _, body = _get(port, "/greet?name=a%20b")
+    assert (status, body) == (200, b'{"data":"Hello a b!"}')
+    status, _, body = _get(port, "/plans/pro")
+    assert status == 200
+    assert json.loads(body) == {"data": {"tier": "pro", "rps": 10000}}
+    status, _, body = _get(port, "/plans/nope")
+    assert status == 404
+    assert body == b'{"error":{"message":"key not found"}}'
+    # POST /order: JSON field binding
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=5)
+    payload = json.dumps({"item": "widget", "qty": 3, "note": "asap"},
+                         separators=(",", ":"))
+    conn.request("POST", "/order", body=payload,
+                 headers={"Content-Type": "application/json"})
+    r = conn.getresponse()
+    assert r.status == 200
+    assert r.read() == (b'{"data":{"item":"widget","qty":3,'
+                        b'"note":"asap"}}')
+    conn.close()
+    # /cached without redis configured -> 500 envelope
+    status, _, body = _get(port, "/cached/x")
+    assert status == 500
