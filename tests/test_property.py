@@ -229,3 +229,92 @@ def test_protobuf_nested_repeated_roundtrip(items, names, kind):
     assert [(d["id"], d["tag"]) for d in out["items"]] == items
     assert out["names"] == names
     assert out["kind"] == kind
+
+
+# ---- round-2 mirrors: JSON field scan, pct decode, query lookup ------------
+
+_JKEY = st.text(alphabet="abcdefghijklmnop_-0123456789", min_size=1,
+                max_size=12)
+_JVAL = st.recursive(
+    st.one_of(st.none(), st.booleans(),
+              st.integers(min_value=-10**12, max_value=10**12),
+              st.floats(allow_nan=False, allow_infinity=False,
+                        width=32),
+              st.text(max_size=30)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(_JKEY, children, max_size=4)),
+    max_leaves=8)
+
+
+@SET
+@given(st.dictionaries(_JKEY, _JVAL, min_size=0, max_size=8),
+       st.sampled_from([(",", ":"), (", ", ": ")]))
+def test_json_top_fields_extracts_true_values(obj, seps):
+    """json_top_fields_py's raw value spans must json-decode to the
+    true field values for any serializable object body (the device
+    scan is byte-equality-pinned against this mirror)."""
+    import json as _json
+
+    from gofr_amd import ops
+    body = _json.dumps(obj, separators=seps).encode()
+    fields = ops.json_top_fields_py(body)
+    if len(obj) > ops.MAX_JSON_FIELDS:
+        return  # scan caps at 8 fields (callers splice empty beyond)
+    got = {}
+    for key, vs, vl in fields:
+        got[key.decode()] = _json.loads(body[vs:vs + vl])
+    assert got == obj
+
+
+@SET
+@given(st.lists(st.one_of(
+    st.binary(min_size=1, max_size=4).filter(lambda b: b"%" not in b),
+    st.integers(min_value=0, max_value=255).map(
+        lambda v: f"%{v:02x}".encode()),
+    st.integers(min_value=0, max_value=255).map(
+        lambda v: f"%{v:02X}".encode())), max_size=12))
+def test_pct_decode_matches_unquote_on_valid_escapes(parts):
+    from urllib.parse import unquote_to_bytes
+
+    from gofr_amd import ops
+    raw = b"".join(parts)
+    got = ops.pct_decode(raw)
+    assert got == unquote_to_bytes(raw)
+
+
+@SET
+@given(st.binary(max_size=24))
+def test_pct_decode_rejects_exactly_invalid_escapes(raw):
+    """None iff some '%' is not followed by two hex digits (the kernel
+    then routes the request to the host parser's unquote leniency)."""
+    from gofr_amd import ops
+    hexd = set(b"0123456789abcdefABCDEF")
+    valid = True
+    i = 0
+    while i < len(raw):
+        if raw[i] == 0x25:
+            if i + 2 >= len(raw) or raw[i + 1] not in hexd \
+                    or raw[i + 2] not in hexd:
+                valid = False
+                break
+            i += 3
+        else:
+            i += 1
+    got = ops.pct_decode(raw)
+    assert (got is not None) == valid
+
+
+@SET
+@given(st.lists(st.tuples(
+    st.text(alphabet="abcxyz_123", min_size=1, max_size=6),
+    st.text(alphabet="abc+%20xyz", max_size=8)), max_size=6),
+    st.text(alphabet="abcxyz_123", min_size=1, max_size=6))
+def test_q_find_first_value_semantics(pairs, probe):
+    """q_find returns the FIRST value for a key (reference
+    Param semantics: http/request.go:28-30) and None for misses."""
+    from gofr_amd import ops
+    q = "&".join(f"{k}={v}" for k, v in pairs).encode()
+    got = ops.q_find_py(q, probe.encode())
+    want = next((v.encode() for k, v in pairs if k == probe), None)
+    assert got == want
