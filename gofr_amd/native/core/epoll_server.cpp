@@ -550,6 +550,13 @@ private:
         if (!staged_.empty()) {
             // one lock per readiness event, not per sliced request
             std::lock_guard<std::mutex> lk(mu_);
+            // backstop: a protocol nobody harvests (e.g. HTTP/1.1 sent
+            // to a gRPC-only listener) must not grow the queue without
+            // bound — drop the connection instead
+            if (ready_.size() + staged_.size() > (1u << 18)) {
+                staged_.clear();
+                return false;
+            }
             for (auto& r : staged_) ready_.push_back(std::move(r));
             staged_.clear();
         }
@@ -718,6 +725,10 @@ private:
         }
         h2.streams.erase(it);
         std::lock_guard<std::mutex> lk(mu_);
+        if (gready_.size() > (1u << 18)) {
+            c.close_after_write = true;  // unconsumed-protocol backstop
+            return;
+        }
         gready_.push_back(std::move(g));
     }
 
