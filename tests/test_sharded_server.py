@@ -107,3 +107,80 @@ def test_sharded_gpuserver_gloo(tmp_path, world, mport, sport):
     assert proc.returncode == 0, out[-4000:]
     for r in range(world):
         assert f"SRANK{r}_OK 24" in out, out[-4000:]
+
+
+@pytest.mark.timeout(120)
+def test_sharded_backlog_spill(tmp_path):
+    """shard_chunk smaller than the burst: overfull owner blocks spill
+    to the C++ backlog and drain over subsequent cycles — every
+    request still gets its response."""
+    script = tmp_path / "worker.py"
+    script.write_text(r"""
+import json
+import os
+import socket
+import sys
+import time
+
+sys.path.insert(0, %(repo)r)
+
+import torch.distributed as dist
+
+import gofr_amd
+from gofr_amd import handlers
+from gofr_amd.config import MapConfig
+from gofr_amd.engine import GPUServer
+
+
+def main():
+    dist.init_process_group("gloo")
+    port = int(os.environ["GOFR_TEST_PORT"])
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/greet", handlers.static_json("hi"))
+    srv = GPUServer(app, port, batch_window_us=1000, world=1, rank=0,
+                    shard_chunk=2)  # 2 slots per cycle -> forced spill
+    srv.start()
+    time.sleep(0.3)
+    conn = socket.create_connection(("127.0.0.1", port), timeout=30)
+    burst = b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n" * 16
+    conn.sendall(burst)  # 16 pipelined requests vs 2-slot cycles
+    got = 0
+    buf = b""
+    while got < 16:
+        chunk = conn.recv(65536)
+        assert chunk, "server closed early"
+        buf += chunk
+        while True:
+            i = buf.find(b"\r\n\r\n")
+            if i < 0:
+                break
+            head = buf[:i]
+            clen = 0
+            for line in head.split(b"\r\n")[1:]:
+                if line.lower().startswith(b"content-length:"):
+                    clen = int(line.split(b":")[1])
+            if len(buf) < i + 4 + clen:
+                break
+            assert head.startswith(b"HTTP/1.1 200"), head[:40]
+            assert buf[i + 4:i + 4 + clen] == b'{"data":"hi"}'
+            buf = buf[i + 4 + clen:]
+            got += 1
+    conn.close()
+    print(f"SPILL_OK {got}", flush=True)
+    srv.stop()
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+main()
+""" % {"repo": REPO})
+    env = dict(os.environ)
+    env["GOFR_TEST_PORT"] = "18427"
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "1", "--master-addr", "127.0.0.1",
+         "--master-port", "29547", str(script)],
+        capture_output=True, text=True, timeout=110, env=env, cwd=REPO)
+    out = proc.stdout + proc.stderr
+    assert proc.returncode == 0, out[-3000:]
+    assert "SPILL_OK 16" in out, out[-3000:]
