@@ -318,3 +318,40 @@ def test_q_find_first_value_semantics(pairs, probe):
     got = ops.q_find_py(q, probe.encode())
     want = next((v.encode() for k, v in pairs if k == probe), None)
     assert got == want
+
+
+@SET
+@given(st.lists(st.sampled_from(
+    ["plain", "b%20c", "%41%42", "x%2Fy", "pct%25", "u%C3%A9", "%ff"]),
+    min_size=0, max_size=4),
+    st.one_of(st.none(), st.sampled_from(["k=%20v", "a=b+c&d="])))
+def test_parse_mirror_percent_decode_agrees_with_unquote(segs, query):
+    """The in-place path decode (kernel mirror) must byte-match
+    urllib's unquote for any valid-escape path, with the query span
+    left RAW (decoded only at splice time)."""
+    from urllib.parse import unquote_to_bytes
+
+    import numpy as np
+
+    from gofr_amd import ops
+    from gofr_amd.http.router import Router
+
+    path = "/" + "/".join(segs)
+    target = path + (f"?{query}" if query else "")
+    raw = f"GET {target} HTTP/1.1\r\nHost: h\r\n\r\n".encode()
+    r = Router()
+    r.add("GET", "/plain", lambda c: None)
+    trie = r.compile()
+    tab = np.asarray([[ops.HK_HOST, 0, 0, 200]], np.int32).reshape(-1)
+    buf = np.frombuffer(raw, np.uint8).copy()  # decode mutates in place
+    fields = ops.cpu_parse_route(buf, np.asarray([0], np.int64),
+                                 np.asarray([len(raw)], np.int32),
+                                 trie, tab)
+    f = fields[0]
+    got_path = bytes(buf[f[ops.FI_PATH_OFF]:
+                         f[ops.FI_PATH_OFF] + f[ops.FI_PATH_LEN]])
+    assert got_path == unquote_to_bytes(path)
+    assert not (f[ops.FI_FLAGS] & ops.FL_NEEDS_HOST)
+    got_q = bytes(buf[f[ops.FI_QUERY_OFF]:
+                      f[ops.FI_QUERY_OFF] + f[ops.FI_QUERY_LEN]])
+    assert got_q == (query or "").encode()  # query stays raw
