@@ -31,7 +31,9 @@ from test_datasources import MiniRedis  # noqa: E402
 
 
 class FastMini(MiniRedis):
-    """MiniRedis + MGET (the batch trampoline's command)."""
+    """MiniRedis + MGET with BULK command parsing (the native RESP
+    array parser instead of readline-per-arg — a 65k-arg MGET command
+    costs ~1 ms to parse instead of ~40)."""
 
     def _dispatch(self, args):
         if args[0].upper() == "MGET":
@@ -45,6 +47,59 @@ class FastMini(MiniRedis):
                     out.append(b"$%d\r\n%s\r\n" % (len(b), b))
             return b"".join(out)
         return super()._dispatch(args)
+
+    def _conn(self, conn):
+        import numpy as np
+
+        from gofr_amd import _core
+        maxi = 1 << 18
+        offs = np.zeros(maxi, np.int64)
+        lens = np.zeros(maxi, np.int32)
+        buf = bytearray()
+        nil = b"$-1\r\n"
+        try:
+            while True:
+                chunk = conn.recv(1 << 20)
+                if not chunk:
+                    return
+                buf += chunk
+                while True:
+                    arr = np.frombuffer(buf, np.uint8)
+                    nargs, done = _core.resp_parse_array(
+                        arr.ctypes.data, len(buf), maxi,
+                        offs.ctypes.data, lens.ctypes.data)
+                    del arr
+                    if not done:
+                        break
+                    consumed = int(offs[nargs - 1]) + \
+                        max(0, int(lens[nargs - 1])) + 2
+                    raw = bytes(buf[:consumed])
+                    del buf[:consumed]
+                    cmd = raw[int(offs[0]):int(offs[0]) +
+                              int(lens[0])].upper()
+                    if cmd == b"MGET":
+                        ol = offs[1:nargs].tolist()
+                        ll = lens[1:nargs].tolist()
+                        data = self.data
+                        parts = [b"*%d\r\n" % (nargs - 1)]
+                        for o, ln in zip(ol, ll):
+                            v = data.get(raw[o:o + ln].decode("latin-1"))
+                            if v is None:
+                                parts.append(nil)
+                            else:
+                                vb = v.encode()
+                                parts.append(b"$%d\r\n%s\r\n"
+                                             % (len(vb), vb))
+                        conn.sendall(b"".join(parts))
+                    else:
+                        args = [raw[int(offs[i]):int(offs[i]) +
+                                    int(lens[i])].decode("latin-1")
+                                for i in range(nargs)]
+                        conn.sendall(self._dispatch(args))
+        except (OSError, ValueError):
+            pass
+        finally:
+            conn.close()
 
 
 def run(eng, payloads, steps, warmup):

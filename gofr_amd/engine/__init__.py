@@ -113,6 +113,9 @@ class RouteProgram:
         self.handler_tab = np.asarray(rows, np.int32).reshape(-1)
         self.n_routes = len(rows)
         self.blob = bytes(blob)
+        self.fields_hook_mask = np.array(
+            [hasattr(h, "__gofr_batch_fields__")
+             for h in self.py_handlers], bool)
         self.kv_tab = (np.concatenate(kv_rows) if kv_rows
                        else np.full(6, -1, np.int32))
         self.kv_blob = bytes(kv_blob) if kv_blob else b"\0"
@@ -1083,11 +1086,27 @@ class BatchEngine:
             blob.extend(body)
             host_tab[r] = (off, len(body), status, _CT_IDS.get(ct, 0))
 
-        for r in range(n):
-            if fields[r][ops.FI_KIND] != ops.HK_HOST:
-                continue
-            if fields[r][ops.FI_FLAGS] & ops.FL_AUTH_FAIL:
-                continue  # auth middleware blocks the handler
+        # vectorized row triage (a per-row Python classification loop
+        # measured ~1.5 us x batch): fields-hook rows never touch the
+        # host parser at all — the handler reads the spans the GPU
+        # already extracted
+        F2 = np.asarray(fields).reshape(n, ops.NF)
+        kind_col = F2[:, ops.FI_KIND]
+        flags_col = F2[:, ops.FI_FLAGS]
+        route_col = F2[:, ops.FI_ROUTE]
+        host_mask = (kind_col == ops.HK_HOST) & \
+            ((flags_col & ops.FL_AUTH_FAIL) == 0)
+        hook_mask = self.program.fields_hook_mask
+        fast_mask = host_mask & (route_col >= 0) & \
+            (route_col < len(hook_mask)) & \
+            ((flags_col & (ops.FL_ERR_PARSE | ops.FL_NEEDS_HOST)) == 0)
+        fast_mask &= hook_mask[np.clip(route_col, 0,
+                                       len(hook_mask) - 1)]
+        fields_groups: dict[int, list] = {}
+        for route_id in np.unique(route_col[fast_mask]).tolist():
+            fields_groups[int(route_id)] = np.nonzero(
+                fast_mask & (route_col == route_id))[0].tolist()
+        for r in np.nonzero(host_mask & ~fast_mask)[0].tolist():
             o = int(offs[r])
             raw = np.asarray(reqs[o:o + int(lens[r])]).tobytes()
             try:
@@ -1119,11 +1138,44 @@ class BatchEngine:
                 body = b'{"error":{"message":"malformed request"}}'
                 ct = "application/json"
             emit(r, status, body, ct)
+        for route_id, rows in fields_groups.items():
+            bfn = self.program.py_handlers[
+                route_id].__gofr_batch_fields__
+            try:
+                packed = bfn(self.app, reqs, offs, fields, rows)
+                tag, pblob, ptab = packed
+                assert tag == "packed" and len(ptab) == len(rows)
+                base = len(blob)
+                blob.extend(pblob)
+                adj = np.asarray(ptab, np.int32).copy()
+                adj[:, 0] += base
+                host_tab[np.asarray(rows, np.int64)] = adj
+            except Exception as e:  # noqa: BLE001 — recovery to 500s
+                import json as _json
+                msg = _json.dumps(
+                    {"error": {"message": str(e) or "batch failed"}},
+                    separators=(",", ":")).encode()
+                for r in rows:
+                    emit(r, 500, msg, "application/json")
         for route_id, group in batch_groups.items():
             bfn = self.program.py_handlers[route_id].__gofr_batch__
             requests = [req for _, req in group]
             try:
                 results = bfn(self.app, requests)
+                if (isinstance(results, tuple) and len(results) == 3
+                        and results[0] == "packed"):
+                    # packed protocol: one blob + int32 [n,4] table
+                    # (off, len, status, ct) — appended wholesale, no
+                    # per-row Python emit
+                    _, pblob, ptab = results
+                    base = len(blob)
+                    blob.extend(pblob)
+                    rows = np.fromiter((r for r, _ in group), np.int64,
+                                       len(group))
+                    adj = np.asarray(ptab, np.int32).copy()
+                    adj[:, 0] += base
+                    host_tab[rows] = adj
+                    continue
                 assert len(results) == len(group)
             except Exception as e:  # noqa: BLE001 — recovery to 500s
                 import json as _json
