@@ -1327,6 +1327,17 @@ class GPUServer:
         stats = os.environ.get("GOFR_SERVE_STATS") == "1"
         st = {"harvest": 0.0, "submit": 0.0, "complete": 0.0,
               "send": 0.0, "cycles": 0, "reqs": 0}
+        # adaptive batch deadline (GOFR_ADAPTIVE_WINDOW=1 — SURVEY §7's
+        # central latency/throughput tension): light load keeps the
+        # latency-first return-on-first-drain harvest; sustained
+        # under-filled cycles switch to a FILL deadline (min_fill +
+        # growing window, capped) so batches arrive fuller; near-full
+        # cycles mean the queues are deep and the window shrinks back.
+        window = float(self.batch_window_us)
+        w_min = float(self.batch_window_us)
+        w_max = float(os.environ.get("GOFR_MAX_WINDOW_US", "2000"))
+        adaptive = os.environ.get("GOFR_ADAPTIVE_WINDOW", "0") == "1"
+        min_fill = 1
         while not self._stop.is_set():
             progressed = False
             if free:
@@ -1338,7 +1349,18 @@ class GPUServer:
                 n, nbytes = self._core.harvest(
                     ln.p_reqs.data_ptr(), bufcap,
                     ln.p_req_off.data_ptr(), ln.p_req_len.data_ptr(),
-                    conn_ids[L].ctypes.data, maxn, self.batch_window_us)
+                    conn_ids[L].ctypes.data, maxn, int(window),
+                    int(min_fill))
+                if adaptive and n:
+                    if n >= maxn - (maxn >> 3):  # deep queues
+                        window = max(w_min, window * 0.7)
+                        min_fill = 1
+                    elif n >= maxn >> 3:         # moderate: batch up
+                        min_fill = maxn >> 1
+                        window = min(w_max, window * 1.3 + 20)
+                    else:                        # light: latency first
+                        min_fill = 1
+                        window = w_min
                 if stats:
                     st["harvest"] += time.perf_counter() - t0
                 if n:

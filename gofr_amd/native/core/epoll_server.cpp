@@ -1038,10 +1038,14 @@ public:
     // dense. The armed serving loop ships the whole fixed-size ring
     // every batch, so segmented placement costs no extra bus bytes.
     // Blocks up to window_us for the FIRST request. GIL released.
+    // min_fill > 1 turns the window into a FILL deadline: keep
+    // draining until min_fill requests arrived or window_us elapsed
+    // (the adaptive batcher's throughput mode); min_fill <= 1 is the
+    // latency-first return-on-first-drain behavior.
     std::pair<int, long> harvest(uintptr_t buf_ptr, long buf_cap,
                                  uintptr_t off_ptr, uintptr_t len_ptr,
                                  uintptr_t conn_ptr, int max_n,
-                                 int window_us) {
+                                 int window_us, int min_fill = 1) {
         py::gil_scoped_release rel;
         uint8_t* buf = (uint8_t*)buf_ptr;
         int64_t* offs = (int64_t*)off_ptr;
@@ -1115,7 +1119,9 @@ public:
                 const long end = (long)r * seg_bytes + s.used;
                 if (end > nbytes) nbytes = end;
             }
-            if (n > 0 || std::chrono::steady_clock::now() >= deadline)
+            if (n >= (min_fill > 1 ? min_fill : 1) ||
+                (n > 0 && min_fill <= 1) ||
+                std::chrono::steady_clock::now() >= deadline)
                 break;
             std::this_thread::sleep_for(std::chrono::microseconds(50));
         }
@@ -1386,7 +1392,10 @@ PYBIND11_MODULE(_core, m) {
         .def("stop", &EpollServer::stop)
         .def("port", &EpollServer::port)
         .def("ready_count", &EpollServer::ready_count)
-        .def("harvest", &EpollServer::harvest)
+        .def("harvest", &EpollServer::harvest,
+             py::arg("buf_ptr"), py::arg("buf_cap"), py::arg("off_ptr"),
+             py::arg("len_ptr"), py::arg("conn_ptr"), py::arg("max_n"),
+             py::arg("window_us"), py::arg("min_fill") = 1)
         .def("harvest_slots", &EpollServer::harvest_slots)
         .def("send", &EpollServer::send)
         .def("register_grpc_path", &EpollServer::register_grpc_path)

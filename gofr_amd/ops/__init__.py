@@ -661,9 +661,15 @@ def _json_body_valid(body: bytes) -> bool:
 
 # ---- template / KV / JSON-bind mirrors (gofr_kernels.hip helpers) ----------
 
+_HEXV = {c: i for i, c in enumerate(b"0123456789abcdef")}
+_HEXV.update({c: i for i, c in enumerate(b"0123456789ABCDEF")})
+
+
 def pct_decode(s: bytes):
     """Mirror of pct_decode_inplace: strict %XX decode ('+' untouched).
-    Returns None on an invalid escape (kernel: host fallback)."""
+    Returns None on an invalid escape (kernel: host fallback). STRICT
+    hex only — Python's int(x, 16) accepts ' 1'/'+1', the kernel's
+    hexval does not (divergence caught by the 20k fuzz campaign)."""
     out = bytearray()
     i = 0
     n = len(s)
@@ -672,10 +678,11 @@ def pct_decode(s: bytes):
         if c == 0x25:  # %
             if i + 2 >= n:
                 return None
-            try:
-                c = int(s[i + 1:i + 3].decode("latin-1"), 16)
-            except ValueError:
+            hi = _HEXV.get(s[i + 1])
+            lo = _HEXV.get(s[i + 2])
+            if hi is None or lo is None:
                 return None
+            c = (hi << 4) | lo
             i += 2
         out.append(c)
         i += 1
@@ -721,11 +728,11 @@ def splice_py(src: bytes, mode: int) -> bytes:
             if c == 0x2B:  # '+'
                 c = 0x20
             elif c == 0x25 and i + 2 < n:
-                try:
-                    c = int(src[i + 1:i + 3].decode("latin-1"), 16)
+                hi = _HEXV.get(src[i + 1])
+                lo = _HEXV.get(src[i + 2])
+                if hi is not None and lo is not None:
+                    c = (hi << 4) | lo
                     i += 2
-                except ValueError:
-                    pass
         if mode & TM_JESC:
             if c in (0x22, 0x5C):  # '"' '\\'
                 out.append(0x5C)
