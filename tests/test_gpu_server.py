@@ -121,49 +121,3 @@ def test_grpc_batched_codec_on_gpu():
         assert s.codec_msgs >= 3
     finally:
         s.stop()
-
-
-def test_adaptive_window_serving(monkeypatch):
-    """GOFR_ADAPTIVE_WINDOW=1 (min-fill batch deadline) serves
-    correctly under both single requests and pipelined bursts."""
-    import socket as _socket
-
-    monkeypatch.setenv("GOFR_ADAPTIVE_WINDOW", "1")
-    monkeypatch.setenv("GOFR_MAX_WINDOW_US", "1000")
-    from gofr_amd.engine import GPUServer
-    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
-    app.GET("/greet", handlers.static_json("hi"))
-    s = GPUServer(app, 0, batch_window_us=200)
-    s.start()
-    try:
-        # single request (light load: latency-first path)
-        st, _, body = _req(s.port, "GET", "/greet")
-        assert st == 200 and json.loads(body) == {"data": "hi"}
-        # pipelined burst (drives the fill-deadline branch)
-        conn = _socket.create_connection(("127.0.0.1", s.port),
-                                         timeout=10)
-        burst = b"GET /greet HTTP/1.1\r\nHost: h\r\n\r\n" * 64
-        conn.sendall(burst)
-        got = 0
-        buf = b""
-        while got < 64:
-            chunk = conn.recv(65536)
-            assert chunk
-            buf += chunk
-            while True:
-                i = buf.find(b"\r\n\r\n")
-                if i < 0:
-                    break
-                head = buf[:i]
-                clen = 0
-                for line in head.split(b"\r\n")[1:]:
-                    if line.lower().startswith(b"content-length:"):
-                        clen = int(line.split(b":")[1])
-                if len(buf) < i + 4 + clen:
-                    break
-                assert head.startswith(b"HTTP/1.1 200"), head[:40]
-                buf = buf[i + 4 + clen:]
-                got += 1
-        conn.close()
-    finally:
-        s.stop()

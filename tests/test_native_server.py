@@ -232,3 +232,60 @@ def test_lifecycle_cycles_leak_free():
             base = (fds(), threading.active_count())
     assert fds() - base[0] <= 4, "fd leak across server lifecycles"
     assert threading.active_count() - base[1] <= 2, "thread leak"
+
+
+def test_harvest_min_fill_deadline():
+    """min_fill > 1 turns the harvest window into a FILL deadline
+    (the adaptive batcher's throughput mode): a short burst below
+    min_fill waits out the window; a burst past min_fill returns as
+    soon as the fill target is met."""
+    import socket
+    import time
+
+    import numpy as np
+
+    from gofr_amd import _core
+
+    core = _core.EpollServer(0, 1 << 20, 2)
+    core.start()
+    try:
+        conn = socket.create_connection(("127.0.0.1", core.port()),
+                                        timeout=10)
+        req = b"GET /x HTTP/1.1\r\nHost: h\r\n\r\n"
+        buf = np.zeros(1 << 20, np.uint8)
+        offs = np.zeros(256, np.int64)
+        lens = np.zeros(256, np.int32)
+        cids = np.zeros(256, np.uint64)
+
+        conn.sendall(req * 3)
+        time.sleep(0.05)
+        t0 = time.perf_counter()
+        n, nb = core.harvest(buf.ctypes.data, 1 << 20, offs.ctypes.data,
+                             lens.ctypes.data, cids.ctypes.data, 256,
+                             100_000, 8)  # want 8, only 3 queued
+        waited = time.perf_counter() - t0
+        assert n == 3
+        assert waited >= 0.08, waited  # waited out the fill window
+
+        conn.sendall(req * 64)
+        t0 = time.perf_counter()
+        total = 0
+        while total < 32:
+            n, nb = core.harvest(buf.ctypes.data, 1 << 20,
+                                 offs.ctypes.data, lens.ctypes.data,
+                                 cids.ctypes.data, 256, 1_000_000, 32)
+            total += n
+        waited = time.perf_counter() - t0
+        assert waited < 0.5, waited  # fill target met early
+        # min_fill=1 keeps the latency-first return-on-first behavior
+        conn.sendall(req)
+        n = 0
+        t0 = time.perf_counter()
+        while n == 0 and time.perf_counter() - t0 < 5:
+            n, nb = core.harvest(buf.ctypes.data, 1 << 20,
+                                 offs.ctypes.data, lens.ctypes.data,
+                                 cids.ctypes.data, 256, 200_000, 1)
+        assert n >= 1
+        conn.close()
+    finally:
+        core.stop()
