@@ -355,3 +355,47 @@ def test_parse_mirror_percent_decode_agrees_with_unquote(segs, query):
     got_q = bytes(buf[f[ops.FI_QUERY_OFF]:
                       f[ops.FI_QUERY_OFF] + f[ops.FI_QUERY_LEN]])
     assert got_q == (query or "").encode()  # query stays raw
+
+
+_VALTXT = st.text(
+    alphabet=st.characters(min_codepoint=0x20, max_codepoint=0x7E,
+                           blacklist_characters="/?#& %+"),
+    min_size=0, max_size=12)
+
+
+@SET
+@given(_VALTXT, _VALTXT,
+       st.dictionaries(_JKEY, _JVAL, min_size=0, max_size=4))
+def test_template_engine_dispatch_parity_fuzz(pid, qval, body_obj):
+    """The HK_TEMPLATE mirror (engine) and the handler's Python body
+    (CPU transport dispatch) must render identical bytes for arbitrary
+    path params, query values and JSON bodies."""
+    import json as _json
+
+    import gofr_amd
+    from gofr_amd import handlers
+    from gofr_amd.config import MapConfig
+    from gofr_amd.engine import BatchEngine
+    from gofr_amd.http.request import parse_request_bytes
+    from gofr_amd.server import dispatch
+
+    app = gofr_amd.New(config=MapConfig({"LOG_LEVEL": "FATAL"}))
+    app.GET("/user/{id}", handlers.template_json(
+        '{"data":{"id":"', ("path", 0), '","q":"', ("query", "q"),
+        '"}}'))
+    app.POST("/order", handlers.template_json(
+        '{"data":{"f":', ("jfield", "f"), "}}"))
+    eng = BatchEngine(app)
+    body = _json.dumps({"f": body_obj}, separators=(",", ":")).encode()
+    raws = [
+        (f"GET /user/{pid or 'x'}?q={qval} HTTP/1.1\r\n"
+         "Host: h\r\n\r\n").encode(),
+        (f"POST /order HTTP/1.1\r\nHost: h\r\n"
+         f"Content-Type: application/json\r\n"
+         f"Content-Length: {len(body)}\r\n\r\n").encode() + body,
+    ]
+    outs = eng.process(list(raws))
+    for raw, out in zip(raws, outs):
+        _, _, ebody = out.partition(b"\r\n\r\n")
+        resp = dispatch(app, parse_request_bytes(raw))
+        assert resp.body == ebody, (raw, resp.body, ebody)
