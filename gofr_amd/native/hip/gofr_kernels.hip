@@ -378,8 +378,12 @@ __device__ void parse_one(uint8_t* __restrict__ reqs,
             body_off = lf_pos[j_empty] + 1;
             hdr_last = j_empty - 1;
         } else if (nlf - 1 > 63) {
-            hdr_last = 63;       // >63 header lines: parse the cap,
-        } else {                 // body falls back to request end
+            // >63 header lines before any blank: beyond the one-line-
+            // per-lane budget — let the host parser serve it fully
+            // (r1 silently truncated the body here)
+            flags |= FL_NEEDS_HOST;
+            hdr_last = 63;
+        } else {
             flags |= FL_ERR_PARSE;  // no blank line before the LFs ran out
             hdr_last = navail - 1;
         }

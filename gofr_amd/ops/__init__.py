@@ -494,7 +494,8 @@ def cpu_parse_route(reqs: np.ndarray, req_off: np.ndarray,
             body_off = lfs[j_empty] + 1
             hdr_last = j_empty - 1
         elif nlf - 1 > 63:
-            hdr_last = 63  # >63 header lines: body falls back to end
+            flags |= FL_NEEDS_HOST  # header-line budget: host serves
+            hdr_last = 63
         else:
             flags |= FL_ERR_PARSE  # LFs ran out before a blank line
             hdr_last = min(nlf - 1, 63)

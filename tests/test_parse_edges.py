@@ -35,24 +35,25 @@ def test_max_62_headers_with_blank_at_63_parse():
                f[ops.FI_BODY_LEN]] == b"xyz"
 
 
-def test_beyond_63_headers_body_falls_to_request_end():
-    """>63 header lines: the cap parses the first 63 and the body
-    falls back to the request end (the r1 serial guard's behavior)."""
+def test_beyond_63_headers_routes_to_host():
+    """>63 header lines before the blank: beyond the one-line-per-lane
+    budget — the request routes to the host parser (r1 silently
+    truncated the body here)."""
     hdrs = "".join(f"H{i}: v{i}\r\n" for i in range(80))
     raw = (f"GET /a HTTP/1.1\r\n{hdrs}\r\nbody").encode()
     f, _ = parse_one(raw)
-    # blank line beyond the examined range -> body_off stays at len
-    assert f[ops.FI_BODY_LEN] == 0
+    assert f[ops.FI_FLAGS] & ops.FL_NEEDS_HOST
+    assert f[ops.FI_KIND] == ops.HK_HOST
 
 
-def test_header_after_64th_line_not_seen():
-    """A Content-Length buried past the 63-line cap is not parsed
-    (matches the kernel's one-lane-per-line budget)."""
+def test_header_after_64th_line_routes_to_host():
+    """A Content-Length buried past the 63-line budget: the request
+    routes to the host parser rather than being half-parsed."""
     hdrs = "".join(f"H{i}: v{i}\r\n" for i in range(70))
     raw = (f"GET /a HTTP/1.1\r\n{hdrs}Content-Length: 5\r\n"
            "\r\nhello").encode()
     f, _ = parse_one(raw)
-    assert f[ops.FI_CLEN] == 0
+    assert f[ops.FI_FLAGS] & ops.FL_NEEDS_HOST
 
 
 def test_bare_lf_blank_line_ends_headers():
