@@ -185,7 +185,10 @@ class Redis:
             self._sock.sendall(payload)
             buf = bytearray()
             while True:
-                chunk = self._sock.recv(1 << 20)
+                # read via the SAME BufferedReader execute() uses — a
+                # raw recv would bypass bytes it already buffered and
+                # desync the stream (caught by the thread-safety test)
+                chunk = self._rfile.read1(1 << 20)
                 if not chunk:
                     raise RedisError("connection closed mid-reply")
                 buf += chunk

@@ -98,3 +98,45 @@ def test_redis_batch_handler_error_recovers(mini):
     head, _, body = out.partition(b"\r\n\r\n")
     assert head.startswith(b"HTTP/1.1 500"), head[:40]
     assert b"error" in body
+
+
+def test_redis_client_thread_safety(mini):
+    """Concurrent serve threads share the client: interleaved
+    execute/MGet/mget_spans must serialize correctly on the command
+    lock (no cross-talk between replies)."""
+    import threading
+
+    for i in range(64):
+        mini.data[f"t:{i}"] = str(i)
+    r = Redis("127.0.0.1", mini.port)
+    r.connect()
+    errors = []
+
+    def worker(k):
+        try:
+            for i in range(40):
+                if i % 3 == 0:
+                    v = r.Get(f"t:{(k * 7 + i) % 64}")
+                    assert v == str((k * 7 + i) % 64), v
+                elif i % 3 == 1:
+                    keys = [f"t:{(k + j) % 64}" for j in range(8)]
+                    vals = r.MGet(keys)
+                    assert list(vals) == \
+                        [str((k + j) % 64) for j in range(8)]
+                else:
+                    keys = [f"t:{(k + j) % 64}" for j in range(8)]
+                    reply, offs, lens = r.mget_spans(keys)
+                    got = [reply[int(offs[j]):int(offs[j]) +
+                                 int(lens[j])].decode()
+                           for j in range(8)]
+                    assert got == [str((k + j) % 64) for j in range(8)]
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    ts = [threading.Thread(target=worker, args=(k,)) for k in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=60)
+    assert not errors, errors
+    r.close()
